@@ -1,0 +1,11 @@
+"""Model zoo: Llama-3 (flagship), GPT-2, Mixtral-style MoE (see moe/)."""
+
+from .llama import (LlamaConfig, LlamaForCausalLM, llama3_8b, llama3_70b,
+                    llama_tiny, llama_mini)
+from .gpt2 import GPT2Config, GPT2ForCausalLM, gpt2_small, gpt2_tiny
+
+__all__ = [
+    "LlamaConfig", "LlamaForCausalLM", "llama3_8b", "llama3_70b",
+    "llama_tiny", "llama_mini", "GPT2Config", "GPT2ForCausalLM",
+    "gpt2_small", "gpt2_tiny",
+]

@@ -1,0 +1,210 @@
+"""Llama-family decoder (Llama-3 shapes) built on the CDNA4 fused ops.
+
+This is the framework's flagship training model for the headline benchmark
+(Llama-3-8B ZeRO-3 bf16 on MI355X — BASELINE.json). Random-init weights,
+synthetic data; architecture matches Llama-3 (GQA attention, RoPE
+theta=500000, SwiGLU MLP, RMSNorm, tied-off lm_head).
+
+Hot ops: RMSNorm / RoPE / SwiGLU are the hand-written HIP kernels in
+``deepspeed_amd.ops``; GEMMs go through torch (hipBLASLt); attention uses
+torch SDPA (flash backend on ROCm) with an explicit-math fallback.
+"""
+
+import math
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.norms import RMSNorm
+from ..ops.rope import apply_rope, rope_tables
+from ..ops.swiglu import swiglu
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    max_seq_len: int = 8192
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    tie_embeddings: bool = False
+    initializer_range: float = 0.02
+
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.num_heads
+
+
+def llama3_8b():
+    return LlamaConfig()
+
+
+def llama3_70b():
+    return LlamaConfig(hidden_size=8192, intermediate_size=28672,
+                       num_layers=80, num_heads=64, num_kv_heads=8)
+
+
+def llama_tiny():
+    """Small config for CPU tests."""
+    return LlamaConfig(vocab_size=512, hidden_size=64, intermediate_size=128,
+                       num_layers=2, num_heads=4, num_kv_heads=2,
+                       max_seq_len=128)
+
+
+def llama_mini():
+    """~0.5B for single-GPU smoke runs."""
+    return LlamaConfig(vocab_size=32000, hidden_size=1024,
+                       intermediate_size=2816, num_layers=8, num_heads=16,
+                       num_kv_heads=4, max_seq_len=4096)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.num_heads = cfg.num_heads
+        self.num_kv_heads = cfg.num_kv_heads
+        self.head_dim = cfg.head_dim
+        h = cfg.hidden_size
+        self.q_proj = nn.Linear(h, cfg.num_heads * self.head_dim, bias=False)
+        self.k_proj = nn.Linear(h, cfg.num_kv_heads * self.head_dim, bias=False)
+        self.v_proj = nn.Linear(h, cfg.num_kv_heads * self.head_dim, bias=False)
+        self.o_proj = nn.Linear(cfg.num_heads * self.head_dim, h, bias=False)
+
+    def forward(self, x, cos, sin, positions=None, kv_cache=None,
+                attention_fn=None):
+        B, S, _ = x.shape
+        q = self.q_proj(x).view(B, S, self.num_heads, self.head_dim)
+        k = self.k_proj(x).view(B, S, self.num_kv_heads, self.head_dim)
+        v = self.v_proj(x).view(B, S, self.num_kv_heads, self.head_dim)
+        q = apply_rope(q, cos, sin, positions)
+        k = apply_rope(k, cos, sin, positions)
+
+        # [B, H, S, D] for SDPA
+        q = q.transpose(1, 2)
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        if kv_cache is not None:
+            k, v = kv_cache.update(self.layer_idx, k, v)
+        if attention_fn is not None:
+            o = attention_fn(q, k, v)
+        else:
+            o = sdpa_gqa(q, k, v, causal=kv_cache is None)
+        o = o.transpose(1, 2).reshape(B, S, -1)
+        return self.o_proj(o)
+
+
+def sdpa_gqa(q, k, v, causal=True):
+    """SDPA with grouped-query support (expands KV if enable_gqa missing)."""
+    try:
+        return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
+                                              enable_gqa=True)
+    except (TypeError, RuntimeError):
+        rep = q.shape[1] // k.shape[1]
+        if rep > 1:
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+        return F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        h, i = cfg.hidden_size, cfg.intermediate_size
+        self.gate_proj = nn.Linear(h, i, bias=False)
+        self.up_proj = nn.Linear(h, i, bias=False)
+        self.down_proj = nn.Linear(i, h, bias=False)
+
+    def forward(self, x):
+        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig, layer_idx: int):
+        super().__init__()
+        self.input_layernorm = RMSNorm(cfg.hidden_size, eps=cfg.rms_eps)
+        self.self_attn = LlamaAttention(cfg)
+        self.self_attn.layer_idx = layer_idx
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, eps=cfg.rms_eps)
+        self.mlp = LlamaMLP(cfg)
+
+    def forward(self, x, cos, sin, positions=None, kv_cache=None):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin, positions,
+                               kv_cache)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg, i) for i in range(cfg.num_layers)])
+        self.norm = RMSNorm(cfg.hidden_size, eps=cfg.rms_eps)
+
+    def forward(self, input_ids, positions=None, kv_cache=None):
+        x = self.embed_tokens(input_ids)
+        cos, sin = rope_tables(self.cfg.head_dim, self.cfg.max_seq_len,
+                               self.cfg.rope_theta, device=x.device)
+        for layer in self.layers:
+            x = layer(x, cos, sin, positions, kv_cache)
+        return self.norm(x)
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.model = LlamaModel(cfg)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        std = self.cfg.initializer_range
+        if isinstance(module, nn.Linear):
+            module.weight.data.normal_(0.0, std)
+            if module.bias is not None:
+                module.bias.data.zero_()
+        elif isinstance(module, nn.Embedding):
+            module.weight.data.normal_(0.0, std)
+
+    def forward(self, input_ids, labels=None, positions=None, kv_cache=None):
+        hidden = self.model(input_ids, positions, kv_cache)
+        if labels is not None:
+            # chunked fp32 cross-entropy: avoids materializing the full
+            # [B*S, vocab] fp32 logits (vocab=128256 -> ~2 GB per 4k tokens)
+            return chunked_cross_entropy(hidden, self.lm_head.weight, labels)
+        return self.lm_head(hidden)
+
+    def num_parameters(self):
+        return sum(p.numel() for p in self.parameters())
+
+
+def chunked_cross_entropy(hidden, lm_weight, labels, chunk_tokens=8192):
+    """loss = CE(hidden @ W^T, labels), computed in token chunks so the fp32
+    logits never exceed chunk_tokens x vocab."""
+    B, S, H = hidden.shape
+    hidden = hidden.view(-1, H)
+    labels = labels.view(-1)
+    n = hidden.shape[0]
+    total = hidden.new_zeros((), dtype=torch.float32)
+    count = hidden.new_zeros((), dtype=torch.float32)
+    for s in range(0, n, chunk_tokens):
+        e = min(s + chunk_tokens, n)
+        logits = F.linear(hidden[s:e], lm_weight).float()
+        tgt = labels[s:e]
+        total = total + F.cross_entropy(logits, tgt, ignore_index=-100,
+                                        reduction="sum")
+        count = count + (tgt != -100).sum()
+    return total / count.clamp(min=1)

@@ -1,0 +1,13 @@
+"""Hand-written CDNA4 (gfx950) HIP ops with CPU torch fallbacks."""
+
+from ._loader import has_ext, get_ext
+from .adam import FusedAdam, fused_adam_step, multi_tensor_adam_available
+from .norms import RMSNorm, FusedLayerNorm, rms_norm, layer_norm
+from .rope import apply_rope, rope_tables
+from .swiglu import swiglu, geglu
+
+__all__ = [
+    "has_ext", "get_ext", "FusedAdam", "fused_adam_step",
+    "multi_tensor_adam_available", "RMSNorm", "FusedLayerNorm", "rms_norm",
+    "layer_norm", "apply_rope", "rope_tables", "swiglu", "geglu",
+]

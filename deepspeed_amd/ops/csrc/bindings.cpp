@@ -1,0 +1,167 @@
+// Python bindings for the deepspeed_amd CDNA4 kernels.
+//
+// Uses torch-ROCm's native HIP stream API (c10::hip) — no CUDA shims.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <tuple>
+
+extern "C" void ds_fused_adam_flat(float* p, const void* g, int grad_dtype,
+                                   float* m, float* v, void* p16, long long n,
+                                   float lr, float beta1, float beta2, float eps,
+                                   float weight_decay, int step, float inv_scale,
+                                   int adamw, void* stream);
+extern "C" void ds_norm_fwd(const void* x, const void* w, const void* b,
+                            void* y, float* invrms, float* mean, int rows,
+                            int H, float eps, int ln, int dtype, void* stream);
+extern "C" void ds_norm_bwd(const void* dy, const void* x, const void* w,
+                            const float* invrms, const float* mean, void* dx,
+                            float* dw, float* db, int rows, int H, int ln,
+                            int dtype, void* stream);
+extern "C" void ds_rope(void* x, const float* cos_table, const float* sin_table,
+                        const int* positions, long long batch, long long seq,
+                        long long heads, long long dim, int bwd, int dtype,
+                        void* stream);
+extern "C" void ds_gated_act_fwd(const void* gate, const void* up, void* out,
+                                 long long n, int act, int dtype, void* stream);
+extern "C" void ds_gated_act_bwd(const void* dout, const void* gate,
+                                 const void* up, void* dgate, void* dup,
+                                 long long n, int act, int dtype, void* stream);
+
+namespace {
+
+int dtype_code(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return 0;
+    case at::kBFloat16: return 1;
+    case at::kHalf: return 2;
+    default:
+      TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+}
+
+void* cur_stream() {
+  return reinterpret_cast<void*>(c10::hip::getCurrentHIPStream().stream());
+}
+
+void fused_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                     c10::optional<at::Tensor> p16, double lr, double beta1,
+                     double beta2, double eps, double weight_decay,
+                     int64_t step, double inv_scale, bool adamw) {
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous() &&
+              v.is_contiguous(), "fused_adam_flat: tensors must be contiguous");
+  TORCH_CHECK(p.scalar_type() == at::kFloat && m.scalar_type() == at::kFloat &&
+              v.scalar_type() == at::kFloat, "p/m/v must be fp32");
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() &&
+              p.numel() == v.numel(), "numel mismatch");
+  void* p16_ptr = nullptr;
+  if (p16.has_value() && p16->defined()) {
+    TORCH_CHECK(p16->scalar_type() == at::kBFloat16 && p16->is_contiguous() &&
+                p16->numel() == p.numel(), "p16 must be contiguous bf16");
+    p16_ptr = p16->data_ptr();
+  }
+  ds_fused_adam_flat(p.data_ptr<float>(), g.data_ptr(), dtype_code(g),
+                     m.data_ptr<float>(), v.data_ptr<float>(), p16_ptr,
+                     p.numel(), (float)lr, (float)beta1, (float)beta2,
+                     (float)eps, (float)weight_decay, (int)step,
+                     (float)inv_scale, adamw ? 1 : 0, cur_stream());
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> norm_fwd(
+    at::Tensor x, at::Tensor w, c10::optional<at::Tensor> b, double eps,
+    bool layernorm) {
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous(), "must be contiguous");
+  const int64_t H = x.size(-1);
+  const int64_t rows = x.numel() / H;
+  TORCH_CHECK(w.numel() == H, "weight shape mismatch");
+  auto y = at::empty_like(x);
+  auto f32opts = x.options().dtype(at::kFloat);
+  auto invrms = at::empty({rows}, f32opts);
+  auto mean = layernorm ? at::empty({rows}, f32opts) : at::empty({0}, f32opts);
+  const void* bptr = nullptr;
+  if (b.has_value() && b->defined()) {
+    TORCH_CHECK(b->is_contiguous() && b->numel() == H, "bias shape mismatch");
+    bptr = b->data_ptr();
+  }
+  ds_norm_fwd(x.data_ptr(), w.data_ptr(), bptr, y.data_ptr(),
+              invrms.data_ptr<float>(),
+              layernorm ? mean.data_ptr<float>() : nullptr, (int)rows, (int)H,
+              (float)eps, layernorm ? 1 : 0, dtype_code(x), cur_stream());
+  return {y, invrms, mean};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> norm_bwd(
+    at::Tensor dy, at::Tensor x, at::Tensor w, at::Tensor invrms,
+    c10::optional<at::Tensor> mean, bool layernorm) {
+  TORCH_CHECK(dy.is_contiguous() && x.is_contiguous() && w.is_contiguous(),
+              "must be contiguous");
+  const int64_t H = x.size(-1);
+  const int64_t rows = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto f32opts = x.options().dtype(at::kFloat);
+  auto dw = at::zeros({H}, f32opts);
+  auto db = layernorm ? at::zeros({H}, f32opts) : at::empty({0}, f32opts);
+  ds_norm_bwd(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+              invrms.data_ptr<float>(),
+              layernorm ? mean->data_ptr<float>() : nullptr, dx.data_ptr(),
+              dw.data_ptr<float>(), layernorm ? db.data_ptr<float>() : nullptr,
+              (int)rows, (int)H, layernorm ? 1 : 0, dtype_code(x),
+              cur_stream());
+  return {dx, dw, db};
+}
+
+void rope(at::Tensor x, at::Tensor cos_table, at::Tensor sin_table,
+          c10::optional<at::Tensor> positions, bool backward) {
+  TORCH_CHECK(x.dim() == 4 && x.is_contiguous(), "rope expects [B,S,H,D]");
+  TORCH_CHECK(cos_table.scalar_type() == at::kFloat &&
+              sin_table.scalar_type() == at::kFloat, "tables must be fp32");
+  const int64_t B = x.size(0), S = x.size(1), Hh = x.size(2), D = x.size(3);
+  TORCH_CHECK(cos_table.size(-1) == D / 2, "table dim mismatch");
+  TORCH_CHECK(cos_table.size(0) >= S || positions.has_value(),
+              "cos table shorter than sequence");
+  const int* pos_ptr = nullptr;
+  if (positions.has_value() && positions->defined()) {
+    TORCH_CHECK(positions->scalar_type() == at::kInt &&
+                positions->is_contiguous() && positions->numel() == B * S,
+                "positions must be int32 [B,S]");
+    pos_ptr = positions->data_ptr<int>();
+  }
+  ds_rope(x.data_ptr(), cos_table.data_ptr<float>(),
+          sin_table.data_ptr<float>(), pos_ptr, B, S, Hh, D, backward ? 1 : 0,
+          dtype_code(x), cur_stream());
+}
+
+at::Tensor gated_act_fwd(at::Tensor gate, at::Tensor up, int64_t act) {
+  TORCH_CHECK(gate.is_contiguous() && up.is_contiguous() &&
+              gate.sizes() == up.sizes(), "shape mismatch");
+  auto out = at::empty_like(gate);
+  ds_gated_act_fwd(gate.data_ptr(), up.data_ptr(), out.data_ptr(),
+                   gate.numel(), (int)act, dtype_code(gate), cur_stream());
+  return out;
+}
+
+std::tuple<at::Tensor, at::Tensor> gated_act_bwd(at::Tensor dout,
+                                                 at::Tensor gate,
+                                                 at::Tensor up, int64_t act) {
+  TORCH_CHECK(dout.is_contiguous() && gate.is_contiguous() &&
+              up.is_contiguous(), "must be contiguous");
+  auto dgate = at::empty_like(gate);
+  auto dup = at::empty_like(up);
+  ds_gated_act_bwd(dout.data_ptr(), gate.data_ptr(), up.data_ptr(),
+                   dgate.data_ptr(), dup.data_ptr(), dout.numel(), (int)act,
+                   dtype_code(gate), cur_stream());
+  return {dgate, dup};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_adam_flat", &fused_adam_flat,
+        "Fused Adam/AdamW on flat fp32 master + 16-bit grad shard");
+  m.def("norm_fwd", &norm_fwd, "RMSNorm/LayerNorm forward");
+  m.def("norm_bwd", &norm_bwd, "RMSNorm/LayerNorm backward");
+  m.def("rope", &rope, "Rotary position embedding (in-place)");
+  m.def("gated_act_fwd", &gated_act_fwd, "SwiGLU/GeGLU forward");
+  m.def("gated_act_bwd", &gated_act_bwd, "SwiGLU/GeGLU backward");
+}

@@ -1,0 +1,533 @@
+"""Core training engine.
+
+Capability parity with the reference's ``deepspeed/runtime/engine.py``
+(DeepSpeedEngine :184 — forward :1926 / backward :2085 / step :2282 /
+save_checkpoint :3218 / load_checkpoint :2872 / allreduce fallback :2611),
+re-designed for one-process-per-GPU over RCCL/xGMI on MI355X.
+
+Precision policy (simpler than the reference's three optimizer stacks):
+* fp32 + ZeRO stage 0 -> plain DDP-style bucketed all-reduce + user optimizer
+* bf16/fp16 at stage 0/1 -> ZeRO-1 partitioned fp32-master optimizer
+  (this is what the reference's BF16_Optimizer does)
+* stage 2 -> ZeRO-2 (gradient partitioning via reduce-scatter)
+* stage 3 -> ZeRO-3 (parameter partitioning; see zero/stage3.py)
+"""
+
+import os
+import shutil
+from typing import Callable, Optional
+
+import torch
+
+from .. import accel
+from .. import comm as dist
+from ..config import Config
+from ..parallel import groups
+from ..utils.logging import log_dist, logger
+from ..utils.timer import (SynchronizedWallClockTimer, ThroughputTimer,
+                           FORWARD_GLOBAL_TIMER, BACKWARD_GLOBAL_TIMER,
+                           STEP_GLOBAL_TIMER)
+from . import lr_schedules
+from .fp16.loss_scaler import create_loss_scaler, LossScaler
+from .utils import (DummyOptim, get_global_norm_of_tensors,
+                    clip_tensors_by_global_norm, see_memory_usage)
+from .zero.stage12 import ZeroStage12Optimizer
+
+MEMORY_OPT_ALLREDUCE_SIZE = 500_000_000
+
+
+class Engine(torch.nn.Module):
+    """Training wrapper returned by :func:`deepspeed_amd.initialize`."""
+
+    def __init__(self,
+                 model: torch.nn.Module,
+                 optimizer: Optional[torch.optim.Optimizer] = None,
+                 model_parameters=None,
+                 lr_scheduler=None,
+                 config: Optional[Config] = None,
+                 mpu=None,
+                 dont_change_device: bool = False):
+        super().__init__()
+        self.module = model
+        self.client_optimizer = optimizer
+        self.client_lr_scheduler = lr_scheduler
+        self.config = config if config is not None else Config(None)
+        self.mpu = mpu
+        self.global_steps = 0
+        self.global_samples = 0
+        self.micro_steps = 0
+        self.skipped_steps = 0
+        self.gradient_accumulation_steps = self.config.gradient_accumulation_steps
+        self.train_micro_batch_size_per_gpu = self.config.train_micro_batch_size_per_gpu
+        self.train_batch_size = self.config.train_batch_size
+        self._is_gradient_accumulation_boundary = True
+        self.inside_no_sync = False
+
+        if not dist.is_initialized():
+            dist.init_distributed()
+        if mpu is not None:
+            groups.set_mpu(mpu)
+        self.dp_group = groups.get_data_parallel_group()
+        self.dp_world_size = dist.get_world_size(self.dp_group)
+        self.seq_parallel_world_size = groups.get_sequence_parallel_world_size()
+
+        self.device = accel.current_device() if dont_change_device else \
+            accel.device_for(accel.local_rank_from_env())
+        self.local_rank = accel.local_rank_from_env()
+        self.global_rank = dist.get_rank()
+        self.world_size = dist.get_world_size()
+
+        self.timers = SynchronizedWallClockTimer()
+        self.tput_timer = ThroughputTimer(
+            batch_size=self.train_batch_size,
+            steps_per_output=self.config.steps_per_print)
+        self.wall_clock_breakdown = self.config.wall_clock_breakdown
+
+        if self.config.comms_logger.enabled:
+            dist.configure_comms_logger(enabled=True,
+                                        verbose=self.config.comms_logger.verbose)
+
+        self.monitor = None
+        if self.config.monitor.enabled:
+            from ..monitor.monitor import MonitorMaster
+            self.monitor = MonitorMaster(self.config.monitor)
+
+        self.flops_profiler = None
+        if self.config.flops_profiler.enabled:
+            from ..profiling.flops_profiler import FlopsProfiler
+            self.flops_profiler = FlopsProfiler(self.module)
+
+        self.zero_stage = self.config.zero.stage
+        self.dtype = self.config.dtype
+
+        self._configure_distributed_model(dont_change_device)
+        self._configure_optimizer(model_parameters)
+        self._configure_lr_scheduler()
+
+        see_memory_usage("engine initialized", force=self.config.memory_breakdown)
+
+    # ------------------------------------------------------------------ setup
+
+    def _configure_distributed_model(self, dont_change_device):
+        if self.zero_stage < 3:
+            if self.dtype != torch.float32:
+                self.module.to(self.dtype)
+            if not dont_change_device:
+                self.module.to(self.device)
+            self._broadcast_model()
+        else:
+            # ZeRO-3: params may already be partitioned (zero.Init); the
+            # stage-3 optimizer handles placement and bcast.
+            if not dont_change_device:
+                self.module.to(self.device)
+
+    def _broadcast_model(self):
+        """Replicate rank-0 weights across the DP group (reference :1183)."""
+        if self.dp_world_size == 1:
+            return
+        for p in self.module.parameters():
+            if torch.is_tensor(p):
+                dist.broadcast(p.data, src=0, group=self.dp_group)
+        for b in self.module.buffers():
+            if torch.is_tensor(b) and b.numel() > 0 and b.dtype.is_floating_point:
+                dist.broadcast(b.data, src=0, group=self.dp_group)
+
+    def _configure_basic_optimizer(self, model_parameters):
+        cfg = self.config.optimizer
+        if cfg is None:
+            return None
+        params = dict(cfg.params)
+        name = cfg.type.lower()
+        if model_parameters is None:
+            model_parameters = [p for p in self.module.parameters() if p.requires_grad]
+        if name in ("adam", "adamw", "fusedadam"):
+            adam_w = params.pop("adam_w_mode", name != "adam")
+            if name == "adamw":
+                adam_w = True
+            from ..ops.adam import FusedAdam
+            return FusedAdam(model_parameters, adam_w_mode=adam_w, **params)
+        if name == "sgd":
+            return torch.optim.SGD(model_parameters, **params)
+        if name == "lion":
+            from ..ops.lion import Lion
+            return Lion(model_parameters, **params)
+        if name == "adagrad":
+            return torch.optim.Adagrad(model_parameters, **params)
+        raise ValueError(f"unsupported optimizer type {cfg.type}")
+
+    def _configure_optimizer(self, model_parameters):
+        basic = self.client_optimizer or self._configure_basic_optimizer(model_parameters)
+        self.basic_optimizer = basic
+        if basic is None:
+            self.optimizer = None
+            return
+
+        stage = self.zero_stage
+        mixed = self.dtype in (torch.float16, torch.bfloat16)
+        if stage == 3:
+            from .zero.stage3 import ZeroStage3Optimizer
+            self.optimizer = ZeroStage3Optimizer(
+                module=self.module,
+                init_optimizer=basic,
+                dp_group=self.dp_group,
+                config=self.config,
+                loss_scaler=self._make_loss_scaler(),
+            )
+        elif stage in (1, 2) or mixed:
+            eff_stage = stage if stage in (1, 2) else 1
+            offload = self.config.zero.offload_optimizer.device == "cpu"
+            self.optimizer = ZeroStage12Optimizer(
+                init_optimizer=basic,
+                stage=eff_stage,
+                dp_group=self.dp_group,
+                reduce_bucket_size=self.config.zero.reduce_bucket_size,
+                allgather_bucket_size=self.config.zero.allgather_bucket_size,
+                overlap_comm=self.config.zero.overlap_comm,
+                clip_grad=self.config.gradient_clipping,
+                loss_scaler=self._make_loss_scaler(),
+                communication_dtype=self._comm_dtype(),
+                gradient_predivide_factor=self.config.gradient_predivide_factor,
+                cpu_offload=offload,
+                offload_pin_memory=self.config.zero.offload_optimizer.pin_memory,
+                mpu=self.mpu,
+            )
+        else:
+            self.optimizer = basic  # fp32 stage 0: engine handles allreduce
+
+    def _make_loss_scaler(self):
+        if self.dtype == torch.float16:
+            return create_loss_scaler(self.config.fp16)
+        return LossScaler(1.0)
+
+    def _comm_dtype(self):
+        name = self.config.communication_data_type
+        if name is None:
+            return None
+        return {"fp32": torch.float32, "fp16": torch.float16,
+                "bf16": torch.bfloat16}[name]
+
+    def _configure_lr_scheduler(self):
+        if self.client_lr_scheduler is not None:
+            self.lr_scheduler = self.client_lr_scheduler
+            return
+        cfg = self.config.scheduler
+        if cfg is None or self.optimizer is None:
+            self.lr_scheduler = None
+            return
+        self.lr_scheduler = lr_schedules.get_scheduler(cfg.type, self.optimizer,
+                                                       cfg.params)
+
+    # -------------------------------------------------------------- accessors
+
+    def get_lr(self):
+        if self.optimizer is None:
+            return []
+        return [g["lr"] for g in self.optimizer.param_groups]
+
+    def get_global_grad_norm(self):
+        if hasattr(self.optimizer, "get_global_grad_norm"):
+            return self.optimizer.get_global_grad_norm()
+        return getattr(self, "_fallback_grad_norm", 0.0)
+
+    @property
+    def loss_scale(self):
+        if hasattr(self.optimizer, "loss_scale"):
+            return self.optimizer.loss_scale
+        return 1.0
+
+    def zero_optimization(self):
+        return self.zero_stage > 0
+
+    def is_gradient_accumulation_boundary(self) -> bool:
+        return self._is_gradient_accumulation_boundary
+
+    def set_gradient_accumulation_boundary(self, is_boundary: bool):
+        self._is_gradient_accumulation_boundary = is_boundary
+        if hasattr(self.optimizer, "is_gradient_accumulation_boundary"):
+            self.optimizer.is_gradient_accumulation_boundary = is_boundary
+
+    def train(self, mode=True):
+        self.module.train(mode)
+        return self
+
+    def eval(self):
+        self.module.eval()
+        return self
+
+    # ------------------------------------------------------------ hot path
+
+    def forward(self, *inputs, **kwargs):
+        if self.wall_clock_breakdown:
+            self.timers(FORWARD_GLOBAL_TIMER).start()
+        if self.flops_profiler is not None and \
+                self.global_steps == self.config.flops_profiler.profile_step and \
+                self.micro_steps % self.gradient_accumulation_steps == 0:
+            self.flops_profiler.start_profile(ignore_list=None)
+        loss = self.module(*inputs, **kwargs)
+        if self.flops_profiler is not None and \
+                self.global_steps == self.config.flops_profiler.profile_step and \
+                self.micro_steps % self.gradient_accumulation_steps == 0:
+            self.flops_profiler.stop_profile()
+            if self.global_rank == 0:
+                self.flops_profiler.print_model_profile(
+                    profile_step=self.global_steps,
+                    top_modules=self.config.flops_profiler.top_modules,
+                    detailed=self.config.flops_profiler.detailed,
+                    output_file=self.config.flops_profiler.output_file)
+            self.flops_profiler.end_profile()
+        if self.wall_clock_breakdown:
+            self.timers(FORWARD_GLOBAL_TIMER).stop()
+        return loss
+
+    def backward(self, loss, retain_graph=False, scale_wrt_gas=True):
+        boundary = (self.micro_steps + 1) % self.gradient_accumulation_steps == 0 \
+            and not self.inside_no_sync
+        self.set_gradient_accumulation_boundary(boundary)
+
+        if self.wall_clock_breakdown:
+            self.timers(BACKWARD_GLOBAL_TIMER).start()
+        if self.tput_timer:
+            self.tput_timer.start()
+
+        if self.gradient_accumulation_steps > 1 and scale_wrt_gas:
+            loss = loss / self.gradient_accumulation_steps
+        if self.seq_parallel_world_size > 1:
+            loss = loss / self.seq_parallel_world_size
+
+        if hasattr(self.optimizer, "backward"):
+            self.optimizer.backward(loss, retain_graph=retain_graph)
+        else:
+            loss.backward(retain_graph=retain_graph)
+
+        if boundary:
+            self.allreduce_gradients()
+
+        if self.wall_clock_breakdown:
+            self.timers(BACKWARD_GLOBAL_TIMER).stop()
+        self.micro_steps += 1
+        self.global_samples += self.train_micro_batch_size_per_gpu
+        return loss
+
+    def allreduce_gradients(self):
+        """At the GAS boundary: flush the ZeRO reducer, or run the bucketed
+        DDP-fallback all-reduce for stage-0 fp32 (reference :2611)."""
+        if hasattr(self.optimizer, "reduce_gradients"):
+            self.optimizer.reduce_gradients()
+        elif self.dp_world_size > 1:
+            self._buffered_allreduce_fallback()
+
+    @torch.no_grad()
+    def _buffered_allreduce_fallback(self, elements_per_buffer=MEMORY_OPT_ALLREDUCE_SIZE):
+        grads = [p.grad for p in self.module.parameters()
+                 if p.grad is not None]
+        bucket, bucket_elems = [], 0
+        from torch._utils import _flatten_dense_tensors, _unflatten_dense_tensors
+
+        def flush():
+            nonlocal bucket, bucket_elems
+            if not bucket:
+                return
+            flat = _flatten_dense_tensors(bucket)
+            flat.div_(self.dp_world_size)
+            dist.all_reduce(flat, group=self.dp_group)
+            for g, synced in zip(bucket, _unflatten_dense_tensors(flat, bucket)):
+                g.copy_(synced)
+            bucket, bucket_elems = [], 0
+
+        for g in grads:
+            bucket.append(g)
+            bucket_elems += g.numel()
+            if bucket_elems >= elements_per_buffer:
+                flush()
+        flush()
+
+    def step(self, lr_kwargs=None):
+        if not self._is_gradient_accumulation_boundary:
+            return
+        if self.wall_clock_breakdown:
+            self.timers(STEP_GLOBAL_TIMER).start()
+
+        if self.optimizer is not None and not isinstance(self.optimizer, DummyOptim):
+            if isinstance(self.optimizer, torch.optim.Optimizer) and \
+                    not hasattr(self.optimizer, "reduce_gradients"):
+                # plain fp32 stage-0 path: clip + step + zero
+                if self.config.gradient_clipping > 0:
+                    params = [p for p in self.module.parameters() if p.grad is not None]
+                    norm = get_global_norm_of_tensors(
+                        [p.grad for p in params], group=None)
+                    self._fallback_grad_norm = float(norm)
+                    clip_tensors_by_global_norm([p.grad for p in params],
+                                                self.config.gradient_clipping, norm)
+                self.optimizer.step()
+                self.optimizer.zero_grad()
+            else:
+                self.optimizer.step()
+                self.optimizer.zero_grad()
+
+        overflow = bool(getattr(self.optimizer, "overflow", False))
+        if overflow:
+            self.skipped_steps += 1
+        elif self.lr_scheduler is not None:
+            try:
+                self.lr_scheduler.step(**(lr_kwargs or {}))
+            except TypeError:
+                self.lr_scheduler.step()
+
+        self.global_steps += 1
+        if self.tput_timer:
+            self.tput_timer.stop(global_step=True)
+        if self.monitor is not None and self.global_rank == 0:
+            self.monitor.write_events([
+                ("Train/lr", self.get_lr()[0] if self.get_lr() else 0.0,
+                 self.global_steps)])
+        if self.wall_clock_breakdown:
+            self.timers(STEP_GLOBAL_TIMER).stop()
+            if self.global_steps % self.config.steps_per_print == 0:
+                self.timers.log([FORWARD_GLOBAL_TIMER, BACKWARD_GLOBAL_TIMER,
+                                 STEP_GLOBAL_TIMER])
+        self.set_gradient_accumulation_boundary(True)
+
+    class _NoSync:
+        def __init__(self, engine):
+            self.engine = engine
+
+        def __enter__(self):
+            self.engine.inside_no_sync = True
+
+        def __exit__(self, *a):
+            self.engine.inside_no_sync = False
+            return False
+
+    def no_sync(self):
+        """Context manager: disable gradient reduction inside (reference :2065)."""
+        return Engine._NoSync(self)
+
+    # ----------------------------------------------------------- data loader
+
+    def deepspeed_io(self, dataset, batch_size=None, num_workers=0, collate_fn=None):
+        from .dataloader import RepeatingLoader, build_dataloader
+        return build_dataloader(
+            dataset,
+            batch_size=batch_size or self.train_micro_batch_size_per_gpu,
+            world_size=self.dp_world_size,
+            rank=dist.get_rank(self.dp_group),
+            num_workers=num_workers,
+            collate_fn=collate_fn)
+
+    # ----------------------------------------------------------- checkpoints
+
+    def _ckpt_tag(self, tag):
+        return tag if tag is not None else f"global_step{self.global_steps}"
+
+    def _model_ckpt_name(self, dirname):
+        mp_rank = groups.get_tensor_parallel_rank()
+        return os.path.join(dirname, f"mp_rank_{mp_rank:02d}_model_states.pt")
+
+    def _zero_ckpt_name(self, dirname):
+        mp_rank = groups.get_tensor_parallel_rank()
+        dp_rank = dist.get_rank(self.dp_group)
+        return os.path.join(
+            dirname, f"zero_pp_rank_{dp_rank}_mp_rank_{mp_rank:02d}_optim_states.pt")
+
+    def save_checkpoint(self, save_dir, tag=None, client_state=None,
+                        save_latest=True, exclude_frozen_parameters=False):
+        """Reference layout: <dir>/<tag>/mp_rank_XX_model_states.pt +
+        zero_pp_rank_R_mp_rank_XX_optim_states.pt + <dir>/latest."""
+        tag = self._ckpt_tag(tag)
+        ckpt_dir = os.path.join(save_dir, tag)
+        os.makedirs(ckpt_dir, exist_ok=True)
+        dp_rank = dist.get_rank(self.dp_group)
+
+        if dp_rank == 0:
+            module_sd = self.module_state_dict(
+                exclude_frozen_parameters=exclude_frozen_parameters)
+            state = {
+                "module": module_sd,
+                "buffer_names": [n for n, _ in self.module.named_buffers()],
+                "lr_scheduler": (self.lr_scheduler.state_dict()
+                                 if self.lr_scheduler is not None and
+                                 hasattr(self.lr_scheduler, "state_dict") else None),
+                "global_steps": self.global_steps,
+                "global_samples": self.global_samples,
+                "skipped_steps": self.skipped_steps,
+                "dp_world_size": self.dp_world_size,
+                "ds_config": self.config.raw,
+                "client_state": client_state or {},
+            }
+            torch.save(state, self._model_ckpt_name(ckpt_dir))
+
+        if self.optimizer is not None and hasattr(self.optimizer, "state_dict") and \
+                not isinstance(self.optimizer, DummyOptim):
+            opt_state = {"optimizer_state_dict": self.optimizer.state_dict()}
+            torch.save(opt_state, self._zero_ckpt_name(ckpt_dir))
+
+        dist.barrier()
+        if save_latest and self.global_rank == 0:
+            with open(os.path.join(save_dir, "latest"), "w") as f:
+                f.write(tag)
+        dist.barrier()
+        return True
+
+    def module_state_dict(self, exclude_frozen_parameters=False):
+        if self.zero_stage == 3 and hasattr(self.optimizer, "full_state_dict"):
+            sd = self.optimizer.full_state_dict()
+        else:
+            sd = self.module.state_dict()
+        if exclude_frozen_parameters:
+            frozen = {n for n, p in self.module.named_parameters()
+                      if not p.requires_grad}
+            sd = {k: v for k, v in sd.items() if k not in frozen}
+        return sd
+
+    def load_checkpoint(self, load_dir, tag=None, load_module_strict=True,
+                        load_optimizer_states=True, load_lr_scheduler_states=True,
+                        load_module_only=False):
+        if tag is None:
+            latest = os.path.join(load_dir, "latest")
+            if not os.path.exists(latest):
+                logger.warning(f"no 'latest' file in {load_dir}")
+                return None, {}
+            with open(latest) as f:
+                tag = f.read().strip()
+        ckpt_dir = os.path.join(load_dir, tag)
+        model_path = self._model_ckpt_name(ckpt_dir)
+        state = torch.load(model_path, map_location="cpu", weights_only=False)
+
+        if self.zero_stage == 3 and hasattr(self.optimizer, "load_full_state_dict"):
+            self.optimizer.load_full_state_dict(state["module"],
+                                                strict=load_module_strict)
+        else:
+            self.module.load_state_dict(state["module"], strict=load_module_strict)
+        self.global_steps = state.get("global_steps", 0)
+        self.global_samples = state.get("global_samples", 0)
+        self.skipped_steps = state.get("skipped_steps", 0)
+
+        if not load_module_only:
+            if load_lr_scheduler_states and self.lr_scheduler is not None and \
+                    state.get("lr_scheduler") is not None:
+                self.lr_scheduler.load_state_dict(state["lr_scheduler"])
+            zero_path = self._zero_ckpt_name(ckpt_dir)
+            if os.path.exists(zero_path) and self.optimizer is not None and \
+                    hasattr(self.optimizer, "load_state_dict"):
+                opt_state = torch.load(zero_path, map_location="cpu",
+                                       weights_only=False)
+                if hasattr(self.optimizer, "loss_scaler"):
+                    self.optimizer.load_state_dict(
+                        opt_state["optimizer_state_dict"],
+                        load_optimizer_states=load_optimizer_states)
+                else:
+                    self.optimizer.load_state_dict(opt_state["optimizer_state_dict"])
+        # ZeRO-1/2 load re-broadcasts masters into the 16-bit params; for
+        # stage 0 re-broadcast from rank 0 for determinism.
+        if not hasattr(self.optimizer, "loss_scaler"):
+            self._broadcast_model()
+        return ckpt_dir, state.get("client_state", {})
+
+    def save_16bit_model(self, save_dir, save_filename="pytorch_model.bin",
+                         exclude_frozen_parameters=False):
+        sd = self.module_state_dict(exclude_frozen_parameters)
+        if self.global_rank == 0:
+            os.makedirs(save_dir, exist_ok=True)
+            torch.save(sd, os.path.join(save_dir, save_filename))
+        dist.barrier()
+        return True

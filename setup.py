@@ -1,0 +1,40 @@
+"""Build the in-tree HIP extension for gfx950 (MI355X).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The resulting deepspeed_amd/ops/_C*.so is git-ignored but ships to the GPU
+box with the repo snapshot.
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+CSRC = os.path.join("deepspeed_amd", "ops", "csrc")
+
+ext = CUDAExtension(
+    name="deepspeed_amd.ops._C",
+    sources=[
+        os.path.join(CSRC, "bindings.cpp"),
+        os.path.join(CSRC, "adam.hip"),
+        os.path.join(CSRC, "norms.hip"),
+        os.path.join(CSRC, "rope.hip"),
+        os.path.join(CSRC, "swiglu.hip"),
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+    },
+)
+
+setup(
+    name="deepspeed_amd",
+    version="0.1.0",
+    packages=["deepspeed_amd"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

@@ -1,0 +1,180 @@
+"""ZeRO-1/2 numerical parity vs plain torch mixed-precision training
+(reference contract: tests/unit/runtime/zero/test_zero.py stage 1/2)."""
+
+import copy
+
+import pytest
+import torch
+
+from .common import run_distributed, run_local
+
+
+class TinyNet(torch.nn.Module):
+    def __init__(self, hidden=32, nlayers=3, seed=7):
+        super().__init__()
+        torch.manual_seed(seed)
+        self.layers = torch.nn.ModuleList(
+            [torch.nn.Linear(hidden, hidden) for _ in range(nlayers)])
+        self.out = torch.nn.Linear(hidden, 1)
+
+    def forward(self, x, labels=None):
+        for l in self.layers:
+            x = torch.tanh(l(x))
+        y = self.out(x)
+        if labels is not None:
+            return torch.nn.functional.mse_loss(y.float(), labels.float())
+        return y
+
+
+def _reference_mixed_precision_loop(model_fp32, data, lr, steps, gas=1,
+                                    dtype=torch.bfloat16, clip=0.0,
+                                    world=1):
+    """fp32 master AdamW with low-precision fwd/bwd — the semantics ZeRO-1/2
+    must reproduce (modulo reduction order)."""
+    master = copy.deepcopy(model_fp32).float()
+    work = copy.deepcopy(model_fp32).to(dtype)
+    opt = torch.optim.AdamW(master.parameters(), lr=lr)
+    losses = []
+    it = iter(data)
+    for _ in range(steps):
+        for p in work.parameters():
+            p.grad = None
+        micro_losses = []
+        for _ in range(gas):
+            xs, ys = next(it)
+            loss = work(xs.to(dtype), labels=ys) / gas
+            loss.backward()
+            micro_losses.append(loss.item() * gas)
+        losses.append(sum(micro_losses) / len(micro_losses))
+        # emulate DP averaging over identical data => grads identical
+        with torch.no_grad():
+            grads = []
+            for pw, pm in zip(work.parameters(), master.parameters()):
+                g = pw.grad.float()
+                pm.grad = g.clone()
+                grads.append(pm.grad)
+            if clip > 0:
+                norm = torch.norm(torch.stack([g.norm() for g in grads]))
+                coef = min(1.0, clip / (norm.item() + 1e-6))
+                for g in grads:
+                    g.mul_(coef)
+        opt.step()
+        with torch.no_grad():
+            for pw, pm in zip(work.parameters(), master.parameters()):
+                pw.copy_(pm.to(dtype))
+    return losses, master
+
+
+def _make_data(n, hidden=32, bs=4, seed=3):
+    torch.manual_seed(seed)
+    return [(torch.randn(bs, hidden), torch.randn(bs, 1)) for _ in range(n)]
+
+
+def _zero_worker(rank, world, stage, gas, clip, dtype_name):
+    import deepspeed_amd
+    dtype = {"bf16": torch.bfloat16, "fp32": torch.float32}[dtype_name]
+    lr, steps = 1e-2, 5
+    model = TinyNet()
+    ref_model = copy.deepcopy(model)
+    data = _make_data(steps * gas)
+
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "gradient_accumulation_steps": gas,
+        "gradient_clipping": clip,
+        "zero_optimization": {"stage": stage, "reduce_bucket_size": 500,
+                              "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": lr}},
+    }
+    if dtype == torch.bfloat16:
+        config["bf16"] = {"enabled": True}
+    engine, opt, _, _ = deepspeed_amd.initialize(model=model, config=config)
+
+    it = iter(data)
+    engine_losses = []
+    for _ in range(steps):
+        micro = []
+        for _ in range(gas):
+            xs, ys = next(it)
+            xs = xs.to(engine.device).to(dtype)
+            ys = ys.to(engine.device)
+            loss = engine(xs, labels=ys)
+            engine.backward(loss)
+            micro.append(loss.item())
+        engine.step()
+        engine_losses.append(sum(micro) / len(micro))
+
+    if dtype == torch.float32:
+        # fp32 still goes through ZeRO-1 fp32 master path when stage>=1
+        ref_losses, ref_master = _reference_mixed_precision_loop(
+            ref_model, data, lr, steps, gas, torch.float32, clip)
+    else:
+        ref_losses, ref_master = _reference_mixed_precision_loop(
+            ref_model, data, lr, steps, gas, dtype, clip)
+
+    for a, b in zip(engine_losses, ref_losses):
+        assert abs(a - b) < 2e-2, f"loss diverged: {engine_losses} vs {ref_losses}"
+    # weight parity (bf16 tolerance)
+    for p_engine, p_ref in zip(engine.module.parameters(),
+                               ref_master.parameters()):
+        assert torch.allclose(p_engine.float().cpu(), p_ref.to(p_engine.dtype).float(),
+                              atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("stage", [1, 2])
+def test_zero_stage_bf16_parity_ws1(stage):
+    run_local(_zero_worker, args=(stage, 1, 0.0, "bf16"))
+
+
+@pytest.mark.parametrize("stage", [1, 2])
+def test_zero_stage_bf16_parity_ws2(stage):
+    run_distributed(_zero_worker, world_size=2, args=(stage, 1, 0.0, "bf16"))
+
+
+def test_zero_gas_parity_ws2():
+    run_distributed(_zero_worker, world_size=2, args=(2, 3, 0.0, "bf16"))
+
+
+def test_zero_clip_parity_ws2():
+    run_distributed(_zero_worker, world_size=2, args=(2, 1, 0.5, "bf16"))
+
+
+def test_zero_fp32_parity_ws2():
+    run_distributed(_zero_worker, world_size=2, args=(1, 1, 0.0, "fp32"))
+
+
+def _different_data_worker(rank, world):
+    """With different per-rank data, ZeRO grads = average over ranks."""
+    import deepspeed_amd
+    lr, steps = 1e-2, 3
+    model = TinyNet()
+    ref_model = copy.deepcopy(model)
+    all_data = [_make_data(steps, seed=100 + r) for r in range(world)]
+
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "zero_optimization": {"stage": 2, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": lr}},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    for i in range(steps):
+        xs, ys = all_data[rank][i]
+        loss = engine(xs.float(), labels=ys)
+        engine.backward(loss)
+        engine.step()
+
+    # reference: single-process with grads averaged over both ranks' batches
+    master = copy.deepcopy(ref_model).float()
+    opt = torch.optim.AdamW(master.parameters(), lr=lr)
+    for i in range(steps):
+        opt.zero_grad()
+        loss = sum(master(all_data[r][i][0], labels=all_data[r][i][1])
+                   for r in range(world)) / world
+        loss.backward()
+        opt.step()
+    for p_engine, p_ref in zip(engine.module.parameters(), master.parameters()):
+        assert torch.allclose(p_engine.float().cpu(), p_ref, atol=1e-4, rtol=1e-3)
+
+
+def test_zero_dp_grad_averaging_ws2():
+    run_distributed(_different_data_worker, world_size=2)
