@@ -1,0 +1,79 @@
+"""GPU end-to-end engine tests: a Llama-mini model steps through the engine
+on an MI355X with the HIP extension loaded (no eager fallback allowed).
+
+Reference contract: tests/unit/runtime/zero/test_zero.py (stage 1/2 training
+loop on GPU) in microsoft/DeepSpeed.
+"""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _init_env():
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29531")
+    os.environ.setdefault("RANK", "0")
+    os.environ.setdefault("LOCAL_RANK", "0")
+    os.environ.setdefault("WORLD_SIZE", "1")
+
+
+def _build_engine(stage, tmp=None, overlap=True):
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaConfig, LlamaForCausalLM
+    torch.manual_seed(11)
+    cfg = LlamaConfig(vocab_size=1024, hidden_size=256, intermediate_size=512,
+                      num_layers=2, num_heads=4, num_kv_heads=2,
+                      max_seq_len=256)
+    model = LlamaForCausalLM(cfg)
+    engine, opt, _, _ = deepspeed_amd.initialize(
+        model=model,
+        config={
+            "train_micro_batch_size_per_gpu": 2,
+            "gradient_accumulation_steps": 1,
+            "bf16": {"enabled": True},
+            "gradient_clipping": 1.0,
+            "zero_optimization": {"stage": stage, "overlap_comm": overlap},
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        })
+    return engine, cfg
+
+
+@pytest.mark.parametrize("stage", [1, 2])
+def test_llama_mini_train_step(stage):
+    from deepspeed_amd.ops import has_ext
+    assert has_ext()
+    _init_env()
+    engine, cfg = _build_engine(stage)
+    torch.manual_seed(0)
+    losses = []
+    for _ in range(5):
+        ids = torch.randint(0, cfg.vocab_size, (2, 128), device="cuda:0")
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses)))
+    # random-label LM loss should move from its init value under training
+    assert losses[-1] < losses[0] + 0.5
+
+
+def test_checkpoint_save_load_gpu(tmp_path):
+    _init_env()
+    engine, cfg = _build_engine(2)
+    torch.manual_seed(1)
+    for _ in range(3):
+        ids = torch.randint(0, cfg.vocab_size, (2, 64), device="cuda:0")
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+    engine.save_checkpoint(str(tmp_path), tag="t3")
+    w0 = {k: v.clone() for k, v in engine.module.state_dict().items()}
+
+    engine2, _ = _build_engine(2)
+    engine2.load_checkpoint(str(tmp_path), tag="t3")
+    for k, v in engine2.module.state_dict().items():
+        torch.testing.assert_close(v, w0[k], rtol=0, atol=0)
