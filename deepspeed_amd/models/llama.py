@@ -183,26 +183,27 @@ class LlamaForCausalLM(nn.Module):
         hidden = self.model(input_ids, positions, kv_cache)
         if labels is not None:
             # chunked fp32 cross-entropy: avoids materializing the full
-            # [B*S, vocab] fp32 logits (vocab=128256 -> ~2 GB per 4k tokens)
-            return chunked_cross_entropy(hidden, self.lm_head.weight, labels)
+            # [B*S, vocab] fp32 logits (vocab=128256 -> ~2 GB per 4k tokens).
+            # Routed through the lm_head MODULE so ZeRO-3 fetch hooks fire.
+            return chunked_cross_entropy(hidden, self.lm_head, labels)
         return self.lm_head(hidden)
 
     def num_parameters(self):
         return sum(p.numel() for p in self.parameters())
 
 
-def chunked_cross_entropy(hidden, lm_weight, labels, chunk_tokens=8192):
-    """loss = CE(hidden @ W^T, labels), computed in token chunks so the fp32
-    logits never exceed chunk_tokens x vocab."""
+def chunked_cross_entropy(hidden, lm_head, labels, chunk_tokens=8192):
+    """loss = CE(lm_head(hidden), labels), computed in token chunks so the
+    fp32 logits never exceed chunk_tokens x vocab."""
     B, S, H = hidden.shape
-    hidden = hidden.view(-1, H)
-    labels = labels.view(-1)
+    hidden = hidden.reshape(-1, H)
+    labels = labels.reshape(-1)
     n = hidden.shape[0]
     total = hidden.new_zeros((), dtype=torch.float32)
     count = hidden.new_zeros((), dtype=torch.float32)
     for s in range(0, n, chunk_tokens):
         e = min(s + chunk_tokens, n)
-        logits = F.linear(hidden[s:e], lm_weight).float()
+        logits = lm_head(hidden[s:e]).float()
         tgt = labels[s:e]
         total = total + F.cross_entropy(logits, tgt, ignore_index=-100,
                                         reduction="sum")
