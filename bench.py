@@ -26,7 +26,7 @@ def parse_args():
     p.add_argument("--micro-batch", type=int, default=2)
     p.add_argument("--gas", type=int, default=1)
     p.add_argument("--zero-stage", type=int,
-                   default=int(os.environ.get("BENCH_ZERO_STAGE", 2)))
+                   default=int(os.environ.get("BENCH_ZERO_STAGE", 3)))
     p.add_argument("--model", type=str, default="llama3-8b",
                    choices=["llama3-8b", "llama3-70b", "llama-mini", "tiny"])
     p.add_argument("--offload", action="store_true",

@@ -42,7 +42,7 @@ def _build_engine(stage, tmp=None, overlap=True):
     return engine, cfg
 
 
-@pytest.mark.parametrize("stage", [1, 2])
+@pytest.mark.parametrize("stage", [1, 2, 3])
 def test_llama_mini_train_step(stage):
     from deepspeed_amd.ops import has_ext
     assert has_ext()

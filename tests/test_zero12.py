@@ -114,25 +114,35 @@ def _zero_worker(rank, world, stage, gas, clip, dtype_name):
 
     for a, b in zip(engine_losses, ref_losses):
         assert abs(a - b) < 2e-2, f"loss diverged: {engine_losses} vs {ref_losses}"
-    # weight parity (bf16 tolerance)
-    for p_engine, p_ref in zip(engine.module.parameters(),
-                               ref_master.parameters()):
-        assert torch.allclose(p_engine.float().cpu(), p_ref.to(p_engine.dtype).float(),
-                              atol=3e-2, rtol=3e-2)
+    # weight parity (bf16 tolerance); stage 3 params are partitioned, so
+    # compare the rank-0 consolidated state dict instead
+    if stage == 3:
+        sd = engine.optimizer.get_full_state_dict()
+        if rank == 0:
+            ref_sd = ref_master.state_dict()
+            for name, t in sd.items():
+                assert torch.allclose(t.float().cpu(), ref_sd[name].to(t.dtype).float(),
+                                      atol=3e-2, rtol=3e-2), f"param {name} diverged"
+    else:
+        for p_engine, p_ref in zip(engine.module.parameters(),
+                                   ref_master.parameters()):
+            assert torch.allclose(p_engine.float().cpu(), p_ref.to(p_engine.dtype).float(),
+                                  atol=3e-2, rtol=3e-2)
 
 
-@pytest.mark.parametrize("stage", [1, 2])
+@pytest.mark.parametrize("stage", [1, 2, 3])
 def test_zero_stage_bf16_parity_ws1(stage):
     run_local(_zero_worker, args=(stage, 1, 0.0, "bf16"))
 
 
-@pytest.mark.parametrize("stage", [1, 2])
+@pytest.mark.parametrize("stage", [1, 2, 3])
 def test_zero_stage_bf16_parity_ws2(stage):
     run_distributed(_zero_worker, world_size=2, args=(stage, 1, 0.0, "bf16"))
 
 
-def test_zero_gas_parity_ws2():
-    run_distributed(_zero_worker, world_size=2, args=(2, 3, 0.0, "bf16"))
+@pytest.mark.parametrize("stage", [2, 3])
+def test_zero_gas_parity_ws2(stage):
+    run_distributed(_zero_worker, world_size=2, args=(stage, 3, 0.0, "bf16"))
 
 
 def test_zero_clip_parity_ws2():
