@@ -77,6 +77,9 @@ def main():
                                  "weight_decay": 0.1}},
     }
 
+    if args.activation_checkpointing:
+        model.model.gradient_checkpointing_enable()
+
     engine, _, _, _ = deepspeed_amd.initialize(model=model, config=ds_config)
     device = engine.device
 

@@ -150,14 +150,26 @@ class LlamaModel(nn.Module):
         self.layers = nn.ModuleList(
             [LlamaDecoderLayer(cfg, i) for i in range(cfg.num_layers)])
         self.norm = RMSNorm(cfg.hidden_size, eps=cfg.rms_eps)
+        self.gradient_checkpointing = False
 
     def forward(self, input_ids, positions=None, kv_cache=None):
         x = self.embed_tokens(input_ids)
         cos, sin = rope_tables(self.cfg.head_dim, self.cfg.max_seq_len,
                                self.cfg.rope_theta, device=x.device)
+        recompute = (self.gradient_checkpointing and self.training
+                     and torch.is_grad_enabled())
+        if recompute:
+            from ..runtime.activation_checkpointing import checkpoint
         for layer in self.layers:
-            x = layer(x, cos, sin, positions, kv_cache)
+            if recompute:
+                x = checkpoint(lambda x_, l=layer: l(x_, cos, sin, positions,
+                                                     kv_cache), x)
+            else:
+                x = layer(x, cos, sin, positions, kv_cache)
         return self.norm(x)
+
+    def gradient_checkpointing_enable(self):
+        self.gradient_checkpointing = True
 
 
 class LlamaForCausalLM(nn.Module):

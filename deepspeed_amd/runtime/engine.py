@@ -92,6 +92,9 @@ class Engine(torch.nn.Module):
             from ..monitor.monitor import MonitorMaster
             self.monitor = MonitorMaster(self.config.monitor)
 
+        from .activation_checkpointing import configure as _ac_configure
+        _ac_configure(self.config.activation_checkpointing, mpu=self.mpu)
+
         self.flops_profiler = None
         if self.config.flops_profiler.enabled:
             from ..profiling.flops_profiler import FlopsProfiler
