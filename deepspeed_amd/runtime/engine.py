@@ -103,6 +103,18 @@ class Engine(torch.nn.Module):
         self.zero_stage = self.config.zero.stage
         self.dtype = self.config.dtype
 
+        # MoE: create EP / expert-DP groups before optimizer partitioning
+        from ..moe.layer import MoE, has_moe_layers
+        self.has_moe_layers = has_moe_layers(self.module)
+        if self.has_moe_layers:
+            for m in self.module.modules():
+                if isinstance(m, MoE):
+                    m.set_deepspeed_parallelism()
+            if self.zero_stage == 3:
+                raise ValueError("MoE expert parallelism composes with ZeRO "
+                                 "stages 0-2 (expert params are partitioned "
+                                 "over the expert-DP group, not ZeRO-3)")
+
         self._configure_distributed_model(dont_change_device)
         self._configure_optimizer(model_parameters)
         self._configure_lr_scheduler()
@@ -125,11 +137,21 @@ class Engine(torch.nn.Module):
                 self.module.to(self.device)
 
     def _broadcast_model(self):
-        """Replicate rank-0 weights across the DP group (reference :1183)."""
+        """Replicate rank-0 weights across the DP group; expert params are
+        broadcast over their expert-DP group instead, since each EP rank owns
+        different experts (reference :1183)."""
         if self.dp_world_size == 1:
             return
+        from ..parallel import groups as pgroups
         for p in self.module.parameters():
-            if torch.is_tensor(p):
+            if not torch.is_tensor(p):
+                continue
+            if getattr(p, "allreduce", True) is False:
+                g = pgroups.get_expert_data_parallel_group(p.group_name)
+                if dist.get_world_size(g) > 1:
+                    src = torch.distributed.get_global_rank(g, 0)
+                    dist.broadcast(p.data, src=src, group=g)
+            else:
                 dist.broadcast(p.data, src=0, group=self.dp_group)
         for b in self.module.buffers():
             if torch.is_tensor(b) and b.numel() > 0 and b.dtype.is_floating_point:
@@ -143,6 +165,14 @@ class Engine(torch.nn.Module):
         name = cfg.type.lower()
         if model_parameters is None:
             model_parameters = [p for p in self.module.parameters() if p.requires_grad]
+        if self.has_moe_layers:
+            from ..moe.layer import \
+                split_params_into_different_moe_groups_for_optimizer
+            if isinstance(model_parameters, list) and model_parameters and \
+                    not isinstance(model_parameters[0], dict):
+                model_parameters = [{"params": model_parameters}]
+            model_parameters = \
+                split_params_into_different_moe_groups_for_optimizer(model_parameters)
         if name in ("adam", "adamw", "fusedadam"):
             adam_w = params.pop("adam_w_mode", name != "adam")
             if name == "adamw":

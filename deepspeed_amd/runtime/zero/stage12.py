@@ -43,7 +43,7 @@ from ..utils import ALIGNMENT, get_global_norm_of_tensors
 class _Bucket:
     __slots__ = ("group_idx", "index", "flat", "grad_flat", "params", "offsets",
                  "numel", "shard_size", "pending", "reduced", "comm_event",
-                 "master_offset")
+                 "master_offset", "pg", "pg_rank", "pg_world")
 
     def __init__(self, group_idx, index, numel, shard_size):
         self.group_idx = group_idx
@@ -58,6 +58,9 @@ class _Bucket:
         self.reduced = False
         self.comm_event = None
         self.master_offset = 0  # offset of this bucket's shard in the group master
+        self.pg = None          # process group this bucket reduces over
+        self.pg_rank = 0
+        self.pg_world = 1
 
 
 class ZeroStage12Optimizer:
@@ -128,6 +131,16 @@ class ZeroStage12Optimizer:
 
     # ------------------------------------------------------------------ setup
 
+    def _group_pg(self, group):
+        """Process group a param group partitions/reduces over: the expert
+        data-parallel group for MoE expert groups (tagged by the engine via
+        split_params_into_different_moe_groups_for_optimizer), else the DP
+        group (reference stage_1_and_2.py real_dp_process_group)."""
+        if group.get("moe"):
+            from ...parallel import groups as pgroups
+            return pgroups.get_expert_data_parallel_group(group["name"])
+        return self.dp_group
+
     def _build_flat_buffers(self):
         for gi, group in enumerate(self.optimizer.param_groups):
             params = [p for p in group["params"] if p.requires_grad]
@@ -138,6 +151,9 @@ class ZeroStage12Optimizer:
                 self.group_shard_numel.append(0)
                 continue
             self.group_params.append(params)
+            pg = self._group_pg(group)
+            pg_world = dist.get_world_size(pg)
+            pg_rank = dist.get_rank(pg)
             device = params[0].device
             dtype = params[0].dtype
             # Reverse order: backward produces grads for the last-constructed
@@ -145,15 +161,16 @@ class ZeroStage12Optimizer:
             ordered = list(reversed(params))
             group_buckets: List[_Bucket] = []
             cur_params, cur_offsets, cur_numel = [], [], 0
-            align = ALIGNMENT * self.world_size
+            align = ALIGNMENT * pg_world
 
             def close_bucket():
                 nonlocal cur_params, cur_offsets, cur_numel
                 if not cur_params:
                     return
                 padded = math.ceil(cur_numel / align) * align
-                b = _Bucket(gi, len(group_buckets), padded, padded // self.world_size)
+                b = _Bucket(gi, len(group_buckets), padded, padded // pg_world)
                 b.params, b.offsets = cur_params, cur_offsets
+                b.pg, b.pg_rank, b.pg_world = pg, pg_rank, pg_world
                 group_buckets.append(b)
                 cur_params, cur_offsets, cur_numel = [], [], 0
 
@@ -186,7 +203,7 @@ class ZeroStage12Optimizer:
                 master = master.pin_memory()
                 owned = owned.pin_memory()
             for b in group_buckets:
-                src = b.flat[self.rank * b.shard_size:(self.rank + 1) * b.shard_size]
+                src = b.flat[b.pg_rank * b.shard_size:(b.pg_rank + 1) * b.shard_size]
                 master[b.master_offset:b.master_offset + b.shard_size].copy_(
                     src.float() if not self.cpu_offload else src.float().cpu())
             self.group_masters.append(master)
@@ -256,14 +273,14 @@ class ZeroStage12Optimizer:
 
         def _issue():
             if self.stage == 1:
-                h = dist.all_reduce(grad, group=self.dp_group, async_op=True)
-                my = grad[self.rank * bucket.shard_size:
-                          (self.rank + 1) * bucket.shard_size]
+                h = dist.all_reduce(grad, group=bucket.pg, async_op=True)
+                my = grad[bucket.pg_rank * bucket.shard_size:
+                          (bucket.pg_rank + 1) * bucket.shard_size]
                 self._inflight.append((h, bucket, my, shard_dst))
             else:
                 recv = torch.empty(bucket.shard_size, dtype=grad.dtype,
                                    device=grad.device)
-                h = dist.reduce_scatter_tensor(recv, grad, group=self.dp_group,
+                h = dist.reduce_scatter_tensor(recv, grad, group=bucket.pg,
                                                async_op=True)
                 self._inflight.append((h, bucket, recv, shard_dst))
 
@@ -382,8 +399,8 @@ class ZeroStage12Optimizer:
                     continue
                 out16 = None
                 if b.flat.dtype == torch.bfloat16:
-                    out16 = b.flat[self.rank * b.shard_size:
-                                   (self.rank + 1) * b.shard_size]
+                    out16 = b.flat[b.pg_rank * b.shard_size:
+                                   (b.pg_rank + 1) * b.shard_size]
                 else:
                     wrote_params = False
                 segments.append((b.master_offset, b.shard_size, out16))
@@ -411,17 +428,17 @@ class ZeroStage12Optimizer:
         for b in self.buckets:
             master = self.group_masters[b.group_idx]
             src = master.data[b.master_offset:b.master_offset + b.shard_size]
-            dst = b.flat[self.rank * b.shard_size:(self.rank + 1) * b.shard_size]
+            dst = b.flat[b.pg_rank * b.shard_size:(b.pg_rank + 1) * b.shard_size]
             dst.copy_(src, non_blocking=self.cpu_offload)
 
     def _allgather_params(self):
-        if self.world_size == 1:
-            return
         handles = []
         for b in self.buckets:
-            shard = b.flat[self.rank * b.shard_size:(self.rank + 1) * b.shard_size]
+            if b.pg_world == 1:
+                continue
+            shard = b.flat[b.pg_rank * b.shard_size:(b.pg_rank + 1) * b.shard_size]
             h = dist.all_gather_into_tensor(b.flat, shard.contiguous(),
-                                            group=self.dp_group, async_op=True)
+                                            group=b.pg, async_op=True)
             handles.append(h)
         for h in handles:
             if h is not None:
@@ -498,8 +515,11 @@ class ZeroStage12Optimizer:
                 shard = master.data[b.master_offset:b.master_offset + b.shard_size]
                 shard = shard.to(b.flat.device)
                 full = torch.empty(b.numel, dtype=torch.float32, device=b.flat.device)
-                dist.all_gather_into_tensor(full, shard.contiguous(),
-                                            group=self.dp_group)
+                if b.pg_world > 1:
+                    dist.all_gather_into_tensor(full, shard.contiguous(),
+                                                group=b.pg)
+                else:
+                    full.copy_(shard)
                 bucket_fulls[b.index] = full
             full_by_group.append(bucket_fulls)
         if self.rank != 0:

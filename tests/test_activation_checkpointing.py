@@ -58,8 +58,7 @@ def test_checkpoint_rng_replay():
     torch.testing.assert_close(x_grad, x.grad)
 
 
-def test_checkpoint_with_zero2_engine():
-    """Engine-level: checkpointed training matches non-checkpointed weights."""
+def _ckpt_zero2_worker(rank, world):
     import deepspeed_amd
 
     def make():
@@ -87,3 +86,12 @@ def test_checkpoint_with_zero2_engine():
         results.append([p.detach().clone() for p in engine.module.parameters()])
     for a, b in zip(results[0], results[1]):
         torch.testing.assert_close(a, b, rtol=1e-5, atol=1e-6)
+
+
+def test_checkpoint_with_zero2_engine():
+    """Engine-level: checkpointed training matches non-checkpointed weights.
+
+    Runs in a child process (run_local): initializing gloo in the pytest main
+    process would deadlock later fork-based multi-process tests."""
+    from .common import run_local
+    run_local(_ckpt_zero2_worker)

@@ -1,0 +1,179 @@
+"""MoE / expert-parallel tests (reference contract: tests/unit/moe/test_moe.py).
+
+Covers: top-k gating invariants, MOELayer parity with dense expert
+computation, EP=2 cross-rank parity vs local EP=1, and a full Mixtral
+train-step through the engine with ZeRO-2 + expert-DP gradient reduction.
+"""
+
+import copy
+
+import pytest
+import torch
+
+from .common import run_distributed, run_local
+
+
+def _seed_experts(moe_layer, ep_rank, num_local):
+    """Deterministic per-GLOBAL-expert weights so EP layouts are comparable."""
+    for j, e in enumerate(moe_layer.deepspeed_moe.experts.local_experts):
+        g = torch.Generator().manual_seed(1000 + ep_rank * num_local + j)
+        for p in e.parameters():
+            with torch.no_grad():
+                p.copy_(torch.randn(p.shape, generator=g) * 0.05)
+
+
+# --------------------------------------------------------------------- gating
+
+def test_topk_gating_invariants():
+    from deepspeed_amd.moe.sharded_moe import topkgating
+    torch.manual_seed(0)
+    T, E, k = 64, 4, 2
+    logits = torch.randn(T, E)
+    l_aux, route = topkgating(logits, k=k, capacity_factor=1.0, min_capacity=4)
+    C = route["capacity"]
+    assert C == max(int(k * T / E), 4)
+    # no expert slot used twice
+    assert route["dest"].unique().numel() == route["dest"].numel()
+    assert (route["dest"] < E * C).all()
+    # kept gates per token sum to ~1 when no route of the token was dropped
+    gates = torch.zeros(T).index_add_(0, route["token"], route["gate"])
+    full = torch.bincount(route["token"], minlength=T) == k
+    assert torch.allclose(gates[full], torch.ones(full.sum()), atol=1e-5)
+    assert l_aux.item() > 0
+
+
+def test_topk_gating_drops_overflow():
+    from deepspeed_amd.moe.sharded_moe import topkgating
+    torch.manual_seed(0)
+    T, E = 32, 4
+    logits = torch.zeros(T, E)
+    logits[:, 0] = 10.0  # all tokens want expert 0
+    _, route = topkgating(logits, k=1, capacity_factor=1.0, min_capacity=4)
+    C = route["capacity"]
+    assert route["token"].numel() == C  # only capacity tokens kept
+    # no drops when drop_tokens=False (capacity grows)
+    _, route2 = topkgating(logits, k=1, capacity_factor=1.0, min_capacity=4,
+                           drop_tokens=False)
+    assert route2["token"].numel() == T
+
+
+# ------------------------------------------------------------ local MOELayer
+
+def _moe_local_worker(rank, world):
+    from deepspeed_amd.moe import MoE
+    from deepspeed_amd.models.llama import LlamaMLP, LlamaConfig
+    torch.manual_seed(7)
+    h = 32
+    lc = LlamaConfig(hidden_size=h, intermediate_size=64)
+    moe = MoE(hidden_size=h, expert=LlamaMLP(lc), num_experts=1, ep_size=1,
+              k=1, capacity_factor=8.0)
+    _seed_experts(moe, 0, 1)
+    x = torch.randn(2, 8, h, requires_grad=True)
+    out, l_aux, counts = moe(x)
+    # single expert, huge capacity: every token routed, gate weight == 1
+    ref = moe.deepspeed_moe.experts.local_experts[0](x)
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+    out.sum().backward()
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+
+
+def test_moe_single_expert_identity():
+    run_local(_moe_local_worker)
+
+
+# ------------------------------------------------------- EP=2 vs EP=1 parity
+
+def _moe_ep_worker(rank, world):
+    from deepspeed_amd.moe import MoE
+    from deepspeed_amd.models.llama import LlamaMLP, LlamaConfig
+    h, E = 32, 4
+    lc = LlamaConfig(hidden_size=h, intermediate_size=64)
+
+    def build(ep):
+        torch.manual_seed(21)  # same gate weights everywhere
+        return MoE(hidden_size=h, expert=LlamaMLP(lc), num_experts=E,
+                   ep_size=ep, k=2, capacity_factor=4.0)
+
+    moe_ep = build(2)
+    _seed_experts(moe_ep, rank, E // 2)
+
+    moe_local = build(1)
+    for j in range(E):
+        g = torch.Generator().manual_seed(1000 + j)
+        for p in moe_local.deepspeed_moe.experts.local_experts[j].parameters():
+            with torch.no_grad():
+                p.copy_(torch.randn(p.shape, generator=g) * 0.05)
+
+    torch.manual_seed(100 + rank)  # different tokens per rank
+    x = torch.randn(2, 8, h)
+    out_ep, _, _ = moe_ep(x)
+    out_local, _, _ = moe_local(x)
+    torch.testing.assert_close(out_ep, out_local, rtol=1e-4, atol=1e-5)
+
+
+def test_moe_ep2_matches_local():
+    run_distributed(_moe_ep_worker, world_size=2)
+
+
+# ------------------------------------------------------ engine + ZeRO-2 + EP
+
+def _mixtral_train_worker(rank, world, stage):
+    import deepspeed_amd
+    from deepspeed_amd import comm as dist
+    from deepspeed_amd.models import MixtralForCausalLM, mixtral_tiny
+
+    torch.manual_seed(17)
+    cfg = mixtral_tiny(ep_size=2, num_experts=4)
+    model = MixtralForCausalLM(cfg)
+    config = {
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": stage, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+    }
+    engine, opt, _, _ = deepspeed_amd.initialize(model=model, config=config)
+
+    # expert param groups must exist and reduce over the expert-DP group
+    moe_groups = [g for g in opt.optimizer.param_groups if g.get("moe")]
+    assert len(moe_groups) == 1
+    assert all(getattr(p, "allreduce", True) is False
+               for p in engine.module.parameters()
+               if getattr(p, "group_name", None))
+
+    torch.manual_seed(50 + rank)
+    losses = []
+    for _ in range(3):
+        ids = torch.randint(0, cfg.vocab_size, (2, 32))
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses)))
+
+    # dense params must be identical across ranks after the step
+    for p in engine.module.parameters():
+        if getattr(p, "allreduce", True) is False:
+            continue
+        ref = p.data.clone()
+        dist.broadcast(ref, src=0)
+        assert torch.equal(ref, p.data), "dense param diverged across ranks"
+
+
+@pytest.mark.parametrize("stage", [1, 2])
+def test_mixtral_ep2_zero_train(stage):
+    run_distributed(_mixtral_train_worker, world_size=2, args=(stage,))
+
+
+def test_split_params_into_moe_groups():
+    from deepspeed_amd.moe import \
+        split_params_into_different_moe_groups_for_optimizer
+
+    p1 = torch.nn.Parameter(torch.zeros(2))
+    p2 = torch.nn.Parameter(torch.zeros(2))
+    p2.allreduce = False
+    p2.group_name = "ep_size_2"
+    groups = split_params_into_different_moe_groups_for_optimizer(
+        {"params": [p1, p2], "lr": 0.1})
+    assert len(groups) == 2
+    assert groups[0]["params"] == [p1] and not groups[0].get("moe")
+    assert groups[1]["params"] == [p2] and groups[1]["moe"]
+    assert groups[1]["name"] == "ep_size_2" and groups[1]["lr"] == 0.1
