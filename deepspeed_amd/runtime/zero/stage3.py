@@ -52,8 +52,8 @@ FREE, INFLIGHT, AVAILABLE = 0, 1, 2
 
 class _Unit:
     __slots__ = ("index", "name", "module", "params", "offsets", "numel",
-                 "shard_size", "shard", "full", "grad_full", "status",
-                 "handle", "persist", "group_idx", "master_offset",
+                 "shard_size", "shard", "full", "grad_full", "grad_seen",
+                 "status", "handle", "persist", "group_idx", "master_offset",
                  "pending_grads", "trainable", "release_pending",
                  "in_backward")
 
@@ -68,6 +68,7 @@ class _Unit:
         self.shard: Optional[torch.Tensor] = None
         self.full: Optional[torch.Tensor] = None
         self.grad_full: Optional[torch.Tensor] = None
+        self.grad_seen: Optional[List[bool]] = None
         self.status = FREE
         self.handle = None
         self.persist = False
@@ -208,8 +209,12 @@ class ZeroStage3Optimizer:
                 dist.broadcast(full, src=dist.get_global_rank(self.dp_group, 0)
                                if hasattr(dist, "get_global_rank") else 0,
                                group=self.dp_group)
-            u.shard = full[self.rank * u.shard_size:
-                           (self.rank + 1) * u.shard_size].clone()
+            if self.world_size > 1:
+                u.shard = full[self.rank * u.shard_size:
+                               (self.rank + 1) * u.shard_size].clone()
+            else:
+                # ws=1: the shard IS the full buffer — keep it, no clone
+                u.shard = full
             for p in u.params:
                 p.ds_shape = p.shape
                 p.ds_numel = p.numel()
@@ -276,13 +281,15 @@ class ZeroStage3Optimizer:
     def _launch_gather(self, u: _Unit):
         if u.status != FREE:
             return
-        u.full = torch.empty(u.numel, dtype=self._dtype, device=self._device)
         if self.world_size > 1:
+            u.full = torch.empty(u.numel, dtype=self._dtype,
+                                 device=self._device)
             u.handle = dist.all_gather_into_tensor(u.full, u.shard,
                                                    group=self.dp_group,
                                                    async_op=True)
         else:
-            u.full.copy_(u.shard)
+            # ws=1: shard covers the whole unit — alias, no alloc, no copy
+            u.full = u.shard
             u.handle = None
         u.status = INFLIGHT
 
@@ -327,10 +334,10 @@ class ZeroStage3Optimizer:
             h3 = mod.register_full_backward_pre_hook(self._pre_backward_hook)
             self._hooks += [h1, h2, h3]
         for u in self.units:
-            for p in u.params:
+            for i, (p, off) in enumerate(zip(u.params, u.offsets)):
                 if p.requires_grad:
                     self._hooks.append(p.register_post_accumulate_grad_hook(
-                        self._make_grad_hook(u)))
+                        self._make_grad_hook(u, i, off)))
             u.pending_grads = sum(1 for p in u.params if p.requires_grad)
 
     def _units_for(self, mod):
@@ -380,32 +387,34 @@ class ZeroStage3Optimizer:
             self._prefetch(self._rtrace, self._bwd_cursor)
         for u in units:
             self._make_available(u)
-            self._ensure_grad_views(u)
+            self._ensure_grad_buffer(u)
         self._drain_inflight_rs(limit=self._max_inflight_rs)
 
-    def _ensure_grad_views(self, u: _Unit):
+    def _ensure_grad_buffer(self, u: _Unit):
+        """Allocate the unit's flat grad landing buffer UNZEROED. Grads are
+        copied in by the post-accumulate hook (first arrival = copy, later
+        arrivals = add), which removes the per-microstep fill kernel and the
+        autograd `p.grad +=` read-modify-write of the preset-view scheme."""
         if not u.trainable:
             return
         if u.grad_full is None:
-            u.grad_full = torch.zeros(u.numel, dtype=self._dtype,
+            u.grad_full = torch.empty(u.numel, dtype=self._dtype,
                                       device=self._device)
-        for p, off in zip(u.params, u.offsets):
-            if p.requires_grad:
-                g = u.grad_full[off:off + p.ds_numel].view(p.ds_shape)
-                if p.grad is None or p.grad.data_ptr() != g.data_ptr():
-                    p.grad = g
+            u.grad_seen = [False] * len(u.params)
 
-    def _make_grad_hook(self, u: _Unit):
+    def _make_grad_hook(self, u: _Unit, idx: int, off: int):
         def hook(param):
-            # autograd accumulated into the preset view; defensive fold-in
             if u.grad_full is None:
-                return
-            off = next(o for p, o in zip(u.params, u.offsets) if p is param)
-            expected = u.grad_full[off:off + param.ds_numel]
-            if param.grad is not None and \
-                    param.grad.data_ptr() != expected.data_ptr():
-                expected.add_(param.grad.detach().reshape(-1))
-                param.grad = expected.view(param.ds_shape)
+                self._ensure_grad_buffer(u)
+            if param.grad is not None:
+                g = param.grad.detach().reshape(-1)
+                dst = u.grad_full[off:off + param.ds_numel]
+                if u.grad_seen[idx]:
+                    dst.add_(g)
+                else:
+                    dst.copy_(g)
+                    u.grad_seen[idx] = True
+                param.grad = None
             u.pending_grads -= 1
             if u.pending_grads == 0:
                 self._reduce_unit(u)
@@ -418,18 +427,27 @@ class ZeroStage3Optimizer:
         if u.grad_full is None:
             return
         grad = u.grad_full
-        grad.div_(self.world_size)
+        # landing buffer was left unzeroed: clear spans that never received a
+        # grad this microstep (frozen/unused params) and the alignment tail
+        for (p, off, seen) in zip(u.params, u.offsets, u.grad_seen):
+            if not seen:
+                grad[off:off + p.ds_numel].zero_()
+        tail = u.offsets[-1] + u.params[-1].ds_numel
+        if tail < u.numel:
+            grad[tail:].zero_()
         if self.world_size > 1:
+            grad.div_(self.world_size)
             recv = torch.empty(u.shard_size, dtype=grad.dtype,
                                device=grad.device)
             h = dist.reduce_scatter_tensor(recv, grad, group=self.dp_group,
                                            async_op=True)
         else:
-            recv = grad[self.rank * u.shard_size:(self.rank + 1) * u.shard_size]
+            recv = grad
             h = None
         self._inflight_rs.append((h, recv, u, grad))
         # params not needed anymore this micro-step
         u.grad_full = None
+        u.grad_seen = None
         for p in u.params:
             p.grad = None
         u.in_backward = False
@@ -448,7 +466,9 @@ class ZeroStage3Optimizer:
                 if self.cpu_offload:
                     dst.add_(recv.float().cpu())
                 else:
-                    dst.add_(recv.to(torch.float32))
+                    # mixed-dtype add_: the cast fuses into the add kernel
+                    # instead of materializing a separate fp32 temp
+                    dst.add_(recv)
 
     # -------------------------------------------------------------- prefetch
 
@@ -615,8 +635,7 @@ class ZeroStage3Optimizer:
                 if self.world_size > 1:
                     handles.append(dist.all_gather_into_tensor(
                         u.full, u.shard, group=self.dp_group, async_op=True))
-                else:
-                    u.full.copy_(u.shard)
+                # ws=1: full aliases shard — the step updated it in place
         for h in handles:
             if h is not None:
                 h.wait()
