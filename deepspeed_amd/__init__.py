@@ -62,12 +62,21 @@ def initialize(args=None,
         from .parallel import groups
         groups.initialize_sequence_parallel(sp)
 
-    engine = Engine(model=model,
-                    optimizer=optimizer,
-                    model_parameters=model_parameters,
-                    lr_scheduler=lr_scheduler,
-                    config=cfg,
-                    mpu=mpu)
+    from .runtime.pipe.module import PipelineModule
+    if isinstance(model, PipelineModule):
+        from .runtime.pipe.engine import PipelineEngine
+        engine = PipelineEngine(model=model,
+                                optimizer=optimizer,
+                                model_parameters=model_parameters,
+                                lr_scheduler=lr_scheduler,
+                                config=cfg)
+    else:
+        engine = Engine(model=model,
+                        optimizer=optimizer,
+                        model_parameters=model_parameters,
+                        lr_scheduler=lr_scheduler,
+                        config=cfg,
+                        mpu=mpu)
 
     dataloader = None
     if training_data is not None:
