@@ -122,10 +122,14 @@ def fused_adam_step(optimizer, group, master_param, grad_shard, combined_scale,
     shard with 16-bit grads consumed in place.
 
     ``segments`` (optional) is a list of ``(offset, numel, out16_or_None)``
-    covering the shard; when given, the kernel runs per segment and fuses
-    the fp32->16-bit param-shard write (out16 is the bucket's shard view),
-    saving a full extra read+write pass over the shard. Returns False if
-    this optimizer cannot be fused (caller falls back to torch).
+    or ``(offset, numel, out16_or_None, grad_tensor)`` covering the shard;
+    when given, the kernel runs per segment and fuses the fp32->16-bit
+    param-shard write (out16 is the bucket's shard view), saving a full
+    extra read+write pass over the shard. A 4th element overrides the grad
+    source for that segment (ZeRO-3 direct-grad mode feeds the 16-bit
+    reduce-scatter output straight in — ``grad_shard`` may then be empty).
+    Returns False if this optimizer cannot be fused (caller falls back to
+    torch).
     """
     if not isinstance(optimizer, _ADAM_TYPES):
         return False
@@ -146,17 +150,19 @@ def fused_adam_step(optimizer, group, master_param, grad_shard, combined_scale,
     grads = grad_shard.view(-1)
     m = state["exp_avg"].view(-1)
     v = state["exp_avg_sq"].view(-1)
-    for off, n, out16 in segments:
+    for seg in segments:
+        off, n, out16 = seg[0], seg[1], seg[2]
+        g = seg[3].view(-1) if len(seg) > 3 else grads[off:off + n]
         sl = slice(off, off + n)
         if out16 is not None and out16.dtype != torch.bfloat16:
             out16 = None  # kernel only fuses bf16 writes; caller copies
         if ext is not None and master_param.is_cuda:
-            ext.fused_adam_flat(master[sl], grads[sl], m[sl], v[sl],
+            ext.fused_adam_flat(master[sl], g, m[sl], v[sl],
                                 out16.view(-1) if out16 is not None else None,
                                 lr, beta1, beta2, eps, wd, state["step"],
                                 inv_scale, adamw)
         else:
-            _torch_adam_step(master[sl], grads[sl], m[sl], v[sl], lr, beta1,
+            _torch_adam_step(master[sl], g, m[sl], v[sl], lr, beta1,
                              beta2, eps, wd, state["step"], adamw, inv_scale)
             if out16 is not None:
                 out16.view(-1).copy_(master[sl])

@@ -53,9 +53,9 @@ FREE, INFLIGHT, AVAILABLE = 0, 1, 2
 class _Unit:
     __slots__ = ("index", "name", "module", "params", "offsets", "numel",
                  "shard_size", "shard", "full", "grad_full", "grad_seen",
-                 "status", "handle", "persist", "group_idx", "master_offset",
-                 "pending_grads", "trainable", "release_pending",
-                 "in_backward")
+                 "grad_shard", "status", "handle", "persist", "group_idx",
+                 "master_offset", "pending_grads", "trainable",
+                 "release_pending", "in_backward")
 
     def __init__(self, index, name, module):
         self.index = index
@@ -69,6 +69,7 @@ class _Unit:
         self.full: Optional[torch.Tensor] = None
         self.grad_full: Optional[torch.Tensor] = None
         self.grad_seen: Optional[List[bool]] = None
+        self.grad_shard: Optional[torch.Tensor] = None  # direct-grad mode
         self.status = FREE
         self.handle = None
         self.persist = False
@@ -112,6 +113,12 @@ class ZeroStage3Optimizer:
         self.custom_loss_scaler = False
         self.micro_step_id = 0
         self._max_inflight_rs = 4
+        # Without gradient accumulation the fp32 grad accumulator is pure
+        # overhead: keep each unit's 16-bit reduce-scatter output and feed it
+        # straight to the fused Adam kernel (which casts in-register). Saves
+        # a full zero-fill + cast-add pass over the shard every step.
+        self.direct_grad = (config.gradient_accumulation_steps == 1
+                            and not self.cpu_offload)
 
         self.units: List[_Unit] = []
         self.param_to_unit: Dict[torch.nn.Parameter, _Unit] = {}
@@ -241,7 +248,11 @@ class ZeroStage3Optimizer:
         master_dev = torch.device("cpu") if self.cpu_offload else self._device
         for gi in range(ngroups):
             m = torch.empty(totals[gi], dtype=torch.float32, device=master_dev)
-            g = torch.zeros(totals[gi], dtype=torch.float32, device=master_dev)
+            if self.direct_grad:
+                g = torch.empty(0, dtype=torch.float32, device=master_dev)
+            else:
+                g = torch.zeros(totals[gi], dtype=torch.float32,
+                                device=master_dev)
             if self.cpu_offload and self.offload_pin_memory and accel.available():
                 m, g = m.pin_memory(), g.pin_memory()
             self.group_masters.append(m)
@@ -461,6 +472,10 @@ class ZeroStage3Optimizer:
             if h is not None:
                 h.wait()
             if u.group_idx >= 0:
+                if self.direct_grad:
+                    # keep the 16-bit RS output; Adam consumes it directly
+                    u.grad_shard = recv
+                    continue
                 dst = self.group_owned_grads[u.group_idx][
                     u.master_offset:u.master_offset + u.shard_size]
                 if self.cpu_offload:
@@ -542,7 +557,11 @@ class ZeroStage3Optimizer:
         self._drain_inflight_rs(limit=0)
 
         scale = self.loss_scaler.loss_scale
-        owned = [g for g in self.group_owned_grads if g.numel() > 0]
+        if self.direct_grad:
+            owned = [u.grad_shard for u in self.units
+                     if u.grad_shard is not None and u.grad_shard.numel() > 0]
+        else:
+            owned = [g for g in self.group_owned_grads if g.numel() > 0]
         norm_sq = None
         if owned:
             norms = torch._foreach_norm(owned, 2.0)
@@ -597,7 +616,14 @@ class ZeroStage3Optimizer:
                 out16 = u.shard if u.shard.dtype == torch.bfloat16 else None
                 if out16 is None:
                     wrote_params = False
-                segments.append((u.master_offset, u.shard_size, out16))
+                if self.direct_grad:
+                    g16 = u.grad_shard
+                    if g16 is None:  # unit never produced grads this step
+                        g16 = torch.zeros(u.shard_size, dtype=self._dtype,
+                                          device=self._device)
+                    segments.append((u.master_offset, u.shard_size, out16, g16))
+                else:
+                    segments.append((u.master_offset, u.shard_size, out16))
             ok = self.fused_adam_fn(self.optimizer, group, master,
                                     self.group_owned_grads[gi],
                                     combined_scale, segments=segments)
@@ -607,6 +633,8 @@ class ZeroStage3Optimizer:
 
     def _torch_step(self, combined_scale):
         from ...ops.adam import _torch_adam_step  # noqa
+        if self.direct_grad:
+            self._materialize_owned_grads()
         for gi, group in enumerate(self.optimizer.param_groups):
             master = self.group_masters[gi]
             if master.numel() == 0:
@@ -641,9 +669,30 @@ class ZeroStage3Optimizer:
                 h.wait()
 
     def _zero_owned_grads(self):
+        if self.direct_grad:
+            for u in self.units:
+                u.grad_shard = None
         for g in self.group_owned_grads:
             if g.numel():
                 g.zero_()
+
+    def _materialize_owned_grads(self):
+        """direct-grad fallback for the unfused torch step: scatter the
+        per-unit 16-bit shards into fp32 group accumulators."""
+        for gi in range(len(self.group_owned_grads)):
+            if self.group_owned_grads[gi].numel() != self.group_shard_numel[gi]:
+                self.group_owned_grads[gi] = torch.zeros(
+                    self.group_shard_numel[gi], dtype=torch.float32,
+                    device=self._device if not self.cpu_offload
+                    else torch.device("cpu"))
+            else:
+                self.group_owned_grads[gi].zero_()
+        for u in self.units:
+            if u.group_idx < 0 or u.grad_shard is None:
+                continue
+            dst = self.group_owned_grads[u.group_idx][
+                u.master_offset:u.master_offset + u.shard_size]
+            dst.copy_(u.grad_shard)
 
     def zero_grad(self, set_to_none: bool = False):
         pass  # transient grad buffers; fp32 accumulators zeroed in step()
