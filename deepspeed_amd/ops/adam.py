@@ -150,6 +150,20 @@ def fused_adam_step(optimizer, group, master_param, grad_shard, combined_scale,
     grads = grad_shard.view(-1)
     m = state["exp_avg"].view(-1)
     v = state["exp_avg_sq"].view(-1)
+    use_cpu_ext = (ext is not None and not master_param.is_cuda
+                   and hasattr(ext, "cpu_adam_flat"))
+    host16 = None
+    if use_cpu_ext and any(len(s) > 2 and s[2] is not None and s[2].is_cuda
+                           for s in segments):
+        # ZeRO-Offload: the host Adam writes updated params as bf16 into a
+        # cached pinned buffer; the H2D ships half the bytes of an fp32
+        # master copy and needs no GPU cast kernel.
+        host16 = state.get("host_bf16")
+        if host16 is None or host16.numel() != master.numel():
+            host16 = torch.empty(master.numel(), dtype=torch.bfloat16)
+            if torch.cuda.is_available():
+                host16 = host16.pin_memory()
+            state["host_bf16"] = host16
     for seg in segments:
         off, n, out16 = seg[0], seg[1], seg[2]
         g = seg[3].view(-1) if len(seg) > 3 else grads[off:off + n]
@@ -161,6 +175,14 @@ def fused_adam_step(optimizer, group, master_param, grad_shard, combined_scale,
                                 out16.view(-1) if out16 is not None else None,
                                 lr, beta1, beta2, eps, wd, state["step"],
                                 inv_scale, adamw)
+        elif use_cpu_ext and not g.is_cuda:
+            w16 = None
+            if out16 is not None:
+                w16 = host16[sl] if out16.is_cuda else out16.view(-1)
+            ext.cpu_adam_flat(master[sl], g, m[sl], v[sl], w16, lr, beta1,
+                              beta2, eps, wd, state["step"], inv_scale, adamw)
+            if out16 is not None and out16.is_cuda:
+                out16.view(-1).copy_(host16[sl], non_blocking=True)
         else:
             _torch_adam_step(master[sl], g, m[sl], v[sl], lr, beta1,
                              beta2, eps, wd, state["step"], adamw, inv_scale)

@@ -28,6 +28,11 @@ extern "C" void ds_gated_act_fwd(const void* gate, const void* up, void* out,
 extern "C" void ds_gated_act_bwd(const void* dout, const void* gate,
                                  const void* up, void* dgate, void* dup,
                                  long long n, int act, int dtype, void* stream);
+extern "C" void ds_cpu_adam_flat(float* p, const void* g, int grad_dtype,
+                                 float* m, float* v, void* p16, long long n,
+                                 float lr, float beta1, float beta2, float eps,
+                                 float weight_decay, int step, float inv_scale,
+                                 int adamw);
 
 namespace {
 
@@ -66,6 +71,31 @@ void fused_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                      p.numel(), (float)lr, (float)beta1, (float)beta2,
                      (float)eps, (float)weight_decay, (int)step,
                      (float)inv_scale, adamw ? 1 : 0, cur_stream());
+}
+
+void cpu_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                   c10::optional<at::Tensor> p16, double lr, double beta1,
+                   double beta2, double eps, double weight_decay, int64_t step,
+                   double inv_scale, bool adamw) {
+  TORCH_CHECK(!p.is_cuda() && !g.is_cuda(), "cpu_adam_flat: host tensors only");
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous() &&
+              v.is_contiguous(), "cpu_adam_flat: tensors must be contiguous");
+  TORCH_CHECK(p.scalar_type() == at::kFloat && m.scalar_type() == at::kFloat &&
+              v.scalar_type() == at::kFloat, "p/m/v must be fp32");
+  TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel() &&
+              p.numel() == v.numel(), "numel mismatch");
+  void* p16_ptr = nullptr;
+  if (p16.has_value() && p16->defined()) {
+    TORCH_CHECK(p16->scalar_type() == at::kBFloat16 && p16->is_contiguous() &&
+                !p16->is_cuda() && p16->numel() == p.numel(),
+                "p16 must be contiguous host bf16");
+    p16_ptr = p16->data_ptr();
+  }
+  ds_cpu_adam_flat(p.data_ptr<float>(), g.data_ptr(), dtype_code(g),
+                   m.data_ptr<float>(), v.data_ptr<float>(), p16_ptr,
+                   p.numel(), (float)lr, (float)beta1, (float)beta2, (float)eps,
+                   (float)weight_decay, (int)step, (float)inv_scale,
+                   adamw ? 1 : 0);
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> norm_fwd(
@@ -159,6 +189,8 @@ std::tuple<at::Tensor, at::Tensor> gated_act_bwd(at::Tensor dout,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam_flat", &fused_adam_flat,
         "Fused Adam/AdamW on flat fp32 master + 16-bit grad shard");
+  m.def("cpu_adam_flat", &cpu_adam_flat,
+        "Host Adam/AdamW on flat pinned fp32 master (ZeRO-Offload)");
   m.def("norm_fwd", &norm_fwd, "RMSNorm/LayerNorm forward");
   m.def("norm_bwd", &norm_bwd, "RMSNorm/LayerNorm backward");
   m.def("rope", &rope, "Rotary position embedding (in-place)");

@@ -277,10 +277,15 @@ class ZeroStage3Optimizer:
                 self.group_masters[gi] = mp
 
     def _try_fused_adam(self):
-        if self.cpu_offload:
-            return None
         try:
             from ...ops.adam import multi_tensor_adam_available, fused_adam_step
+            if self.cpu_offload:
+                # offload step runs on host: needs the AVX cpu_adam op
+                from ...ops._loader import get_ext
+                ext = get_ext()
+                if ext is not None and hasattr(ext, "cpu_adam_flat"):
+                    return fused_adam_step
+                return None
             if multi_tensor_adam_available():
                 return fused_adam_step
         except Exception:

@@ -20,15 +20,17 @@ ext = CUDAExtension(
     name="deepspeed_amd.ops._C",
     sources=[
         os.path.join(CSRC, "bindings.cpp"),
+        os.path.join(CSRC, "cpu_adam.cpp"),
         os.path.join(CSRC, "adam.hip"),
         os.path.join(CSRC, "norms.hip"),
         os.path.join(CSRC, "rope.hip"),
         os.path.join(CSRC, "swiglu.hip"),
     ],
     extra_compile_args={
-        "cxx": ["-O3", "-std=c++17"],
+        "cxx": ["-O3", "-std=c++17", "-fopenmp", "-mavx2", "-mfma"],
         "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
     },
+    extra_link_args=["-fopenmp"],
 )
 
 setup(
