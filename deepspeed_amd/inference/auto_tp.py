@@ -1,0 +1,126 @@
+"""AutoTP — automatic tensor-parallel sharding of Linear layers for
+inference (reference: deepspeed/module_inject/auto_tp.py AutoTP :192,
+layers.py LinearLayer :370 / LinearAllreduce :300).
+
+Column-parallel layers shard out_features (no comm); row-parallel layers
+shard in_features and all-reduce the partial outputs over the TP group —
+on the MI355X node that all-reduce runs over xGMI.
+
+The generic policy shards by module-name suffix. For attention projections
+the column split must respect head boundaries: q/k/v are sharded by whole
+heads (their out_features are head multiples, so an even world split
+preserves heads as long as heads % tp == 0, asserted by the engine).
+"""
+
+import re
+from typing import Iterable, Optional
+
+import torch
+import torch.nn as nn
+
+from .. import comm as dist
+
+
+class LinearLayer(nn.Module):
+    """Column-parallel linear: holds rows [r*out/P, (r+1)*out/P) of the
+    weight; outputs this rank's slice of the features."""
+
+    def __init__(self, weight: torch.Tensor, bias: Optional[torch.Tensor]):
+        super().__init__()
+        self.weight = nn.Parameter(weight, requires_grad=False)
+        self.bias = nn.Parameter(bias, requires_grad=False) \
+            if bias is not None else None
+
+    def forward(self, x):
+        return torch.nn.functional.linear(x, self.weight, self.bias)
+
+
+class LinearAllreduce(nn.Module):
+    """Row-parallel linear: holds columns of the weight; all-reduces the
+    partial product over the TP group (reference layers.py:300)."""
+
+    def __init__(self, weight: torch.Tensor, bias: Optional[torch.Tensor],
+                 group):
+        super().__init__()
+        self.weight = nn.Parameter(weight, requires_grad=False)
+        self.bias = nn.Parameter(bias, requires_grad=False) \
+            if bias is not None else None
+        self.group = group
+
+    def forward(self, x):
+        y = torch.nn.functional.linear(x, self.weight)
+        if dist.get_world_size(self.group) > 1:
+            dist.all_reduce(y, group=self.group)
+        if self.bias is not None:
+            y = y + self.bias
+        return y
+
+
+# module-name suffixes -> parallel style for common decoder architectures
+COLUMN_PATTERNS = (r"q_proj$", r"k_proj$", r"v_proj$", r"gate_proj$",
+                   r"up_proj$", r"w1$", r"w3$", r"c_attn$", r"wqkv$",
+                   r"fc1$", r"gate_up_proj$")
+ROW_PATTERNS = (r"o_proj$", r"down_proj$", r"w2$", r"c_proj$", r"fc2$",
+                r"dense$", r"out_proj$")
+
+
+def _match(name: str, patterns: Iterable[str]) -> bool:
+    return any(re.search(p, name) for p in patterns)
+
+
+@torch.no_grad()
+def shard_model(model: nn.Module, tp_group, tp_rank: int, tp_size: int,
+                column_patterns=COLUMN_PATTERNS, row_patterns=ROW_PATTERNS):
+    """Replace matching nn.Linear modules with TP-sharded versions in place.
+
+    Returns the number of modules sharded. Idempotent on already-replaced
+    modules (they are no longer nn.Linear).
+    """
+    if tp_size == 1:
+        return 0
+    replaced = 0
+    for parent_name, parent in list(model.named_modules()):
+        for child_name, child in list(parent._modules.items()):
+            if not isinstance(child, nn.Linear):
+                continue
+            full = f"{parent_name}.{child_name}" if parent_name else child_name
+            W = child.weight.data
+            b = child.bias.data if child.bias is not None else None
+            if _match(full, column_patterns):
+                out = W.size(0)
+                assert out % tp_size == 0, \
+                    f"{full}: out_features {out} not divisible by tp {tp_size}"
+                sl = slice(tp_rank * out // tp_size,
+                           (tp_rank + 1) * out // tp_size)
+                new = LinearLayer(W[sl].clone(),
+                                  b[sl].clone() if b is not None else None)
+            elif _match(full, row_patterns):
+                inp = W.size(1)
+                assert inp % tp_size == 0, \
+                    f"{full}: in_features {inp} not divisible by tp {tp_size}"
+                sl = slice(tp_rank * inp // tp_size,
+                           (tp_rank + 1) * inp // tp_size)
+                # bias applied once, after the all-reduce
+                new = LinearAllreduce(W[:, sl].clone(),
+                                      b.clone() if b is not None else None,
+                                      tp_group)
+            else:
+                continue
+            parent._modules[child_name] = new
+            replaced += 1
+    return replaced
+
+
+def shard_attention_heads(model: nn.Module, tp_rank: int, tp_size: int):
+    """Patch head counts on attention modules that track them
+    (reference auto_tp.py shard head counts via mp sizing)."""
+    if tp_size == 1:
+        return
+    for mod in model.modules():
+        for attr in ("num_heads", "num_kv_heads", "num_attention_heads",
+                     "num_key_value_heads"):
+            n = getattr(mod, attr, None)
+            if isinstance(n, int) and n > 0:
+                assert n % tp_size == 0, \
+                    f"{attr}={n} not divisible by tp={tp_size}"
+                setattr(mod, attr, n // tp_size)

@@ -1,0 +1,142 @@
+"""Inference engine — TP sharding + KV-cache decoding + generate()
+(reference: deepspeed/inference/engine.py InferenceEngine :40,
+init_inference deepspeed/__init__.py:291).
+
+MI355X-first shape: the model's hot ops are already the framework's HIP
+kernels (RMSNorm/RoPE/SwiGLU) and hipBLASLt GEMMs; this engine adds the
+serving mechanics — tensor-parallel sharding over the xGMI mesh
+(see auto_tp.py), a contiguous KV cache, and a sampling loop.
+"""
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from .. import accel
+from .. import comm as dist
+from ..parallel import groups
+from ..utils.logging import log_dist
+from .auto_tp import shard_attention_heads, shard_model
+from .kv_cache import StaticKVCache
+
+
+@dataclass
+class InferenceConfig:
+    dtype: object = torch.bfloat16
+    tensor_parallel: dict = field(default_factory=dict)  # {"tp_size": N}
+    max_out_tokens: int = 1024
+    min_out_tokens: int = 1
+    replace_with_kernel_inject: bool = False  # model already uses HIP ops
+    checkpoint: Optional[str] = None
+    enable_cuda_graph: bool = False  # hipGraph capture of the decode step
+
+    @property
+    def tp_size(self) -> int:
+        if isinstance(self.tensor_parallel, dict):
+            return int(self.tensor_parallel.get("tp_size", 1))
+        return int(getattr(self.tensor_parallel, "tp_size", 1))
+
+
+class InferenceEngine(nn.Module):
+    def __init__(self, model: nn.Module, config: InferenceConfig):
+        super().__init__()
+        self.module = model
+        self.config = config
+        self._kv = None
+        self._graph = None
+
+        tp = config.tp_size
+        if tp > 1:
+            if not dist.is_initialized():
+                dist.init_distributed()
+            groups.initialize_tensor_parallel(tp)
+            self.tp_group = groups.get_tensor_parallel_group()
+            self.tp_rank = groups.get_tensor_parallel_rank()
+            n = shard_model(self.module, self.tp_group, self.tp_rank, tp)
+            shard_attention_heads(self.module, self.tp_rank, tp)
+            log_dist(f"AutoTP: sharded {n} linears over tp={tp}")
+        else:
+            self.tp_group = None
+            self.tp_rank = 0
+
+        if isinstance(config.dtype, str):
+            config.dtype = {"fp32": torch.float32, "fp16": torch.float16,
+                            "bf16": torch.bfloat16}[config.dtype]
+        if config.dtype != torch.float32:
+            self.module.to(config.dtype)
+        if accel.available():
+            self.module.to(accel.current_device())
+        self.device = next(self.module.parameters()).device
+        self.module.eval()
+
+    def forward(self, *args, **kwargs):
+        with torch.no_grad():
+            return self.module(*args, **kwargs)
+
+    # --------------------------------------------------------------- generate
+
+    def _model_geometry(self):
+        m = self.module
+        cfg = getattr(m, "cfg", None)
+        assert cfg is not None, "generate() needs a model with .cfg geometry"
+        kv_heads = getattr(cfg, "num_kv_heads", None) or cfg.num_heads
+        tp = self.config.tp_size
+        return (cfg.num_layers, kv_heads // max(tp, 1) if tp > 1 else kv_heads,
+                cfg.head_dim, cfg.max_seq_len)
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
+                 do_sample: bool = False, temperature: float = 1.0,
+                 top_k: int = 0, eos_token_id: Optional[int] = None):
+        """KV-cached autoregressive generation (greedy or sampled).
+
+        input_ids: [B, S] prompt. Returns [B, S + new] including the prompt.
+        """
+        m = self.module
+        B, S = input_ids.shape
+        layers, kv_heads, head_dim, max_seq = self._model_geometry()
+        total = min(S + max_new_tokens, max_seq)
+        self._kv = StaticKVCache(layers, B, kv_heads, total, head_dim,
+                                 dtype=self.config.dtype
+                                 if self.config.dtype != torch.float32
+                                 else torch.float32,
+                                 device=self.device)
+        input_ids = input_ids.to(self.device)
+        out = input_ids
+
+        # prefill
+        positions = torch.arange(S, device=self.device,
+                                 dtype=torch.int32).expand(B, S).contiguous()
+        logits = m(input_ids, positions=positions, kv_cache=self._kv)
+        self._kv.advance()
+        next_tok = self._select(logits[:, -1], do_sample, temperature, top_k)
+        out = torch.cat([out, next_tok], dim=1)
+        finished = torch.zeros(B, dtype=torch.bool, device=self.device)
+
+        for _ in range(max_new_tokens - 1):
+            if out.size(1) >= total:
+                break
+            if eos_token_id is not None:
+                finished |= next_tok.squeeze(1) == eos_token_id
+                if bool(finished.all()):
+                    break
+            pos = torch.full((B, 1), self._kv.cur_len, device=self.device,
+                             dtype=torch.int32)
+            logits = m(next_tok, positions=pos, kv_cache=self._kv)
+            self._kv.advance()
+            next_tok = self._select(logits[:, -1], do_sample, temperature,
+                                    top_k)
+            out = torch.cat([out, next_tok], dim=1)
+        return out
+
+    def _select(self, logits, do_sample, temperature, top_k):
+        if not do_sample:
+            return logits.argmax(dim=-1, keepdim=True)
+        logits = logits.float() / max(temperature, 1e-5)
+        if top_k > 0:
+            kth = logits.topk(top_k, dim=-1).values[..., -1, None]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        probs = torch.softmax(logits, dim=-1)
+        return torch.multinomial(probs, 1)

@@ -96,7 +96,9 @@ class LlamaAttention(nn.Module):
         if attention_fn is not None:
             o = attention_fn(q, k, v)
         else:
-            o = sdpa_gqa(q, k, v, causal=kv_cache is None)
+            # prefill (q spans the whole kv prefix) is causal; decode steps
+            # (q_len < kv_len) attend to the full cached prefix
+            o = sdpa_gqa(q, k, v, causal=q.size(2) == k.size(2))
         o = o.transpose(1, 2).reshape(B, S, -1)
         return self.o_proj(o)
 
