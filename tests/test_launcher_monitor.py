@@ -1,0 +1,65 @@
+"""Launcher + monitor tests (reference contract:
+tests/unit/launcher/test_ds_arguments.py, tests/unit/monitor/test_monitor.py).
+"""
+
+import os
+import subprocess
+import sys
+
+import torch
+
+
+def test_hostfile_parse(tmp_path):
+    from deepspeed_amd.launcher.runner import parse_hostfile
+    hf = tmp_path / "hostfile"
+    hf.write_text("nodeA slots=8\n# comment\nnodeB slots=4  # trailing\n\n")
+    hosts = parse_hostfile(str(hf))
+    assert hosts == {"nodeA": 8, "nodeB": 4}
+
+
+def test_launcher_spawns_ranks(tmp_path):
+    """End-to-end: the runner must set RANK/LOCAL_RANK/WORLD_SIZE."""
+    script = tmp_path / "probe.py"
+    script.write_text(
+        "import os\n"
+        f"open(os.path.join({str(tmp_path)!r}, 'r' + os.environ['RANK']),"
+        " 'w').write(' '.join(\n"
+        "    [os.environ['RANK'], os.environ['LOCAL_RANK'],"
+        " os.environ['WORLD_SIZE']]))\n")
+    out = subprocess.run(
+        [sys.executable, "-m", "deepspeed_amd.launcher.runner",
+         "--num_gpus", "2", "--master_port", "29871", str(script)],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert (tmp_path / "r0").read_text() == "0 0 2"
+    assert (tmp_path / "r1").read_text() == "1 1 2"
+
+
+def test_launcher_propagates_failure(tmp_path):
+    script = tmp_path / "boom.py"
+    script.write_text("import os, sys; sys.exit(3 if os.environ['RANK'] == '1' else 0)\n")
+    out = subprocess.run(
+        [sys.executable, "-m", "deepspeed_amd.launcher.runner",
+         "--num_gpus", "2", str(script)],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 3
+
+
+def test_csv_monitor(tmp_path):
+    from deepspeed_amd.monitor.monitor import CsvMonitor, MonitorMaster
+    from deepspeed_amd.config import MonitorConfig
+    m = CsvMonitor(output_path=str(tmp_path), job_name="j")
+    m.write_events([("Train/loss", 1.5, 1), ("Train/loss", 1.2, 2),
+                    ("Train/lr", 0.1, 1)])
+    m.close()
+    loss = (tmp_path / "j" / "Train_loss.csv").read_text().strip().splitlines()
+    assert loss == ["1,1.5", "2,1.2"]
+
+    mm = MonitorMaster(MonitorConfig(
+        enabled=True,
+        csv_monitor={"enabled": True, "output_path": str(tmp_path),
+                     "job_name": "k"}))
+    mm.write_events([("a/b", 3.0, 7)])
+    assert (tmp_path / "k" / "a_b.csv").read_text().strip() == "7,3.0"
