@@ -75,6 +75,10 @@ class ZeroConfig(ConfigModel):
     sub_group_size: int = 1_000_000_000_000
     zero_hpz_partition_size: int = 1
     round_robin_gradients: bool = False
+    # ZeRO++-style quantized weight all-gather (qwZ): ship int8 + group
+    # scales over xGMI instead of bf16 (opt-in; changes forward numerics)
+    zero_quantized_weights: bool = False
+    zero_quantization_group_size: int = 2048
     # fp32 grad accumulation buffer (stage 1/2)
     fp32_grad_accum: bool = False
 

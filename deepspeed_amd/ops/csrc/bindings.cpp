@@ -33,6 +33,12 @@ extern "C" void ds_cpu_adam_flat(float* p, const void* g, int grad_dtype,
                                  float lr, float beta1, float beta2, float eps,
                                  float weight_decay, int step, float inv_scale,
                                  int adamw);
+extern "C" void ds_groupwise_quant(const void* x, int dtype, void* q,
+                                   float* scales, long long n, int group_size,
+                                   int bits, void* stream);
+extern "C" void ds_groupwise_dequant(const void* q, const float* scales,
+                                     void* out, int dtype, long long n,
+                                     int group_size, int bits, void* stream);
 
 namespace {
 
@@ -184,6 +190,35 @@ std::tuple<at::Tensor, at::Tensor> gated_act_bwd(at::Tensor dout,
   return {dgate, dup};
 }
 
+std::tuple<at::Tensor, at::Tensor> groupwise_quant(at::Tensor x,
+                                                   int64_t group_size,
+                                                   int64_t bits) {
+  TORCH_CHECK(x.is_contiguous() && x.is_cuda(), "x must be contiguous GPU");
+  TORCH_CHECK(bits == 8 || bits == 4, "bits must be 4 or 8");
+  TORCH_CHECK(group_size % 2 == 0, "group_size must be even");
+  const long long n = x.numel();
+  const long long groups = (n + group_size - 1) / group_size;
+  const long long qbytes = bits == 8 ? n : (n + 1) / 2;
+  auto q = at::empty({qbytes}, x.options().dtype(at::kChar));
+  auto scales = at::empty({groups}, x.options().dtype(at::kFloat));
+  ds_groupwise_quant(x.data_ptr(), dtype_code(x), q.data_ptr(),
+                     scales.data_ptr<float>(), n, (int)group_size, (int)bits,
+                     cur_stream());
+  return {q, scales};
+}
+
+at::Tensor groupwise_dequant(at::Tensor q, at::Tensor scales, int64_t numel,
+                             int64_t group_size, int64_t bits,
+                             at::ScalarType dtype) {
+  TORCH_CHECK(q.is_contiguous() && q.is_cuda() && scales.is_contiguous(),
+              "q/scales must be contiguous GPU");
+  auto out = at::empty({numel}, q.options().dtype(dtype));
+  ds_groupwise_dequant(q.data_ptr(), scales.data_ptr<float>(), out.data_ptr(),
+                       dtype_code(out), numel, (int)group_size, (int)bits,
+                       cur_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -196,4 +231,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope", &rope, "Rotary position embedding (in-place)");
   m.def("gated_act_fwd", &gated_act_fwd, "SwiGLU/GeGLU forward");
   m.def("gated_act_bwd", &gated_act_bwd, "SwiGLU/GeGLU backward");
+  m.def("groupwise_quant", &groupwise_quant,
+        "Groupwise symmetric int8/int4 quantization");
+  m.def("groupwise_dequant", &groupwise_dequant,
+        "Groupwise symmetric int8/int4 dequantization");
 }

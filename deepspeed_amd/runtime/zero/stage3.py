@@ -119,6 +119,10 @@ class ZeroStage3Optimizer:
         # a full zero-fill + cast-add pass over the shard every step.
         self.direct_grad = (config.gradient_accumulation_steps == 1
                             and not self.cpu_offload)
+        # qwZ: quantized weight all-gather (int8 + group scales over xGMI)
+        self.quantized_weights = bool(zc.zero_quantized_weights) \
+            and self.world_size > 1
+        self.quant_group_size = int(zc.zero_quantization_group_size)
 
         self.units: List[_Unit] = []
         self.param_to_unit: Dict[torch.nn.Parameter, _Unit] = {}
@@ -294,10 +298,30 @@ class ZeroStage3Optimizer:
 
     # ------------------------------------------------------------ fetch/release
 
+    def _unit_quant_group(self, u: _Unit) -> int:
+        g = self.quant_group_size
+        while g > 2 and u.shard_size % g:
+            g //= 2
+        return g
+
     def _launch_gather(self, u: _Unit):
         if u.status != FREE:
             return
-        if self.world_size > 1:
+        if self.world_size > 1 and self.quantized_weights:
+            from ...ops.quantizer import quantize
+            gs = self._unit_quant_group(u)
+            q, s = quantize(u.shard, gs, bits=8)
+            q_full = torch.empty(q.numel() * self.world_size, dtype=q.dtype,
+                                 device=q.device)
+            s_full = torch.empty(s.numel() * self.world_size,
+                                 dtype=torch.float32, device=q.device)
+            h1 = dist.all_gather_into_tensor(q_full, q, group=self.dp_group,
+                                             async_op=True)
+            h2 = dist.all_gather_into_tensor(s_full, s, group=self.dp_group,
+                                             async_op=True)
+            u.full = None
+            u.handle = ("qwz", h1, h2, q_full, s_full, gs)
+        elif self.world_size > 1:
             u.full = torch.empty(u.numel, dtype=self._dtype,
                                  device=self._device)
             u.handle = dist.all_gather_into_tensor(u.full, u.shard,
@@ -313,7 +337,17 @@ class ZeroStage3Optimizer:
         if u.status == AVAILABLE:
             return
         assert u.status == INFLIGHT, f"unit {u.name} not in flight"
-        if u.handle is not None:
+        if isinstance(u.handle, tuple):  # qwZ quantized gather
+            from ...ops.quantizer import dequantize
+            _, h1, h2, q_full, s_full, gs = u.handle
+            if h1 is not None:
+                h1.wait()
+            if h2 is not None:
+                h2.wait()
+            u.full = dequantize(q_full, s_full, u.numel, gs, bits=8,
+                                dtype=self._dtype)
+            u.handle = None
+        elif u.handle is not None:
             u.handle.wait()
             u.handle = None
         for p, off in zip(u.params, u.offsets):

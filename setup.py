@@ -22,6 +22,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "cpu_adam.cpp"),
         os.path.join(CSRC, "adam.hip"),
+        os.path.join(CSRC, "quantize.hip"),
         os.path.join(CSRC, "norms.hip"),
         os.path.join(CSRC, "rope.hip"),
         os.path.join(CSRC, "swiglu.hip"),
