@@ -435,6 +435,19 @@ class Engine(torch.nn.Module):
         """Context manager: disable gradient reduction inside (reference :2065)."""
         return Engine._NoSync(self)
 
+    def compile(self, backend="inductor", compile_kwargs=None):
+        """torch.compile the wrapped module in place (reference :3820).
+        ZeRO-3's .data-swapping hooks are graph breaks by construction, so
+        compilation is per-submodule-region under stage 3."""
+        self.module = torch.compile(self.module, backend=backend,
+                                    **(compile_kwargs or {}))
+        self._is_compiled = True
+        return self
+
+    @property
+    def is_compiled(self) -> bool:
+        return getattr(self, "_is_compiled", False)
+
     # ----------------------------------------------------------- data loader
 
     def deepspeed_io(self, dataset, batch_size=None, num_workers=0, collate_fn=None):

@@ -1,0 +1,112 @@
+"""Aux subsystem tests: elasticity math, curriculum scheduler, progressive
+layer drop, eigenvalue power iteration, engine.compile
+(reference contracts: tests/unit/elasticity/test_elastic.py,
+tests/unit/runtime/test_data_efficiency.py, tests/unit/runtime/test_pld.py).
+"""
+
+import math
+
+import pytest
+import torch
+
+
+def test_elastic_config_basic():
+    from deepspeed_amd.elasticity import compute_elastic_config
+    ds_config = {"elasticity": {
+        "enabled": True, "max_train_batch_size": 2000,
+        "micro_batch_sizes": [2, 4, 6], "min_gpus": 1, "max_gpus": 10000,
+        "prefer_larger_batch": True}}
+    batch, gpus = compute_elastic_config(ds_config)
+    assert batch <= 2000 and batch % 2 == 0
+    # every valid gpu count divides batch by some micro batch
+    for g in gpus[:20]:
+        assert any(batch % (m * g) == 0 for m in (2, 4, 6))
+
+
+def test_elastic_config_world_size():
+    from deepspeed_amd.elasticity import (ElasticityError,
+                                          compute_elastic_config)
+    ds_config = {"elasticity": {
+        "enabled": True, "max_train_batch_size": 100,
+        "micro_batch_sizes": [4], "min_gpus": 1, "max_gpus": 16}}
+    batch, gpus, micro = compute_elastic_config(ds_config, world_size=4,
+                                                return_microbatch=True)
+    assert 4 in gpus and batch % (micro * 4) == 0
+    with pytest.raises(ElasticityError):
+        compute_elastic_config({"elasticity": {"enabled": False}})
+
+
+def test_curriculum_fixed_linear():
+    from deepspeed_amd.runtime.data_pipeline import CurriculumScheduler
+    s = CurriculumScheduler({
+        "curriculum_type": "fixed_linear", "min_difficulty": 8,
+        "max_difficulty": 64,
+        "schedule_config": {"total_curriculum_step": 100,
+                            "difficulty_step": 8}})
+    assert s.update_difficulty(0) == 8
+    mid = s.update_difficulty(50)
+    assert 8 < mid < 64 and mid % 8 == 0
+    assert s.update_difficulty(100) == 64
+    assert s.update_difficulty(10_000) == 64
+
+
+def test_curriculum_fixed_discrete():
+    from deepspeed_amd.runtime.data_pipeline import CurriculumScheduler
+    s = CurriculumScheduler({
+        "curriculum_type": "fixed_discrete", "min_difficulty": 2,
+        "max_difficulty": 10,
+        "schedule_config": {"difficulty": [2, 6, 10],
+                            "max_step": [10, 20]}})
+    assert s.update_difficulty(5) == 2
+    assert s.update_difficulty(15) == 6
+    assert s.update_difficulty(25) == 10
+
+
+def test_progressive_layer_drop():
+    from deepspeed_amd.runtime.progressive_layer_drop import \
+        ProgressiveLayerDrop
+    pld = ProgressiveLayerDrop(theta=0.5, gamma=0.001)
+    assert pld.get_theta() == 1.0
+    pld.update_state(0)
+    assert abs(pld.get_theta() - 1.0) < 1e-6
+    pld.update_state(10_000)
+    assert 0.5 <= pld.get_theta() < 0.51  # decays toward theta
+    assert pld.get_state()["progressive_layer_drop"]
+
+
+def test_eigenvalue_power_iteration():
+    from deepspeed_amd.runtime.eigenvalue import Eigenvalue
+    torch.manual_seed(0)
+    lin = torch.nn.Linear(8, 1, bias=False)
+    lin._deepspeed_eigenvalue_block = True
+    X = torch.randn(256, 8)
+    loss = (lin(X) ** 2).mean()
+    loss.backward(create_graph=True)
+    ev = Eigenvalue(max_iter=500, tol=1e-5).compute_eigenvalue(lin)
+    # quadratic loss: Hessian = 2/N X^T X; top eigenvalue known
+    H = 2 * X.T @ X / X.shape[0]
+    expect = torch.linalg.eigvalsh(H).max().item()
+    assert len(ev) == 1
+    assert abs(ev[0] - expect) / expect < 0.05
+
+
+def test_engine_compile_flag():
+    from .common import run_local
+
+    def worker(rank, world):
+        import deepspeed_amd
+        from deepspeed_amd.models import GPT2ForCausalLM, gpt2_tiny
+        model = GPT2ForCausalLM(gpt2_tiny())
+        engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+            "train_micro_batch_size_per_gpu": 1,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+        assert not engine.is_compiled
+        engine.compile(backend="eager")  # inductor needs a compiler rig
+        assert engine.is_compiled
+        ids = torch.randint(0, 128, (1, 16))
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+        assert torch.isfinite(loss)
+
+    run_local(worker)
