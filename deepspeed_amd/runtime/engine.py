@@ -226,6 +226,9 @@ class Engine(torch.nn.Module):
             )
         else:
             self.optimizer = basic  # fp32 stage 0: engine handles allreduce
+        if hasattr(self.optimizer, "annotate_param_names"):
+            # names feed the universal-checkpoint layout manifest
+            self.optimizer.annotate_param_names(self.module)
 
     def _make_loss_scaler(self):
         if self.dtype == torch.float16:
@@ -527,7 +530,7 @@ class Engine(torch.nn.Module):
 
     def load_checkpoint(self, load_dir, tag=None, load_module_strict=True,
                         load_optimizer_states=True, load_lr_scheduler_states=True,
-                        load_module_only=False):
+                        load_module_only=False, load_universal=False):
         if tag is None:
             latest = os.path.join(load_dir, "latest")
             if not os.path.exists(latest):
@@ -547,6 +550,15 @@ class Engine(torch.nn.Module):
         self.global_steps = state.get("global_steps", 0)
         self.global_samples = state.get("global_samples", 0)
         self.skipped_steps = state.get("skipped_steps", 0)
+
+        if load_universal:
+            # elastic load: per-param fp32 state sliced at THIS world size
+            from ..checkpoint.universal import load_universal as _load_usd
+            usd = _load_usd(os.path.join(load_dir, f"{tag}_universal"))
+            assert hasattr(self.optimizer, "load_universal_state_dict"), \
+                "universal load requires a ZeRO optimizer"
+            self.optimizer.load_universal_state_dict(self.module, usd)
+            return ckpt_dir, state.get("client_state", {})
 
         if not load_module_only:
             if load_lr_scheduler_states and self.lr_scheduler is not None and \

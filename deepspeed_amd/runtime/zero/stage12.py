@@ -487,8 +487,73 @@ class ZeroStage12Optimizer:
             "fp32_flat_groups": [m.data if m.numel() else m for m in
                                  self.group_masters],
             "base_optimizer_state": self.optimizer.state_dict(),
+            # layout manifest: lets the offline universal-checkpoint
+            # converter reassemble per-param fp32 state without the model
+            # (reference: checkpoint/ds_to_universal.py)
+            "layout": self.layout_manifest(),
         }
         return sd
+
+    def layout_manifest(self):
+        buckets = []
+        for b in self.buckets:
+            buckets.append({
+                "group_idx": b.group_idx,
+                "master_offset": b.master_offset,
+                "shard_size": b.shard_size,
+                "numel": b.numel,
+                "pg_world": b.pg_world,
+                "pg_rank": b.pg_rank,
+                "params": [(getattr(p, "_ds_name", None), off, p.numel(),
+                            tuple(p.shape))
+                           for p, off in zip(b.params, b.offsets)],
+            })
+        return buckets
+
+    def annotate_param_names(self, module):
+        """Stamp parameter names used by the universal layout manifest."""
+        for n, p in module.named_parameters():
+            p._ds_name = n
+
+    @torch.no_grad()
+    def load_universal_state_dict(self, module, usd):
+        """Load a universal (per-param fp32 master + optimizer state)
+        checkpoint at ANY data-parallel world size: for every bucket param,
+        copy the overlap of its flat span with this rank's shard."""
+        self.annotate_param_names(module)
+        name_of = {p: n for n, p in module.named_parameters()}
+        for gi, group in enumerate(self.optimizer.param_groups):
+            master_p = self.group_masters[gi]
+            if master_p.numel() == 0:
+                continue
+            state = self.optimizer.state.setdefault(master_p, {})
+            if "exp_avg" not in state:
+                state["exp_avg"] = torch.zeros_like(master_p,
+                                                    dtype=torch.float32)
+                state["exp_avg_sq"] = torch.zeros_like(master_p,
+                                                       dtype=torch.float32)
+            state["step"] = usd.get("step", 0)
+        for b in self.buckets:
+            master = self.group_masters[b.group_idx]
+            st = self.optimizer.state[master]
+            dsts = {"param": master.data, "exp_avg": st["exp_avg"],
+                    "exp_avg_sq": st["exp_avg_sq"]}
+            lo = b.pg_rank * b.shard_size        # my shard span in the bucket
+            hi = lo + b.shard_size
+            for p, off in zip(b.params, b.offsets):
+                name = name_of.get(p)
+                if name is None or name not in usd["param"]:
+                    continue
+                a, z = max(off, lo), min(off + p.numel(), hi)
+                if a >= z:
+                    continue
+                for kind, dst in dsts.items():
+                    src = usd[kind][name].reshape(-1)
+                    dst[b.master_offset + (a - lo):
+                        b.master_offset + (z - lo)].copy_(
+                        src[a - off:z - off])
+        self._copy_masters_to_params()
+        self._allgather_params()
 
     def load_state_dict(self, sd, load_optimizer_states=True):
         assert sd["world_size"] == self.world_size, \

@@ -1,0 +1,78 @@
+"""Universal checkpoint tests (reference contract:
+tests/unit/checkpoint/test_universal_checkpoint.py): save at DP=2, convert
+offline, resume at DP=1 with identical optimizer math.
+"""
+
+import os
+
+import torch
+
+from .common import run_distributed, run_local
+
+_CONFIG = {
+    "train_micro_batch_size_per_gpu": 2,
+    "zero_optimization": {"stage": 1, "overlap_comm": False},
+    "optimizer": {"type": "AdamW",
+                  "params": {"lr": 1e-3, "weight_decay": 0.01}},
+}
+
+
+def _build_engine():
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    torch.manual_seed(31)
+    model = LlamaForCausalLM(llama_tiny())
+    engine, opt, _, _ = deepspeed_amd.initialize(model=model,
+                                                 config=dict(_CONFIG))
+    return engine, opt
+
+
+def _same_batches(n):
+    g = torch.Generator().manual_seed(77)
+    return [torch.randint(0, 512, (2, 32), generator=g) for _ in range(n)]
+
+
+def _train(engine, batches):
+    for ids in batches:
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+    return loss
+
+
+def _save_worker(rank, world, tmp):
+    engine, _ = _build_engine()
+    _train(engine, _same_batches(2))  # identical batch on both ranks
+    engine.save_checkpoint(tmp, tag="step2")
+    # continue one more step; dump resulting fp32 masters for comparison
+    _train(engine, _same_batches(3)[2:])
+    fp32 = engine.optimizer.get_fp32_state_dict(engine.module)
+    if rank == 0:
+        torch.save(fp32, os.path.join(tmp, "ref_after3.pt"))
+
+
+def _resume_worker(rank, world, tmp):
+    engine, opt = _build_engine()
+    engine.load_checkpoint(tmp, tag="step2", load_universal=True)
+    _train(engine, _same_batches(3)[2:])
+    fp32 = opt.get_fp32_state_dict(engine.module)
+    ref = torch.load(os.path.join(tmp, "ref_after3.pt"), weights_only=False)
+    assert set(fp32) == set(ref)
+    for k in ref:
+        torch.testing.assert_close(fp32[k], ref[k], rtol=1e-5, atol=1e-6), k
+
+
+def test_universal_checkpoint_dp2_to_dp1(tmp_path):
+    tmp = str(tmp_path)
+    run_distributed(_save_worker, world_size=2, args=(tmp,))
+
+    from deepspeed_amd.checkpoint import ds_to_universal
+    out = ds_to_universal(tmp, tag="step2")
+    usd = torch.load(os.path.join(out, "universal_optim_states.pt"),
+                     weights_only=False)
+    assert usd["step"] == 2
+    assert "model.layers.0.self_attn.q_proj.weight" in usd["param"]
+    # universal params must equal the gathered fp32 truth at save time
+    assert all(v.dtype == torch.float32 for v in usd["param"].values())
+
+    run_local(_resume_worker, args=(tmp,))
