@@ -118,6 +118,9 @@ def argparse_suppress():
 
 
 def zero_init(config=None, **kwargs):
-    """ZeRO-3 construction-time partitioning context (reference zero.Init)."""
+    """ZeRO-3 construction-time context (reference zero.Init)."""
     from .runtime.zero.partition import Init
     return Init(config=config, **kwargs)
+
+
+from .runtime import zero  # noqa: E402  (deepspeed.zero parity namespace)

@@ -94,49 +94,59 @@ class InferenceEngine(nn.Module):
 
         input_ids: [B, S] prompt. Returns [B, S + new] including the prompt.
         """
-        m = self.module
-        B, S = input_ids.shape
         layers, kv_heads, head_dim, max_seq = self._model_geometry()
-        total = min(S + max_new_tokens, max_seq)
-        self._kv = StaticKVCache(layers, B, kv_heads, total, head_dim,
-                                 dtype=self.config.dtype
-                                 if self.config.dtype != torch.float32
-                                 else torch.float32,
-                                 device=self.device)
-        input_ids = input_ids.to(self.device)
-        out = input_ids
+        return kv_generate(self.module, input_ids.to(self.device),
+                           n_layers=layers, kv_heads=kv_heads,
+                           head_dim=head_dim, max_seq=max_seq,
+                           dtype=(self.config.dtype
+                                  if self.config.dtype != torch.float32
+                                  else torch.float32),
+                           max_new_tokens=max_new_tokens,
+                           do_sample=do_sample, temperature=temperature,
+                           top_k=top_k, eos_token_id=eos_token_id)
 
-        # prefill
-        positions = torch.arange(S, device=self.device,
-                                 dtype=torch.int32).expand(B, S).contiguous()
-        logits = m(input_ids, positions=positions, kv_cache=self._kv)
-        self._kv.advance()
-        next_tok = self._select(logits[:, -1], do_sample, temperature, top_k)
-        out = torch.cat([out, next_tok], dim=1)
-        finished = torch.zeros(B, dtype=torch.bool, device=self.device)
 
-        for _ in range(max_new_tokens - 1):
-            if out.size(1) >= total:
+def _select_token(logits, do_sample, temperature, top_k):
+    if not do_sample:
+        return logits.argmax(dim=-1, keepdim=True)
+    logits = logits.float() / max(temperature, 1e-5)
+    if top_k > 0:
+        kth = logits.topk(top_k, dim=-1).values[..., -1, None]
+        logits = logits.masked_fill(logits < kth, float("-inf"))
+    probs = torch.softmax(logits, dim=-1)
+    return torch.multinomial(probs, 1)
+
+
+@torch.no_grad()
+def kv_generate(module, input_ids, *, n_layers, kv_heads, head_dim, max_seq,
+                dtype, max_new_tokens=32, do_sample=False, temperature=1.0,
+                top_k=0, eos_token_id=None):
+    """Shared KV-cached generation loop (used by InferenceEngine and the
+    hybrid RLHF engine)."""
+    device = input_ids.device
+    B, S = input_ids.shape
+    total = min(S + max_new_tokens, max_seq)
+    kv = StaticKVCache(n_layers, B, kv_heads, total, head_dim, dtype=dtype,
+                       device=device)
+    out = input_ids
+    positions = torch.arange(S, device=device,
+                             dtype=torch.int32).expand(B, S).contiguous()
+    logits = module(input_ids, positions=positions, kv_cache=kv)
+    kv.advance()
+    next_tok = _select_token(logits[:, -1], do_sample, temperature, top_k)
+    out = torch.cat([out, next_tok], dim=1)
+    finished = torch.zeros(B, dtype=torch.bool, device=device)
+
+    for _ in range(max_new_tokens - 1):
+        if out.size(1) >= total:
+            break
+        if eos_token_id is not None:
+            finished |= next_tok.squeeze(1) == eos_token_id
+            if bool(finished.all()):
                 break
-            if eos_token_id is not None:
-                finished |= next_tok.squeeze(1) == eos_token_id
-                if bool(finished.all()):
-                    break
-            pos = torch.full((B, 1), self._kv.cur_len, device=self.device,
-                             dtype=torch.int32)
-            logits = m(next_tok, positions=pos, kv_cache=self._kv)
-            self._kv.advance()
-            next_tok = self._select(logits[:, -1], do_sample, temperature,
-                                    top_k)
-            out = torch.cat([out, next_tok], dim=1)
-        return out
-
-    def _select(self, logits, do_sample, temperature, top_k):
-        if not do_sample:
-            return logits.argmax(dim=-1, keepdim=True)
-        logits = logits.float() / max(temperature, 1e-5)
-        if top_k > 0:
-            kth = logits.topk(top_k, dim=-1).values[..., -1, None]
-            logits = logits.masked_fill(logits < kth, float("-inf"))
-        probs = torch.softmax(logits, dim=-1)
-        return torch.multinomial(probs, 1)
+        pos = torch.full((B, 1), kv.cur_len, device=device, dtype=torch.int32)
+        logits = module(next_tok, positions=pos, kv_cache=kv)
+        kv.advance()
+        next_tok = _select_token(logits[:, -1], do_sample, temperature, top_k)
+        out = torch.cat([out, next_tok], dim=1)
+    return out

@@ -149,6 +149,9 @@ class ZeroStage3Optimizer:
         self._register_hooks()
         self.fused_adam_fn = self._try_fused_adam()
 
+        from .partition import register_stage3
+        register_stage3(self)
+
         n_persist = sum(1 for u in self.units if u.persist)
         log_dist(f"ZeRO stage 3: world={self.world_size} units={len(self.units)} "
                  f"(persistent={n_persist}) "
@@ -772,20 +775,31 @@ class ZeroStage3Optimizer:
             self.modifier_rank = modifier_rank
 
         def __enter__(self):
+            # pin as persistent for the duration so the forward-hook release
+            # machinery cannot free the buffers mid-context (e.g. when a
+            # hybrid-engine generate() runs forwards inside)
+            self._was_persist = [u.persist for u in self.units]
             for u in self.units:
                 self.opt._launch_gather(u)
+                u.persist = True
             for u in self.units:
                 self.opt._make_available(u)
             return self
 
         def __exit__(self, *exc):
             with torch.no_grad():
-                for u in self.units:
-                    if self.modifier_rank is not None:
+                for u, was in zip(self.units, self._was_persist):
+                    u.persist = was
+                    if u.full is None:
+                        continue
+                    if self.modifier_rank is not None and \
+                            self.opt.world_size > 1:
                         dist.broadcast(u.full, src=self.modifier_rank,
                                        group=self.opt.dp_group)
-                    u.shard.copy_(u.full[self.opt.rank * u.shard_size:
-                                         (self.opt.rank + 1) * u.shard_size])
+                    if u.full is not u.shard:  # ws=1 aliases them
+                        u.shard.copy_(u.full[self.opt.rank * u.shard_size:
+                                             (self.opt.rank + 1) *
+                                             u.shard_size])
                     if not u.persist:
                         self.opt._release(u)
             return False
