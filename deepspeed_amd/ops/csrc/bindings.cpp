@@ -39,6 +39,13 @@ extern "C" void ds_groupwise_quant(const void* x, int dtype, void* q,
 extern "C" void ds_groupwise_dequant(const void* q, const float* scales,
                                      void* out, int dtype, long long n,
                                      int group_size, int bits, void* stream);
+extern "C" void* ds_aio_create(long long block_size, int n_threads);
+extern "C" void ds_aio_destroy(void* h);
+extern "C" int ds_aio_pwrite(void* h, const void* data, long long nbytes,
+                             const char* path);
+extern "C" int ds_aio_pread(void* h, void* data, long long nbytes,
+                            const char* path);
+extern "C" int ds_aio_wait(void* h);
 
 namespace {
 
@@ -219,6 +226,42 @@ at::Tensor groupwise_dequant(at::Tensor q, at::Tensor scales, int64_t numel,
   return out;
 }
 
+class AioHandle {
+ public:
+  AioHandle(int64_t block_size, int n_threads)
+      : h_(ds_aio_create(block_size, n_threads)) {}
+  ~AioHandle() { ds_aio_destroy(h_); }
+  void async_pwrite(at::Tensor t, const std::string& path) {
+    TORCH_CHECK(!t.is_cuda() && t.is_contiguous(),
+                "aio: host contiguous tensors only");
+    keep_.push_back(t);  // hold storage until wait()
+    TORCH_CHECK(ds_aio_pwrite(h_, t.data_ptr(),
+                              t.numel() * t.element_size(),
+                              path.c_str()) == 0, "aio pwrite setup failed");
+  }
+  void async_pread(at::Tensor t, const std::string& path) {
+    TORCH_CHECK(!t.is_cuda() && t.is_contiguous(),
+                "aio: host contiguous tensors only");
+    keep_.push_back(t);
+    TORCH_CHECK(ds_aio_pread(h_, t.data_ptr(),
+                             t.numel() * t.element_size(),
+                             path.c_str()) == 0, "aio pread setup failed");
+  }
+  int wait() {
+    int e;
+    {
+      pybind11::gil_scoped_release nogil;
+      e = ds_aio_wait(h_);
+    }
+    keep_.clear();
+    return e;
+  }
+
+ private:
+  void* h_;
+  std::vector<at::Tensor> keep_;
+};
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -235,4 +278,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Groupwise symmetric int8/int4 quantization");
   m.def("groupwise_dequant", &groupwise_dequant,
         "Groupwise symmetric int8/int4 dequantization");
+  pybind11::class_<AioHandle>(m, "AioHandle")
+      .def(pybind11::init<int64_t, int>(),
+           pybind11::arg("block_size") = 1 << 20,
+           pybind11::arg("n_threads") = 8)
+      .def("async_pwrite", &AioHandle::async_pwrite)
+      .def("async_pread", &AioHandle::async_pread)
+      .def("wait", &AioHandle::wait);
 }

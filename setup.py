@@ -21,6 +21,7 @@ ext = CUDAExtension(
     sources=[
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "cpu_adam.cpp"),
+        os.path.join(CSRC, "aio.cpp"),
         os.path.join(CSRC, "adam.hip"),
         os.path.join(CSRC, "quantize.hip"),
         os.path.join(CSRC, "norms.hip"),
