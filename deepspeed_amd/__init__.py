@@ -124,3 +124,9 @@ def zero_init(config=None, **kwargs):
 
 
 from .runtime import zero  # noqa: E402  (deepspeed.zero parity namespace)
+
+
+def tp_model_init(model, tp_size: int, dtype=None, config=None, **kwargs):
+    """Tensor-parallel training init (reference deepspeed/__init__.py:369)."""
+    from .runtime.tensor_parallel import tp_model_init as _tp
+    return _tp(model, tp_size, dtype)
