@@ -353,9 +353,26 @@ class Engine(torch.nn.Module):
             self._buffered_allreduce_fallback()
 
     @torch.no_grad()
+    def _sparse_allreduce(self, grad):
+        """Average a sparse gradient (sparse embeddings) across DP by
+        all-gathering indices+values (reference engine.py:2627)."""
+        grad = grad.coalesce()
+        parts = [None] * self.dp_world_size
+        dist.all_gather_object(parts, (grad.indices().cpu(),
+                                       grad.values().cpu()),
+                               group=self.dp_group)
+        idx = torch.cat([p[0] for p in parts], dim=1).to(grad.device)
+        val = torch.cat([p[1] for p in parts], dim=0).to(grad.device)
+        return torch.sparse_coo_tensor(idx, val / self.dp_world_size,
+                                       grad.shape).coalesce()
+
+    @torch.no_grad()
     def _buffered_allreduce_fallback(self, elements_per_buffer=MEMORY_OPT_ALLREDUCE_SIZE):
+        for p in self.module.parameters():
+            if p.grad is not None and p.grad.is_sparse:
+                p.grad = self._sparse_allreduce(p.grad)
         grads = [p.grad for p in self.module.parameters()
-                 if p.grad is not None]
+                 if p.grad is not None and not p.grad.is_sparse]
         bucket, bucket_elems = [], 0
         from torch._utils import _flatten_dense_tensors, _unflatten_dense_tensors
 
