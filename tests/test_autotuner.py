@@ -1,0 +1,54 @@
+"""Autotuner tests (reference contract: tests/unit/autotuning/test_autotuning.py)."""
+
+import torch
+
+
+def test_autotuner_picks_best_with_fake_runner():
+    from deepspeed_amd.autotuning import Autotuner
+
+    # synthetic cost model: throughput grows with mb until OOM at 16;
+    # stage 2 is best at the chosen mb
+    def runner(cfg):
+        mb = cfg["train_micro_batch_size_per_gpu"]
+        stage = cfg["zero_optimization"]["stage"]
+        if mb >= 16:
+            return 0.0  # OOM
+        base = {0: 90.0, 1: 95.0, 2: 120.0, 3: 100.0}[stage]
+        return base * min(mb, 8) / 8
+
+    tuner = Autotuner({"optimizer": {"type": "AdamW", "params": {}}},
+                      runner=runner)
+    best = tuner.tune()
+    assert best["train_micro_batch_size_per_gpu"] == 8
+    assert best["zero_optimization"]["stage"] == 2
+    assert tuner.best_metric == 120.0
+    assert len(tuner.results) >= 5
+
+
+def test_autotuner_end_to_end_inprocess():
+    """Default runner: real engine steps on a tiny model (CPU)."""
+    from .common import run_local
+
+    def worker(rank, world):
+        from deepspeed_amd.autotuning import Autotuner
+        from deepspeed_amd.models import GPT2ForCausalLM, gpt2_tiny
+
+        def model_factory():
+            torch.manual_seed(0)
+            return GPT2ForCausalLM(gpt2_tiny())
+
+        def sample_factory(mb):
+            ids = torch.randint(0, 128, (mb, 16))
+            return (ids,), {"labels": ids}
+
+        tuner = Autotuner(
+            {"optimizer": {"type": "AdamW", "params": {"lr": 1e-4}}},
+            micro_batch_sizes=[1, 2], zero_stages=[1, 2],
+            model_factory=model_factory, sample_factory=sample_factory,
+            steps=2)
+        best = tuner.tune()
+        assert best["train_micro_batch_size_per_gpu"] in (1, 2)
+        assert best["zero_optimization"]["stage"] in (1, 2)
+        assert tuner.best_metric > 0
+
+    run_local(worker)
