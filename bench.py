@@ -23,7 +23,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=8)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--seq-len", type=int, default=4096)
-    p.add_argument("--micro-batch", type=int, default=2)
+    p.add_argument("--micro-batch", type=int, default=8)
     p.add_argument("--gas", type=int, default=1)
     p.add_argument("--zero-stage", type=int,
                    default=int(os.environ.get("BENCH_ZERO_STAGE", 3)))
