@@ -110,3 +110,30 @@ def test_engine_compile_flag():
         assert torch.isfinite(loss)
 
     run_local(worker)
+
+
+def test_curriculum_data_sampler():
+    from deepspeed_amd.runtime.data_pipeline.curriculum_scheduler import \
+        CurriculumScheduler
+    from deepspeed_amd.runtime.data_pipeline.data_sampler import \
+        DeepSpeedDataSampler
+    diffs = [10, 20, 30, 40] * 25  # 100 samples
+    sched = CurriculumScheduler({
+        "curriculum_type": "fixed_discrete", "min_difficulty": 10,
+        "max_difficulty": 40,
+        "schedule_config": {"difficulty": [10, 40], "max_step": [1]}})
+    s = DeepSpeedDataSampler(diffs, sched, batch_size=4, dp_rank=0, dp_size=2)
+    batches = list(s)
+    assert all(len(b) == 4 for b in batches)
+    # first batches: only difficulty-10 samples are eligible
+    assert all(diffs[i] == 10 for i in batches[0])
+    # later batches include harder samples (difficulty opens at step 2)
+    assert any(diffs[i] > 10 for b in batches[2:] for i in b)
+    # DP disjointness: rank 1 gets different indices for the same epoch/step
+    s1 = DeepSpeedDataSampler(diffs, CurriculumScheduler({
+        "curriculum_type": "fixed_discrete", "min_difficulty": 10,
+        "max_difficulty": 40,
+        "schedule_config": {"difficulty": [10, 40], "max_step": [1]}}),
+        batch_size=4, dp_rank=1, dp_size=2)
+    b1 = next(iter(s1))
+    assert not set(batches[0]) & set(b1)
