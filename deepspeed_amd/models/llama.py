@@ -87,6 +87,14 @@ class LlamaAttention(nn.Module):
         q = apply_rope(q, cos, sin, positions)
         k = apply_rope(k, cos, sin, positions)
 
+        if kv_cache is None and attention_fn is None:
+            from ..ops.attention import flash_attn_available, flash_attn_fwd
+            if flash_attn_available(q, k):
+                # hand-written CDNA4 MFMA flash kernel, BSHD layout — skips
+                # the transpose entirely (inference/no-grad prefill path)
+                o = flash_attn_fwd(q, k, v, causal=True)
+                return self.o_proj(o.reshape(B, S, -1))
+
         # [B, H, S, D] for SDPA
         q = q.transpose(1, 2)
         k = k.transpose(1, 2)

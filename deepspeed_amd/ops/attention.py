@@ -1,0 +1,47 @@
+"""Flash-attention forward (hand-written CDNA4 MFMA kernel).
+
+See csrc/attention.hip for the kernel design (probe-verified MFMA
+layouts, swapped-QK^T online softmax). This wrapper owns the V transpose
+(the kernel consumes vt[B,Hkv,D,S] so its PV fragments are contiguous
+16-byte loads) and the eligibility check, falling back to torch SDPA where
+the kernel does not apply (training backward, D!=128, ragged seq, CPU).
+"""
+
+import math
+from typing import Optional
+
+import torch
+
+from ._loader import get_ext
+
+
+def flash_attn_available(q: torch.Tensor, k: torch.Tensor) -> bool:
+    return (get_ext() is not None and q.is_cuda
+            and q.dtype == torch.bfloat16
+            and q.size(-1) == 128 and q.size(1) % 32 == 0
+            and q.size(1) == k.size(1)
+            and not torch.is_grad_enabled())
+
+
+def flash_attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   causal: bool = True,
+                   scale: Optional[float] = None) -> torch.Tensor:
+    """q,k,v: [B, S, H(kv), D] bf16 -> o [B, S, H, D]. Inference/prefill
+    only (no autograd)."""
+    ext = get_ext()
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.size(-1))
+    vt = v.permute(0, 2, 3, 1).contiguous()  # [B, Hkv, D, S]
+    return ext.flash_attn_fwd(q.contiguous(), k.contiguous(), vt, scale,
+                              causal)
+
+
+def sdpa_reference(q, k, v, causal=True, scale=None):
+    """Plain fp32 reference in the same [B,S,H,D] layout (for tests)."""
+    qt, kt, vt = (t.transpose(1, 2).float() for t in (q, k, v))
+    rep = qt.shape[1] // kt.shape[1]
+    if rep > 1:
+        kt = kt.repeat_interleave(rep, dim=1)
+        vt = vt.repeat_interleave(rep, dim=1)
+    o = torch.nn.functional.scaled_dot_product_attention(
+        qt, kt, vt, is_causal=causal, scale=scale)
+    return o.transpose(1, 2).to(q.dtype)

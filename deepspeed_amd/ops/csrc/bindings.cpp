@@ -46,6 +46,9 @@ extern "C" int ds_aio_pwrite(void* h, const void* data, long long nbytes,
 extern "C" int ds_aio_pread(void* h, void* data, long long nbytes,
                             const char* path);
 extern "C" int ds_aio_wait(void* h);
+extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
+                             void* o, int B, int S, int H, int Hkv,
+                             float scale, int causal, void* stream);
 
 namespace {
 
@@ -226,6 +229,24 @@ at::Tensor groupwise_dequant(at::Tensor q, at::Tensor scales, int64_t numel,
   return out;
 }
 
+at::Tensor flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor vt,
+                          double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
+              "flash_fwd: bf16 GPU tensors only");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && vt.is_contiguous(),
+              "flash_fwd: contiguous tensors required");
+  const int B = q.size(0), S = q.size(1), H = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128, "flash_fwd: head_dim must be 128");
+  TORCH_CHECK(S % 32 == 0, "flash_fwd: seq must be a multiple of 32");
+  TORCH_CHECK(k.size(1) == S && vt.size(3) == S && vt.size(2) == D &&
+              vt.size(1) == Hkv && H % Hkv == 0, "flash_fwd: shape mismatch");
+  auto o = at::empty_like(q);
+  ds_flash_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(), B, S,
+               H, Hkv, (float)scale, causal ? 1 : 0, cur_stream());
+  return o;
+}
+
 class AioHandle {
  public:
   AioHandle(int64_t block_size, int n_threads)
@@ -278,6 +299,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Groupwise symmetric int8/int4 quantization");
   m.def("groupwise_dequant", &groupwise_dequant,
         "Groupwise symmetric int8/int4 dequantization");
+  m.def("flash_attn_fwd", &flash_attn_fwd,
+        "MFMA flash-attention forward (bf16, D=128, GQA, causal)");
   pybind11::class_<AioHandle>(m, "AioHandle")
       .def(pybind11::init<int64_t, int>(),
            pybind11::arg("block_size") = 1 << 20,
