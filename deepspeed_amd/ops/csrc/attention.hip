@@ -3,25 +3,26 @@
 // Replaces the aotriton SDPA forward on the inference/prefill path
 // (reference analogue: csrc/transformer/inference softmax+GEMM chain and
 // inference/v2 blocked flash). Structure follows the CDNA4 guide's
-// verified flash ladder (cdna_hip_programming.md §attention): swapped
-// QK^T so softmax is per-lane, online-softmax rescale, MFMA everywhere.
+// verified flash ladder (cdna_hip_programming.md §attention).
 //
-// Design (v1, correctness-first):
-// * one WAVE per 32 query rows; grid (ceil(S/32), H, B), 64 threads.
-//   No LDS at all — K is read as A-fragments straight from global (the
-//   guide's pitfall #7: at these sizes L2 serves K/V better than staging),
-//   V is consumed from a pre-transposed copy vt[B,Hkv,D,S] so its
-//   A-fragments are contiguous 16-byte loads.
+// v2 structure (8-wave K/V reuse):
+// * 512-thread workgroup = 8 waves x 32 query rows (Q tile = 256 rows);
+//   grid (ceil(S/256), H, B). K and V^T tiles are staged in LDS ONCE per
+//   kv step and consumed by all 8 waves — global K/V traffic drops 8x vs
+//   the naive per-wave version (which measured HBM-bound at 164 TF).
+// * XOR swizzle on the LDS tiles (guide §6 G4: `byte ^= ((row&7)<<4)`)
+//   so the MFMA fragment reads (row-strided ds_read_b128) don't bank
+//   conflict.
 // * mfma_f32_32x32x16_bf16 with probe-verified layouts
 //   (scripts/mfma_probe.hip, run on gfx950 2026-08-20):
 //     A: row = lane&31, k = reg + 8*(lane>>5)
 //     B: col = lane&31, k = reg + 8*(lane>>5)
 //     C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
-// * swapped S^T = mfma(A=K, B=Q): each lane then owns 16 score entries of
-//   ONE query column -> row max/sum needs 15 VALU ops + one shfl_xor(32).
+// * swapped S^T = mfma(A=K, B=Q): each lane owns 16 score entries of ONE
+//   query column -> row max/sum is 15 VALU ops + one shfl_xor(32).
 // * P re-layout for PV (C/D -> B fragment) is a register permutation
 //   within the (lane, lane^32) pair: reg' = (kv&3) + 4*(kv>>3),
-//   cross-half values fetched with shfl_xor(32).
+//   cross-half values via shfl_xor(32).
 //
 // Layouts: q,k [B,S,H(kv),128] bf16; vt [B,Hkv,128,S] bf16; o [B,S,H,128].
 // Requires D=128, S % 32 == 0 (wrapper falls back to SDPA otherwise).
@@ -31,20 +32,14 @@
 namespace {
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8s;
+typedef __attribute__((ext_vector_type(8))) short lds_chunk;  // 16 B
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
-constexpr int QB = 32;   // query rows per wave
-constexpr int KB = 32;   // kv rows per tile
-constexpr int D = 128;   // head dim
-
-__device__ __forceinline__ float bf2f(short s) {
-  union {
-    float f;
-    unsigned u;
-  } c;
-  c.u = ((unsigned)(unsigned short)s) << 16;
-  return c.f;
-}
+constexpr int QB = 32;    // query rows per wave
+constexpr int NWAVE = 8;  // waves per workgroup
+constexpr int QTILE = QB * NWAVE;
+constexpr int KB = 32;  // kv rows per tile
+constexpr int D = 128;  // head dim
 
 __device__ __forceinline__ short f2bf(float f) {
   union {
@@ -56,29 +51,46 @@ __device__ __forceinline__ short f2bf(float f) {
   return (short)(r >> 16);
 }
 
+// LDS tile helpers: tiles stored as 16-byte chunks with an XOR swizzle on
+// the chunk index so row-strided fragment reads spread across banks.
+// K tile: [KB][D] bf16 = 32 rows x 16 chunks. V^T tile: [D][KB] = 128 x 4.
+__device__ __forceinline__ int k_sw(int row, int chunk) {
+  return row * 16 + (chunk ^ (row & 7));
+}
+__device__ __forceinline__ int v_sw(int row, int chunk) {
+  return row * 4 + (chunk ^ (row & 3));
+}
+
 template <bool CAUSAL>
-__global__ __launch_bounds__(64) void flash_fwd_kernel(
+__global__ __launch_bounds__(QB * NWAVE) void flash_fwd_kernel(
     const short* __restrict__ q,   // [B, S, H, D]
     const short* __restrict__ k,   // [B, S, Hkv, D]
     const short* __restrict__ vt,  // [B, Hkv, D, S]
     short* __restrict__ o,         // [B, S, H, D]
     const int B, const int S, const int H, const int Hkv,
     const float scale) {
-  const int lane = threadIdx.x;
-  const int q0 = blockIdx.x * QB;
+  __shared__ lds_chunk kt_lds[KB * 16];   // 8 KB
+  __shared__ lds_chunk vt_lds[D * 4];     // 8 KB
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int q0b = blockIdx.x * QTILE;  // block's first q row
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int hkv = h / (H / Hkv);
-  if (q0 >= S) return;
 
-  const int col = lane & 31;        // my query row within the tile
-  const int half = lane >> 5;       // k-slot half
-  const int qrow = q0 + col;
+  const int col = lane & 31;  // my query row within the wave / kv row in A
+  const int half = lane >> 5;
+  const int q0w = q0b + wid * QB;
+  const int qrow = q0w + col;
+  const bool q_ok = qrow < S;
+  const int qload = q_ok ? qrow : S - 1;
 
   // ---- preload Q as B-fragments: qf[kk] holds d = kk*16 + 8*half + [0,8)
   bf16x8s qf[8];
   {
-    const short* qp = q + (((long long)b * S + qrow) * H + h) * D;
+    const short* qp = q + (((long long)b * S + qload) * H + h) * D;
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk)
       qf[kk] = *(const bf16x8s*)(qp + kk * 16 + 8 * half);
@@ -88,18 +100,36 @@ __global__ __launch_bounds__(64) void flash_fwd_kernel(
   float l_run = 0.f;
   f32x16 oacc[4] = {};  // O^T accumulators, one per 32-wide d block
 
-  const int kv_end = CAUSAL ? min(S, q0 + QB) : S;
-  const long long k_bh = ((long long)b * S) * Hkv + hkv;  // row stride below
+  const int kv_end_blk = CAUSAL ? min(S, q0b + QTILE) : S;
+  const int kv_end_wave = CAUSAL ? min(S, q0w + QB) : S;
+  const long long k_base = (((long long)b * S) * Hkv + hkv) * D;
   const short* vtp = vt + (((long long)b * Hkv + hkv) * D) * S;
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+  for (int kv0 = 0; kv0 < kv_end_blk; kv0 += KB) {
+    // ---- cooperative stage: K tile (512 chunks) + V^T tile (512 chunks)
+    __syncthreads();
+    {
+      // K: chunk g -> row g/16, chunk g%16; global row kv0+row
+      const int row = tid >> 4, c = tid & 15;
+      const int kvr = min(kv0 + row, S - 1);
+      kt_lds[k_sw(row, c)] = *(const bf16x8s*)(
+          k + k_base + (long long)kvr * Hkv * D + c * 8);
+      // V^T: chunk g -> row g/4 (=d), chunk g%4; global col kv0 + c*8
+      const int vrow = tid >> 2, vc = tid & 3;
+      const int kvc = min(kv0 + vc * 8, S - 8);  // S%8==0 guaranteed
+      vt_lds[v_sw(vrow, vc)] = *(const bf16x8s*)(
+          vtp + (long long)vrow * S + kvc);
+    }
+    __syncthreads();
+
+    if (kv0 >= kv_end_wave) continue;  // past my diagonal: barriers only
+
     // ---- S^T = K · Q^T : A = K rows (kv), B = Q cols (q)
     f32x16 st = {};
-    const short* kp = k + ((k_bh + (long long)kv0 * Hkv) * D);
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {
-      bf16x8s kf = *(const bf16x8s*)(kp + (long long)col * Hkv * D +
-                                     kk * 16 + 8 * half);
+      // A slot: row=col, d chunk = kk*2 + half (8 bf16 each)
+      bf16x8s kf = kt_lds[k_sw(col, kk * 2 + half)];
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[kk], st, 0, 0, 0);
     }
 
@@ -110,7 +140,7 @@ __global__ __launch_bounds__(64) void flash_fwd_kernel(
     for (int r = 0; r < 16; ++r) {
       const int kvl = (r & 3) + 8 * (r >> 2) + 4 * half;
       float s = st[r] * scale;
-      if (CAUSAL && kv0 + kvl > qrow) s = -1e30f;
+      if ((CAUSAL && kv0 + kvl > qrow) || kv0 + kvl >= S) s = -1e30f;
       p[r] = s;
       mt = fmaxf(mt, s);
     }
@@ -148,13 +178,12 @@ __global__ __launch_bounds__(64) void flash_fwd_kernel(
     // ---- O^T += V^T · P : A = V^T rows (d), k = kv; B = P cols (q)
 #pragma unroll
     for (int dblk = 0; dblk < 4; ++dblk) {
-      const short* vp = vtp + (long long)(dblk * 32 + col) * S + kv0;
-      // rescale accumulated output by alpha (once per kv tile)
 #pragma unroll
       for (int r = 0; r < 16; ++r) oacc[dblk][r] *= alpha;
 #pragma unroll
       for (int kk2 = 0; kk2 < 2; ++kk2) {
-        bf16x8s vf = *(const bf16x8s*)(vp + kk2 * 16 + 8 * half);
+        // A slot: row = dblk*32+col, kv chunk = kk2*2 + half
+        bf16x8s vf = vt_lds[v_sw(dblk * 32 + col, kk2 * 2 + half)];
         oacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             vf, pf[kk2], oacc[dblk], 0, 0, 0);
       }
@@ -162,7 +191,8 @@ __global__ __launch_bounds__(64) void flash_fwd_kernel(
   }
 
   // ---- epilogue: O[q][d] = O^T / l
-  const float inv_l = 1.f / l_run;
+  if (!q_ok) return;
+  const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
   short* op = o + (((long long)b * S + qrow) * H + h) * D;
 #pragma unroll
   for (int dblk = 0; dblk < 4; ++dblk) {
@@ -179,14 +209,14 @@ __global__ __launch_bounds__(64) void flash_fwd_kernel(
 extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
                              void* o, int B, int S, int H, int Hkv,
                              float scale, int causal, void* stream) {
-  dim3 grid((S + QB - 1) / QB, H, B);
+  dim3 grid((S + QTILE - 1) / QTILE, H, B);
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   if (causal)
-    hipLaunchKernelGGL((flash_fwd_kernel<true>), grid, dim3(64), 0, st,
-                       (const short*)q, (const short*)k, (const short*)vt,
+    hipLaunchKernelGGL((flash_fwd_kernel<true>), grid, dim3(QB * NWAVE), 0,
+                       st, (const short*)q, (const short*)k, (const short*)vt,
                        (short*)o, B, S, H, Hkv, scale);
   else
-    hipLaunchKernelGGL((flash_fwd_kernel<false>), grid, dim3(64), 0, st,
-                       (const short*)q, (const short*)k, (const short*)vt,
+    hipLaunchKernelGGL((flash_fwd_kernel<false>), grid, dim3(QB * NWAVE), 0,
+                       st, (const short*)q, (const short*)k, (const short*)vt,
                        (short*)o, B, S, H, Hkv, scale);
 }
