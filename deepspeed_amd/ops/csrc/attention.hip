@@ -61,7 +61,8 @@ __device__ __forceinline__ int v_sw(int row, int chunk) {
   return row * 4 + (chunk ^ (row & 3));
 }
 
-template <bool CAUSAL>
+// VAR ablation (debug): bit0 = K from LDS, bit1 = V from LDS (3 = normal)
+template <bool CAUSAL, int VAR = 3>
 __global__ __launch_bounds__(QB * NWAVE) void flash_fwd_kernel(
     const short* __restrict__ q,   // [B, S, H, D]
     const short* __restrict__ k,   // [B, S, Hkv, D]
@@ -129,7 +130,13 @@ __global__ __launch_bounds__(QB * NWAVE) void flash_fwd_kernel(
 #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {
       // A slot: row=col, d chunk = kk*2 + half (8 bf16 each)
-      bf16x8s kf = kt_lds[k_sw(col, kk * 2 + half)];
+      bf16x8s kf;
+      if (VAR & 1)
+        kf = kt_lds[k_sw(col, kk * 2 + half)];
+      else
+        kf = *(const bf16x8s*)(k + k_base +
+                               (long long)min(kv0 + col, S - 1) * Hkv * D +
+                               kk * 16 + 8 * half);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[kk], st, 0, 0, 0);
     }
 
@@ -183,7 +190,12 @@ __global__ __launch_bounds__(QB * NWAVE) void flash_fwd_kernel(
 #pragma unroll
       for (int kk2 = 0; kk2 < 2; ++kk2) {
         // A slot: row = dblk*32+col, kv chunk = kk2*2 + half
-        bf16x8s vf = vt_lds[v_sw(dblk * 32 + col, kk2 * 2 + half)];
+        bf16x8s vf;
+        if (VAR & 2)
+          vf = vt_lds[v_sw(dblk * 32 + col, kk2 * 2 + half)];
+        else
+          vf = *(const bf16x8s*)(vtp + (long long)(dblk * 32 + col) * S +
+                                 kv0 + kk2 * 16 + 8 * half);
         oacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             vf, pf[kk2], oacc[dblk], 0, 0, 0);
       }
@@ -219,4 +231,36 @@ extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
     hipLaunchKernelGGL((flash_fwd_kernel<false>), grid, dim3(QB * NWAVE), 0,
                        st, (const short*)q, (const short*)k, (const short*)vt,
                        (short*)o, B, S, H, Hkv, scale);
+}
+
+extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
+                                 void* o, int B, int S, int H, int Hkv,
+                                 float scale, int variant, void* stream) {
+  dim3 grid((S + QTILE - 1) / QTILE, H, B);
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  switch (variant) {
+    case 0:
+      hipLaunchKernelGGL((flash_fwd_kernel<true, 0>), grid,
+                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         (const short*)k, (const short*)vt, (short*)o, B, S,
+                         H, Hkv, scale);
+      break;
+    case 1:
+      hipLaunchKernelGGL((flash_fwd_kernel<true, 1>), grid,
+                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         (const short*)k, (const short*)vt, (short*)o, B, S,
+                         H, Hkv, scale);
+      break;
+    case 2:
+      hipLaunchKernelGGL((flash_fwd_kernel<true, 2>), grid,
+                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         (const short*)k, (const short*)vt, (short*)o, B, S,
+                         H, Hkv, scale);
+      break;
+    default:
+      hipLaunchKernelGGL((flash_fwd_kernel<true, 3>), grid,
+                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         (const short*)k, (const short*)vt, (short*)o, B, S,
+                         H, Hkv, scale);
+  }
 }

@@ -49,6 +49,9 @@ extern "C" int ds_aio_wait(void* h);
 extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
                              void* o, int B, int S, int H, int Hkv,
                              float scale, int causal, void* stream);
+extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
+                                 void* o, int B, int S, int H, int Hkv,
+                                 float scale, int variant, void* stream);
 
 namespace {
 
@@ -299,6 +302,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Groupwise symmetric int8/int4 quantization");
   m.def("groupwise_dequant", &groupwise_dequant,
         "Groupwise symmetric int8/int4 dequantization");
+  m.def("flash_attn_fwd_dbg",
+        [](at::Tensor q, at::Tensor k, at::Tensor vt, double scale,
+           int64_t variant) {
+          auto o = at::empty_like(q);
+          ds_flash_fwd_dbg(q.data_ptr(), k.data_ptr(), vt.data_ptr(),
+                           o.data_ptr(), q.size(0), q.size(1), q.size(2),
+                           k.size(2), (float)scale, (int)variant,
+                           cur_stream());
+          return o;
+        },
+        "flash fwd ablation (bit0 K-LDS, bit1 V-LDS)");
   m.def("flash_attn_fwd", &flash_attn_fwd,
         "MFMA flash-attention forward (bf16, D=128, GQA, causal)");
   pybind11::class_<AioHandle>(m, "AioHandle")
