@@ -1,3 +1,4 @@
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import math, torch
 from deepspeed_amd.ops import _C
 from deepspeed_amd.ops.attention import sdpa_reference
