@@ -38,6 +38,7 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 constexpr int QB = 32;    // query rows per wave
 constexpr int NWAVE = 8;  // waves per workgroup
 constexpr int QTILE = QB * NWAVE;
+constexpr int TPB = NWAVE * 64;  // 8 waves x 64 lanes
 constexpr int KB = 32;  // kv rows per tile
 constexpr int D = 128;  // head dim
 
@@ -63,7 +64,7 @@ __device__ __forceinline__ int v_sw(int row, int chunk) {
 
 // VAR ablation (debug): bit0 = K from LDS, bit1 = V from LDS (3 = normal)
 template <bool CAUSAL, int VAR = 3>
-__global__ __launch_bounds__(QB * NWAVE) void flash_fwd_kernel(
+__global__ __launch_bounds__(TPB) void flash_fwd_kernel(
     const short* __restrict__ q,   // [B, S, H, D]
     const short* __restrict__ k,   // [B, S, Hkv, D]
     const short* __restrict__ vt,  // [B, Hkv, D, S]
@@ -224,11 +225,11 @@ extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
   dim3 grid((S + QTILE - 1) / QTILE, H, B);
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   if (causal)
-    hipLaunchKernelGGL((flash_fwd_kernel<true>), grid, dim3(QB * NWAVE), 0,
+    hipLaunchKernelGGL((flash_fwd_kernel<true>), grid, dim3(TPB), 0,
                        st, (const short*)q, (const short*)k, (const short*)vt,
                        (short*)o, B, S, H, Hkv, scale);
   else
-    hipLaunchKernelGGL((flash_fwd_kernel<false>), grid, dim3(QB * NWAVE), 0,
+    hipLaunchKernelGGL((flash_fwd_kernel<false>), grid, dim3(TPB), 0,
                        st, (const short*)q, (const short*)k, (const short*)vt,
                        (short*)o, B, S, H, Hkv, scale);
 }
@@ -241,25 +242,25 @@ extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
   switch (variant) {
     case 0:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 0>), grid,
-                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         dim3(TPB), 0, st, (const short*)q,
                          (const short*)k, (const short*)vt, (short*)o, B, S,
                          H, Hkv, scale);
       break;
     case 1:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 1>), grid,
-                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         dim3(TPB), 0, st, (const short*)q,
                          (const short*)k, (const short*)vt, (short*)o, B, S,
                          H, Hkv, scale);
       break;
     case 2:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 2>), grid,
-                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         dim3(TPB), 0, st, (const short*)q,
                          (const short*)k, (const short*)vt, (short*)o, B, S,
                          H, Hkv, scale);
       break;
     default:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 3>), grid,
-                         dim3(QB * NWAVE), 0, st, (const short*)q,
+                         dim3(TPB), 0, st, (const short*)q,
                          (const short*)k, (const short*)vt, (short*)o, B, S,
                          H, Hkv, scale);
   }
