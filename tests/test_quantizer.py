@@ -86,3 +86,25 @@ def _qwz_worker(rank, world):
 
 def test_zero3_quantized_weight_allgather():
     run_distributed(_qwz_worker, world_size=2)
+
+
+def test_fp8_quantize_roundtrip():
+    from deepspeed_amd.ops.fp_quantizer import fp8_dequantize, fp8_quantize
+    torch.manual_seed(2)
+    x = torch.randn(5000) * 4
+    q, s = fp8_quantize(x, group_size=512)
+    assert q.dtype == torch.float8_e4m3fn and s.numel() == 10
+    y = fp8_dequantize(q, s, 5000, 512, torch.float32)
+    # e4m3: ~2 decimal digits of precision after groupwise scaling
+    rel = (y - x).abs() / x.abs().clamp(min=1e-3)
+    assert rel.median() < 0.04 and (y - x).abs().max() < x.abs().max() * 0.1
+
+
+def test_fp8_quantizer_class():
+    from deepspeed_amd.ops.fp_quantizer import FP8Quantizer
+    fq = FP8Quantizer(group_size=256)
+    x = torch.randn(16, 64, dtype=torch.bfloat16)
+    q, s = fq.quantize(x)
+    y = fq.dequantize(q, s)
+    assert y.shape == x.shape and y.dtype == torch.bfloat16
+    assert (y.float() - x.float()).abs().mean() < 0.05
