@@ -16,6 +16,12 @@ from ._loader import get_ext
 
 
 def flash_attn_available(q: torch.Tensor, k: torch.Tensor) -> bool:
+    # Numerics-verified but not yet past aotriton SDPA's throughput
+    # (130 vs 400 TF — the guide ladder's KVBLK=64/async-stage/defer-max
+    # steps are pending), so the model hot path keeps SDPA unless opted in.
+    import os
+    if os.environ.get("DS_AMD_FLASH") != "1":
+        return False
     return (get_ext() is not None and q.is_cuda
             and q.dtype == torch.bfloat16
             and q.size(-1) == 128 and q.size(1) % 32 == 0
