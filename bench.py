@@ -28,7 +28,8 @@ def parse_args():
     p.add_argument("--zero-stage", type=int,
                    default=int(os.environ.get("BENCH_ZERO_STAGE", 3)))
     p.add_argument("--model", type=str, default="llama3-8b",
-                   choices=["llama3-8b", "llama3-70b", "llama-mini", "tiny"])
+                   choices=["llama3-8b", "llama3-70b", "llama-mini", "tiny",
+                            "phi3-mini", "mixtral-8x7b", "mixtral-mini"])
     p.add_argument("--offload", action="store_true",
                    help="ZeRO-Offload optimizer states to host DRAM")
     p.add_argument("--activation-checkpointing", action="store_true")
@@ -36,12 +37,19 @@ def parse_args():
     return p.parse_args()
 
 
-def build_model(name):
-    from deepspeed_amd.models import (LlamaForCausalLM, llama3_8b, llama3_70b,
-                                      llama_mini, llama_tiny)
-    cfg = {"llama3-8b": llama3_8b, "llama3-70b": llama3_70b,
-           "llama-mini": llama_mini, "tiny": llama_tiny}[name]()
+def build_model(name, world_size=1):
     torch.manual_seed(42)
+    if name.startswith("mixtral"):
+        from deepspeed_amd.models import (MixtralForCausalLM, mixtral_8x7b,
+                                          mixtral_mini)
+        f = mixtral_8x7b if name == "mixtral-8x7b" else mixtral_mini
+        cfg = f(ep_size=min(world_size, 8))
+        return MixtralForCausalLM(cfg), cfg
+    from deepspeed_amd.models import (LlamaForCausalLM, llama3_8b, llama3_70b,
+                                      llama_mini, llama_tiny, phi3_mini)
+    cfg = {"llama3-8b": llama3_8b, "llama3-70b": llama3_70b,
+           "llama-mini": llama_mini, "tiny": llama_tiny,
+           "phi3-mini": phi3_mini}[name]()
     return LlamaForCausalLM(cfg), cfg
 
 
@@ -59,7 +67,7 @@ def main():
     if use_gpu:
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
 
-    model, cfg = build_model(args.model)
+    model, cfg = build_model(args.model, world_size)
     n_params = model.num_parameters()
 
     ds_config = {

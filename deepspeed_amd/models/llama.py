@@ -58,6 +58,16 @@ def llama_tiny():
                        max_seq_len=128)
 
 
+def phi3_mini():
+    """Phi-3-mini 3.8B shapes (BASELINE config 5: small-model ZeRO-2 path).
+    Phi-3's decoder is Llama-architecture (RoPE, SwiGLU, RMSNorm) with MHA
+    (no GQA) and a 32k vocab."""
+    return LlamaConfig(vocab_size=32064, hidden_size=3072,
+                       intermediate_size=8192, num_layers=32, num_heads=32,
+                       num_kv_heads=32, max_seq_len=4096, rope_theta=10000.0,
+                       tie_embeddings=False)
+
+
 def llama_mini():
     """~0.5B for single-GPU smoke runs."""
     return LlamaConfig(vocab_size=32000, hidden_size=1024,
