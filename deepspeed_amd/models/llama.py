@@ -97,6 +97,12 @@ class LlamaAttention(nn.Module):
         q = apply_rope(q, cos, sin, positions)
         k = apply_rope(k, cos, sin, positions)
 
+        if getattr(self, "sp_group", None) is not None and kv_cache is None:
+            # Ulysses: all-to-all scatters heads / gathers the sequence so
+            # each rank runs full-sequence causal attention on H/P heads
+            o = self._ulysses_attn(q, k, v)
+            return self.o_proj(o.reshape(B, S, -1))
+
         if kv_cache is None and attention_fn is None:
             from ..ops.attention import flash_attn_available, flash_attn_fwd
             if flash_attn_available(q, k):
@@ -119,6 +125,32 @@ class LlamaAttention(nn.Module):
             o = sdpa_gqa(q, k, v, causal=q.size(2) == k.size(2))
         o = o.transpose(1, 2).reshape(B, S, -1)
         return self.o_proj(o)
+
+
+def _bshd_causal_attention(q, k, v):
+    """Local attention in [b, s, H, d] layout (Ulysses local_attn)."""
+    q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+    return sdpa_gqa(q, k, v, causal=True).transpose(1, 2)
+
+
+def enable_ulysses(model, sp_group=None):
+    """Wire Ulysses sequence parallelism into every attention module.
+    Requires num_heads % sp_world == 0 and num_kv_heads % sp_world == 0."""
+    from ..parallel import groups
+    from ..sequence import DistributedAttention
+    sp_group = sp_group or groups.get_sequence_parallel_group()
+    assert sp_group is not None, "initialize_sequence_parallel first"
+    import torch.distributed as td
+    P = td.get_world_size(sp_group)
+    for mod in model.modules():
+        if isinstance(mod, LlamaAttention):
+            assert mod.num_heads % P == 0 and mod.num_kv_heads % P == 0, \
+                f"heads {mod.num_heads}/{mod.num_kv_heads} not divisible " \
+                f"by sp={P}"
+            mod.sp_group = sp_group
+            mod._ulysses_attn = DistributedAttention(_bshd_causal_attention,
+                                                     sp_group)
+    return model
 
 
 def sdpa_gqa(q, k, v, causal=True):

@@ -60,8 +60,13 @@ def get_data_parallel_group():
         return _mpu.get_data_parallel_group()
     if _DATA_PARALLEL_GROUP is not None:
         return _DATA_PARALLEL_GROUP
-    if _SEQUENCE_DATA_PARALLEL_GROUP is not None:
-        return _SEQUENCE_DATA_PARALLEL_GROUP
+    if _SEQUENCE_PARALLEL_GROUP is not None:
+        # Ulysses: the model is replicated over SP and each SP rank holds a
+        # different sequence chunk, so ZeRO must shard AND average grads
+        # over the full DPxSP mesh — the average over world equals
+        # (1/dp)sum_dp (1/sp)sum_sp d(local mean loss), i.e. the exact
+        # gradient of the global token-mean loss with NO extra scaling.
+        return _ensure_world_group()
     return _ensure_world_group()
 
 

@@ -327,8 +327,9 @@ class Engine(torch.nn.Module):
 
         if self.gradient_accumulation_steps > 1 and scale_wrt_gas:
             loss = loss / self.gradient_accumulation_steps
-        if self.seq_parallel_world_size > 1:
-            loss = loss / self.seq_parallel_world_size
+        # Ulysses note: NO extra loss scaling for SP — ZeRO averages grads
+        # over the full DPxSP mesh (see parallel/groups.py), which already
+        # yields the gradient of the global token-mean loss.
 
         if hasattr(self.optimizer, "backward"):
             self.optimizer.backward(loss, retain_graph=retain_graph)

@@ -79,7 +79,9 @@ def _sp_groups_worker(rank, world):
     from deepspeed_amd.parallel import groups
     groups.initialize_sequence_parallel(world)
     assert groups.get_sequence_parallel_world_size() == world
-    assert groups.get_data_parallel_world_size() == 1
+    # ZeRO shards/averages over the full DPxSP mesh under Ulysses (the
+    # model replicates over SP but sees different sequence chunks)
+    assert groups.get_data_parallel_world_size() == world
 
 
 def test_sp_group_topology():
@@ -99,3 +101,63 @@ def _shard_adapter_worker(rank, world):
 
 def test_sp_dataloader_shard():
     run_distributed(_shard_adapter_worker, world_size=2)
+
+
+def _ulysses_model_worker(rank, world):
+    """End-to-end: Llama with Ulysses SP=2 trains to the same weights as a
+    single-process run on the full sequence."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    from deepspeed_amd.models.llama import enable_ulysses
+    from deepspeed_amd.parallel import groups
+
+    groups.initialize_sequence_parallel(world)
+    torch.manual_seed(19)
+    model = LlamaForCausalLM(llama_tiny())
+    enable_ulysses(model)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": 1, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+    })
+
+    # reference: identical model, full sequence, single process
+    torch.manual_seed(19)
+    ref = LlamaForCausalLM(llama_tiny())
+    opt_ref = torch.optim.AdamW(ref.parameters(), lr=1e-3)
+
+    torch.manual_seed(91)  # same full batch everywhere
+    S = 32
+    sl = slice(rank * S // world, (rank + 1) * S // world)
+    losses, ref_losses = [], []
+    for _ in range(3):
+        ids = torch.randint(0, 512, (2, S + 1))
+        x, y = ids[:, :-1], ids[:, 1:]
+        pos = torch.arange(S, dtype=torch.int32).expand(2, S)
+        loss = engine(x[:, sl].contiguous(), labels=y[:, sl].contiguous(),
+                      positions=pos[:, sl].contiguous())
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss)
+
+        l2 = ref(x, labels=y)
+        l2.backward()
+        opt_ref.step()
+        opt_ref.zero_grad()
+        ref_losses.append(l2.item())
+
+    # mean of SP-rank losses == full-sequence loss
+    import torch.distributed as td
+    for lr_, want in zip(losses, ref_losses):
+        t = lr_.detach().clone()
+        td.all_reduce(t)
+        assert abs(t.item() / world - want) < 1e-4, (t.item() / world, want)
+
+    # trained weights match the single-process reference
+    for (n, p), (_, pr) in zip(engine.module.named_parameters(),
+                               ref.named_parameters()):
+        torch.testing.assert_close(p, pr, rtol=1e-4, atol=2e-4), n
+
+
+def test_ulysses_llama_end_to_end():
+    run_distributed(_ulysses_model_worker, world_size=2)
