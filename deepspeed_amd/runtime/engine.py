@@ -115,6 +115,10 @@ class Engine(torch.nn.Module):
                                  "stages 0-2 (expert params are partitioned "
                                  "over the expert-DP group, not ZeRO-3)")
 
+        from .checkpoint_engine import create_checkpoint_engine
+        self.checkpoint_engine = create_checkpoint_engine(
+            self.config.raw.get("checkpoint", {}).get("engine", "torch"))
+
         self._configure_distributed_model(dont_change_device)
         self._configure_optimizer(model_parameters)
         self._configure_lr_scheduler()
@@ -521,15 +525,16 @@ class Engine(torch.nn.Module):
                 "ds_config": self.config.raw,
                 "client_state": client_state or {},
             }
-            torch.save(state, self._model_ckpt_name(ckpt_dir))
+            self.checkpoint_engine.save(state, self._model_ckpt_name(ckpt_dir))
 
         if self.optimizer is not None and hasattr(self.optimizer, "state_dict") and \
                 not isinstance(self.optimizer, DummyOptim):
             opt_state = {"optimizer_state_dict": self.optimizer.state_dict()}
-            torch.save(opt_state, self._zero_ckpt_name(ckpt_dir))
+            self.checkpoint_engine.save(opt_state, self._zero_ckpt_name(ckpt_dir))
 
         dist.barrier()
-        if save_latest and self.global_rank == 0:
+        ok = self.checkpoint_engine.commit(tag)
+        if save_latest and self.global_rank == 0 and ok:
             with open(os.path.join(save_dir, "latest"), "w") as f:
                 f.write(tag)
         dist.barrier()
@@ -558,7 +563,7 @@ class Engine(torch.nn.Module):
                 tag = f.read().strip()
         ckpt_dir = os.path.join(load_dir, tag)
         model_path = self._model_ckpt_name(ckpt_dir)
-        state = torch.load(model_path, map_location="cpu", weights_only=False)
+        state = self.checkpoint_engine.load(model_path, map_location="cpu")
 
         if self.zero_stage == 3 and hasattr(self.optimizer, "load_full_state_dict"):
             self.optimizer.load_full_state_dict(state["module"],
@@ -585,8 +590,8 @@ class Engine(torch.nn.Module):
             zero_path = self._zero_ckpt_name(ckpt_dir)
             if os.path.exists(zero_path) and self.optimizer is not None and \
                     hasattr(self.optimizer, "load_state_dict"):
-                opt_state = torch.load(zero_path, map_location="cpu",
-                                       weights_only=False)
+                opt_state = self.checkpoint_engine.load(zero_path,
+                                                        map_location="cpu")
                 if hasattr(self.optimizer, "loss_scaler"):
                     self.optimizer.load_state_dict(
                         opt_state["optimizer_state_dict"],

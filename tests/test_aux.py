@@ -137,3 +137,19 @@ def test_curriculum_data_sampler():
         batch_size=4, dp_rank=1, dp_size=2)
     b1 = next(iter(s1))
     assert not set(batches[0]) & set(b1)
+
+
+def test_checkpoint_engines(tmp_path):
+    from deepspeed_amd.runtime.checkpoint_engine import (
+        AsyncTorchCheckpointEngine, TorchCheckpointEngine,
+        create_checkpoint_engine)
+    sd = {"w": torch.randn(100), "step": 7}
+    for eng in (TorchCheckpointEngine(), AsyncTorchCheckpointEngine()):
+        p = str(tmp_path / f"{type(eng).__name__}.pt")
+        eng.create("tag")
+        eng.save(sd, p)
+        assert eng.commit("tag")
+        back = eng.load(p)
+        assert torch.equal(back["w"], sd["w"]) and back["step"] == 7
+    assert isinstance(create_checkpoint_engine("async"),
+                      AsyncTorchCheckpointEngine)
