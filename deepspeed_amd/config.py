@@ -75,6 +75,9 @@ class ZeroConfig(ConfigModel):
     sub_group_size: int = 1_000_000_000_000
     zero_hpz_partition_size: int = 1
     round_robin_gradients: bool = False
+    # MiCS: shard params over sub-groups of this size, replicate across
+    # sub-groups with a hierarchical gradient all-reduce (0 = off)
+    mics_shard_size: int = 0
     # ZeRO++-style quantized weight all-gather (qwZ): ship int8 + group
     # scales over xGMI instead of bf16 (opt-in; changes forward numerics)
     zero_quantized_weights: bool = False

@@ -96,6 +96,32 @@ class ZeroStage3Optimizer:
         self.dp_group = dp_group
         self.world_size = dist.get_world_size(dp_group)
         self.rank = dist.get_rank(dp_group)
+        # MiCS (reference zero/mics.py MiCS_Optimizer): partition params
+        # over SUB-groups of mics_shard_size ranks and replicate across
+        # sub-groups; gathers run inside the small shard group while
+        # gradients take one extra all-reduce over the replica group.
+        self.replica_group = None
+        self.replica_world = 1
+        mics = int(getattr(config.zero, "mics_shard_size", 0) or 0)
+        if 1 <= mics < self.world_size:
+            full_world = self.world_size
+            grank = dist.get_rank(dp_group)
+            assert full_world % mics == 0, \
+                f"world {full_world} not divisible by mics_shard_size {mics}"
+            for start in range(0, full_world, mics):
+                ranks = list(range(start, start + mics))
+                g = dist.new_group(ranks)
+                if grank in ranks:
+                    shard_group = g
+            for off in range(mics):
+                ranks = list(range(off, full_world, mics))
+                g = dist.new_group(ranks)
+                if grank in ranks:
+                    self.replica_group = g
+            self.dp_group = shard_group
+            self.world_size = mics
+            self.rank = grank % mics
+            self.replica_world = full_world // mics
         self.mpu = mpu
         self.loss_scaler = loss_scaler or LossScaler(1.0)
         self._config_dtype = (config.dtype
@@ -219,7 +245,10 @@ class ZeroStage3Optimizer:
             tail = u.offsets[-1] + u.params[-1].numel()
             if tail < u.numel:
                 full[tail:].zero_()
-            if self.world_size > 1:
+            if self.replica_group is not None:
+                # MiCS: replicas must start identical too
+                dist.broadcast(full, src=0)
+            elif self.world_size > 1:
                 dist.broadcast(full, src=dist.get_global_rank(self.dp_group, 0)
                                if hasattr(dist, "get_global_rank") else 0,
                                group=self.dp_group)
@@ -514,6 +543,10 @@ class ZeroStage3Optimizer:
             if h is not None:
                 h.wait()
             if u.group_idx >= 0:
+                if self.replica_group is not None:
+                    # MiCS phase 2: average this shard across replicas
+                    dist.all_reduce(recv, group=self.replica_group)
+                    recv = recv / self.replica_world if recv.dtype.is_floating_point else recv
                 if self.direct_grad:
                     # keep the 16-bit RS output; Adam consumes it directly
                     u.grad_shard = recv
