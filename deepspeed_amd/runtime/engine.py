@@ -460,6 +460,67 @@ class Engine(torch.nn.Module):
         """Context manager: disable gradient reduction inside (reference :2065)."""
         return Engine._NoSync(self)
 
+    # ------------------------------------------------------ state offloading
+
+    def _optimizer_state_tensors(self):
+        """Yield (container, key, tensor) for every optimizer-state tensor
+        (ZeRO flat masters/accumulators + inner Adam moments)."""
+        opt = self.optimizer
+        if opt is None:
+            return
+        for attr in ("group_masters", "group_owned_grads"):
+            lst = getattr(opt, attr, None)
+            if lst is not None:
+                for i, t in enumerate(lst):
+                    if torch.is_tensor(t) and t.numel():
+                        yield lst, i, t
+        inner = getattr(opt, "optimizer", opt)
+        for state in getattr(inner, "state", {}).values():
+            for k, v in state.items():
+                if torch.is_tensor(v) and v.numel():
+                    yield state, k, v
+
+    @torch.no_grad()
+    def offload_states(self, include=None, device="cpu", pin_memory=True,
+                       non_blocking=False):
+        """Move optimizer states to host DRAM to free HBM between phases
+        (reference engine.py:3844) — e.g. before a long generate() in RLHF."""
+        moved = 0
+        for container, key, t in list(self._optimizer_state_tensors()):
+            if not t.is_cuda:
+                continue
+            host = torch.empty(t.shape, dtype=t.dtype, device=device)
+            if pin_memory and device == "cpu" and accel.available():
+                host = host.pin_memory()
+            host.copy_(t, non_blocking=non_blocking)
+            if isinstance(t, torch.nn.Parameter):
+                # inner-optimizer params are referenced from param_groups
+                # and as state keys: swap storage, keep identity
+                t.data = host
+            else:
+                container[key] = host
+            moved += t.numel() * t.element_size()
+        if accel.available():
+            accel.synchronize()
+            torch.cuda.empty_cache()
+        log_dist(f"offload_states: moved {moved / 1e9:.2f} GB to {device}")
+        return moved
+
+    @torch.no_grad()
+    def reload_states(self, non_blocking=False):
+        """Inverse of offload_states (reference engine.py:3876)."""
+        for container, key, t in list(self._optimizer_state_tensors()):
+            if t.is_cuda:
+                continue
+            dev = torch.empty(t.shape, dtype=t.dtype, device=self.device)
+            dev.copy_(t, non_blocking=non_blocking)
+            if isinstance(t, torch.nn.Parameter):
+                t.data = dev
+            else:
+                container[key] = dev
+        if accel.available():
+            accel.synchronize()
+
     def compile(self, backend="inductor", compile_kwargs=None):
         """torch.compile the wrapped module in place (reference :3820).
         ZeRO-3's .data-swapping hooks are graph breaks by construction, so
