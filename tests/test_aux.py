@@ -153,3 +153,38 @@ def test_checkpoint_engines(tmp_path):
         assert torch.equal(back["w"], sd["w"]) and back["step"] == 7
     assert isinstance(create_checkpoint_engine("async"),
                       AsyncTorchCheckpointEngine)
+
+
+def test_random_ltd():
+    from deepspeed_amd.runtime.data_pipeline.random_ltd import (
+        RandomLayerTokenDrop, RandomLTDScheduler, convert_to_random_ltd)
+    import torch.nn as nn
+
+    sched = RandomLTDScheduler(total_layers=4, random_ltd_layer_num=2,
+                               start_seq=8, max_seq=16, step_size=4,
+                               schedule_steps=10)
+    assert sched.update_seq(0) == 8
+    assert sched.update_seq(5) == 12
+    assert sched.update_seq(10) == 16
+    sched.current_seq = 8
+
+    class Layers(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.layers = nn.ModuleList([nn.Linear(4, 4) for _ in range(4)])
+
+    m = Layers()
+    n = convert_to_random_ltd(m, "layers", sched)
+    assert n == 2 and isinstance(m.layers[1], RandomLayerTokenDrop)
+
+    x = torch.randn(2, 16, 4)
+    m.train()
+    y = m.layers[1](x)
+    assert y.shape == x.shape
+    # dropped positions pass through unchanged: exactly seq-keep untouched
+    untouched = (y == x).all(dim=-1).sum(dim=1)
+    assert (untouched == 16 - 8).all()
+    # eval mode: full sequence processed
+    m.eval()
+    y2 = m.layers[1](x)
+    assert not torch.equal(y2, x)
