@@ -188,3 +188,34 @@ def test_random_ltd():
     m.eval()
     y2 = m.layers[1](x)
     assert not torch.equal(y2, x)
+
+
+def test_data_analyzer_end_to_end(tmp_path):
+    from deepspeed_amd.runtime.data_pipeline.data_analyzer import (
+        DataAnalyzer, load_index_to_metric, metric_seqlen)
+    from deepspeed_amd.runtime.data_pipeline import CurriculumScheduler
+    from deepspeed_amd.runtime.data_pipeline.data_sampler import \
+        DeepSpeedDataSampler
+
+    torch.manual_seed(0)
+    data = [torch.randint(0, 100, (int(l),))
+            for l in torch.randint(4, 33, (40,))]
+    # 2-worker map, then reduce
+    for w in range(2):
+        DataAnalyzer(data, save_path=str(tmp_path), worker_id=w,
+                     num_workers=2).run_map()
+    merged = DataAnalyzer(data, save_path=str(tmp_path),
+                          num_workers=2).run_reduce()
+    diffs = merged["seqlen"]
+    assert diffs == [float(len(s)) for s in data]
+    assert load_index_to_metric(str(tmp_path), "seqlen") == diffs
+
+    # feeds the curriculum sampler directly
+    sched = CurriculumScheduler({
+        "curriculum_type": "fixed_linear", "min_difficulty": 8,
+        "max_difficulty": 32,
+        "schedule_config": {"total_curriculum_step": 4,
+                            "difficulty_step": 8}})
+    sampler = DeepSpeedDataSampler(diffs, sched, batch_size=2)
+    first = next(iter(sampler))
+    assert all(diffs[i] <= 8 for i in first)
