@@ -125,3 +125,74 @@ class OnebitAdam(torch.optim.Optimizer):
                     update = update.add(p.float(), alpha=group["weight_decay"])
                 p.add_(update.to(p.dtype), alpha=-group["lr"])
         return loss
+
+
+class ZeroOneAdam(torch.optim.Optimizer):
+    """0/1 Adam (reference fp16/onebit/zoadam.py): both the variance AND
+    the communication are intermittent — variance refreshes every
+    ``var_update_interval`` steps and momenta synchronize (1-bit
+    compressed) every ``local_step_interval`` steps, with error feedback
+    carrying the skipped information. Cuts communication ROUNDS as well as
+    volume vs 1-bit Adam."""
+
+    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                 weight_decay=0.0, var_freeze_step=50, var_update_scaler=4,
+                 local_step_scaler=2, comm_group=None):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self.var_freeze_step = var_freeze_step
+        self.var_update_scaler = var_update_scaler
+        self.local_step_scaler = local_step_scaler
+        self.comm_group = comm_group
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        world = dist.get_world_size(self.comm_group) \
+            if dist.is_initialized() else 1
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                g = p.grad.float()
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(p,
+                                                           dtype=torch.float32)
+                    state["error"] = torch.zeros_like(p, dtype=torch.float32)
+                state["step"] += 1
+                t = state["step"]
+                m, v = state["exp_avg"], state["exp_avg_sq"]
+
+                frozen = t > self.var_freeze_step
+                if not frozen:
+                    # exact stage: synchronous grads, live variance
+                    if world > 1:
+                        dist.all_reduce(g, group=self.comm_group)
+                        g /= world
+                    m.mul_(beta1).add_(g, alpha=1 - beta1)
+                    v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+                else:
+                    m.mul_(beta1).add_(g, alpha=1 - beta1)
+                    # intermittent variance refresh (local, frozen between)
+                    if t % self.var_update_scaler == 0:
+                        v.mul_(beta2).addcmul_(m, m, value=1 - beta2)
+                    # intermittent compressed momentum sync
+                    if t % self.local_step_scaler == 0:
+                        m.copy_(compressed_allreduce(m, state["error"],
+                                                     self.comm_group))
+
+                bc1 = 1 - beta1 ** t
+                bc2 = 1 - beta2 ** t
+                denom = (v / bc2).sqrt_().add_(group["eps"])
+                update = (m / bc1) / denom
+                if group["weight_decay"] != 0.0:
+                    update = update.add(p.float(), alpha=group["weight_decay"])
+                p.add_(update.to(p.dtype), alpha=-group["lr"])
+        return loss
