@@ -87,19 +87,19 @@ def _zoadam_worker(rank, world):
     from deepspeed_amd.runtime.fp16.onebit import ZeroOneAdam
     torch.manual_seed(3)
     model = torch.nn.Linear(16, 1)
-    opt = ZeroOneAdam(model.parameters(), lr=5e-2, var_freeze_step=5,
+    opt = ZeroOneAdam(model.parameters(), lr=5e-2, var_freeze_step=15,
                       var_update_scaler=4, local_step_scaler=2)
     torch.manual_seed(200)
     X = torch.randn(64, 16)
     y = X @ torch.randn(16, 1)
     losses = []
-    for _ in range(40):
+    for _ in range(80):
         loss = torch.nn.functional.mse_loss(model(X), y)
         opt.zero_grad()
         loss.backward()
         opt.step()
         losses.append(loss.item())
-    assert losses[-1] < 0.1 * losses[0], losses[::8]
+    assert losses[-1] < 0.1 * losses[0], losses[::16]
     import torch.distributed as td
     w = model.weight.detach().clone()
     peers = [torch.empty_like(w) for _ in range(world)]
