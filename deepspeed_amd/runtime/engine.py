@@ -335,6 +335,7 @@ class Engine(torch.nn.Module):
         # over the full DPxSP mesh (see parallel/groups.py), which already
         # yields the gradient of the global token-mean loss.
 
+        self._last_loss = float(loss.detach().float())
         if hasattr(self.optimizer, "backward"):
             self.optimizer.backward(loss, retain_graph=retain_graph)
         else:
@@ -435,9 +436,18 @@ class Engine(torch.nn.Module):
         if self.tput_timer:
             self.tput_timer.stop(global_step=True)
         if self.monitor is not None and self.global_rank == 0:
-            self.monitor.write_events([
-                ("Train/lr", self.get_lr()[0] if self.get_lr() else 0.0,
-                 self.global_steps)])
+            events = [("Train/lr", self.get_lr()[0] if self.get_lr() else 0.0,
+                       self.global_steps),
+                      ("Train/loss_scale", float(self.loss_scale),
+                       self.global_steps)]
+            if getattr(self, "_last_loss", None) is not None:
+                events.append(("Train/loss", self._last_loss,
+                               self.global_steps))
+            if self.tput_timer and self.tput_timer.avg_samples_per_sec():
+                events.append(("Train/samples_per_sec",
+                               self.tput_timer.avg_samples_per_sec(),
+                               self.global_steps))
+            self.monitor.write_events(events)
         if self.wall_clock_breakdown:
             self.timers(STEP_GLOBAL_TIMER).stop()
             if self.global_steps % self.config.steps_per_print == 0:

@@ -63,3 +63,30 @@ def test_csv_monitor(tmp_path):
                      "job_name": "k"}))
     mm.write_events([("a/b", 3.0, 7)])
     assert (tmp_path / "k" / "a_b.csv").read_text().strip() == "7,3.0"
+
+
+def test_engine_writes_monitor_events(tmp_path):
+    from .common import run_local
+
+    def worker(rank, world, tmp=str(tmp_path)):
+        import deepspeed_amd
+        from deepspeed_amd.models import GPT2ForCausalLM, gpt2_tiny
+        model = GPT2ForCausalLM(gpt2_tiny())
+        engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+            "train_micro_batch_size_per_gpu": 1,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "monitor_config": {"enabled": True,
+                               "csv_monitor": {"enabled": True,
+                                               "output_path": tmp,
+                                               "job_name": "m"}}})
+        ids = torch.randint(0, 128, (1, 16))
+        for _ in range(2):
+            loss = engine(ids, labels=ids)
+            engine.backward(loss)
+            engine.step()
+
+    run_local(worker)
+    files = {f.name for f in (tmp_path / "m").iterdir()}
+    assert "Train_loss.csv" in files and "Train_lr.csv" in files
+    rows = (tmp_path / "m" / "Train_loss.csv").read_text().strip().splitlines()
+    assert len(rows) == 2
