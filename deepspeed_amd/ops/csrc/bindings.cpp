@@ -49,6 +49,10 @@ extern "C" int ds_aio_wait(void* h);
 extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
                              void* o, int B, int S, int H, int Hkv,
                              float scale, int causal, void* stream);
+extern "C" void ds_cpu_lion_flat(float* p, const void* g, int grad_dtype,
+                                 float* m, void* p16, long long n, float lr,
+                                 float beta1, float beta2, float weight_decay,
+                                 float inv_scale);
 extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
                                  void* o, int B, int S, int H, int Hkv,
                                  float scale, int variant, void* stream);
@@ -293,6 +297,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused Adam/AdamW on flat fp32 master + 16-bit grad shard");
   m.def("cpu_adam_flat", &cpu_adam_flat,
         "Host Adam/AdamW on flat pinned fp32 master (ZeRO-Offload)");
+  m.def("cpu_lion_flat",
+        [](at::Tensor p, at::Tensor g, at::Tensor m,
+           c10::optional<at::Tensor> p16, double lr, double beta1,
+           double beta2, double weight_decay, double inv_scale) {
+          TORCH_CHECK(!p.is_cuda() && p.is_contiguous() && g.is_contiguous()
+                      && m.is_contiguous(), "cpu_lion: host contiguous");
+          TORCH_CHECK(p.scalar_type() == at::kFloat &&
+                      m.scalar_type() == at::kFloat, "p/m must be fp32");
+          void* p16_ptr = nullptr;
+          if (p16.has_value() && p16->defined()) p16_ptr = p16->data_ptr();
+          ds_cpu_lion_flat(p.data_ptr<float>(), g.data_ptr(), dtype_code(g),
+                           m.data_ptr<float>(), p16_ptr, p.numel(), (float)lr,
+                           (float)beta1, (float)beta2, (float)weight_decay,
+                           (float)inv_scale);
+        },
+        "Host Lion on flat fp32 master");
   m.def("norm_fwd", &norm_fwd, "RMSNorm/LayerNorm forward");
   m.def("norm_bwd", &norm_bwd, "RMSNorm/LayerNorm backward");
   m.def("rope", &rope, "Rotary position embedding (in-place)");

@@ -144,3 +144,67 @@ extern "C" void ds_cpu_adam_flat(float* p, const void* g, int grad_dtype,
     }
   }
 }
+
+// ---- CPU Lion (reference csrc/lion/cpu_lion_impl.cpp): p -= lr*(sign(c)
+// + wd*p), m = b2*m + (1-b2)*g, with c = b1*m + (1-b1)*g. Same fused
+// streaming-pass design as Adam above.
+namespace {
+
+template <typename GradLoad, bool WRITE_BF16>
+void lion_loop(float* __restrict__ p, GradLoad gload, float* __restrict__ m,
+               uint16_t* __restrict__ p16, int64_t n, float lr, float beta1,
+               float beta2, float wd, float inv_scale) {
+#pragma omp parallel for schedule(static)
+  for (int64_t c = 0; c < n; c += 4096) {
+    const int64_t end = c + 4096 < n ? c + 4096 : n;
+#pragma omp simd
+    for (int64_t i = c; i < end; ++i) {
+      const float g = gload(i) * inv_scale;
+      const float u = beta1 * m[i] + (1.f - beta1) * g;
+      const float s = u > 0.f ? 1.f : (u < 0.f ? -1.f : 0.f);
+      p[i] -= lr * (s + wd * p[i]);
+      m[i] = beta2 * m[i] + (1.f - beta2) * g;
+      if (WRITE_BF16) p16[i] = f32_to_bf16_rne(p[i]);
+    }
+  }
+}
+
+template <typename GradLoad>
+void lion_dispatch(float* p, GradLoad gload, float* m, uint16_t* p16,
+                   int64_t n, float lr, float b1, float b2, float wd,
+                   float inv_scale) {
+  if (p16)
+    lion_loop<GradLoad, true>(p, gload, m, p16, n, lr, b1, b2, wd, inv_scale);
+  else
+    lion_loop<GradLoad, false>(p, gload, m, p16, n, lr, b1, b2, wd,
+                               inv_scale);
+}
+
+}  // namespace
+
+extern "C" void ds_cpu_lion_flat(float* p, const void* g, int grad_dtype,
+                                 float* m, void* p16, long long n, float lr,
+                                 float beta1, float beta2, float weight_decay,
+                                 float inv_scale) {
+  uint16_t* p16t = reinterpret_cast<uint16_t*>(p16);
+  switch (grad_dtype) {
+    case 0: {
+      const float* gf = reinterpret_cast<const float*>(g);
+      lion_dispatch(p, [gf](int64_t i) { return gf[i]; }, m, p16t, n, lr,
+                    beta1, beta2, weight_decay, inv_scale);
+      break;
+    }
+    case 1: {
+      const uint16_t* gb = reinterpret_cast<const uint16_t*>(g);
+      lion_dispatch(p, [gb](int64_t i) { return bf16_to_f32(gb[i]); }, m,
+                    p16t, n, lr, beta1, beta2, weight_decay, inv_scale);
+      break;
+    }
+    case 2: {
+      const uint16_t* gh = reinterpret_cast<const uint16_t*>(g);
+      lion_dispatch(p, [gh](int64_t i) { return f16_to_f32(gh[i]); }, m, p16t,
+                    n, lr, beta1, beta2, weight_decay, inv_scale);
+      break;
+    }
+  }
+}

@@ -54,3 +54,30 @@ def test_cpu_adam_multi_step_convergence():
         ext.cpu_adam_flat(p, g, m, v, None, 1e-1, 0.9, 0.99, 1e-8, 0.0,
                           step, 1.0, True)
     assert (p - target).abs().mean() < 0.05
+
+
+@needs_ext
+@pytest.mark.parametrize("grad_dtype", [torch.float32, torch.bfloat16])
+def test_cpu_lion_matches_torch(grad_dtype):
+    """Host Lion vs the framework's torch Lion reference."""
+    ext = get_ext()
+    if not hasattr(ext, "cpu_lion_flat"):
+        pytest.skip("cpu_lion_flat not in extension")
+    torch.manual_seed(4)
+    n = 50_001
+    p = torch.randn(n)
+    g = (torch.randn(n) * 0.1).to(grad_dtype)
+    m = torch.rand(n) * 0.01
+    p2, m2 = p.clone(), m.clone()
+    lr, b1, b2, wd = 1e-3, 0.9, 0.99, 0.1
+
+    ext.cpu_lion_flat(p, g, m, None, lr, b1, b2, wd, 1.0)
+
+    # torch reference (lion update rule)
+    gf = g.float()
+    u = b1 * m2 + (1 - b1) * gf
+    p2 -= lr * (u.sign() + wd * p2)
+    m2.mul_(b2).add_(gf, alpha=1 - b2)
+
+    torch.testing.assert_close(p, p2, rtol=1e-6, atol=1e-7)
+    torch.testing.assert_close(m, m2, rtol=1e-6, atol=1e-7)
