@@ -188,6 +188,9 @@ class Engine(torch.nn.Module):
         if name == "lion":
             from ..ops.lion import Lion
             return Lion(model_parameters, **params)
+        if name in ("lamb", "fusedlamb"):
+            from ..ops.lamb import FusedLamb
+            return FusedLamb(model_parameters, **params)
         if name == "adagrad":
             return torch.optim.Adagrad(model_parameters, **params)
         raise ValueError(f"unsupported optimizer type {cfg.type}")

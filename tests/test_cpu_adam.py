@@ -81,3 +81,39 @@ def test_cpu_lion_matches_torch(grad_dtype):
 
     torch.testing.assert_close(p, p2, rtol=1e-6, atol=1e-7)
     torch.testing.assert_close(m, m2, rtol=1e-6, atol=1e-7)
+
+
+def test_lamb_trains_and_trust_ratio():
+    from deepspeed_amd.ops.lamb import FusedLamb
+    torch.manual_seed(0)
+    model = torch.nn.Linear(16, 1)
+    opt = FusedLamb(model.parameters(), lr=5e-2, weight_decay=0.01)
+    X = torch.randn(128, 16)
+    y = X @ torch.randn(16, 1)
+    losses = []
+    for _ in range(60):
+        loss = torch.nn.functional.mse_loss(model(X), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert losses[-1] < 0.1 * losses[0], losses[::10]
+
+
+def test_lamb_via_engine_config():
+    from .common import run_local
+
+    def worker(rank, world):
+        import deepspeed_amd
+        from deepspeed_amd.models import GPT2ForCausalLM, gpt2_tiny
+        model = GPT2ForCausalLM(gpt2_tiny())
+        engine, opt, _, _ = deepspeed_amd.initialize(model=model, config={
+            "train_micro_batch_size_per_gpu": 1,
+            "optimizer": {"type": "Lamb", "params": {"lr": 1e-3}}})
+        ids = torch.randint(0, 128, (1, 16))
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+        assert torch.isfinite(loss)
+
+    run_local(worker)
