@@ -188,3 +188,44 @@ def _different_data_worker(rank, world):
 
 def test_zero_dp_grad_averaging_ws2():
     run_distributed(_different_data_worker, world_size=2)
+
+
+def _fp16_overflow_worker(rank, world):
+    """fp16 dynamic loss scaling: an inf gradient must skip the step, halve
+    the scale, and recover (reference contract:
+    tests/unit/runtime/half_precision/test_fp16.py)."""
+    import deepspeed_amd
+
+    torch.manual_seed(1)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 1))
+    engine, opt, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 4,
+        "fp16": {"enabled": True, "initial_scale_power": 8,
+                 "hysteresis": 1},
+        "zero_optimization": {"stage": 1, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+    })
+    assert engine.loss_scale == 2 ** 8
+    before = [p.detach().clone() for p in engine.module.parameters()]
+
+    # poison one forward to overflow fp16 grads
+    x = torch.randn(4, 8).half()
+    loss = engine(x).sum() * 1e30
+    engine.backward(loss)
+    engine.step()
+    assert engine.skipped_steps == 1
+    assert engine.loss_scale < 2 ** 8  # scale backed off
+    for p, b in zip(engine.module.parameters(), before):
+        assert torch.equal(p.detach(), b), "params must not move on overflow"
+
+    # normal step proceeds and params move
+    loss = engine(x).sum()
+    engine.backward(loss)
+    engine.step()
+    moved = any(not torch.equal(p.detach(), b)
+                for p, b in zip(engine.module.parameters(), before))
+    assert moved and engine.skipped_steps == 1
+
+
+def test_fp16_overflow_skip_and_recover():
+    run_local(_fp16_overflow_worker)
