@@ -7,7 +7,7 @@ allowed scale points, so a job can resume at a different world size without
 changing optimization behavior.
 """
 
-from typing import Dict, List, Tuple
+from typing import Dict, List
 
 LATEST_ELASTICITY_VERSION = 0.2
 

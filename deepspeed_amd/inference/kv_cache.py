@@ -10,7 +10,7 @@ nothing until far larger batch x context than a single node serves — the
 contiguous layout keeps SDPA reads fully coalesced instead.
 """
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

@@ -10,9 +10,7 @@ Hot ops: RMSNorm / RoPE / SwiGLU are the hand-written HIP kernels in
 torch SDPA (flash backend on ROCm) with an explicit-math fallback.
 """
 
-import math
 from dataclasses import dataclass
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -3,9 +3,6 @@ deepspeed/runtime/data_pipeline/curriculum_scheduler.py CurriculumScheduler
 — fixed_linear / fixed_root / fixed_discrete / custom schedules over a
 difficulty metric such as sequence length)."""
 
-import math
-
-
 class CurriculumScheduler:
     def __init__(self, config: dict):
         self.state = {}

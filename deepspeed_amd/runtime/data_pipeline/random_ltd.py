@@ -9,8 +9,6 @@ a schedule until full length, cutting pretraining FLOPs ~2x at equal
 quality (reference random-LTD paper).
 """
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 

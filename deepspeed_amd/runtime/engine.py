@@ -14,8 +14,7 @@ Precision policy (simpler than the reference's three optimizer stacks):
 """
 
 import os
-import shutil
-from typing import Callable, Optional
+from typing import Optional
 
 import torch
 
@@ -550,7 +549,7 @@ class Engine(torch.nn.Module):
     # ----------------------------------------------------------- data loader
 
     def deepspeed_io(self, dataset, batch_size=None, num_workers=0, collate_fn=None):
-        from .dataloader import RepeatingLoader, build_dataloader
+        from .dataloader import build_dataloader
         return build_dataloader(
             dataset,
             batch_size=batch_size or self.train_micro_batch_size_per_gpu,

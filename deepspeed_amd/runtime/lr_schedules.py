@@ -6,7 +6,7 @@ state_dict/load_state_dict for checkpointing.
 """
 
 import math
-from typing import List, Union
+from typing import List
 
 VALID_SCHEDULES = ["LRRangeTest", "OneCycle", "WarmupLR", "WarmupDecayLR",
                    "WarmupCosineLR"]

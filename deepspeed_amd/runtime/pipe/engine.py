@@ -12,9 +12,6 @@ Differences from the reference, by MI355X design:
   tied grads are correct without a separate dead-bucket path.
 """
 
-from types import MethodType
-from typing import Optional
-
 import torch
 
 from ... import comm as dist

@@ -28,16 +28,15 @@ owned shard is valid) — half the xGMI traffic.
 """
 
 import math
-from collections import defaultdict
 from typing import Dict, List, Optional
 
 import torch
 
 from ... import accel
 from ... import comm as dist
-from ...utils.logging import log_dist, logger
+from ...utils.logging import log_dist
 from ..fp16.loss_scaler import LossScalerBase, LossScaler
-from ..utils import ALIGNMENT, get_global_norm_of_tensors
+from ..utils import ALIGNMENT
 
 
 class _Bucket:

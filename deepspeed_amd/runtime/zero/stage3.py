@@ -36,14 +36,13 @@ prefetch) — re-designed MI355X-first rather than ported:
 """
 
 import math
-from collections import defaultdict
 from typing import Dict, List, Optional
 
 import torch
 
 from ... import accel
 from ... import comm as dist
-from ...utils.logging import log_dist, logger
+from ...utils.logging import log_dist
 from ..fp16.loss_scaler import LossScalerBase, LossScaler
 from ..utils import ALIGNMENT
 
