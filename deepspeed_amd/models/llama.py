@@ -113,14 +113,20 @@ class LlamaAttention(nn.Module):
         q = q.transpose(1, 2)
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
+        attn_mask = None
         if kv_cache is not None:
             k, v = kv_cache.update(self.layer_idx, k, v)
+            # ragged caches expose a per-sequence validity mask (slots in
+            # one batch sit at different positions)
+            attn_mask = getattr(kv_cache, "last_mask", None)
         if attention_fn is not None:
             o = attention_fn(q, k, v)
         else:
             # prefill (q spans the whole kv prefix) is causal; decode steps
             # (q_len < kv_len) attend to the full cached prefix
-            o = sdpa_gqa(q, k, v, causal=q.size(2) == k.size(2))
+            o = sdpa_gqa(q, k, v,
+                         causal=attn_mask is None and q.size(2) == k.size(2),
+                         attn_mask=attn_mask)
         o = o.transpose(1, 2).reshape(B, S, -1)
         return self.o_proj(o)
 
@@ -151,17 +157,19 @@ def enable_ulysses(model, sp_group=None):
     return model
 
 
-def sdpa_gqa(q, k, v, causal=True):
+def sdpa_gqa(q, k, v, causal=True, attn_mask=None):
     """SDPA with grouped-query support (expands KV if enable_gqa missing)."""
     try:
-        return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
+        return F.scaled_dot_product_attention(q, k, v, attn_mask=attn_mask,
+                                              is_causal=causal,
                                               enable_gqa=True)
     except (TypeError, RuntimeError):
         rep = q.shape[1] // k.shape[1]
         if rep > 1:
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
-        return F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+        return F.scaled_dot_product_attention(q, k, v, attn_mask=attn_mask,
+                                              is_causal=causal)
 
 
 class LlamaMLP(nn.Module):
