@@ -304,3 +304,35 @@ def get_comms_logger() -> Optional[CommsLogger]:
 def log_summary():
     if _comms_logger is not None and get_rank() == 0:
         logger.info("\n" + _comms_logger.summary())
+
+
+def reduce_scatter_coalesced(tensors, group=None):
+    """Reduce-scatter MANY tensors in one collective (reference:
+    deepspeed/runtime/comm/coalesced_collectives.py:157): per-rank chunks of
+    every tensor are interleaved into one flat buffer so a single
+    reduce_scatter_tensor covers them all. Returns this rank's partition of
+    each tensor (flat, ceil(numel/world) sized, zero-padded)."""
+    world = get_world_size(group)
+    rank = get_rank(group)
+    import math as _math
+    part_sizes = [_math.ceil(t.numel() / world) for t in tensors]
+    total = sum(part_sizes)
+    device, dtype = tensors[0].device, tensors[0].dtype
+    flat = torch.zeros(total * world, dtype=dtype, device=device)
+    # layout: [rank0: t0_part, t1_part, ...][rank1: ...] ...
+    for r in range(world):
+        off = r * total
+        for t, ps in zip(tensors, part_sizes):
+            src = t.reshape(-1)[r * ps:(r + 1) * ps]
+            flat[off:off + src.numel()].copy_(src)
+            off += ps
+    recv = torch.empty(total, dtype=dtype, device=device)
+    if world > 1:
+        reduce_scatter_tensor(recv, flat, group=group)
+    else:
+        recv.copy_(flat)
+    out, off = [], 0
+    for ps in part_sizes:
+        out.append(recv[off:off + ps])
+        off += ps
+    return out

@@ -77,3 +77,26 @@ def test_single_process_noop_paths():
     dist.all_gather_into_tensor(out, torch.ones(4))
     assert torch.allclose(out, torch.ones(4))
     assert dist.get_rank() == 0 and dist.get_world_size() == 1
+
+
+def _rs_coalesced_worker(rank, world):
+    from deepspeed_amd import comm as dist
+    torch.manual_seed(0)  # same tensors everywhere
+    a = torch.arange(10, dtype=torch.float32) + rank   # rank-divergent
+    b = torch.arange(7, dtype=torch.float32) * (rank + 1)
+    parts = dist.reduce_scatter_coalesced([a, b])
+    # expected: sum over ranks, my slice (ceil sizes, zero-padded)
+    a_sum = torch.stack([torch.arange(10, dtype=torch.float32) + r
+                         for r in range(world)]).sum(0)
+    b_sum = torch.stack([torch.arange(7, dtype=torch.float32) * (r + 1)
+                         for r in range(world)]).sum(0)
+    pa, pb = 5, 4
+    want_a = a_sum[rank * pa:(rank + 1) * pa]
+    b_pad = torch.cat([b_sum, torch.zeros(world * pb - 7)])
+    want_b = b_pad[rank * pb:(rank + 1) * pb]
+    torch.testing.assert_close(parts[0], want_a)
+    torch.testing.assert_close(parts[1], want_b)
+
+
+def test_reduce_scatter_coalesced():
+    run_distributed(_rs_coalesced_worker, world_size=2)
