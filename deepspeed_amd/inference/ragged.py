@@ -88,8 +88,13 @@ class Request:
     max_new_tokens: int = 64
     eos_token_id: Optional[int] = None
     slot: int = -1
+    prefilled: int = 0                        # prompt tokens already in KV
     generated: List[int] = field(default_factory=list)
     done: bool = False
+
+    @property
+    def in_prefill(self) -> bool:
+        return self.prefilled < self.prompt.numel()
 
     @property
     def tokens(self) -> List[int]:
@@ -102,7 +107,12 @@ class ContinuousBatcher:
     in a single batched forward (reference engine_v2 put/schedule loop)."""
 
     def __init__(self, model, max_slots: int = 8, max_seq: int = None,
-                 dtype=torch.float32, device=None):
+                 dtype=torch.float32, device=None,
+                 prefill_chunk: int = 0):
+        """``prefill_chunk`` > 0 bounds prompt tokens prefetched per step
+        (Dynamic SplitFuse: long prompts stream in across iterations so the
+        decode batch's latency stays flat; reference inference/v2
+        scheduling). 0 = whole prompt in one step."""
         self.model = model
         cfg = model.cfg
         self.device = device or next(model.parameters()).device
@@ -114,6 +124,7 @@ class ContinuousBatcher:
         self.free_slots = deque(range(max_slots))
         self.pending: deque = deque()
         self.active: Dict[int, Request] = {}
+        self.prefill_chunk = prefill_chunk
 
     def put(self, req: Request):
         self.pending.append(req)
@@ -128,29 +139,43 @@ class ContinuousBatcher:
         self.cache.end_step()
         return logits
 
+    def _prefill_some(self, req: Request):
+        """Advance one request's prefill by up to prefill_chunk tokens; on
+        completion, sample its first token."""
+        S = req.prompt.numel()
+        chunk = S - req.prefilled if self.prefill_chunk <= 0 \
+            else min(self.prefill_chunk, S - req.prefilled)
+        lo, hi = req.prefilled, req.prefilled + chunk
+        pos = torch.arange(lo, hi, dtype=torch.int32,
+                           device=self.device).unsqueeze(0)
+        logits = self._forward(
+            req.prompt[lo:hi].view(1, -1).to(self.device), [req.slot],
+            chunk, pos)
+        req.prefilled = hi
+        if not req.in_prefill:
+            req.generated.append(int(logits[0, -1].argmax()))
+
     @torch.no_grad()
     def step(self) -> List[Request]:
         """One scheduling iteration; returns requests finished this step."""
-        # 1) prefill newly admitted requests (one forward each — lengths
-        #    differ; SplitFuse-style chunked co-batching is future work)
+        # 1) admit queued requests into free slots
         while self.pending and self.free_slots:
             req = self.pending.popleft()
             req.slot = self.free_slots.popleft()
             self.cache.free(req.slot)
             self.cache.lens[req.slot] = 0
-            S = req.prompt.numel()
-            pos = torch.arange(S, dtype=torch.int32,
-                               device=self.device).unsqueeze(0)
-            logits = self._forward(req.prompt.view(1, -1).to(self.device),
-                                   [req.slot], S, pos)
-            tok = int(logits[0, -1].argmax())
-            req.generated.append(tok)
             self.active[req.uid] = req
+        # 1b) advance prefills (bounded per step when prefill_chunk is set)
+        for req in self.active.values():
+            if req.in_prefill:
+                self._prefill_some(req)
 
         finished = []
         # 2) retire sequences that hit eos/max BEFORE the decode batch
         for uid in list(self.active):
             req = self.active[uid]
+            if req.in_prefill:
+                continue
             if (req.eos_token_id is not None and req.generated and
                     req.generated[-1] == req.eos_token_id) or \
                     len(req.generated) >= req.max_new_tokens or \
@@ -160,9 +185,10 @@ class ContinuousBatcher:
                 self.free_slots.append(req.slot)
                 del self.active[uid]
 
-        # 3) one batched decode step for every active sequence
-        if self.active:
-            reqs = list(self.active.values())
+        # 3) one batched decode step for every fully-prefilled sequence
+        decode = [r for r in self.active.values() if not r.in_prefill]
+        if decode:
+            reqs = decode
             slots = [r.slot for r in reqs]
             last = torch.tensor([[r.generated[-1]] for r in reqs],
                                 device=self.device)

@@ -155,3 +155,24 @@ def test_continuous_batcher_eos_frees_slot():
                    eos_token_id=first))
     done2 = b2.run_to_completion()
     assert len(done2[0].generated) <= 2
+
+
+def test_continuous_batcher_splitfuse_chunked_prefill():
+    """Bounded per-step prefill (Dynamic SplitFuse): chunk-streamed prompts
+    must generate exactly the same tokens."""
+    import deepspeed_amd
+    from deepspeed_amd.inference.ragged import ContinuousBatcher, Request
+    model, cfg = _model(seed=6)
+    engine = deepspeed_amd.init_inference(model, dtype=torch.float32)
+    g = torch.Generator().manual_seed(4)
+    prompts = [torch.randint(0, cfg.vocab_size, (int(n),), generator=g)
+               for n in (11, 6)]
+    want = [engine.generate(p.view(1, -1), max_new_tokens=4)[0, p.numel():]
+            .tolist() for p in prompts]
+
+    batcher = ContinuousBatcher(model, max_slots=4, prefill_chunk=3)
+    for i, p in enumerate(prompts):
+        batcher.put(Request(uid=i, prompt=p, max_new_tokens=4))
+    done = {r.uid: r for r in batcher.run_to_completion()}
+    for uid in range(2):
+        assert done[uid].generated == want[uid], (uid, done[uid].generated)
