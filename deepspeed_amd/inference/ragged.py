@@ -108,7 +108,7 @@ class ContinuousBatcher:
 
     def __init__(self, model, max_slots: int = 8, max_seq: int = None,
                  dtype=torch.float32, device=None,
-                 prefill_chunk: int = 0):
+                 prefill_chunk: int = 0, cache_cls=None):
         """``prefill_chunk`` > 0 bounds prompt tokens prefetched per step
         (Dynamic SplitFuse: long prompts stream in across iterations so the
         decode batch's latency stays flat; reference inference/v2
@@ -118,9 +118,10 @@ class ContinuousBatcher:
         self.device = device or next(model.parameters()).device
         max_seq = max_seq or cfg.max_seq_len
         kv_heads = getattr(cfg, "num_kv_heads", None) or cfg.num_heads
-        self.cache = RaggedKVCache(cfg.num_layers, max_slots, kv_heads,
-                                   max_seq, cfg.head_dim, dtype=dtype,
-                                   device=self.device)
+        cache_cls = cache_cls or RaggedKVCache
+        self.cache = cache_cls(cfg.num_layers, max_slots, kv_heads,
+                               max_seq, cfg.head_dim, dtype=dtype,
+                               device=self.device)
         self.free_slots = deque(range(max_slots))
         self.pending: deque = deque()
         self.active: Dict[int, Request] = {}
@@ -182,6 +183,7 @@ class ContinuousBatcher:
                     int(self.cache.lens[req.slot]) + 1 >= self.cache.max_seq:
                 req.done = True
                 finished.append(req)
+                self.cache.free(req.slot)   # return KV blocks to the pool
                 self.free_slots.append(req.slot)
                 del self.active[uid]
 
@@ -208,3 +210,83 @@ class ContinuousBatcher:
                 break
             out.extend(self.step())
         return out
+
+
+class PagedKVCache:
+    """Blocked KV cache (reference: inference/v2/ragged blocked KV +
+    csrc block allocator): storage is a shared pool of fixed-size blocks
+    [num_blocks, Hkv, block_size, D]; each slot owns a growable block list,
+    so HBM is committed per ~block instead of per max_seq slot. Same
+    begin_step/update/end_step protocol as RaggedKVCache, so it drops into
+    ContinuousBatcher unchanged; reads materialize the active slots'
+    blocks into a padded [n, Hkv, L, D] view per step."""
+
+    def __init__(self, num_layers: int, num_slots: int, kv_heads: int,
+                 max_seq: int, head_dim: int, dtype=torch.float32,
+                 device="cpu", block_size: int = 64, num_blocks: int = None):
+        self.block_size = block_size
+        blocks_per_slot = (max_seq + block_size - 1) // block_size
+        self.num_blocks = num_blocks or num_slots * blocks_per_slot
+        shape = (self.num_blocks, kv_heads, block_size, head_dim)
+        self.k = [torch.zeros(shape, dtype=dtype, device=device)
+                  for _ in range(num_layers)]
+        self.v = [torch.zeros(shape, dtype=dtype, device=device)
+                  for _ in range(num_layers)]
+        self.free_blocks = list(range(self.num_blocks - 1, -1, -1))
+        self.block_table: Dict[int, List[int]] = {s: [] for s in
+                                                  range(num_slots)}
+        self.lens = torch.zeros(num_slots, dtype=torch.long)
+        self.max_seq = max_seq
+        self.device = device
+        self.last_mask: Optional[torch.Tensor] = None
+
+    def free(self, slot: int):
+        self.free_blocks.extend(reversed(self.block_table[slot]))
+        self.block_table[slot] = []
+        self.lens[slot] = 0
+
+    def _ensure_blocks(self, slot: int, upto: int):
+        need = (upto + self.block_size - 1) // self.block_size
+        while len(self.block_table[slot]) < need:
+            assert self.free_blocks, "paged KV pool exhausted"
+            self.block_table[slot].append(self.free_blocks.pop())
+
+    def begin_step(self, slots: List[int], q_len: int):
+        self._slots = list(slots)
+        self._q_len = q_len
+        starts = self.lens[torch.as_tensor(self._slots)]
+        L = int((starts + q_len).max())
+        for s, st in zip(self._slots, starts.tolist()):
+            self._ensure_blocks(s, st + q_len)
+        pos = torch.arange(L, device=self.device)
+        qpos = (starts.to(self.device)[:, None, None] +
+                torch.arange(q_len, device=self.device)[None, :, None])
+        self.last_mask = (pos[None, None, :] <= qpos).unsqueeze(1)
+        self._L = L
+        self._starts = starts
+
+    def update(self, layer_idx: int, k: torch.Tensor, v: torch.Tensor):
+        n, Hkv, q_len, D = k.shape
+        bs = self.block_size
+        for i, s in enumerate(self._slots):
+            st = int(self._starts[i])
+            for j in range(q_len):
+                p = st + j
+                blk = self.block_table[s][p // bs]
+                self.k[layer_idx][blk, :, p % bs] = k[i, :, j]
+                self.v[layer_idx][blk, :, p % bs] = v[i, :, j]
+        # materialize padded views: gather each slot's blocks
+        nblk = (self._L + bs - 1) // bs
+        idx = torch.tensor([ (self.block_table[s] + [0] * nblk)[:nblk]
+                             for s in self._slots ], dtype=torch.long,
+                           device=self.device)              # [n, nblk]
+        kb = self.k[layer_idx][idx]                          # [n,nblk,Hkv,bs,D]
+        vb = self.v[layer_idx][idx]
+        kb = kb.permute(0, 2, 1, 3, 4).reshape(n, Hkv, nblk * bs, D)
+        vb = vb.permute(0, 2, 1, 3, 4).reshape(n, Hkv, nblk * bs, D)
+        return kb[:, :, :self._L], vb[:, :, :self._L]
+
+    def end_step(self):
+        self.lens[torch.as_tensor(self._slots)] += self._q_len
+        self._slots = None
+        self.last_mask = None
