@@ -201,3 +201,37 @@ def test_paged_kv_cache_matches_slot_cache():
     b2.put(Request(uid=0, prompt=prompts[0], max_new_tokens=2))
     b2.run_to_completion()
     assert len(b2.cache.free_blocks) == before  # all blocks recycled
+
+
+def test_inference_server_http():
+    """FastAPI front-end: concurrent requests share decode batches and
+    return the same tokens as direct generation."""
+    import deepspeed_amd
+    from concurrent.futures import ThreadPoolExecutor
+    from fastapi.testclient import TestClient
+    from deepspeed_amd.inference.server import InferenceServer, build_app
+
+    model, cfg = _model(seed=6)
+    engine = deepspeed_amd.init_inference(model, dtype=torch.float32)
+    g = torch.Generator().manual_seed(8)
+    prompts = [torch.randint(0, cfg.vocab_size, (int(n),), generator=g)
+               for n in (6, 10, 4)]
+    want = [engine.generate(p.view(1, -1), max_new_tokens=5)[0, p.numel():]
+            .tolist() for p in prompts]
+
+    server = InferenceServer(model, max_slots=4).start()
+    try:
+        client = TestClient(build_app(server))
+        assert client.get("/health").json()["status"] == "ok"
+
+        def call(p):
+            r = client.post("/generate", json={"token_ids": p.tolist(),
+                                               "max_new_tokens": 5})
+            assert r.status_code == 200
+            return r.json()["generated"]
+
+        with ThreadPoolExecutor(3) as ex:
+            got = list(ex.map(call, prompts))
+        assert got == want
+    finally:
+        server.stop()
