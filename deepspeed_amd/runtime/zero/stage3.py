@@ -163,6 +163,7 @@ class ZeroStage3Optimizer:
         self._bwd_cursor = 0
         self._recording = True
 
+        self._trace_misses = 0
         self._inflight_rs = []   # (handle, recv, unit, grad_full_ref)
         self._pending_release: List[_Unit] = []
         self._hooks = []
@@ -568,7 +569,12 @@ class ZeroStage3Optimizer:
             while i < len(t) and t[i] != u.index:
                 i += 1
             if i < len(t):
+                # skipped entries = module calls the trace predicted but the
+                # model did not make (dynamic control flow)
+                self._trace_misses += i - self._fwd_cursor
                 self._fwd_cursor = i + 1
+            else:
+                self._trace_misses += 1  # unit not in trace at all
 
     def _advance_bwd_cursor(self, units):
         t = self._rtrace
@@ -600,6 +606,20 @@ class ZeroStage3Optimizer:
                     dedup.append(idx)
             self._trace = dedup
             self._rtrace = list(reversed(dedup))
+        elif self._trace_complete and self._trace_misses > \
+                max(4, len(self._trace) // 4):
+            # trace invalidation (reference partitioned_param_coordinator
+            # trace-mismatch handling): the model's call order diverged from
+            # the recorded trace — prefetch was fetching the wrong units.
+            # Correctness is unaffected (pre-forward hooks always gather on
+            # demand) but memory/bandwidth churns, so re-record next step.
+            log_dist(f"ZeRO-3: prefetch trace invalidated "
+                     f"({self._trace_misses} misses) — re-recording")
+            self._trace = []
+            self._rtrace = []
+            self._trace_complete = False
+            self._recording = True
+        self._trace_misses = 0
         self._fwd_cursor = 0
         self._bwd_cursor = 0
 
