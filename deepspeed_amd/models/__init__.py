@@ -3,12 +3,16 @@
 from .llama import (LlamaConfig, LlamaForCausalLM, llama3_8b, llama3_70b,
                     llama_tiny, llama_mini, phi3_mini)
 from .gpt2 import GPT2Config, GPT2ForCausalLM, gpt2_small, gpt2_tiny
+from .bert import (BertConfig, BertForPreTraining, BertModel, bert_base,
+                   bert_large, bert_tiny)
 from .mixtral import (MixtralConfig, MixtralForCausalLM, mixtral_8x7b,
                       mixtral_tiny, mixtral_mini)
 
 __all__ = [
     "LlamaConfig", "LlamaForCausalLM", "llama3_8b", "llama3_70b",
     "llama_tiny", "llama_mini", "phi3_mini", "GPT2Config", "GPT2ForCausalLM",
-    "gpt2_small", "gpt2_tiny", "MixtralConfig", "MixtralForCausalLM",
+    "gpt2_small", "gpt2_tiny", "BertConfig", "BertModel",
+    "BertForPreTraining", "bert_base", "bert_large", "bert_tiny",
+    "MixtralConfig", "MixtralForCausalLM",
     "mixtral_8x7b", "mixtral_tiny", "mixtral_mini",
 ]
