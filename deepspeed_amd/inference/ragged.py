@@ -194,8 +194,10 @@ class ContinuousBatcher:
             slots = [r.slot for r in reqs]
             last = torch.tensor([[r.generated[-1]] for r in reqs],
                                 device=self.device)
-            pos = self.cache.lens[torch.as_tensor(slots)] \
-                .to(torch.int32).unsqueeze(1)
+            lens = self.cache.lens
+            idx = torch.as_tensor(slots, device=lens.device)
+            pos = lens[idx].to(device=self.device,
+                               dtype=torch.int32).unsqueeze(1)
             logits = self._forward(last, slots, 1, pos)
             toks = logits[:, -1].argmax(dim=-1)
             for r, t in zip(reqs, toks.tolist()):
@@ -254,7 +256,8 @@ class PagedKVCache:
     def begin_step(self, slots: List[int], q_len: int):
         self._slots = list(slots)
         self._q_len = q_len
-        starts = self.lens[torch.as_tensor(self._slots)]
+        starts = self.lens[torch.as_tensor(self._slots,
+                                           device=self.lens.device)]
         L = int((starts + q_len).max())
         for s, st in zip(self._slots, starts.tolist()):
             self._ensure_blocks(s, st + q_len)
@@ -287,6 +290,7 @@ class PagedKVCache:
         return kb[:, :, :self._L], vb[:, :, :self._L]
 
     def end_step(self):
-        self.lens[torch.as_tensor(self._slots)] += self._q_len
+        idx = torch.as_tensor(self._slots, device=self.lens.device)
+        self.lens[idx] += self._q_len
         self._slots = None
         self.last_mask = None
