@@ -373,7 +373,7 @@ def test_zero3_dynamic_control_flow():
     run_distributed(_z3_dynamic_worker, world_size=2)
 
 
-def _z3_zoo_worker(rank, world, arch):
+def _z3_zoo_worker(rank, world, arch, stage=3):
     """ZeRO-3 must match plain single-process training for awkward module
     graphs: tied params, frozen params, params unused in some step, nested
     containers (reference contract: test_zero_context*/test_ds_initialize
@@ -418,7 +418,7 @@ def _z3_zoo_worker(rank, world, arch):
     model = Zoo(arch)
     engine, opt, _, _ = deepspeed_amd.initialize(model=model, config={
         "train_micro_batch_size_per_gpu": 2,
-        "zero_optimization": {"stage": 3, "overlap_comm": False,
+        "zero_optimization": {"stage": stage, "overlap_comm": False,
                               "stage3_param_persistence_threshold": 0},
         "optimizer": {"type": "AdamW",
                       "params": {"lr": 1e-2, "weight_decay": 0.01}}})
@@ -445,7 +445,10 @@ def _z3_zoo_worker(rank, world, arch):
     for g, w in zip(got, want):
         assert abs(g - w) < 1e-4, (arch, got, want)
 
-    fp32 = opt.get_full_state_dict(dtype=torch.float32)
+    if stage == 3:
+        fp32 = opt.get_full_state_dict(dtype=torch.float32)
+    else:
+        fp32 = opt.get_fp32_state_dict(engine.module)
     if rank == 0:
         for n, p in ref.named_parameters():
             if arch == "unused" and n.startswith("extra."):
@@ -453,7 +456,11 @@ def _z3_zoo_worker(rank, world, arch):
                 # never receive grads still get weight decay through the
                 # zero-grad Adam step; torch.AdamW skips them entirely
                 continue
-            torch.testing.assert_close(fp32[n].float(), p.detach(),
+            if n not in fp32:  # stage-1/2: frozen params stay in the module
+                got = dict(engine.module.named_parameters())[n].detach()
+            else:
+                got = fp32[n].float()
+            torch.testing.assert_close(got, p.detach(),
                                        rtol=1e-4, atol=1e-5,
                                        msg=f"{arch}/{n}")
         if arch == "frozen":
@@ -468,3 +475,8 @@ def _z3_zoo_worker(rank, world, arch):
 def test_zero3_module_zoo():
     for arch in ("plain", "tied", "frozen", "unused"):
         run_distributed(_z3_zoo_worker, world_size=2, args=(arch,))
+
+
+def test_zero2_module_zoo():
+    for arch in ("tied", "frozen", "unused"):
+        run_distributed(_z3_zoo_worker, world_size=2, args=(arch, 2))
