@@ -219,3 +219,40 @@ def test_data_analyzer_end_to_end(tmp_path):
     sampler = DeepSpeedDataSampler(diffs, sched, batch_size=2)
     first = next(iter(sampler))
     assert all(diffs[i] <= 8 for i in first)
+
+
+def test_sparse_attention_layouts():
+    from deepspeed_amd.ops.sparse_attention import (BigBirdSparsityConfig,
+                                                    DenseSparsityConfig,
+                                                    FixedSparsityConfig,
+                                                    SparseSelfAttention)
+    import torch.nn.functional as F
+    H, B, S, D, blk = 2, 2, 64, 8, 16
+    q = torch.randn(B, H, S, D)
+    k = torch.randn(B, H, S, D)
+    v = torch.randn(B, H, S, D)
+
+    # dense config == plain attention
+    dense = SparseSelfAttention(DenseSparsityConfig(H, blk))
+    torch.testing.assert_close(dense(q, k, v),
+                               F.scaled_dot_product_attention(q, k, v))
+
+    fixed = FixedSparsityConfig(H, blk, num_local_blocks=2,
+                                attention="unidirectional")
+    layout = fixed.make_layout(S)
+    n = S // blk
+    assert layout.shape == (H, n, n)
+    assert not layout[:, 0, 1].any()          # causal: no future blocks
+    assert layout[:, 3, 2].all()              # local window
+    out = SparseSelfAttention(fixed)(q, k, v)
+    assert out.shape == q.shape and torch.isfinite(out).all()
+    # causal masked-out blocks change nothing: perturb a far-future key
+    k2 = k.clone()
+    k2[:, :, -1] += 100.0
+    out2 = SparseSelfAttention(fixed)(q, k2, v)
+    torch.testing.assert_close(out[:, :, :16], out2[:, :, :16])
+
+    bb = BigBirdSparsityConfig(H, blk, num_random_blocks=1)
+    lb = bb.make_layout(S)
+    assert lb[:, 2, 1].all() and lb[:, 2, 3].all()   # sliding window
+    assert lb[:, :, 0].all() and lb[:, 0, :].all()   # global block
