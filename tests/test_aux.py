@@ -256,3 +256,18 @@ def test_sparse_attention_layouts():
     lb = bb.make_layout(S)
     assert lb[:, 2, 1].all() and lb[:, 2, 3].all()   # sliding window
     assert lb[:, :, 0].all() and lb[:, 0, :].all()   # global block
+
+
+def test_zero_memory_estimators():
+    from deepspeed_amd.runtime.zero import (
+        estimate_zero2_model_states_mem_needs,
+        estimate_zero3_model_states_mem_needs)
+    P = 8_000_000_000
+    gpu3, cpu3 = estimate_zero3_model_states_mem_needs(P, 8, 1)
+    # 8B over 8 GPUs: (2+2+12)*1e9*1.5 = 24 GB/GPU
+    assert abs(gpu3 - 24e9) / 24e9 < 0.01 and cpu3 == 0
+    gpu3o, cpu3o = estimate_zero3_model_states_mem_needs(P, 8, 1,
+                                                         cpu_offload=True)
+    assert gpu3o < gpu3 and cpu3o > 100e9  # states land in host DRAM
+    gpu2, _ = estimate_zero2_model_states_mem_needs(P, 8, 1)
+    assert gpu2 > gpu3  # full replicas of params+grads
