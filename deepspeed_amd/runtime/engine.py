@@ -337,7 +337,10 @@ class Engine(torch.nn.Module):
         # over the full DPxSP mesh (see parallel/groups.py), which already
         # yields the gradient of the global token-mean loss.
 
-        self._last_loss = float(loss.detach().float())
+        # keep as a tensor: .item() here would force a device sync every
+        # microstep and stall CPU run-ahead; materialized only when the
+        # monitor actually writes
+        self._last_loss = loss.detach()
         if hasattr(self.optimizer, "backward"):
             self.optimizer.backward(loss, retain_graph=retain_graph)
         else:
@@ -443,7 +446,7 @@ class Engine(torch.nn.Module):
                       ("Train/loss_scale", float(self.loss_scale),
                        self.global_steps)]
             if getattr(self, "_last_loss", None) is not None:
-                events.append(("Train/loss", self._last_loss,
+                events.append(("Train/loss", float(self._last_loss),
                                self.global_steps))
             if self.tput_timer and self.tput_timer.avg_samples_per_sec():
                 events.append(("Train/samples_per_sec",
