@@ -25,6 +25,8 @@ from .utils.logging import logger, log_dist
 # convenience re-exports
 from .ops.adam import FusedAdam  # noqa: F401
 from .ops.norms import RMSNorm, FusedLayerNorm  # noqa: F401
+from .ops.transformer import (DeepSpeedTransformerConfig,  # noqa: F401
+                              DeepSpeedTransformerLayer)
 
 
 def initialize(args=None,

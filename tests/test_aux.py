@@ -271,3 +271,27 @@ def test_zero_memory_estimators():
     assert gpu3o < gpu3 and cpu3o > 100e9  # states land in host DRAM
     gpu2, _ = estimate_zero2_model_states_mem_needs(P, 8, 1)
     assert gpu2 > gpu3  # full replicas of params+grads
+
+
+def test_deepspeed_transformer_layer():
+    from deepspeed_amd import (DeepSpeedTransformerConfig,
+                               DeepSpeedTransformerLayer)
+    torch.manual_seed(0)
+    cfg = DeepSpeedTransformerConfig(hidden_size=64, intermediate_size=128,
+                                     heads=4, attn_dropout_ratio=0.0,
+                                     hidden_dropout_ratio=0.0,
+                                     pre_layer_norm=True, seed=11)
+    layer = DeepSpeedTransformerLayer(cfg)
+    x = torch.randn(2, 16, 64, requires_grad=True)
+    mask = torch.ones(2, 16, dtype=torch.long)
+    y = layer(x, mask)
+    assert y.shape == x.shape
+    y.sum().backward()
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+    # post-LN variant runs too
+    cfg2 = DeepSpeedTransformerConfig(hidden_size=64, intermediate_size=128,
+                                      heads=4, pre_layer_norm=False,
+                                      attn_dropout_ratio=0.0,
+                                      hidden_dropout_ratio=0.0)
+    y2 = DeepSpeedTransformerLayer(cfg2)(x)
+    assert y2.shape == x.shape
