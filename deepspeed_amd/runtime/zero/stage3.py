@@ -923,7 +923,72 @@ class ZeroStage3Optimizer:
             "fp32_flat_groups": [m.data if m.numel() else m
                                  for m in self.group_masters],
             "base_optimizer_state": self.optimizer.state_dict(),
+            # layout manifest for the offline universal-checkpoint converter
+            # (checkpoint/universal.py) — same schema as stage 1/2
+            "layout": self.layout_manifest(),
         }
+
+    def annotate_param_names(self, module):
+        for n, p in module.named_parameters():
+            p._ds_name = n
+
+    def layout_manifest(self):
+        units = []
+        for u in self.units:
+            units.append({
+                "group_idx": u.group_idx,
+                "master_offset": u.master_offset,
+                "shard_size": u.shard_size,
+                "numel": u.numel,
+                "pg_world": self.world_size,
+                "pg_rank": self.rank,
+                "params": [(getattr(p, "_ds_name", None), off, p.ds_numel,
+                            tuple(p.ds_shape))
+                           for p, off in zip(u.params, u.offsets)],
+            })
+        return units
+
+    @torch.no_grad()
+    def load_universal_state_dict(self, module, usd):
+        """Elastic ZeRO-3 load: slice per-param fp32 universal state into
+        this world size's unit shards (mirrors stage12's)."""
+        self.annotate_param_names(module)
+        for gi, group in enumerate(self.optimizer.param_groups):
+            master_p = self.group_masters[gi]
+            if master_p.numel() == 0:
+                continue
+            st = self.optimizer.state.setdefault(master_p, {})
+            if "exp_avg" not in st:
+                st["exp_avg"] = torch.zeros_like(master_p,
+                                                 dtype=torch.float32)
+                st["exp_avg_sq"] = torch.zeros_like(master_p,
+                                                    dtype=torch.float32)
+            st["step"] = usd.get("step", 0)
+        for u in self.units:
+            if u.group_idx < 0 or not u.trainable:
+                continue
+            master = self.group_masters[u.group_idx]
+            st = self.optimizer.state[master]
+            dsts = {"param": master.data, "exp_avg": st["exp_avg"],
+                    "exp_avg_sq": st["exp_avg_sq"]}
+            lo = self.rank * u.shard_size
+            hi = lo + u.shard_size
+            for p, off in zip(u.params, u.offsets):
+                name = getattr(p, "_ds_name", None)
+                if name is None or name not in usd["param"]:
+                    continue
+                a, z = max(off, lo), min(off + p.ds_numel, hi)
+                if a >= z:
+                    continue
+                for kind, dst in dsts.items():
+                    src = usd[kind][name].reshape(-1)
+                    dst[u.master_offset + (a - lo):
+                        u.master_offset + (z - lo)].copy_(src[a - off:z - off])
+        self._copy_masters_to_shards()
+        self._refresh_persistent()
+        for u in self.units:
+            if not u.persist and u.status == AVAILABLE:
+                self._release(u)
 
     def load_state_dict(self, sd, load_optimizer_states=True):
         assert sd["world_size"] == self.world_size, \

@@ -17,13 +17,15 @@ _CONFIG = {
 }
 
 
-def _build_engine():
+def _build_engine(stage=1):
     import deepspeed_amd
     from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    import copy
     torch.manual_seed(31)
     model = LlamaForCausalLM(llama_tiny())
-    engine, opt, _, _ = deepspeed_amd.initialize(model=model,
-                                                 config=dict(_CONFIG))
+    cfg = copy.deepcopy(_CONFIG)
+    cfg["zero_optimization"]["stage"] = stage
+    engine, opt, _, _ = deepspeed_amd.initialize(model=model, config=cfg)
     return engine, opt
 
 
@@ -40,22 +42,28 @@ def _train(engine, batches):
     return loss
 
 
-def _save_worker(rank, world, tmp):
-    engine, _ = _build_engine()
+def _save_worker(rank, world, tmp, stage=1):
+    engine, _ = _build_engine(stage)
     _train(engine, _same_batches(2))  # identical batch on both ranks
     engine.save_checkpoint(tmp, tag="step2")
     # continue one more step; dump resulting fp32 masters for comparison
     _train(engine, _same_batches(3)[2:])
-    fp32 = engine.optimizer.get_fp32_state_dict(engine.module)
+    if hasattr(engine.optimizer, "get_fp32_state_dict"):
+        fp32 = engine.optimizer.get_fp32_state_dict(engine.module)
+    else:
+        fp32 = engine.optimizer.get_full_state_dict(dtype=torch.float32)
     if rank == 0:
         torch.save(fp32, os.path.join(tmp, "ref_after3.pt"))
 
 
-def _resume_worker(rank, world, tmp):
-    engine, opt = _build_engine()
+def _resume_worker(rank, world, tmp, stage=1):
+    engine, opt = _build_engine(stage)
     engine.load_checkpoint(tmp, tag="step2", load_universal=True)
     _train(engine, _same_batches(3)[2:])
-    fp32 = opt.get_fp32_state_dict(engine.module)
+    if hasattr(opt, "get_fp32_state_dict"):
+        fp32 = opt.get_fp32_state_dict(engine.module)
+    else:
+        fp32 = opt.get_full_state_dict(dtype=torch.float32)
     ref = torch.load(os.path.join(tmp, "ref_after3.pt"), weights_only=False)
     assert set(fp32) == set(ref)
     for k in ref:
@@ -76,3 +84,11 @@ def test_universal_checkpoint_dp2_to_dp1(tmp_path):
     assert all(v.dtype == torch.float32 for v in usd["param"].values())
 
     run_local(_resume_worker, args=(tmp,))
+
+
+def test_universal_checkpoint_zero3_dp2_to_dp1(tmp_path):
+    tmp = str(tmp_path)
+    run_distributed(_save_worker, world_size=2, args=(tmp, 3))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="step2")
+    run_local(_resume_worker, args=(tmp, 3))
