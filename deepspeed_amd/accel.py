@@ -146,3 +146,33 @@ def supports_bf16() -> bool:
     if available():
         return torch.cuda.is_bf16_supported()
     return True  # CPU bf16 emulation works for tests
+
+
+def get_accelerator():
+    """Reference-API compatibility (accelerator/real_accelerator.py:51):
+    returns this module — deliberately a single ROCm/MI355X implementation,
+    not an N-backend ABC (SURVEY §7 design stance)."""
+    import sys
+    return sys.modules[__name__]
+
+
+# reference DeepSpeedAccelerator method-name aliases
+def device_count() -> int:
+    import torch
+    return torch.cuda.device_count() if available() else 0
+
+
+def memory_allocated(device=None) -> int:
+    import torch
+    return torch.cuda.memory_allocated(device) if available() else 0
+
+
+def max_memory_allocated(device=None) -> int:
+    import torch
+    return torch.cuda.max_memory_allocated(device) if available() else 0
+
+
+def empty_cache() -> None:
+    import torch
+    if available():
+        torch.cuda.empty_cache()
