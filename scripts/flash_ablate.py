@@ -10,7 +10,7 @@ v = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
 vt = v.permute(0, 2, 3, 1).contiguous()
 ref = sdpa_reference(q, k, v, causal=True).float()
 scale = 1.0 / math.sqrt(D)
-for var in (0, 1, 2, 3):
+for var in (0, 1, 2, 3, 4):
     o = _C.flash_attn_fwd_dbg(q.contiguous(), k.contiguous(), vt, scale, var).float()
     err = (o - ref).abs()
     rel = (err.max() / ref.abs().max()).item()
