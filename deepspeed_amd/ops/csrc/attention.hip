@@ -73,6 +73,7 @@ __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
     const short* __restrict__ k,   // [B, S, Hkv, D]
     const short* __restrict__ vt,  // [B, Hkv, D, S]
     short* __restrict__ o,         // [B, S, H, D]
+    float* __restrict__ lse,       // optional [B, H, S] log-sum-exp
     const int B, const int S, const int H, const int Hkv,
     const float scale) {
   constexpr int KB = KSUB * 32;   // kv rows staged per barrier pair
@@ -218,6 +219,8 @@ __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
   // ---- epilogue: O[q][d] = O^T / l
   if (!q_ok) return;
   const float inv_l = l_run > 0.f ? 1.f / l_run : 0.f;
+  if (lse != nullptr && half == 0)  // one lane half owns the q row
+    lse[((long long)b * H + h) * S + qrow] = m_run + __logf(l_run);
   short* op = o + (((long long)b * S + qrow) * H + h) * D;
 #pragma unroll
   for (int dblk = 0; dblk < 4; ++dblk) {
@@ -232,18 +235,18 @@ __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
 }  // namespace
 
 extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
-                             void* o, int B, int S, int H, int Hkv,
-                             float scale, int causal, void* stream) {
+                             void* o, void* lse, int B, int S, int H,
+                             int Hkv, float scale, int causal, void* stream) {
   dim3 grid((S + QTILE - 1) / QTILE, H, B);
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
   if (causal)
     hipLaunchKernelGGL((flash_fwd_kernel<true>), grid, dim3(TPB), 0, st,
                        (const short*)q, (const short*)k, (const short*)vt,
-                       (short*)o, B, S, H, Hkv, scale);
+                       (short*)o, (float*)lse, B, S, H, Hkv, scale);
   else
     hipLaunchKernelGGL((flash_fwd_kernel<false>), grid, dim3(TPB), 0, st,
                        (const short*)q, (const short*)k, (const short*)vt,
-                       (short*)o, B, S, H, Hkv, scale);
+                       (short*)o, (float*)lse, B, S, H, Hkv, scale);
 }
 
 // variants 0-3: K/V LDS ablation at KSUB=1; variant 4: KVBLK=64 (KSUB=2)
@@ -256,26 +259,27 @@ extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
     case 0:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 0>), grid, dim3(TPB), 0, st,
                          (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, B, S, H, Hkv, scale);
+                         (short*)o, nullptr, B, S, H, Hkv, scale);
       break;
     case 1:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 1>), grid, dim3(TPB), 0, st,
                          (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, B, S, H, Hkv, scale);
+                         (short*)o, nullptr, B, S, H, Hkv, scale);
       break;
     case 2:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 2>), grid, dim3(TPB), 0, st,
                          (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, B, S, H, Hkv, scale);
+                         (short*)o, nullptr, B, S, H, Hkv, scale);
       break;
     case 4:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 3, 2>), grid, dim3(TPB), 0,
                          st, (const short*)q, (const short*)k,
-                         (const short*)vt, (short*)o, B, S, H, Hkv, scale);
+                         (const short*)vt, (short*)o, nullptr, B, S, H, Hkv,
+                         scale);
       break;
     default:
       hipLaunchKernelGGL((flash_fwd_kernel<true, 3>), grid, dim3(TPB), 0, st,
                          (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, B, S, H, Hkv, scale);
+                         (short*)o, nullptr, B, S, H, Hkv, scale);
   }
 }

@@ -47,8 +47,8 @@ extern "C" int ds_aio_pread(void* h, void* data, long long nbytes,
                             const char* path);
 extern "C" int ds_aio_wait(void* h);
 extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
-                             void* o, int B, int S, int H, int Hkv,
-                             float scale, int causal, void* stream);
+                             void* o, void* lse, int B, int S, int H,
+                             int Hkv, float scale, int causal, void* stream);
 extern "C" void ds_cpu_lion_flat(float* p, const void* g, int grad_dtype,
                                  float* m, void* p16, long long n, float lr,
                                  float beta1, float beta2, float weight_decay,
@@ -249,8 +249,9 @@ at::Tensor flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor vt,
   TORCH_CHECK(k.size(1) == S && vt.size(3) == S && vt.size(2) == D &&
               vt.size(1) == Hkv && H % Hkv == 0, "flash_fwd: shape mismatch");
   auto o = at::empty_like(q);
-  ds_flash_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(), B, S,
-               H, Hkv, (float)scale, causal ? 1 : 0, cur_stream());
+  ds_flash_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(),
+               nullptr, B, S, H, Hkv, (float)scale, causal ? 1 : 0,
+               cur_stream());
   return o;
 }
 
