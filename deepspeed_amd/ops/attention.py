@@ -51,3 +51,19 @@ def sdpa_reference(q, k, v, causal=True, scale=None):
     o = torch.nn.functional.scaled_dot_product_attention(
         qt, kt, vt, is_causal=causal, scale=scale)
     return o.transpose(1, 2).to(q.dtype)
+
+
+def flash_attn_bwd(q, k, v, o, do, lse, causal=True, scale=None):
+    """Backward pass in BHSD ([B,H(kv),S,D] bf16; lse [B,H,S] fp32).
+    Returns (dq, dk, dv). GPU-validation pending (round-2 item) — the
+    tile algebra is CPU-verified in ops/flash_bwd_ref.py."""
+    ext = get_ext()
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.size(-1))
+    delta = (do.float() * o.float()).sum(-1)              # [B,H,S]
+    qt = q.transpose(-1, -2).contiguous()
+    kt = k.transpose(-1, -2).contiguous()
+    dot = do.transpose(-1, -2).contiguous()
+    return ext.flash_attn_bwd(q.contiguous(), k.contiguous(), v.contiguous(),
+                              do.contiguous(), qt, kt, dot,
+                              lse.contiguous().float(), delta.contiguous(),
+                              scale, causal)

@@ -53,6 +53,12 @@ extern "C" void ds_cpu_lion_flat(float* p, const void* g, int grad_dtype,
                                  float* m, void* p16, long long n, float lr,
                                  float beta1, float beta2, float weight_decay,
                                  float inv_scale);
+extern "C" void ds_flash_bwd(const void* q, const void* k, const void* v,
+                             const void* dout, const void* qt,
+                             const void* kt, const void* dot,
+                             const float* lse, const float* delta, void* dq,
+                             void* dk, void* dv, int B, int S, int H,
+                             int Hkv, float scale, int causal, void* stream);
 extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
                                  void* o, int B, int S, int H, int Hkv,
                                  float scale, int variant, void* stream);
@@ -323,6 +329,33 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Groupwise symmetric int8/int4 quantization");
   m.def("groupwise_dequant", &groupwise_dequant,
         "Groupwise symmetric int8/int4 dequantization");
+  m.def("flash_attn_bwd",
+        [](at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout,
+           at::Tensor qt, at::Tensor kt, at::Tensor dot, at::Tensor lse,
+           at::Tensor delta, double scale, bool causal) {
+          TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+                      q.is_contiguous() && k.is_contiguous() &&
+                      v.is_contiguous() && dout.is_contiguous() &&
+                      qt.is_contiguous() && kt.is_contiguous() &&
+                      dot.is_contiguous(), "flash_bwd: bf16 contiguous GPU");
+          TORCH_CHECK(lse.scalar_type() == at::kFloat &&
+                      delta.scalar_type() == at::kFloat, "lse/delta fp32");
+          const int B = q.size(0), H = q.size(1), S = q.size(2);
+          const int Hkv = k.size(1);
+          TORCH_CHECK(q.size(3) == 128 && S % 32 == 0,
+                      "flash_bwd: D=128, S%32==0");
+          auto dq = at::empty_like(q);
+          auto dk = at::empty_like(k);
+          auto dv = at::empty_like(v);
+          ds_flash_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                       dout.data_ptr(), qt.data_ptr(), kt.data_ptr(),
+                       dot.data_ptr(), lse.data_ptr<float>(),
+                       delta.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                       dv.data_ptr(), B, S, H, Hkv, (float)scale,
+                       causal ? 1 : 0, cur_stream());
+          return std::make_tuple(dq, dk, dv);
+        },
+        "MFMA flash-attention backward (BHSD; GPU-validation pending)");
   m.def("flash_attn_fwd_dbg",
         [](at::Tensor q, at::Tensor k, at::Tensor vt, double scale,
            int64_t variant) {
