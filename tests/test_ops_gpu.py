@@ -5,6 +5,8 @@ microsoft/DeepSpeed): same-op comparison within dtype tolerance. Every test
 here requires an MI355X and the in-tree extension (no eager fallback).
 """
 
+import os
+
 import pytest
 import torch
 
