@@ -328,3 +328,45 @@ def test_flash_bwd_reference_matches_autograd():
     torch.testing.assert_close(dq, q.grad, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(dk, k.grad, rtol=1e-4, atol=1e-5)
     torch.testing.assert_close(dv, v.grad, rtol=1e-4, atol=1e-5)
+
+
+def test_flash_bwd_lds_transpose_maps():
+    """Lane-exact simulation of attention_bwd.hip's C/D->A LDS transposes
+    (the only index machinery not shared with the GPU-validated forward)."""
+    import numpy as np
+
+    def cd_row(r, half):
+        return (r & 3) + 8 * (r >> 2) + 4 * half
+
+    T = 32
+    PT = np.arange(T * T).reshape(T, T)  # P^T[kv][q] distinct values
+
+    # C/D fragments: lane holds col=lane&31, rows cd_row(r, lane>>5)
+    lds = np.zeros((T, T), dtype=int)
+    for lane in range(64):
+        col, half = lane & 31, lane >> 5
+        for r in range(16):
+            lds[cd_row(r, half), col] = PT[cd_row(r, half), col]
+    assert (lds == PT).all()
+
+    # kernel-1 A-frag read: row=lane&31 (kv), k-slot q = kk*16+8*half+[0,8)
+    for lane in range(64):
+        row, half = lane & 31, lane >> 5
+        for kk in range(2):
+            frag = lds[row, kk * 16 + 8 * half: kk * 16 + 8 * half + 8]
+            assert (frag == PT[row, kk * 16 + 8 * half:
+                               kk * 16 + 8 * half + 8]).all()
+
+    # kernel-2 transposed store: lds2[q][kv] then read row=q, k=kv
+    lds2 = np.zeros((T, T), dtype=int)
+    for lane in range(64):
+        col, half = lane & 31, lane >> 5
+        for r in range(16):
+            lds2[col, cd_row(r, half)] = PT[cd_row(r, half), col]
+    assert (lds2 == PT.T).all()
+    for lane in range(64):
+        row, half = lane & 31, lane >> 5
+        for kk in range(2):
+            frag = lds2[row, kk * 16 + 8 * half: kk * 16 + 8 * half + 8]
+            assert (frag == PT.T[row, kk * 16 + 8 * half:
+                                 kk * 16 + 8 * half + 8]).all()
