@@ -67,3 +67,35 @@ def flash_attn_bwd(q, k, v, o, do, lse, causal=True, scale=None):
                               do.contiguous(), qt, kt, dot,
                               lse.contiguous().float(), delta.contiguous(),
                               scale, causal)
+
+
+class FlashAttnFunc(torch.autograd.Function):
+    """Training-path flash attention (BSHD in/out). GPU-validation of the
+    backward kernels is a round-2 gate; until then the model keeps SDPA."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        ext = get_ext()
+        scale = scale if scale is not None else 1.0 / math.sqrt(q.size(-1))
+        vt = v.permute(0, 2, 3, 1).contiguous()
+        o, lse = ext.flash_attn_fwd_lse(q.contiguous(), k.contiguous(), vt,
+                                        scale, causal)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal, ctx.scale = causal, scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        # kernels run in BHSD
+        qh, kh, vh, oh, doh = (t.transpose(1, 2).contiguous()
+                               for t in (q, k, v, o, do))
+        dq, dk, dv = flash_attn_bwd(qh, kh, vh, oh, doh, lse,
+                                    causal=ctx.causal, scale=ctx.scale)
+        return (dq.transpose(1, 2), dk.transpose(1, 2),
+                dv.transpose(1, 2), None, None)
+
+
+def flash_attn_func(q, k, v, causal=True, scale=None):
+    """Differentiable flash attention, [B,S,H(kv),D] bf16."""
+    return FlashAttnFunc.apply(q, k, v, causal, scale)

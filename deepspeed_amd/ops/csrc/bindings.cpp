@@ -261,6 +261,18 @@ at::Tensor flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor vt,
   return o;
 }
 
+std::tuple<at::Tensor, at::Tensor> flash_attn_fwd_lse(
+    at::Tensor q, at::Tensor k, at::Tensor vt, double scale, bool causal) {
+  const int B = q.size(0), S = q.size(1), H = q.size(2);
+  const int Hkv = k.size(2);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  ds_flash_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(),
+               lse.data_ptr<float>(), B, S, H, Hkv, (float)scale,
+               causal ? 1 : 0, cur_stream());
+  return {o, lse};
+}
+
 class AioHandle {
  public:
   AioHandle(int64_t block_size, int n_threads)
@@ -329,6 +341,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Groupwise symmetric int8/int4 quantization");
   m.def("groupwise_dequant", &groupwise_dequant,
         "Groupwise symmetric int8/int4 dequantization");
+  m.def("flash_attn_fwd_lse", &flash_attn_fwd_lse,
+        "flash fwd returning (o, logsumexp) for training");
   m.def("flash_attn_bwd",
         [](at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout,
            at::Tensor qt, at::Tensor kt, at::Tensor dot, at::Tensor lse,
