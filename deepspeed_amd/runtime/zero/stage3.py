@@ -178,6 +178,15 @@ class ZeroStage3Optimizer:
         from .partition import register_stage3
         register_stage3(self)
 
+        # honesty about accepted-but-inert knobs (reference parity options
+        # whose MI355X design doesn't need them — see module docstring)
+        if int(zc.sub_group_size) != 1_000_000_000_000:
+            log_dist("ZeRO-3: sub_group_size is accepted but inert (module-"
+                     "unit partitioning already bounds working-set size)")
+        if zc.round_robin_gradients:
+            log_dist("ZeRO-3: round_robin_gradients is inert (per-unit "
+                     "reduce-scatter has no bucket-order imbalance)")
+
         n_persist = sum(1 for u in self.units if u.persist)
         log_dist(f"ZeRO stage 3: world={self.world_size} units={len(self.units)} "
                  f"(persistent={n_persist}) "

@@ -271,3 +271,32 @@ def test_flash_attn_bwd_numerics(shape):
         rel = (got.float() - want.float()).abs().max() / \
             want.float().abs().max()
         assert rel < 5e-2, (name, rel)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("DS_AMD_FLASH_BWD_TEST") != "1",
+                    reason="flash bwd GPU validation pending (round-2)")
+def test_flash_attn_func_autograd():
+    """End-to-end differentiable flash attention vs SDPA autograd (BSHD)."""
+    from deepspeed_amd.ops.attention import flash_attn_func, sdpa_reference
+    B, S, H, Hkv, D = 2, 512, 8, 2, 128
+    torch.manual_seed(1)
+    q = torch.randn(B, S, H, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = flash_attn_func(q, k, v, causal=True)
+    do = torch.randn_like(o)
+    o.backward(do)
+    got = (q.grad.clone(), k.grad.clone(), v.grad.clone())
+
+    q2 = q.detach().clone().requires_grad_(True)
+    k2 = k.detach().clone().requires_grad_(True)
+    v2 = v.detach().clone().requires_grad_(True)
+    ref = sdpa_reference(q2, k2, v2, causal=True)
+    ref.backward(do)
+    for g, w, n in zip(got, (q2.grad, k2.grad, v2.grad), "qkv"):
+        rel = (g.float() - w.float()).abs().max() / w.float().abs().max()
+        assert rel < 5e-2, (n, rel)
