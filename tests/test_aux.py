@@ -370,3 +370,80 @@ def test_flash_bwd_lds_transpose_maps():
             frag = lds2[row, kk * 16 + 8 * half: kk * 16 + 8 * half + 8]
             assert (frag == PT.T[row, kk * 16 + 8 * half:
                                  kk * 16 + 8 * half + 8]).all()
+
+
+def test_flash_bwd_address_math():
+    """Mirror attention_bwd.hip's pointer arithmetic in Python and check
+    every fragment load/store hits the intended (b,h,s,d) element."""
+    B, H, Hkv, S, D = 2, 4, 2, 64, 128
+    G = H // Hkv
+
+    def cd_row(r, half):
+        return (r & 3) + 8 * (r >> 2) + 4 * half
+
+    def bhsd(b, h, s, d, HH):  # flat index of [B, HH, S, D]
+        return ((b * HH + h) * S + s) * D + d
+
+    def bhds(b, h, d, s, HH):  # flat index of [B, HH, D, S] (qt/kt/dot)
+        return ((b * HH + h) * D + d) * S + s
+
+    b, hkv, kv0 = 1, 1, 32
+    for lane in range(0, 64, 17):
+        col, half = lane & 31, lane >> 5
+        kvrow = kv0 + col
+        # K/V A-frag base (kernel 1)
+        base = ((b * Hkv + hkv) * S + kvrow) * D + 8 * half
+        for kk in range(8):
+            for reg in range(8):
+                want = bhsd(b, hkv, kvrow, kk * 16 + 8 * half + reg, Hkv)
+                assert base + kk * 16 + reg == want
+        for g in range(G):
+            h = hkv * G + g
+            qbase = ((b * H + h) * S) * D
+            tbase = ((b * H + h) * D) * S
+            sbase = (b * H + h) * S
+            qs = 32
+            qrow = qs + col
+            # Q/dO B-frags
+            for kk in range(8):
+                for reg in range(8):
+                    assert qbase + qrow * D + kk * 16 + 8 * half + reg == \
+                        bhsd(b, h, qrow, kk * 16 + 8 * half + reg, H)
+            # lse/delta scalars
+            assert sbase + qrow == (b * H + h) * S + qrow
+            # dot/qt B-frags for dV/dK: (col=d, k=q)
+            for dblk in range(4):
+                trow = (dblk * 32 + col) * S + qs
+                for kk in range(2):
+                    for reg in range(8):
+                        q_idx = qs + kk * 16 + 8 * half + reg
+                        assert tbase + trow + kk * 16 + 8 * half + reg == \
+                            bhds(b, h, dblk * 32 + col, q_idx, H)
+        # dk/dv writes: (col=d_local, row=kv)
+        obase = ((b * Hkv + hkv) * S) * D
+        for dblk in range(4):
+            for r in range(16):
+                kvl = cd_row(r, half)
+                assert obase + (kv0 + kvl) * D + dblk * 32 + col == \
+                    bhsd(b, hkv, kv0 + kvl, dblk * 32 + col, Hkv)
+
+    # kernel 2: kt B-frag (col=d, k=kv) and dq writes
+    h, qs = 3, 32
+    hkv2 = h // G
+    ktbase = ((b * Hkv + hkv2) * D) * S
+    for lane in (0, 33, 63):
+        col, half = lane & 31, lane >> 5
+        for dblk in range(4):
+            kv0b = 0
+            trow = ktbase + (dblk * 32 + col) * S + kv0b
+            for kk in range(2):
+                for reg in range(8):
+                    kv_idx = kv0b + kk * 16 + 8 * half + reg
+                    assert trow + kk * 16 + 8 * half + reg == \
+                        bhds(b, hkv2, dblk * 32 + col, kv_idx, Hkv)
+        qbase = ((b * H + h) * S) * D
+        for dblk in range(4):
+            for r in range(16):
+                ql = cd_row(r, half)
+                assert qbase + (qs + ql) * D + dblk * 32 + col == \
+                    bhsd(b, h, qs + ql, dblk * 32 + col, H)
