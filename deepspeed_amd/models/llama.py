@@ -264,7 +264,10 @@ class LlamaForCausalLM(nn.Module):
 
 def chunked_cross_entropy(hidden, lm_head, labels, chunk_tokens=8192):
     """loss = CE(lm_head(hidden), labels), computed in token chunks so the
-    fp32 logits never exceed chunk_tokens x vocab."""
+    fp32 logits never exceed chunk_tokens x vocab. With DS_AMD_FUSED_CE=1
+    the per-chunk softmax+nll runs as the fused bf16 HIP kernel and no
+    fp32 logits are materialized at all."""
+    from ..ops.cross_entropy import fused_cross_entropy_sum
     B, S, H = hidden.shape
     hidden = hidden.reshape(-1, H)
     labels = labels.reshape(-1)
@@ -273,9 +276,8 @@ def chunked_cross_entropy(hidden, lm_head, labels, chunk_tokens=8192):
     count = hidden.new_zeros((), dtype=torch.float32)
     for s in range(0, n, chunk_tokens):
         e = min(s + chunk_tokens, n)
-        logits = lm_head(hidden[s:e]).float()
-        tgt = labels[s:e]
-        total = total + F.cross_entropy(logits, tgt, ignore_index=-100,
-                                        reduction="sum")
-        count = count + (tgt != -100).sum()
+        logits = lm_head(hidden[s:e])
+        l, c = fused_cross_entropy_sum(logits, labels[s:e])
+        total = total + l
+        count = count + c
     return total / count.clamp(min=1)

@@ -59,6 +59,13 @@ extern "C" void ds_flash_bwd(const void* q, const void* k, const void* v,
                              const float* lse, const float* delta, void* dq,
                              void* dk, void* dv, int B, int S, int H,
                              int Hkv, float scale, int causal, void* stream);
+extern "C" void ds_ce_fwd(const void* logits, const long long* labels,
+                          float* loss, float* lse, int N, long long V,
+                          long long ignore_index, void* stream);
+extern "C" void ds_ce_bwd(const void* logits, const long long* labels,
+                          const float* lse, const float* gscale,
+                          void* dlogits, int N, long long V,
+                          long long ignore_index, void* stream);
 extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
                                  void* o, int B, int S, int H, int Hkv,
                                  float scale, int variant, void* stream);
@@ -341,6 +348,33 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Groupwise symmetric int8/int4 quantization");
   m.def("groupwise_dequant", &groupwise_dequant,
         "Groupwise symmetric int8/int4 dequantization");
+  m.def("ce_fwd",
+        [](at::Tensor logits, at::Tensor labels, int64_t ignore_index) {
+          TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() &&
+                      logits.scalar_type() == at::kBFloat16 &&
+                      labels.scalar_type() == at::kLong, "ce_fwd args");
+          const int N = logits.size(0);
+          const long long V = logits.size(1);
+          auto loss = at::empty({N}, logits.options().dtype(at::kFloat));
+          auto lse = at::empty({N}, logits.options().dtype(at::kFloat));
+          ds_ce_fwd(logits.data_ptr(), (const long long*)labels.data_ptr<int64_t>(),
+                    loss.data_ptr<float>(), lse.data_ptr<float>(), N, V,
+                    ignore_index, cur_stream());
+          return std::make_tuple(loss, lse);
+        },
+        "fused CE forward: bf16 logits -> per-token loss + lse");
+  m.def("ce_bwd",
+        [](at::Tensor logits, at::Tensor labels, at::Tensor lse,
+           at::Tensor gscale, int64_t ignore_index) {
+          const int N = logits.size(0);
+          const long long V = logits.size(1);
+          auto d = at::empty_like(logits);
+          ds_ce_bwd(logits.data_ptr(), (const long long*)labels.data_ptr<int64_t>(),
+                    lse.data_ptr<float>(), gscale.data_ptr<float>(),
+                    d.data_ptr(), N, V, ignore_index, cur_stream());
+          return d;
+        },
+        "fused CE backward: dlogits in bf16");
   m.def("flash_attn_fwd_lse", &flash_attn_fwd_lse,
         "flash fwd returning (o, logsumexp) for training");
   m.def("flash_attn_bwd",

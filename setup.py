@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "quantize.hip"),
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "attention_bwd.hip"),
+        os.path.join(CSRC, "cross_entropy.hip"),
         os.path.join(CSRC, "norms.hip"),
         os.path.join(CSRC, "rope.hip"),
         os.path.join(CSRC, "swiglu.hip"),

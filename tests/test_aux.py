@@ -447,3 +447,19 @@ def test_flash_bwd_address_math():
                 ql = cd_row(r, half)
                 assert qbase + (qs + ql) * D + dblk * 32 + col == \
                     bhsd(b, h, qs + ql, dblk * 32 + col, H)
+
+
+def test_fused_ce_fallback_matches_fp32_path():
+    """CPU fallback of fused_cross_entropy_sum == the original fp32 chunked
+    CE math (this path feeds every existing engine test)."""
+    from deepspeed_amd.ops.cross_entropy import fused_cross_entropy_sum
+    torch.manual_seed(0)
+    logits = torch.randn(64, 100)
+    labels = torch.randint(0, 100, (64,))
+    labels[::5] = -100
+    l, c = fused_cross_entropy_sum(logits, labels)
+    ref = torch.nn.functional.cross_entropy(logits.float(), labels,
+                                            ignore_index=-100,
+                                            reduction="sum")
+    torch.testing.assert_close(l, ref)
+    assert c.item() == (labels != -100).sum().item()

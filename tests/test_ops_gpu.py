@@ -300,3 +300,29 @@ def test_flash_attn_func_autograd():
     for g, w, n in zip(got, (q2.grad, k2.grad, v2.grad), "qkv"):
         rel = (g.float() - w.float()).abs().max() / w.float().abs().max()
         assert rel < 5e-2, (n, rel)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(os.environ.get("DS_AMD_FUSED_CE_TEST") != "1",
+                    reason="fused CE GPU validation pending (round-2); "
+                           "set DS_AMD_FUSED_CE_TEST=1 to run")
+def test_fused_cross_entropy_numerics():
+    """Fused bf16 CE (loss + dlogits) vs fp32 torch cross_entropy."""
+    from deepspeed_amd.ops.cross_entropy import _FusedCE
+    torch.manual_seed(0)
+    N, V = 1024, 32000
+    logits = (torch.randn(N, V, device="cuda") * 3).bfloat16() \
+        .requires_grad_(True)
+    labels = torch.randint(0, V, (N,), device="cuda")
+    labels[::7] = -100  # ignore rows
+
+    loss_sum, count = _FusedCE.apply(logits, labels, -100)
+    (loss_sum / count).backward()
+
+    l2 = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(l2, labels, ignore_index=-100,
+                                            reduction="sum")
+    (ref / count).backward()
+    assert abs(loss_sum.item() - ref.item()) / ref.item() < 2e-2
+    err = (logits.grad.float() - l2.grad).abs().max()
+    assert err < 1e-3, err
