@@ -141,6 +141,12 @@ def test_zero_stage_bf16_parity_ws2(stage):
 
 
 @pytest.mark.parametrize("stage", [2, 3])
+def test_zero_stage_bf16_parity_ws4(stage):
+    """ws=4: shard shapes closer to the 8-GPU node the driver benches."""
+    run_distributed(_zero_worker, world_size=4, args=(stage, 1, 0.0, "bf16"))
+
+
+@pytest.mark.parametrize("stage", [2, 3])
 def test_zero_gas_parity_ws2(stage):
     run_distributed(_zero_worker, world_size=2, args=(stage, 3, 0.0, "bf16"))
 
