@@ -168,6 +168,10 @@ class Engine(torch.nn.Module):
         name = cfg.type.lower()
         if model_parameters is None:
             model_parameters = [p for p in self.module.parameters() if p.requires_grad]
+        if isinstance(model_parameters, list) and not model_parameters:
+            # e.g. a parameterless pipeline stage (activation-only layers)
+            log_dist("no trainable parameters on this rank - DummyOptim")
+            return DummyOptim([])
         if self.has_moe_layers:
             from ..moe.layer import \
                 split_params_into_different_moe_groups_for_optimizer
@@ -199,6 +203,10 @@ class Engine(torch.nn.Module):
         self.basic_optimizer = basic
         if basic is None:
             self.optimizer = None
+            return
+
+        if isinstance(basic, DummyOptim):
+            self.optimizer = basic
             return
 
         stage = self.zero_stage

@@ -102,10 +102,14 @@ def empty_cache():
 
 
 class DummyOptim(torch.optim.Optimizer):
-    """Placeholder optimizer when the user trains without one (inference/eval)."""
+    """Placeholder optimizer when the user trains without one (inference/
+    eval, or a parameterless pipeline stage)."""
 
     def __init__(self, params):
-        super().__init__(params, defaults={})
+        params = list(params)
+        if not params:  # torch rejects an empty parameter list
+            params = [torch.nn.Parameter(torch.zeros(1))]
+        super().__init__(params, defaults={"lr": 0.0})
 
     def step(self, closure=None):
         pass

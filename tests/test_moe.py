@@ -211,3 +211,8 @@ def test_mixtral_gpu_train_step():
         losses.append(loss.item())
     assert all(map(torch.isfinite, map(torch.tensor, losses)))
     assert losses[-1] < losses[0]
+
+
+def test_mixtral_ep4_zero_train():
+    """EP=4 over 4 ranks: full-mesh a2a with one expert slice per rank."""
+    run_distributed(_mixtral_train_worker, world_size=4, args=(2,))

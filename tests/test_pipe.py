@@ -145,3 +145,7 @@ def _tied_worker(rank, world):
 
 def test_pipeline_tied_weights():
     run_distributed(_tied_worker, world_size=2)
+
+
+def test_pipeline_4stage_parity():
+    run_distributed(_pipe_parity_worker, world_size=4)
