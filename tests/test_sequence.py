@@ -182,3 +182,50 @@ def test_chunked_prefill_attention_exact():
     # offload variant: same result (host round-trip is a no-op on CPU)
     out = chunked_prefill_attention(q, k, v, chunk_size=32, kv_offload=True)
     torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+
+
+def _ulysses_ws4_worker(rank, world):
+    """SP=4 end-to-end with heads=8/kv=4 (kv heads divisible by sp)."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM
+    from deepspeed_amd.models.llama import LlamaConfig, enable_ulysses
+    from deepspeed_amd.parallel import groups
+
+    groups.initialize_sequence_parallel(world)
+    cfg = LlamaConfig(vocab_size=256, hidden_size=64, intermediate_size=96,
+                      num_layers=2, num_heads=8, num_kv_heads=4,
+                      max_seq_len=64)
+    torch.manual_seed(41)
+    model = LlamaForCausalLM(cfg)
+    enable_ulysses(model)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": 1, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+
+    torch.manual_seed(41)
+    ref = LlamaForCausalLM(cfg)
+    opt_ref = torch.optim.AdamW(ref.parameters(), lr=1e-3)
+
+    torch.manual_seed(90)
+    S = 32
+    sl = slice(rank * S // world, (rank + 1) * S // world)
+    for _ in range(2):
+        ids = torch.randint(0, 256, (2, S + 1))
+        x, y = ids[:, :-1], ids[:, 1:]
+        pos = torch.arange(S, dtype=torch.int32).expand(2, S)
+        loss = engine(x[:, sl].contiguous(), labels=y[:, sl].contiguous(),
+                      positions=pos[:, sl].contiguous())
+        engine.backward(loss)
+        engine.step()
+        l2 = ref(x, labels=y)
+        l2.backward()
+        opt_ref.step()
+        opt_ref.zero_grad()
+    for (n, p), (_, pr) in zip(engine.module.named_parameters(),
+                               ref.named_parameters()):
+        torch.testing.assert_close(p, pr, rtol=2e-4, atol=3e-4), n
+
+
+def test_ulysses_sp4():
+    run_distributed(_ulysses_ws4_worker, world_size=4)

@@ -64,6 +64,8 @@ def _resume_worker(rank, world, tmp, stage=1):
         fp32 = opt.get_fp32_state_dict(engine.module)
     else:
         fp32 = opt.get_full_state_dict(dtype=torch.float32)
+    if rank != 0:
+        return  # consolidated fp32 export lands on rank 0 only
     ref = torch.load(os.path.join(tmp, "ref_after3.pt"), weights_only=False)
     assert set(fp32) == set(ref)
     for k in ref:
@@ -92,3 +94,12 @@ def test_universal_checkpoint_zero3_dp2_to_dp1(tmp_path):
     from deepspeed_amd.checkpoint import ds_to_universal
     ds_to_universal(tmp, tag="step2")
     run_local(_resume_worker, args=(tmp, 3))
+
+
+def test_universal_checkpoint_dp4_to_dp2(tmp_path):
+    """Elastic resume shrinking 4 -> 2 ranks (not just to a single rank)."""
+    tmp = str(tmp_path)
+    run_distributed(_save_worker, world_size=4, args=(tmp, 1))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="step2")
+    run_distributed(_resume_worker, world_size=2, args=(tmp, 1))
