@@ -61,3 +61,39 @@ def _tp_train_worker(rank, world):
 
 def test_tp2_training_parity():
     run_distributed(_tp_train_worker, world_size=2)
+
+
+def _tp4_worker(rank, world):
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM
+    from deepspeed_amd.models.llama import LlamaConfig
+
+    cfg = LlamaConfig(vocab_size=256, hidden_size=64, intermediate_size=96,
+                      num_layers=2, num_heads=8, num_kv_heads=4,
+                      max_seq_len=64)
+    torch.manual_seed(23)
+    model = LlamaForCausalLM(cfg)
+    torch.manual_seed(23)
+    ref = LlamaForCausalLM(cfg)
+
+    deepspeed_amd.tp_model_init(model, tp_size=world)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    opt_ref = torch.optim.AdamW(ref.parameters(), lr=1e-3)
+
+    torch.manual_seed(66)
+    for _ in range(2):
+        ids = torch.randint(0, 256, (2, 16))
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+        l2 = ref(ids, labels=ids)
+        l2.backward()
+        opt_ref.step()
+        opt_ref.zero_grad()
+        assert abs(loss.item() - l2.item()) < 2e-4, (loss.item(), l2.item())
+
+
+def test_tp4_training_parity():
+    run_distributed(_tp4_worker, world_size=4)
