@@ -146,6 +146,11 @@ class Engine(torch.nn.Module):
         if self.dp_world_size == 1:
             return
         from ..parallel import groups as pgroups
+        # src must be the group's OWN first rank: with TP/pipe grids the DP
+        # group need not contain global rank 0
+        dp_src = torch.distributed.get_global_rank(self.dp_group, 0) \
+            if self.dp_group is not None and \
+            self.dp_group is not torch.distributed.group.WORLD else 0
         for p in self.module.parameters():
             if not torch.is_tensor(p):
                 continue
@@ -155,10 +160,10 @@ class Engine(torch.nn.Module):
                     src = torch.distributed.get_global_rank(g, 0)
                     dist.broadcast(p.data, src=src, group=g)
             else:
-                dist.broadcast(p.data, src=0, group=self.dp_group)
+                dist.broadcast(p.data, src=dp_src, group=self.dp_group)
         for b in self.module.buffers():
             if torch.is_tensor(b) and b.numel() > 0 and b.dtype.is_floating_point:
-                dist.broadcast(b.data, src=0, group=self.dp_group)
+                dist.broadcast(b.data, src=dp_src, group=self.dp_group)
 
     def _configure_basic_optimizer(self, model_parameters):
         cfg = self.config.optimizer

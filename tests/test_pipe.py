@@ -149,3 +149,51 @@ def test_pipeline_tied_weights():
 
 def test_pipeline_4stage_parity():
     run_distributed(_pipe_parity_worker, world_size=4)
+
+
+def _pp2dp2_worker(rank, world):
+    """Hybrid PP=2 x DP=2: each stage has two data-parallel replicas; the
+    per-stage DP all-reduce must average their gradients."""
+    import deepspeed_amd
+    from deepspeed_amd.runtime.pipe.module import PipelineModule
+
+    loss_fn = nn.functional.mse_loss
+    net = PipelineModule(_make_layers(), num_stages=2, loss_fn=loss_fn,
+                         partition_method="parameters")
+    assert net.grid.data_parallel_size == 2
+    engine, _, _, _ = deepspeed_amd.initialize(model=net, config=dict(_CONFIG))
+
+    n_steps, mbs = 2, _CONFIG["gradient_accumulation_steps"]
+    dp = net.grid.data_parallel_id
+    data_all = [_make_data(n_steps * mbs, 4, seed=123 + d) for d in range(2)]
+    it = iter(data_all[dp])
+    losses = [engine.train_batch(it).item() for _ in range(n_steps)]
+
+    # reference: grads averaged over BOTH dp streams
+    ref = nn.Sequential(*_make_layers())
+    opt = torch.optim.AdamW(ref.parameters(), lr=1e-3)
+    its = [iter(data_all[d]) for d in range(2)]
+    ref_losses = []
+    for _ in range(n_steps):
+        tot = 0.0
+        for d in range(2):
+            for _ in range(mbs):
+                x, y = next(its[d])
+                loss = loss_fn(ref(x), y)
+                (loss / (mbs * 2)).backward()
+                tot += loss.item()
+        opt.step()
+        opt.zero_grad()
+        ref_losses.append(tot / (mbs * 2))
+
+    for got, want in zip(losses, ref_losses):
+        assert abs(got - want) < 1e-5, (losses, ref_losses)
+    ref_slice = list(ref)[net.part_start:net.part_end]
+    for m, r in zip(net.forward_funcs, ref_slice):
+        if isinstance(m, nn.Module):
+            for pm, pr in zip(m.parameters(), r.parameters()):
+                torch.testing.assert_close(pm, pr, rtol=1e-4, atol=2e-5)
+
+
+def test_pipeline_pp2_dp2_hybrid():
+    run_distributed(_pp2dp2_worker, world_size=4)

@@ -97,3 +97,54 @@ def _tp4_worker(rank, world):
 
 def test_tp4_training_parity():
     run_distributed(_tp4_worker, world_size=4)
+
+
+def _tp2dp2_worker(rank, world):
+    """Hybrid TP=2 x DP=2: TP shards within pairs, ZeRO-1 DP across pairs."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    from deepspeed_amd.parallel import groups
+
+    torch.manual_seed(29)
+    model = LlamaForCausalLM(llama_tiny())
+    torch.manual_seed(29)
+    ref = LlamaForCausalLM(llama_tiny())
+
+    deepspeed_amd.tp_model_init(model, tp_size=2)
+    assert groups.get_tensor_parallel_world_size() == 2
+    assert groups.get_data_parallel_world_size() == 2
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": 1, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    opt_ref = torch.optim.AdamW(ref.parameters(), lr=1e-3)
+
+    dp_rank = rank // 2  # tp pairs are contiguous
+    torch.manual_seed(70)
+    all_ids = [torch.randint(0, 512, (2, 2, 16)) for _ in range(2)]
+    for step_ids in all_ids:
+        ids = step_ids[dp_rank]
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+
+        # reference: grads averaged over both DP batches
+        l_tot = 0.0
+        for d in range(2):
+            l2 = ref(step_ids[d], labels=step_ids[d])
+            (l2 / 2).backward()
+            l_tot += l2.item() / 2
+        opt_ref.step()
+        opt_ref.zero_grad()
+
+    # replicated (non-TP) params must match the reference across all ranks
+    import torch.distributed as td
+    for (n, p), (_, pr) in zip(engine.module.named_parameters(),
+                               ref.named_parameters()):
+        if getattr(p, "tensor_model_parallel", False):
+            continue
+        torch.testing.assert_close(p, pr, rtol=2e-4, atol=3e-4), n
+
+
+def test_tp2_dp2_hybrid():
+    run_distributed(_tp2dp2_worker, world_size=4)
