@@ -336,3 +336,10 @@ def reduce_scatter_coalesced(tensors, group=None):
         out.append(recv[off:off + ps])
         off += ps
     return out
+
+
+def get_global_rank(group=None, group_rank: int = 0) -> int:
+    """Global rank of `group_rank` within `group` (identity for WORLD)."""
+    if group is None or group is torch_dist.group.WORLD:
+        return group_rank
+    return torch_dist.get_global_rank(group, group_rank)

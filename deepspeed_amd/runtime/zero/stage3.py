@@ -258,8 +258,8 @@ class ZeroStage3Optimizer:
                 # MiCS: replicas must start identical too
                 dist.broadcast(full, src=0)
             elif self.world_size > 1:
-                dist.broadcast(full, src=dist.get_global_rank(self.dp_group, 0)
-                               if hasattr(dist, "get_global_rank") else 0,
+                dist.broadcast(full,
+                               src=dist.get_global_rank(self.dp_group, 0),
                                group=self.dp_group)
             if self.world_size > 1:
                 u.shard = full[self.rank * u.shard_size:
