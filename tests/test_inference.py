@@ -235,3 +235,27 @@ def test_inference_server_http():
         assert got == want
     finally:
         server.stop()
+
+
+def test_mixtral_generate():
+    """MoE model through the inference engine: KV-cached greedy generation
+    matches the no-cache rollout (expert routing is deterministic)."""
+    import deepspeed_amd
+    from deepspeed_amd.models import MixtralForCausalLM, mixtral_tiny
+    torch.manual_seed(12)
+    cfg = mixtral_tiny(ep_size=1, num_experts=4)
+    # capacity-based routing is batch-composition-dependent by design;
+    # lift the capacity so cached decode routes identically to full fwd
+    cfg.capacity_factor = 64.0
+    model = MixtralForCausalLM(cfg)
+    engine = deepspeed_amd.init_inference(model, dtype=torch.float32)
+    prompt = torch.randint(0, cfg.vocab_size, (2, 6))
+    out = engine.generate(prompt, max_new_tokens=5)
+    assert out.shape == (2, 11)
+
+    ids = prompt.clone()
+    with torch.no_grad():
+        for _ in range(5):
+            logits = model(ids)
+            ids = torch.cat([ids, logits[:, -1].argmax(-1, keepdim=True)], 1)
+    assert torch.equal(out, ids)
