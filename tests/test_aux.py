@@ -463,3 +463,20 @@ def test_fused_ce_fallback_matches_fp32_path():
                                             reduction="sum")
     torch.testing.assert_close(l, ref)
     assert c.item() == (labels != -100).sum().item()
+
+
+def test_flops_profiler_counts_gemms():
+    from deepspeed_amd.profiling.flops_profiler import FlopsProfiler
+    import torch.nn as nn
+    m = nn.Sequential(nn.Linear(64, 128), nn.ReLU(), nn.Linear(128, 10))
+    prof = FlopsProfiler(m)
+    prof.start_profile()
+    x = torch.randn(4, 64)
+    m(x)
+    flops = prof.get_total_flops()
+    params = prof.get_total_params()
+    prof.end_profile()
+    # 2*B*(64*128 + 128*10) MACs->flops
+    expect = 2 * 4 * (64 * 128 + 128 * 10)
+    assert abs(flops - expect) / expect < 0.05, (flops, expect)
+    assert params == 64 * 128 + 128 + 128 * 10 + 10
