@@ -54,7 +54,7 @@ class _Unit:
                  "shard_size", "shard", "full", "grad_full", "grad_seen",
                  "grad_shard", "status", "handle", "persist", "group_idx",
                  "master_offset", "pending_grads", "trainable",
-                 "release_pending", "in_backward")
+                 "release_pending", "in_backward", "sec_shard")
 
     def __init__(self, index, name, module):
         self.index = index
@@ -78,6 +78,7 @@ class _Unit:
         self.trainable = True
         self.release_pending = False
         self.in_backward = False
+        self.sec_shard: Optional[torch.Tensor] = None  # hpZ secondary shard
 
 
 class ZeroStage3Optimizer:
@@ -148,6 +149,34 @@ class ZeroStage3Optimizer:
         self.quantized_weights = bool(zc.zero_quantized_weights) \
             and self.world_size > 1
         self.quant_group_size = int(zc.zero_quantization_group_size)
+        # hpZ (ZeRO++ hierarchical partitioning, reference
+        # zero/parameter_offload.py + zeropp paper): each rank additionally
+        # keeps a SECONDARY shard of every unit over a small group of
+        # zero_hpz_partition_size consecutive ranks (one node on multi-node
+        # jobs), so forward/backward weight all-gathers traverse only the
+        # intra-node xGMI links; the secondary shards refresh from the
+        # primary ones with one global gather per optimizer step. Gradients
+        # keep the full-world reduce-scatter (primary partitioning is
+        # unchanged). Off (1) by default — on a single 8-GPU node every link
+        # is xGMI so there is nothing to localize.
+        self.hpz_group = None
+        self.hpz_world = 1
+        self.hpz_rank = 0
+        hpz = int(getattr(zc, "zero_hpz_partition_size", 1) or 1)
+        if hpz > 1 and self.replica_group is not None:
+            log_dist("ZeRO-3: zero_hpz_partition_size ignored under MiCS "
+                     "(shard groups are already node-local)")
+        elif 1 < hpz < self.world_size:
+            assert self.world_size % hpz == 0, \
+                f"world {self.world_size} not divisible by hpz size {hpz}"
+            grank = dist.get_rank(dp_group)
+            for start in range(0, dist.get_world_size(dp_group), hpz):
+                ranks = list(range(start, start + hpz))
+                g = dist.new_group(ranks)
+                if grank in ranks:
+                    self.hpz_group = g
+            self.hpz_world = hpz
+            self.hpz_rank = self.rank % hpz
 
         self.units: List[_Unit] = []
         self.param_to_unit: Dict[torch.nn.Parameter, _Unit] = {}
@@ -191,7 +220,9 @@ class ZeroStage3Optimizer:
         log_dist(f"ZeRO stage 3: world={self.world_size} units={len(self.units)} "
                  f"(persistent={n_persist}) "
                  f"shard_elems={sum(self.group_shard_numel)} "
-                 f"prefetch={self.prefetch_bucket_size}")
+                 f"prefetch={self.prefetch_bucket_size}"
+                 + (f" hpz={self.hpz_world}" if self.hpz_group is not None
+                    else ""))
 
     # ------------------------------------------------------------------ setup
 
@@ -264,6 +295,10 @@ class ZeroStage3Optimizer:
             if self.world_size > 1:
                 u.shard = full[self.rank * u.shard_size:
                                (self.rank + 1) * u.shard_size].clone()
+                if self.hpz_group is not None and not u.persist:
+                    ss = u.numel // self.hpz_world
+                    u.sec_shard = full[self.hpz_rank * ss:
+                                       (self.hpz_rank + 1) * ss].clone()
             else:
                 # ws=1: the shard IS the full buffer — keep it, no clone
                 u.shard = full
@@ -339,34 +374,40 @@ class ZeroStage3Optimizer:
 
     # ------------------------------------------------------------ fetch/release
 
-    def _unit_quant_group(self, u: _Unit) -> int:
+    def _unit_quant_group(self, numel: int) -> int:
         g = self.quant_group_size
-        while g > 2 and u.shard_size % g:
+        while g > 2 and numel % g:
             g //= 2
         return g
 
     def _launch_gather(self, u: _Unit):
         if u.status != FREE:
             return
+        # hpZ: gather from the node-local secondary shard over the small
+        # group instead of the primary shard over the whole world
+        if u.sec_shard is not None:
+            src, group, gworld = u.sec_shard, self.hpz_group, self.hpz_world
+        else:
+            src, group, gworld = u.shard, self.dp_group, self.world_size
         if self.world_size > 1 and self.quantized_weights:
             from ...ops.quantizer import quantize
-            gs = self._unit_quant_group(u)
-            q, s = quantize(u.shard, gs, bits=8)
-            q_full = torch.empty(q.numel() * self.world_size, dtype=q.dtype,
+            gs = self._unit_quant_group(src.numel())
+            q, s = quantize(src, gs, bits=8)
+            q_full = torch.empty(q.numel() * gworld, dtype=q.dtype,
                                  device=q.device)
-            s_full = torch.empty(s.numel() * self.world_size,
+            s_full = torch.empty(s.numel() * gworld,
                                  dtype=torch.float32, device=q.device)
-            h1 = dist.all_gather_into_tensor(q_full, q, group=self.dp_group,
+            h1 = dist.all_gather_into_tensor(q_full, q, group=group,
                                              async_op=True)
-            h2 = dist.all_gather_into_tensor(s_full, s, group=self.dp_group,
+            h2 = dist.all_gather_into_tensor(s_full, s, group=group,
                                              async_op=True)
             u.full = None
             u.handle = ("qwz", h1, h2, q_full, s_full, gs)
         elif self.world_size > 1:
             u.full = torch.empty(u.numel, dtype=self._dtype,
                                  device=self._device)
-            u.handle = dist.all_gather_into_tensor(u.full, u.shard,
-                                                   group=self.dp_group,
+            u.handle = dist.all_gather_into_tensor(u.full, src,
+                                                   group=group,
                                                    async_op=True)
         else:
             # ws=1: shard covers the whole unit — alias, no alloc, no copy
@@ -703,6 +744,7 @@ class ZeroStage3Optimizer:
         if not wrote_params:
             self._copy_masters_to_shards()
         self._refresh_persistent()
+        self._refresh_secondary()
         self._zero_owned_grads()
         self.micro_step_id = 0
 
@@ -758,6 +800,23 @@ class ZeroStage3Optimizer:
             master = self.group_masters[u.group_idx]
             src = master.data[u.master_offset:u.master_offset + u.shard_size]
             u.shard.copy_(src, non_blocking=self.cpu_offload)
+
+    def _refresh_secondary(self):
+        """hpZ: re-derive the node-local secondary shards from the updated
+        primary shards — ONE global all-gather per trainable unit per step,
+        after which every fetch this step (fwd + bwd + grad-accum
+        microsteps) is intra-node only."""
+        if self.hpz_group is None:
+            return
+        ss_div = self.hpz_world
+        for u in self.units:
+            if u.sec_shard is None or not u.trainable:
+                continue
+            full = torch.empty(u.numel, dtype=self._dtype, device=self._device)
+            dist.all_gather_into_tensor(full, u.shard, group=self.dp_group)
+            ss = u.numel // ss_div
+            u.sec_shard.copy_(full[self.hpz_rank * ss:(self.hpz_rank + 1) * ss])
+            del full
 
     def _refresh_persistent(self):
         handles = []
@@ -861,6 +920,11 @@ class ZeroStage3Optimizer:
                         u.shard.copy_(u.full[self.opt.rank * u.shard_size:
                                              (self.opt.rank + 1) *
                                              u.shard_size])
+                    if u.sec_shard is not None:
+                        ss = u.numel // self.opt.hpz_world
+                        u.sec_shard.copy_(
+                            u.full[self.opt.hpz_rank * ss:
+                                   (self.opt.hpz_rank + 1) * ss])
                     if not u.persist:
                         self.opt._release(u)
             return False
@@ -995,6 +1059,7 @@ class ZeroStage3Optimizer:
                         u.master_offset + (z - lo)].copy_(src[a - off:z - off])
         self._copy_masters_to_shards()
         self._refresh_persistent()
+        self._refresh_secondary()
         for u in self.units:
             if not u.persist and u.status == AVAILABLE:
                 self._release(u)
@@ -1010,6 +1075,7 @@ class ZeroStage3Optimizer:
             self.optimizer.load_state_dict(sd["base_optimizer_state"])
         self._copy_masters_to_shards()
         self._refresh_persistent()
+        self._refresh_secondary()
         # any materialized unit must see the new weights
         for u in self.units:
             if not u.persist and u.status == AVAILABLE:

@@ -70,7 +70,7 @@ def _make_data(n, hidden=32, bs=4, seed=3):
     return [(torch.randn(bs, hidden), torch.randn(bs, 1)) for _ in range(n)]
 
 
-def _zero_worker(rank, world, stage, gas, clip, dtype_name):
+def _zero_worker(rank, world, stage, gas, clip, dtype_name, hpz=1):
     import deepspeed_amd
     dtype = {"bf16": torch.bfloat16, "fp32": torch.float32}[dtype_name]
     lr, steps = 1e-2, 5
@@ -83,7 +83,8 @@ def _zero_worker(rank, world, stage, gas, clip, dtype_name):
         "gradient_accumulation_steps": gas,
         "gradient_clipping": clip,
         "zero_optimization": {"stage": stage, "reduce_bucket_size": 500,
-                              "overlap_comm": False},
+                              "overlap_comm": False,
+                              "zero_hpz_partition_size": hpz},
         "optimizer": {"type": "AdamW", "params": {"lr": lr}},
     }
     if dtype == torch.bfloat16:
@@ -144,6 +145,54 @@ def test_zero_stage_bf16_parity_ws2(stage):
 def test_zero_stage_bf16_parity_ws4(stage):
     """ws=4: shard shapes closer to the 8-GPU node the driver benches."""
     run_distributed(_zero_worker, world_size=4, args=(stage, 1, 0.0, "bf16"))
+
+
+def test_zero3_hpz_parity_ws4():
+    """hpZ (ZeRO++ hierarchical partitioning): secondary shards over groups
+    of 2, weight gathers intra-group, grads still world-wide RS. gas=3 so
+    several fetch cycles run between secondary refreshes."""
+    run_distributed(_zero_worker, world_size=4, args=(3, 3, 0.0, "bf16", 2))
+
+
+def test_zero3_hpz_qwz_parity_ws4():
+    """hpZ composed with qwZ (quantized weight gathers over the small group)
+    — looser tolerance, int8 weight gather is lossy."""
+    run_distributed(_z3_hpz_qwz_worker, world_size=4)
+
+
+def _z3_hpz_qwz_worker(rank, world):
+    import deepspeed_amd
+    lr, steps, gas = 1e-2, 3, 2
+    model = TinyNet()
+    ref_model = copy.deepcopy(model)
+    data = _make_data(steps * gas)
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "gradient_accumulation_steps": gas,
+        "zero_optimization": {"stage": 3, "overlap_comm": False,
+                              "zero_hpz_partition_size": 2,
+                              "zero_quantized_weights": True,
+                              "zero_quantization_group_size": 64},
+        "optimizer": {"type": "AdamW", "params": {"lr": lr}},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    it = iter(data)
+    losses = []
+    for _ in range(steps):
+        micro = []
+        for _ in range(gas):
+            xs, ys = next(it)
+            loss = engine(xs.to(engine.device).bfloat16(),
+                          labels=ys.to(engine.device))
+            engine.backward(loss)
+            micro.append(loss.item())
+        engine.step()
+        losses.append(sum(micro) / len(micro))
+    ref_losses, _ = _reference_mixed_precision_loop(
+        ref_model, data, lr, steps, gas, torch.bfloat16)
+    for a, b in zip(losses, ref_losses):
+        assert abs(a - b) < 8e-2, (losses, ref_losses)
 
 
 @pytest.mark.parametrize("stage", [2, 3])
