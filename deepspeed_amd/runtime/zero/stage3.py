@@ -36,6 +36,7 @@ prefetch) — re-designed MI355X-first rather than ported:
 """
 
 import math
+import os
 from typing import Dict, List, Optional
 
 import torch
@@ -132,8 +133,26 @@ class ZeroStage3Optimizer:
         self.max_live_parameters = int(zc.stage3_max_live_parameters)
         self.clip_grad = config.gradient_clipping
         self.overlap_comm = zc.overlap_comm
-        self.cpu_offload = zc.offload_optimizer.device == "cpu"
+        # ZeRO-Infinity: "nvme" keeps the fp32 master + Adam moments in swap
+        # files (ops/csrc/aio.cpp thread-pool engine) and streams them
+        # through host RAM one chunk at a time during step(); gradients land
+        # in resident host fp32 accumulators like plain CPU offload.
+        self.nvme_offload = zc.offload_optimizer.device == "nvme"
+        self.cpu_offload = zc.offload_optimizer.device in ("cpu", "nvme")
         self.offload_pin_memory = zc.offload_optimizer.pin_memory
+        self._swapper = None
+        if self.nvme_offload:
+            from ..swap_tensor.swapper import AsyncTensorSwapper
+            aio = config.aio
+            self._swapper = AsyncTensorSwapper(
+                os.path.join(zc.offload_optimizer.nvme_path,
+                             f"zero3_rank{dist.get_rank()}"),
+                block_size=int(aio.block_size),
+                n_threads=int(aio.thread_count))
+            sub = int(zc.sub_group_size)
+            self._swap_chunk = sub if sub < 10**12 else (1 << 26)
+            self._nvme_init = set()   # (gi, ci) chunks whose moments exist
+            self._nvme_step_count = 0
         self.is_gradient_accumulation_boundary = True
         self.overflow = False
         self.custom_loss_scaler = False
@@ -209,9 +228,11 @@ class ZeroStage3Optimizer:
 
         # honesty about accepted-but-inert knobs (reference parity options
         # whose MI355X design doesn't need them — see module docstring)
-        if int(zc.sub_group_size) != 1_000_000_000_000:
+        if int(zc.sub_group_size) != 1_000_000_000_000 and not self.nvme_offload:
             log_dist("ZeRO-3: sub_group_size is accepted but inert (module-"
-                     "unit partitioning already bounds working-set size)")
+                     "unit partitioning already bounds working-set size; "
+                     "with offload_optimizer.device=nvme it sets the swap "
+                     "chunk size)")
         if zc.round_robin_gradients:
             log_dist("ZeRO-3: round_robin_gradients is inert (per-unit "
                      "reduce-scatter has no bucket-order imbalance)")
@@ -327,17 +348,23 @@ class ZeroStage3Optimizer:
             totals[u.group_idx] += u.shard_size
         master_dev = torch.device("cpu") if self.cpu_offload else self._device
         for gi in range(ngroups):
-            m = torch.empty(totals[gi], dtype=torch.float32, device=master_dev)
+            n_resident = 0 if self.nvme_offload else totals[gi]
+            m = torch.empty(n_resident, dtype=torch.float32, device=master_dev)
             if self.direct_grad:
                 g = torch.empty(0, dtype=torch.float32, device=master_dev)
             else:
                 g = torch.zeros(totals[gi], dtype=torch.float32,
                                 device=master_dev)
             if self.cpu_offload and self.offload_pin_memory and accel.available():
-                m, g = m.pin_memory(), g.pin_memory()
+                if m.numel():
+                    m = m.pin_memory()
+                g = g.pin_memory()
             self.group_masters.append(m)
             self.group_owned_grads.append(g)
             self.group_shard_numel.append(totals[gi])
+        if self.nvme_offload:
+            self._nvme_write_initial_masters()
+            return
         for u in self.units:
             if u.group_idx < 0 or not u.trainable:
                 continue
@@ -737,7 +764,10 @@ class ZeroStage3Optimizer:
         self._global_grad_norm = float(global_norm) if global_norm is not None else 0.0
 
         stepped = wrote_params = False
-        if self.fused_adam_fn is not None:
+        if self.nvme_offload:
+            self._nvme_step(combined_scale)
+            stepped = wrote_params = True
+        elif self.fused_adam_fn is not None:
             stepped, wrote_params = self._fused_step(combined_scale)
         if not stepped:
             self._torch_step(combined_scale)
@@ -775,6 +805,120 @@ class ZeroStage3Optimizer:
             if not ok:
                 return False, False
         return True, wrote_params
+
+    # ------------------------------------------------ NVMe (ZeRO-Infinity)
+
+    def _nvme_chunks(self, gi):
+        total = self.group_shard_numel[gi]
+        ci = c0 = 0
+        while c0 < total:
+            yield ci, c0, min(c0 + self._swap_chunk, total)
+            ci += 1
+            c0 += self._swap_chunk
+
+    def _units_in_span(self, gi, c0, c1):
+        for u in self.units:
+            if u.group_idx != gi or not u.trainable:
+                continue
+            a = max(u.master_offset, c0)
+            z = min(u.master_offset + u.shard_size, c1)
+            if a < z:
+                yield u, a, z
+
+    @torch.no_grad()
+    def _nvme_write_initial_masters(self):
+        for gi in range(len(self.group_shard_numel)):
+            for ci, c0, c1 in self._nvme_chunks(gi):
+                chunk = torch.empty(c1 - c0, dtype=torch.float32)
+                for u, a, z in self._units_in_span(gi, c0, c1):
+                    src = u.shard[a - u.master_offset:z - u.master_offset]
+                    chunk[a - c0:z - c0].copy_(src.float().cpu())
+                self._swapper.swap_out(f"g{gi}c{ci}_p", chunk)
+                self._swapper.synchronize()
+                del chunk
+
+    @torch.no_grad()
+    def _nvme_step(self, combined_scale):
+        """Chunked optimizer step over swapped state (reference
+        swap_tensor/optimizer_utils.py OptimizerSwapper): for each chunk of
+        the group flat, read master+moments from NVMe, run the AVX host Adam
+        against the resident fp32 grad slice, write the updated bf16 params
+        straight into the overlapping unit shards, and write master+moments
+        back out. Host RAM high-water mark is one chunk (x3 fp32 + 1 bf16),
+        independent of model size."""
+        from ...ops.adam import _adam_hyperparams, _torch_adam_step
+        from ...ops._loader import get_ext
+        ext = get_ext()
+        have_cpu_adam = ext is not None and hasattr(ext, "cpu_adam_flat")
+        scale = float(combined_scale) if not torch.is_tensor(combined_scale) \
+            else float(combined_scale.item())
+        inv_scale = 1.0 / scale
+        self._nvme_step_count += 1
+        step = self._nvme_step_count
+        for gi, group in enumerate(self.optimizer.param_groups):
+            if self.group_shard_numel[gi] == 0:
+                continue
+            lr, beta1, beta2, eps, wd, adamw = _adam_hyperparams(
+                self.optimizer, group)
+            grads = self.group_owned_grads[gi]
+            for ci, c0, c1 in self._nvme_chunks(gi):
+                n = c1 - c0
+                p = self._swapper.swap_in(f"g{gi}c{ci}_p")
+                if (gi, ci) in self._nvme_init:
+                    m = self._swapper.swap_in(f"g{gi}c{ci}_m")
+                    v = self._swapper.swap_in(f"g{gi}c{ci}_v")
+                    self._swapper.synchronize()
+                else:
+                    self._swapper.synchronize()
+                    m = torch.zeros(n, dtype=torch.float32)
+                    v = torch.zeros(n, dtype=torch.float32)
+                    self._nvme_init.add((gi, ci))
+                g = grads[c0:c1]
+                w16 = torch.empty(n, dtype=torch.bfloat16)
+                if have_cpu_adam:
+                    ext.cpu_adam_flat(p, g, m, v, w16, lr, beta1, beta2,
+                                      eps, wd, step, inv_scale, adamw)
+                else:
+                    _torch_adam_step(p, g, m, v, lr, beta1, beta2, eps,
+                                     wd, step, adamw, inv_scale)
+                    w16.copy_(p)
+                for u, a, z in self._units_in_span(gi, c0, c1):
+                    dst = u.shard[a - u.master_offset:z - u.master_offset]
+                    dst.copy_(w16[a - c0:z - c0], non_blocking=dst.is_cuda)
+                self._swapper.swap_out(f"g{gi}c{ci}_p", p)
+                self._swapper.swap_out(f"g{gi}c{ci}_m", m)
+                self._swapper.swap_out(f"g{gi}c{ci}_v", v)
+                self._swapper.synchronize()
+                del p, m, v, w16
+        if accel.available():
+            torch.cuda.synchronize()
+
+    @torch.no_grad()
+    def _nvme_flat(self, gi, kind):
+        """Materialize one group's fp32 flat (param/exp_avg/exp_avg_sq) from
+        swap — checkpointing only."""
+        out = torch.empty(self.group_shard_numel[gi], dtype=torch.float32)
+        for ci, c0, c1 in self._nvme_chunks(gi):
+            key = f"g{gi}c{ci}_{kind}"
+            if kind != "p" and (gi, ci) not in self._nvme_init:
+                out[c0:c1].zero_()
+                continue
+            self._swapper.swap_in(key, out[c0:c1])
+            self._swapper.synchronize()
+        return out
+
+    @torch.no_grad()
+    def _nvme_load_flat(self, gi, kind, flat):
+        for ci, c0, c1 in self._nvme_chunks(gi):
+            self._swapper.swap_out(f"g{gi}c{ci}_{kind}", flat[c0:c1].clone())
+            self._swapper.synchronize()
+            if kind != "p":
+                self._nvme_init.add((gi, ci))
+        if kind == "p":  # refresh bf16 unit shards from the new master
+            for u, a, z in self._units_in_span(gi, 0,
+                                               self.group_shard_numel[gi]):
+                dst = u.shard[a - u.master_offset:z - u.master_offset]
+                dst.copy_(flat[a:z].to(dst.dtype))
 
     def _torch_step(self, combined_scale):
         from ...ops.adam import _torch_adam_step  # noqa
@@ -964,10 +1108,15 @@ class ZeroStage3Optimizer:
         module = module or self.module
         out = {} if self.rank == 0 else None
         param_names = {p: n for n, p in module.named_parameters()}
+        nvme_masters = {}
+        if self.nvme_offload:
+            nvme_masters = {gi: torch.nn.Parameter(self._nvme_flat(gi, "p"))
+                            for gi in range(len(self.group_shard_numel))}
         for u in self.units:
             if u.group_idx < 0 or not u.trainable:
                 continue
-            master = self.group_masters[u.group_idx]
+            master = nvme_masters.get(u.group_idx,
+                                      self.group_masters[u.group_idx])
             shard = master.data[u.master_offset:
                                 u.master_offset + u.shard_size].to(self._device)
             full = torch.empty(u.numel, dtype=torch.float32, device=self._device)
@@ -988,6 +1137,21 @@ class ZeroStage3Optimizer:
     # ------------------------------------------------------------ checkpoint
 
     def state_dict(self):
+        if self.nvme_offload:
+            ng = len(self.group_shard_numel)
+            return {
+                "stage": 3,
+                "world_size": self.world_size,
+                "rank": self.rank,
+                "loss_scaler": self.loss_scaler.state_dict(),
+                "fp32_flat_groups": [self._nvme_flat(gi, "p")
+                                     for gi in range(ng)],
+                "nvme_moments": [(self._nvme_flat(gi, "m"),
+                                  self._nvme_flat(gi, "v"))
+                                 for gi in range(ng)],
+                "nvme_step": self._nvme_step_count,
+                "layout": self.layout_manifest(),
+            }
         return {
             "stage": 3,
             "world_size": self.world_size,
@@ -1025,6 +1189,10 @@ class ZeroStage3Optimizer:
     def load_universal_state_dict(self, module, usd):
         """Elastic ZeRO-3 load: slice per-param fp32 universal state into
         this world size's unit shards (mirrors stage12's)."""
+        if self.nvme_offload:
+            raise NotImplementedError(
+                "universal-checkpoint load with offload_optimizer.device="
+                "nvme: load on cpu/none offload, re-save, then switch")
         self.annotate_param_names(module)
         for gi, group in enumerate(self.optimizer.param_groups):
             master_p = self.group_masters[gi]
@@ -1068,6 +1236,20 @@ class ZeroStage3Optimizer:
         assert sd["world_size"] == self.world_size, \
             "ZeRO-3 checkpoint reshaping requires the universal checkpoint path"
         self.loss_scaler.load_state_dict(sd["loss_scaler"])
+        if self.nvme_offload:
+            for gi, flat in enumerate(sd["fp32_flat_groups"]):
+                self._nvme_load_flat(gi, "p", flat)
+            if load_optimizer_states and "nvme_moments" in sd:
+                for gi, (m, v) in enumerate(sd["nvme_moments"]):
+                    self._nvme_load_flat(gi, "m", m)
+                    self._nvme_load_flat(gi, "v", v)
+                self._nvme_step_count = sd.get("nvme_step", 0)
+            self._refresh_persistent()
+            self._refresh_secondary()
+            for u in self.units:
+                if not u.persist and u.status == AVAILABLE:
+                    self._release(u)
+            return
         for gi, flat in enumerate(sd["fp32_flat_groups"]):
             if self.group_masters[gi].numel():
                 self.group_masters[gi].data.copy_(flat)
