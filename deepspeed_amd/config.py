@@ -82,6 +82,9 @@ class ZeroConfig(ConfigModel):
     # scales over xGMI instead of bf16 (opt-in; changes forward numerics)
     zero_quantized_weights: bool = False
     zero_quantization_group_size: int = 2048
+    # qgZ: quantized-gradient all-to-all reduce (two-level when
+    # zero_hpz_partition_size groups are set — see runtime/zero/qgz.py)
+    zero_quantized_gradients: bool = False
     # fp32 grad accumulation buffer (stage 1/2)
     fp32_grad_accum: bool = False
 

@@ -178,6 +178,11 @@ class ZeroStage3Optimizer:
         # keep the full-world reduce-scatter (primary partitioning is
         # unchanged). Off (1) by default — on a single 8-GPU node every link
         # is xGMI so there is nothing to localize.
+        # qgZ: quantized-gradient all-to-all reduce (runtime/zero/qgz.py);
+        # two-level (intra-node then inter-node) when hpz groups exist
+        self.quantized_grads = bool(zc.zero_quantized_gradients) \
+            and self.world_size > 1
+        self.qgz_inter_group = None
         self.hpz_group = None
         self.hpz_world = 1
         self.hpz_rank = 0
@@ -196,6 +201,14 @@ class ZeroStage3Optimizer:
                     self.hpz_group = g
             self.hpz_world = hpz
             self.hpz_rank = self.rank % hpz
+            if self.quantized_grads:
+                # inter-node groups: same local index across nodes
+                for off in range(hpz):
+                    ranks = list(range(off, dist.get_world_size(dp_group),
+                                       hpz))
+                    g = dist.new_group(ranks)
+                    if grank in ranks:
+                        self.qgz_inter_group = g
 
         self.units: List[_Unit] = []
         self.param_to_unit: Dict[torch.nn.Parameter, _Unit] = {}
@@ -594,7 +607,15 @@ class ZeroStage3Optimizer:
         tail = u.offsets[-1] + u.params[-1].ds_numel
         if tail < u.numel:
             grad[tail:].zero_()
-        if self.world_size > 1:
+        if self.world_size > 1 and self.quantized_grads:
+            from .qgz import quantized_reduce
+            grad.div_(self.world_size)
+            recv = quantized_reduce(grad, u.shard_size, self.dp_group,
+                                    intra_group=self.hpz_group,
+                                    inter_group=self.qgz_inter_group,
+                                    group_size=self.quant_group_size)
+            h = None
+        elif self.world_size > 1:
             grad.div_(self.world_size)
             recv = torch.empty(u.shard_size, dtype=grad.dtype,
                                device=grad.device)

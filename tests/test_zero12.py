@@ -195,6 +195,82 @@ def _z3_hpz_qwz_worker(rank, world):
         assert abs(a - b) < 8e-2, (losses, ref_losses)
 
 
+def _qgz_unit_worker(rank, world, two_level):
+    """quantized_reduce vs exact reduce-scatter. grads are integers in
+    [-127,127] with a 127 pinned per quant group, so level-1 quantization
+    is exact and only the (bounded) level-2 requant error remains."""
+    import torch.distributed as tdist
+    from deepspeed_amd import comm as dist
+    from deepspeed_amd.runtime.zero.qgz import quantized_reduce
+    shard, gs = 64, 32
+    torch.manual_seed(100 + rank)
+    grad = torch.randint(-126, 127, (world * shard,)).float()
+    grad[::gs] = 127.0
+    intra = inter = None
+    if two_level:
+        S = 2
+        for start in range(0, world, S):
+            g = tdist.new_group(list(range(start, start + S)))
+            if rank in range(start, start + S):
+                intra = g
+        for off in range(S):
+            g = tdist.new_group(list(range(off, world, S)))
+            if rank % S == off:
+                inter = g
+    mine = quantized_reduce(grad.clone(), shard, None, intra_group=intra,
+                            inter_group=inter, group_size=gs)
+    exact = torch.empty(shard)
+    dist.reduce_scatter_tensor(exact, grad.clone())
+    if two_level:
+        # second-hop requant of partial sums: error <= N_nodes * amax/254
+        tol = 2 * exact.abs().max().item() / 254 + 1e-4
+        assert (mine - exact).abs().max().item() <= tol, \
+            (mine - exact).abs().max()
+    else:
+        torch.testing.assert_close(mine, exact)
+
+
+@pytest.mark.parametrize("two_level", [False, True])
+def test_qgz_quantized_reduce_ws4(two_level):
+    run_distributed(_qgz_unit_worker, world_size=4, args=(two_level,))
+
+
+def test_zero3_qgz_training_ws4():
+    """end-to-end ZeRO-3 with hpZ + qgZ (two-level quantized grad reduce)."""
+    run_distributed(_z3_zeropp_worker, world_size=4)
+
+
+def _z3_zeropp_worker(rank, world):
+    import deepspeed_amd
+    lr, steps = 1e-2, 3
+    model = TinyNet()
+    ref_model = copy.deepcopy(model)
+    data = _make_data(steps)
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "zero_optimization": {"stage": 3, "overlap_comm": False,
+                              "zero_hpz_partition_size": 2,
+                              "zero_quantized_gradients": True,
+                              "zero_quantization_group_size": 64},
+        "optimizer": {"type": "AdamW", "params": {"lr": lr}},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    it = iter(data)
+    losses = []
+    for _ in range(steps):
+        xs, ys = next(it)
+        loss = engine(xs.to(engine.device).bfloat16(),
+                      labels=ys.to(engine.device))
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    ref_losses, _ = _reference_mixed_precision_loop(
+        ref_model, data, lr, steps, 1, torch.bfloat16)
+    for a, b in zip(losses, ref_losses):
+        assert abs(a - b) < 8e-2, (losses, ref_losses)
+
+
 def _z3_nvme_worker(rank, world, swap_dir):
     """ZeRO-Infinity: optimizer state on disk, chunked host Adam step."""
     import deepspeed_amd
