@@ -128,6 +128,8 @@ class MonitorConfig(ConfigModel):
     enabled: bool = False
     tensorboard: Dict[str, Any] = Field(default_factory=dict)
     csv_monitor: Dict[str, Any] = Field(default_factory=dict)
+    wandb: Dict[str, Any] = Field(default_factory=dict)
+    comet: Dict[str, Any] = Field(default_factory=dict)
 
 
 class MoEConfig(ConfigModel):
@@ -200,7 +202,14 @@ class Config:
             **g("activation_checkpointing", {}))
         self.comms_logger = CommsLoggerConfig(**g("comms_logger", {}))
         self.flops_profiler = FlopsProfilerConfig(**g("flops_profiler", {}))
-        self.monitor = MonitorConfig(**g("monitor_config", {}))
+        # reference accepts the writer blocks at the top level of ds_config
+        # (tensorboard/wandb/comet/csv_monitor) as well as grouped
+        mon = dict(g("monitor_config", {}))
+        for k in ("tensorboard", "wandb", "comet", "csv_monitor"):
+            blk = g(k, None)
+            if blk is not None and k not in mon:
+                mon[k] = blk
+        self.monitor = MonitorConfig(**mon)
         self.moe = MoEConfig(**g("moe", {}))
         self.pipeline = PipelineConfig(**g("pipeline", {}))
         self.tensor_parallel = TensorParallelConfig(**g("tensor_parallel", {}))

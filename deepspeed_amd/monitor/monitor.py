@@ -70,6 +70,32 @@ class WandbMonitor(Monitor):
             self.wandb.log({tag: value}, step=step)
 
 
+class CometMonitor(Monitor):
+    """Comet experiment writer (reference monitor/comet.py CometMonitor)."""
+
+    def __init__(self, project=None, workspace=None, api_key=None,
+                 experiment_name=None, experiment_key=None, mode=None,
+                 online=None, samples_log_interval=100):
+        import comet_ml  # may raise
+        kwargs = {}
+        if mode:
+            kwargs["mode"] = mode
+        if online is not None:
+            kwargs["online"] = online
+        self.experiment = comet_ml.start(
+            api_key=api_key, project=project, workspace=workspace,
+            experiment_key=experiment_key, **kwargs)
+        if experiment_name:
+            self.experiment.set_name(experiment_name)
+
+    def write_events(self, events: List[Event]):
+        for tag, value, step in events:
+            self.experiment.__internal_api__log_metric__(
+                tag, value, framework="deepspeed_amd", step=step) \
+                if hasattr(self.experiment, "__internal_api__log_metric__") \
+                else self.experiment.log_metric(tag, value, step=step)
+
+
 class MonitorMaster(Monitor):
     """Fans events out to every enabled writer (reference monitor.py:30)."""
 
@@ -81,6 +107,18 @@ class MonitorMaster(Monitor):
                 self.monitors.append(TensorBoardMonitor(**tb))
             except Exception as e:  # tensorboard not installed
                 logger.warning(f"tensorboard writer disabled: {e}")
+        wb = dict(getattr(config, "wandb", {}) or {})
+        if wb.pop("enabled", False):
+            try:
+                self.monitors.append(WandbMonitor(**wb))
+            except Exception as e:  # wandb not installed / offline
+                logger.warning(f"wandb writer disabled: {e}")
+        cm = dict(getattr(config, "comet", {}) or {})
+        if cm.pop("enabled", False):
+            try:
+                self.monitors.append(CometMonitor(**cm))
+            except Exception as e:  # comet_ml not installed / offline
+                logger.warning(f"comet writer disabled: {e}")
         csv_cfg = dict(getattr(config, "csv_monitor", {}) or {})
         if csv_cfg.pop("enabled", False) or not self.monitors:
             self.monitors.append(CsvMonitor(**csv_cfg))

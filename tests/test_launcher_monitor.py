@@ -90,3 +90,24 @@ def test_engine_writes_monitor_events(tmp_path):
     assert "Train_loss.csv" in files and "Train_lr.csv" in files
     rows = (tmp_path / "m" / "Train_loss.csv").read_text().strip().splitlines()
     assert len(rows) == 2
+
+
+def test_top_level_writer_blocks_and_disabled_fallbacks(tmp_path):
+    """Reference accepts tensorboard/wandb/comet blocks at ds_config top
+    level; unavailable writers (wandb/comet not installed) degrade to a
+    warning, never an exception, and CSV still engages as fallback."""
+    from deepspeed_amd.config import Config
+    from deepspeed_amd.monitor.monitor import MonitorMaster, CsvMonitor
+    cfg = Config({
+        "train_micro_batch_size_per_gpu": 1,
+        "wandb": {"enabled": True, "project": "x"},
+        "comet": {"enabled": True, "project": "x"},
+        "csv_monitor": {"enabled": True, "output_path": str(tmp_path),
+                        "job_name": "j"},
+    })
+    assert cfg.monitor.wandb["enabled"]
+    assert cfg.monitor.comet["enabled"]
+    mm = MonitorMaster(cfg.monitor)
+    assert any(isinstance(m, CsvMonitor) for m in mm.monitors)
+    mm.write_events([("a/b", 1.0, 0)])
+    assert (tmp_path / "j" / "a_b.csv").exists()
