@@ -5,6 +5,7 @@ from .adam import FusedAdam, fused_adam_step, multi_tensor_adam_available
 from .norms import RMSNorm, FusedLayerNorm, rms_norm, layer_norm
 from .rope import apply_rope, rope_tables
 from .swiglu import swiglu, geglu
+from .evoformer import DS4Sci_EvoformerAttention, EvoformerAttention
 
 __all__ = [
     "has_ext", "get_ext", "FusedAdam", "fused_adam_step",

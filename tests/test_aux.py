@@ -480,3 +480,32 @@ def test_flops_profiler_counts_gemms():
     expect = 2 * 4 * (64 * 128 + 128 * 10)
     assert abs(flops - expect) / expect < 0.05, (flops, expect)
     assert params == 64 * 128 + 128 + 128 * 10 + 10
+
+
+def test_evoformer_attention_reference_math():
+    """DS4Sci_EvoformerAttention vs an index-loop fp32 reference with both
+    bias patterns (mask bias [B,N,1,1,S] + pair bias [B,1,H,S,S])."""
+    import math
+    from deepspeed_amd.ops.evoformer import DS4Sci_EvoformerAttention
+    torch.manual_seed(0)
+    B, N, S, H, D = 2, 3, 5, 2, 4
+    Q, K, V = (torch.randn(B, N, S, H, D) for _ in range(3))
+    bias1 = torch.randn(B, N, 1, 1, S) * 0.5
+    bias2 = torch.randn(B, 1, H, S, S) * 0.5
+    out = DS4Sci_EvoformerAttention(Q, K, V, [bias1, bias2])
+    assert out.shape == (B, N, S, H, D)
+    scale = 1.0 / math.sqrt(D)
+    for b in range(B):
+        for n in range(N):
+            for h in range(H):
+                for s in range(S):
+                    logit = (Q[b, n, s, h] @ K[b, n, :, h].T) * scale \
+                        + bias1[b, n, 0, 0] + bias2[b, 0, h, s]
+                    p = torch.softmax(logit, -1)
+                    ref = p @ V[b, n, :, h]
+                    torch.testing.assert_close(out[b, n, s, h], ref,
+                                               atol=1e-5, rtol=1e-5)
+    # no-bias and single-bias paths
+    out_nb = DS4Sci_EvoformerAttention(Q, K, V)
+    out_b1 = DS4Sci_EvoformerAttention(Q, K, V, [bias1, None])
+    assert not torch.allclose(out_nb, out_b1)
