@@ -46,13 +46,21 @@ class RaggedKVCache:
     def free(self, slot: int):
         self.lens[slot] = 0
 
-    def begin_step(self, slots: List[int], q_len: int):
+    def begin_step(self, slots: List[int], q_len: int, offsets=None):
         """All sequences in this step share q_len (1 for decode; the prompt
-        length for a single-sequence prefill)."""
+        length for a single-sequence prefill). SplitFuse token packing:
+        pass q_len=1 with one entry per TOKEN — `slots` may repeat and
+        `offsets[i]` places token i at ``lens[slot]+offsets[i]``, so a
+        prefill chunk and the decode batch ride one forward."""
         self._slots = torch.as_tensor(slots, dtype=torch.long,
                                       device=self.device)
         self._q_len = q_len
+        self._packed = offsets is not None
         starts = self.lens[self._slots]             # [n]
+        if self._packed:
+            assert q_len == 1, "packed mode is per-token (q_len 1)"
+            starts = starts + torch.as_tensor(offsets, dtype=torch.long,
+                                              device=self.device)
         n = len(slots)
         L = int((starts + q_len).max())
         # mask[i, 1, qi, j]: query at absolute pos starts_i+qi may see j
@@ -76,7 +84,11 @@ class RaggedKVCache:
                 self.v[layer_idx][sl, :, :self._L])
 
     def end_step(self):
-        self.lens[self._slots] += self._q_len
+        if self._packed:   # repeated slots: one token each
+            self.lens.index_add_(0, self._slots,
+                                 torch.ones_like(self._slots))
+        else:
+            self.lens[self._slots] += self._q_len
         self._slots = None
         self.last_mask = None
 
@@ -108,11 +120,19 @@ class ContinuousBatcher:
 
     def __init__(self, model, max_slots: int = 8, max_seq: int = None,
                  dtype=torch.float32, device=None,
-                 prefill_chunk: int = 0, cache_cls=None):
+                 prefill_chunk: int = 0, cache_cls=None,
+                 token_budget: int = 0):
         """``prefill_chunk`` > 0 bounds prompt tokens prefetched per step
         (Dynamic SplitFuse: long prompts stream in across iterations so the
         decode batch's latency stays flat; reference inference/v2
-        scheduling). 0 = whole prompt in one step."""
+        scheduling). 0 = whole prompt in one step.
+
+        ``token_budget`` > 0 enables TRUE SplitFuse token packing: every
+        iteration issues ONE forward whose batch rows are individual
+        tokens — the whole decode batch plus up to the remaining budget of
+        prefill-chunk tokens — so prefill never stalls decode and each
+        step's work is a constant ~token_budget tokens (reference Dynamic
+        SplitFuse scheduling, inference/v2)."""
         self.model = model
         cfg = model.cfg
         self.device = device or next(model.parameters()).device
@@ -126,6 +146,7 @@ class ContinuousBatcher:
         self.pending: deque = deque()
         self.active: Dict[int, Request] = {}
         self.prefill_chunk = prefill_chunk
+        self.token_budget = token_budget
 
     def put(self, req: Request):
         self.pending.append(req)
@@ -156,23 +177,16 @@ class ContinuousBatcher:
         if not req.in_prefill:
             req.generated.append(int(logits[0, -1].argmax()))
 
-    @torch.no_grad()
-    def step(self) -> List[Request]:
-        """One scheduling iteration; returns requests finished this step."""
-        # 1) admit queued requests into free slots
+    def _admit(self):
         while self.pending and self.free_slots:
             req = self.pending.popleft()
             req.slot = self.free_slots.popleft()
             self.cache.free(req.slot)
             self.cache.lens[req.slot] = 0
             self.active[req.uid] = req
-        # 1b) advance prefills (bounded per step when prefill_chunk is set)
-        for req in self.active.values():
-            if req.in_prefill:
-                self._prefill_some(req)
 
+    def _retire(self) -> List[Request]:
         finished = []
-        # 2) retire sequences that hit eos/max BEFORE the decode batch
         for uid in list(self.active):
             req = self.active[uid]
             if req.in_prefill:
@@ -186,6 +200,72 @@ class ContinuousBatcher:
                 self.cache.free(req.slot)   # return KV blocks to the pool
                 self.free_slots.append(req.slot)
                 del self.active[uid]
+        return finished
+
+    @torch.no_grad()
+    def _step_packed(self) -> List[Request]:
+        """True SplitFuse iteration: one forward, batch rows = tokens."""
+        self._admit()
+        finished = self._retire()
+        slots, offsets, ids, owners = [], [], [], []
+        budget = self.token_budget
+        for r in self.active.values():      # decodes first — latency
+            if not r.in_prefill:
+                slots.append(r.slot)
+                offsets.append(0)
+                ids.append(r.generated[-1])
+                owners.append((r, 0))
+                budget -= 1
+        for r in self.active.values():      # fill with prefill chunks
+            if budget <= 0:
+                break
+            if r.in_prefill:
+                chunk = min(budget, r.prompt.numel() - r.prefilled)
+                if self.prefill_chunk > 0:
+                    chunk = min(chunk, self.prefill_chunk)
+                for j in range(chunk):
+                    slots.append(r.slot)
+                    offsets.append(j)
+                    ids.append(int(r.prompt[r.prefilled + j]))
+                owners.append((r, chunk))
+                budget -= chunk
+        if not slots:
+            return finished
+        T = len(slots)
+        ids_t = torch.tensor(ids, device=self.device).view(T, 1)
+        lens = self.cache.lens
+        pos = (lens[torch.as_tensor(slots, device=lens.device)] +
+               torch.as_tensor(offsets, device=lens.device)).to(
+                   device=self.device, dtype=torch.int32).view(T, 1)
+        self.cache.begin_step(slots, 1, offsets=offsets)
+        logits = self.model(ids_t, positions=pos, kv_cache=self.cache)
+        self.cache.end_step()
+        row = 0
+        for r, chunk in owners:
+            if chunk == 0:                       # decode row
+                r.generated.append(int(logits[row, -1].argmax()))
+                row += 1
+            else:                                # prefill rows
+                row += chunk
+                r.prefilled += chunk
+                if not r.in_prefill:             # prompt complete: sample
+                    r.generated.append(int(logits[row - 1, -1].argmax()))
+        return finished
+
+    @torch.no_grad()
+    def step(self) -> List[Request]:
+        """One scheduling iteration; returns requests finished this step."""
+        if self.token_budget > 0:
+            return self._step_packed()
+        # 1) admit queued requests into free slots
+        self._admit()
+        # 1b) advance prefills (bounded per step when prefill_chunk is set)
+        for req in self.active.values():
+            if req.in_prefill:
+                self._prefill_some(req)
+
+        # 2) retire sequences that hit eos/max BEFORE the decode batch
+        finished = self._retire()
 
         # 3) one batched decode step for every fully-prefilled sequence
         decode = [r for r in self.active.values() if not r.in_prefill]
@@ -253,11 +333,16 @@ class PagedKVCache:
             assert self.free_blocks, "paged KV pool exhausted"
             self.block_table[slot].append(self.free_blocks.pop())
 
-    def begin_step(self, slots: List[int], q_len: int):
+    def begin_step(self, slots: List[int], q_len: int, offsets=None):
         self._slots = list(slots)
         self._q_len = q_len
+        self._packed = offsets is not None
         starts = self.lens[torch.as_tensor(self._slots,
                                            device=self.lens.device)]
+        if self._packed:
+            assert q_len == 1, "packed mode is per-token (q_len 1)"
+            starts = starts + torch.as_tensor(offsets,
+                                              device=self.lens.device)
         L = int((starts + q_len).max())
         for s, st in zip(self._slots, starts.tolist()):
             self._ensure_blocks(s, st + q_len)
@@ -291,6 +376,9 @@ class PagedKVCache:
 
     def end_step(self):
         idx = torch.as_tensor(self._slots, device=self.lens.device)
-        self.lens[idx] += self._q_len
+        if self._packed:
+            self.lens.index_add_(0, idx, torch.ones_like(idx))
+        else:
+            self.lens[idx] += self._q_len
         self._slots = None
         self.last_mask = None

@@ -259,3 +259,32 @@ def test_mixtral_generate():
             logits = model(ids)
             ids = torch.cat([ids, logits[:, -1].argmax(-1, keepdim=True)], 1)
     assert torch.equal(out, ids)
+
+
+def test_continuous_batcher_token_packed_splitfuse():
+    """True SplitFuse token packing: decode batch + prefill chunks share ONE
+    forward per iteration (batch rows = tokens); outputs must equal the
+    sequential KV-cached generate, for both cache types and with requests
+    arriving mid-flight."""
+    import deepspeed_amd
+    from deepspeed_amd.inference.ragged import (ContinuousBatcher,
+                                                PagedKVCache, Request)
+    model, cfg = _model(seed=9)
+    engine = deepspeed_amd.init_inference(model, dtype=torch.float32)
+    g = torch.Generator().manual_seed(11)
+    prompts = [torch.randint(0, cfg.vocab_size, (int(n),), generator=g)
+               for n in (13, 5, 9)]
+    want = [engine.generate(p.view(1, -1), max_new_tokens=5)[0, p.numel():]
+            .tolist() for p in prompts]
+
+    for cache_cls in (None, PagedKVCache):
+        batcher = ContinuousBatcher(model, max_slots=4, token_budget=6,
+                                    cache_cls=cache_cls)
+        batcher.put(Request(uid=0, prompt=prompts[0], max_new_tokens=5))
+        batcher.put(Request(uid=1, prompt=prompts[1], max_new_tokens=5))
+        batcher.step()          # 0/1 mid-prefill when 2 arrives
+        batcher.put(Request(uid=2, prompt=prompts[2], max_new_tokens=5))
+        done = {r.uid: r for r in batcher.run_to_completion()}
+        for uid in range(3):
+            assert done[uid].generated == want[uid], \
+                (cache_cls, uid, done[uid].generated, want[uid])
