@@ -56,6 +56,11 @@ def parse_args(args=None):
     p.add_argument("--module", action="store_true",
                    help="run user_script as a python module (-m)")
     p.add_argument("--no_python", action="store_true")
+    p.add_argument("--max_restarts", type=int, default=0,
+                   help="elastic agent: on worker failure, restart the "
+                        "whole local group up to N times (workers resume "
+                        "from their latest checkpoint; DSAMD_RESTART_COUNT "
+                        "in the env tells them which attempt this is)")
     p.add_argument("user_script", type=str)
     p.add_argument("user_args", nargs=argparse.REMAINDER)
     return p.parse_args(args)
@@ -69,7 +74,7 @@ def device_count():
         return 1
 
 
-def launch_local(args, local_gpu_ids=None):
+def launch_local(args, local_gpu_ids=None, restart_count=0):
     n_local = args.num_gpus if args.num_gpus > 0 else device_count()
     if local_gpu_ids:
         n_local = len(local_gpu_ids)
@@ -84,6 +89,8 @@ def launch_local(args, local_gpu_ids=None):
         env["LOCAL_WORLD_SIZE"] = str(n_local)
         env["MASTER_ADDR"] = args.master_addr
         env["MASTER_PORT"] = str(args.master_port)
+        env["DSAMD_RESTART_COUNT"] = str(restart_count)
+        env["TORCHELASTIC_RESTART_COUNT"] = str(restart_count)
         if local_gpu_ids:
             env["HIP_VISIBLE_DEVICES"] = str(local_gpu_ids[local_rank])
             env["LOCAL_RANK"] = "0"
@@ -172,7 +179,20 @@ def main(argv=None):
         part = args.include.split("@")[-1]
         if ":" in part:
             local_ids = [int(x) for x in part.split(":")[1].split(",")]
-    sys.exit(launch_local(args, local_ids))
+    # elastic agent loop (reference: torch-elastic restart-all semantics +
+    # deepspeed elastic training): a failed worker kills the group; the
+    # whole group relaunches on a fresh rendezvous port and the application
+    # resumes from its latest checkpoint.
+    attempt = 0
+    while True:
+        rc = launch_local(args, local_ids, restart_count=attempt)
+        if rc == 0 or attempt >= args.max_restarts:
+            sys.exit(rc)
+        attempt += 1
+        args.master_port += 1   # stale TCPStore may linger on the old port
+        logger.warning(f"worker group failed (rc={rc}); elastic restart "
+                       f"{attempt}/{args.max_restarts} on port "
+                       f"{args.master_port}")
 
 
 if __name__ == "__main__":

@@ -111,3 +111,29 @@ def test_top_level_writer_blocks_and_disabled_fallbacks(tmp_path):
     assert any(isinstance(m, CsvMonitor) for m in mm.monitors)
     mm.write_events([("a/b", 1.0, 0)])
     assert (tmp_path / "j" / "a_b.csv").exists()
+
+
+def test_launcher_elastic_restart(tmp_path):
+    """Elastic agent: a worker failure restarts the whole group (fresh
+    rendezvous port, DSAMD_RESTART_COUNT bumped); the relaunched attempt
+    succeeds and the job exits 0."""
+    script = tmp_path / "flaky.py"
+    script.write_text(
+        "import os, sys\n"
+        "attempt = os.environ['DSAMD_RESTART_COUNT']\n"
+        f"open(os.path.join({str(tmp_path)!r},"
+        " 'a' + attempt + '_r' + os.environ['RANK']), 'w')"
+        ".write(os.environ['MASTER_PORT'])\n"
+        "sys.exit(5 if (attempt == '0' and os.environ['RANK'] == '1')"
+        " else 0)\n")
+    out = subprocess.run(
+        [sys.executable, "-m", "deepspeed_amd.launcher.runner",
+         "--num_gpus", "2", "--max_restarts", "2",
+         "--master_port", "29881", str(script)],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stdout + out.stderr
+    # both attempts ran, second on a bumped port
+    assert (tmp_path / "a0_r1").read_text() == "29881"
+    assert (tmp_path / "a1_r0").read_text() == "29882"
+    assert (tmp_path / "a1_r1").exists()
