@@ -271,7 +271,7 @@ def _z3_zeropp_worker(rank, world):
         assert abs(a - b) < 8e-2, (losses, ref_losses)
 
 
-def _z3_nvme_worker(rank, world, swap_dir):
+def _z3_nvme_worker(rank, world, swap_dir, gas=1):
     """ZeRO-Infinity: optimizer state on disk, chunked host Adam step."""
     import deepspeed_amd
     pytest_mod = __import__("pytest")
@@ -282,9 +282,10 @@ def _z3_nvme_worker(rank, world, swap_dir):
     lr, steps = 1e-2, 4
     model = TinyNet()
     ref_model = copy.deepcopy(model)
-    data = _make_data(steps)
+    data = _make_data(steps * gas)
     config = {
         "train_micro_batch_size_per_gpu": 4,
+        "gradient_accumulation_steps": gas,
         "zero_optimization": {
             "stage": 3, "overlap_comm": False,
             "sub_group_size": 300,  # force multi-chunk swapping
@@ -298,14 +299,17 @@ def _z3_nvme_worker(rank, world, swap_dir):
     it = iter(data)
     losses = []
     for _ in range(steps):
-        xs, ys = next(it)
-        loss = engine(xs.to(engine.device).bfloat16(),
-                      labels=ys.to(engine.device))
-        engine.backward(loss)
+        micro = []
+        for _ in range(gas):
+            xs, ys = next(it)
+            loss = engine(xs.to(engine.device).bfloat16(),
+                          labels=ys.to(engine.device))
+            engine.backward(loss)
+            micro.append(loss.item())
         engine.step()
-        losses.append(loss.item())
+        losses.append(sum(micro) / len(micro))
     ref_losses, ref_master = _reference_mixed_precision_loop(
-        ref_model, data, lr, steps, 1, torch.bfloat16)
+        ref_model, data, lr, steps, gas, torch.bfloat16)
     for a, b in zip(losses, ref_losses):
         assert abs(a - b) < 2e-2, (losses, ref_losses)
     sd_model = engine.optimizer.get_full_state_dict()
@@ -326,6 +330,12 @@ def _z3_nvme_worker(rank, world, swap_dir):
 
 def test_zero3_nvme_offload_parity_ws2(tmp_path):
     run_distributed(_z3_nvme_worker, world_size=2, args=(str(tmp_path),))
+
+
+def test_zero3_nvme_offload_gas_parity_ws2(tmp_path):
+    """NVMe offload composes with gradient accumulation (grads build up
+    in the resident host fp32 accumulators between swapped steps)."""
+    run_distributed(_z3_nvme_worker, world_size=2, args=(str(tmp_path), 2))
 
 
 @pytest.mark.parametrize("stage", [2, 3])
