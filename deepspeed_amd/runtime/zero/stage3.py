@@ -1118,9 +1118,17 @@ class ZeroStage3Optimizer:
                         out[name] = t.to(dtype) if dtype is not None else t
             del full
         if out is not None:
-            # buffers are not partitioned
-            for n, b in self.module.named_buffers():
-                out[n] = b.detach().clone()
+            # buffers are not partitioned; keep state_dict() semantics —
+            # non-persistent buffers (e.g. HF rotary inv_freq) stay out
+            for mname, mod in self.module.named_modules():
+                for bname, b in mod._buffers.items():
+                    if b is None or \
+                            bname in getattr(mod,
+                                             "_non_persistent_buffers_set",
+                                             ()):
+                        continue
+                    out[f"{mname}.{bname}" if mname else bname] = \
+                        b.detach().clone()
         return out
 
     @torch.no_grad()
