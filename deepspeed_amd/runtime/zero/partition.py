@@ -45,27 +45,36 @@ class Init:
     def __enter__(self):
         if not self.enabled:
             return self
-        self._orig_init = nn.Module.__init__
+        # Parameters are REGISTERED (not constructed) through
+        # Module.__setattr__ / register_parameter — always after the
+        # subclass's super().__init__() has returned — so the cast must
+        # hook registration, not nn.Module.__init__.
+        self._orig_setattr = nn.Module.__setattr__
+        self._orig_register = nn.Module.register_parameter
         target_dtype = self.dtype
 
-        @functools.wraps(self._orig_init)
-        def wrapped(mod, *args, **kwargs):
-            Init._depth = getattr(Init, "_depth", 0) + 1
-            try:
-                self._orig_init(mod, *args, **kwargs)
-            finally:
-                Init._depth -= 1
-            # cast only once construction fully finished (outermost module
-            # sees children already cast; direct params cast here)
-            for p in mod._parameters.values():
-                if p is not None and p.is_floating_point():
-                    p.data = p.data.to(target_dtype)
-        nn.Module.__init__ = wrapped
+        @functools.wraps(self._orig_setattr)
+        def wrapped_setattr(mod, name, value):
+            if isinstance(value, nn.Parameter) and value.is_floating_point() \
+                    and not value.is_meta:
+                value.data = value.data.to(target_dtype)
+            self._orig_setattr(mod, name, value)
+
+        @functools.wraps(self._orig_register)
+        def wrapped_register(mod, name, param):
+            if isinstance(param, nn.Parameter) and \
+                    param.is_floating_point() and not param.is_meta:
+                param.data = param.data.to(target_dtype)
+            self._orig_register(mod, name, param)
+
+        nn.Module.__setattr__ = wrapped_setattr
+        nn.Module.register_parameter = wrapped_register
         return self
 
     def __exit__(self, *exc):
-        if self._orig_init is not None:
-            nn.Module.__init__ = self._orig_init
+        if getattr(self, "_orig_setattr", None) is not None:
+            nn.Module.__setattr__ = self._orig_setattr
+            nn.Module.register_parameter = self._orig_register
         return False
 
 

@@ -114,3 +114,33 @@ def test_hf_gpt2_generate_under_engine():
         got = engine.module.generate(ids, max_new_tokens=8, do_sample=False,
                                      pad_token_id=0)
     assert torch.equal(want, got)
+
+
+def test_zero_init_hf_construction_then_stage3_step():
+    """`with zero.Init(): model = AutoModel...` (reference large-model entry
+    path, partition_parameters.py Init:824): params come out in the target
+    dtype and the model trains under a stage-3 engine."""
+    import deepspeed_amd
+    from deepspeed_amd.runtime.zero.partition import Init
+    from transformers import GPT2Config, GPT2LMHeadModel
+    with Init(dtype=torch.bfloat16):
+        cfg = GPT2Config(n_layer=2, n_embd=64, n_head=2, vocab_size=128,
+                         n_positions=64, bos_token_id=0, eos_token_id=0,
+                         resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0,
+                         attn_implementation="eager")
+        model = GPT2LMHeadModel(cfg)
+    assert all(p.dtype == torch.bfloat16 for p in model.parameters())
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": 3},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "bf16": {"enabled": True},
+    })
+    ids = torch.randint(0, 128, (2, 16), device=engine.device)
+    first = None
+    for _ in range(3):
+        loss = engine(ids, labels=ids).loss
+        engine.backward(loss)
+        engine.step()
+        first = first if first is not None else float(loss)
+    assert float(loss) < first  # memorizing one batch must reduce loss
