@@ -59,13 +59,35 @@ class LinearAllreduce(nn.Module):
 # module-name suffixes -> parallel style for common decoder architectures
 COLUMN_PATTERNS = (r"q_proj$", r"k_proj$", r"v_proj$", r"gate_proj$",
                    r"up_proj$", r"w1$", r"w3$", r"c_attn$", r"wqkv$",
-                   r"fc1$", r"gate_up_proj$")
-ROW_PATTERNS = (r"o_proj$", r"down_proj$", r"w2$", r"c_proj$", r"fc2$",
+                   r"fc1$", r"c_fc$", r"gate_up_proj$")
+ROW_PATTERNS = (r"o_proj$", r"down_proj$", r"w2$", r"attn\.c_proj$",
+                r"mlp\.c_proj$", r"c_proj$", r"fc2$",
                 r"dense$", r"out_proj$")
+# fused projections whose out dim concatenates q|k|v: each rank must take
+# its head slice from EACH third (reference auto_tp.py qkv handling)
+FUSED_QKV_PATTERNS = (r"c_attn$", r"query_key_value$", r"wqkv$")
 
 
 def _match(name: str, patterns: Iterable[str]) -> bool:
     return any(re.search(p, name) for p in patterns)
+
+
+def _is_hf_conv1d(mod) -> bool:
+    """transformers.pytorch_utils.Conv1D: weight [in, out], y = x@W + b
+    (GPT-2 family). Detected structurally so transformers stays optional."""
+    return (type(mod).__name__ == "Conv1D" and hasattr(mod, "weight")
+            and hasattr(mod, "nf"))
+
+
+def _fused_qkv_rows(out: int, tp_rank: int, tp_size: int) -> torch.Tensor:
+    """Row indices of this rank's slice of a fused [3h] qkv out dim."""
+    h = out // 3
+    per = h // tp_size
+    idx = []
+    for third in range(3):
+        base = third * h + tp_rank * per
+        idx.append(torch.arange(base, base + per))
+    return torch.cat(idx)
 
 
 @torch.no_grad()
@@ -81,19 +103,29 @@ def shard_model(model: nn.Module, tp_group, tp_rank: int, tp_size: int,
     replaced = 0
     for parent_name, parent in list(model.named_modules()):
         for child_name, child in list(parent._modules.items()):
-            if not isinstance(child, nn.Linear):
+            is_conv1d = _is_hf_conv1d(child)
+            if not isinstance(child, nn.Linear) and not is_conv1d:
                 continue
             full = f"{parent_name}.{child_name}" if parent_name else child_name
-            W = child.weight.data
+            # Conv1D stores weight transposed ([in, out]); normalize to
+            # linear orientation, shard, and keep serving in linear form
+            W = child.weight.data.t().contiguous() if is_conv1d \
+                else child.weight.data
             b = child.bias.data if child.bias is not None else None
             if _match(full, column_patterns):
                 out = W.size(0)
                 assert out % tp_size == 0, \
                     f"{full}: out_features {out} not divisible by tp {tp_size}"
-                sl = slice(tp_rank * out // tp_size,
-                           (tp_rank + 1) * out // tp_size)
-                new = LinearLayer(W[sl].clone(),
-                                  b[sl].clone() if b is not None else None)
+                if _match(full, FUSED_QKV_PATTERNS) and out % 3 == 0:
+                    rows = _fused_qkv_rows(out, tp_rank, tp_size)
+                    new = LinearLayer(W[rows].clone(),
+                                      b[rows].clone() if b is not None
+                                      else None)
+                else:
+                    sl = slice(tp_rank * out // tp_size,
+                               (tp_rank + 1) * out // tp_size)
+                    new = LinearLayer(W[sl].clone(),
+                                      b[sl].clone() if b is not None else None)
             elif _match(full, row_patterns):
                 inp = W.size(1)
                 assert inp % tp_size == 0, \
@@ -124,3 +156,10 @@ def shard_attention_heads(model: nn.Module, tp_rank: int, tp_size: int):
                 assert n % tp_size == 0, \
                     f"{attr}={n} not divisible by tp={tp_size}"
                 setattr(mod, attr, n // tp_size)
+        # GPT-2-style fused attention: split_size/embed_dim size the
+        # c_attn output split and the head reshape — scale them with tp
+        if hasattr(mod, "c_attn"):
+            for attr in ("split_size", "embed_dim"):
+                n = getattr(mod, attr, None)
+                if isinstance(n, int) and n > 0 and n % tp_size == 0:
+                    setattr(mod, attr, n // tp_size)

@@ -120,6 +120,11 @@ def test_zero_init_hf_construction_then_stage3_step():
     """`with zero.Init(): model = AutoModel...` (reference large-model entry
     path, partition_parameters.py Init:824): params come out in the target
     dtype and the model trains under a stage-3 engine."""
+    from .common import run_local
+    run_local(_zero_init_worker)
+
+
+def _zero_init_worker(rank, world):
     import deepspeed_amd
     from deepspeed_amd.runtime.zero.partition import Init
     from transformers import GPT2Config, GPT2LMHeadModel
@@ -144,3 +149,28 @@ def test_zero_init_hf_construction_then_stage3_step():
         engine.step()
         first = first if first is not None else float(loss)
     assert float(loss) < first  # memorizing one batch must reduce loss
+
+
+def _autotp_gpt2_worker(rank, world):
+    """AutoTP over HF GPT-2: Conv1D layers (transposed weights) + fused
+    c_attn qkv head-slicing; sharded forward must match the unsharded
+    model's logits."""
+    import torch.distributed as tdist
+    from deepspeed_amd.inference.auto_tp import (shard_model,
+                                                 shard_attention_heads)
+    model = _gpt2().eval()
+    ref = copy.deepcopy(model)
+    ids = torch.randint(0, 128, (2, 10), generator=torch.Generator()
+                        .manual_seed(3))
+    with torch.no_grad():
+        want = ref(ids).logits
+    n = shard_model(model, None, rank, world)
+    assert n == 4 * 2  # c_attn + attn.c_proj + mlp.c_fc + mlp.c_proj x2 layers
+    shard_attention_heads(model, rank, world)
+    with torch.no_grad():
+        got = model(ids).logits
+    torch.testing.assert_close(got, want, atol=2e-4, rtol=1e-4)
+
+
+def test_autotp_hf_gpt2_forward_parity_ws2():
+    run_distributed(_autotp_gpt2_worker, world_size=2)
