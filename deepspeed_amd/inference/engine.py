@@ -89,7 +89,8 @@ class InferenceEngine(nn.Module):
     @torch.no_grad()
     def generate(self, input_ids: torch.Tensor, max_new_tokens: int = 32,
                  do_sample: bool = False, temperature: float = 1.0,
-                 top_k: int = 0, eos_token_id: Optional[int] = None):
+                 top_k: int = 0, top_p: float = 1.0,
+                 eos_token_id: Optional[int] = None):
         """KV-cached autoregressive generation (greedy or sampled).
 
         input_ids: [B, S] prompt. Returns [B, S + new] including the prompt.
@@ -103,16 +104,29 @@ class InferenceEngine(nn.Module):
                                   else torch.float32),
                            max_new_tokens=max_new_tokens,
                            do_sample=do_sample, temperature=temperature,
-                           top_k=top_k, eos_token_id=eos_token_id)
+                           top_k=top_k, top_p=top_p,
+                           eos_token_id=eos_token_id)
 
 
-def _select_token(logits, do_sample, temperature, top_k):
+def _select_token(logits, do_sample, temperature, top_k, top_p=1.0):
     if not do_sample:
         return logits.argmax(dim=-1, keepdim=True)
     logits = logits.float() / max(temperature, 1e-5)
     if top_k > 0:
         kth = logits.topk(top_k, dim=-1).values[..., -1, None]
         logits = logits.masked_fill(logits < kth, float("-inf"))
+    if top_p < 1.0:
+        # nucleus sampling: keep the smallest prefix of the sorted
+        # distribution whose mass reaches top_p (always >= 1 token)
+        sorted_logits, sorted_idx = logits.sort(dim=-1, descending=True)
+        sorted_probs = torch.softmax(sorted_logits, dim=-1)
+        cum = sorted_probs.cumsum(dim=-1)
+        # drop token i if the mass BEFORE it already covers top_p
+        drop_sorted = (cum - sorted_probs) >= top_p
+        drop_sorted[..., 0] = False
+        drop = torch.zeros_like(drop_sorted).scatter(-1, sorted_idx,
+                                                     drop_sorted)
+        logits = logits.masked_fill(drop, float("-inf"))
     probs = torch.softmax(logits, dim=-1)
     return torch.multinomial(probs, 1)
 
@@ -120,7 +134,7 @@ def _select_token(logits, do_sample, temperature, top_k):
 @torch.no_grad()
 def kv_generate(module, input_ids, *, n_layers, kv_heads, head_dim, max_seq,
                 dtype, max_new_tokens=32, do_sample=False, temperature=1.0,
-                top_k=0, eos_token_id=None):
+                top_k=0, top_p=1.0, eos_token_id=None):
     """Shared KV-cached generation loop (used by InferenceEngine and the
     hybrid RLHF engine)."""
     device = input_ids.device
@@ -133,7 +147,8 @@ def kv_generate(module, input_ids, *, n_layers, kv_heads, head_dim, max_seq,
                              dtype=torch.int32).expand(B, S).contiguous()
     logits = module(input_ids, positions=positions, kv_cache=kv)
     kv.advance()
-    next_tok = _select_token(logits[:, -1], do_sample, temperature, top_k)
+    next_tok = _select_token(logits[:, -1], do_sample, temperature, top_k,
+                             top_p)
     out = torch.cat([out, next_tok], dim=1)
     finished = torch.zeros(B, dtype=torch.bool, device=device)
 
@@ -147,6 +162,7 @@ def kv_generate(module, input_ids, *, n_layers, kv_heads, head_dim, max_seq,
         pos = torch.full((B, 1), kv.cur_len, device=device, dtype=torch.int32)
         logits = module(next_tok, positions=pos, kv_cache=kv)
         kv.advance()
-        next_tok = _select_token(logits[:, -1], do_sample, temperature, top_k)
+        next_tok = _select_token(logits[:, -1], do_sample, temperature,
+                                 top_k, top_p)
         out = torch.cat([out, next_tok], dim=1)
     return out
