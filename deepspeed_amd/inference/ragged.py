@@ -99,10 +99,20 @@ class Request:
     prompt: torch.Tensor                      # [S] token ids
     max_new_tokens: int = 64
     eos_token_id: Optional[int] = None
+    # per-request sampling params (reference inference/v2 SamplingParams)
+    do_sample: bool = False
+    temperature: float = 1.0
+    top_k: int = 0
+    top_p: float = 1.0
     slot: int = -1
     prefilled: int = 0                        # prompt tokens already in KV
     generated: List[int] = field(default_factory=list)
     done: bool = False
+
+    def select(self, logits: torch.Tensor) -> int:
+        from .engine import _select_token
+        return int(_select_token(logits.view(1, -1), self.do_sample,
+                                 self.temperature, self.top_k, self.top_p))
 
     @property
     def in_prefill(self) -> bool:
@@ -175,7 +185,7 @@ class ContinuousBatcher:
             chunk, pos)
         req.prefilled = hi
         if not req.in_prefill:
-            req.generated.append(int(logits[0, -1].argmax()))
+            req.generated.append(req.select(logits[0, -1]))
 
     def _admit(self):
         while self.pending and self.free_slots:
@@ -243,13 +253,13 @@ class ContinuousBatcher:
         row = 0
         for r, chunk in owners:
             if chunk == 0:                       # decode row
-                r.generated.append(int(logits[row, -1].argmax()))
+                r.generated.append(r.select(logits[row, -1]))
                 row += 1
             else:                                # prefill rows
                 row += chunk
                 r.prefilled += chunk
                 if not r.in_prefill:             # prompt complete: sample
-                    r.generated.append(int(logits[row - 1, -1].argmax()))
+                    r.generated.append(r.select(logits[row - 1, -1]))
         return finished
 
     @torch.no_grad()
@@ -279,9 +289,8 @@ class ContinuousBatcher:
             pos = lens[idx].to(device=self.device,
                                dtype=torch.int32).unsqueeze(1)
             logits = self._forward(last, slots, 1, pos)
-            toks = logits[:, -1].argmax(dim=-1)
-            for r, t in zip(reqs, toks.tolist()):
-                r.generated.append(int(t))
+            for i, r in enumerate(reqs):
+                r.generated.append(r.select(logits[i, -1]))
         return finished
 
     @torch.no_grad()

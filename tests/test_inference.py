@@ -315,3 +315,25 @@ def test_top_p_nucleus_sampling():
     out = engine.generate(ids, max_new_tokens=3, do_sample=True,
                           temperature=0.8, top_p=0.9)
     assert out.shape[1] == 9
+
+
+def test_continuous_batcher_per_request_sampling():
+    """Per-request SamplingParams: a near-zero-temperature sampled request
+    must reproduce the greedy tokens while co-batched with greedy ones."""
+    import deepspeed_amd
+    from deepspeed_amd.inference.ragged import ContinuousBatcher, Request
+    model, cfg = _model(seed=13)
+    engine = deepspeed_amd.init_inference(model, dtype=torch.float32)
+    g = torch.Generator().manual_seed(2)
+    prompts = [torch.randint(0, cfg.vocab_size, (7,), generator=g)
+               for _ in range(2)]
+    want = [engine.generate(p.view(1, -1), max_new_tokens=4)[0, p.numel():]
+            .tolist() for p in prompts]
+    torch.manual_seed(0)
+    batcher = ContinuousBatcher(model, max_slots=4, token_budget=8)
+    batcher.put(Request(uid=0, prompt=prompts[0], max_new_tokens=4))
+    batcher.put(Request(uid=1, prompt=prompts[1], max_new_tokens=4,
+                        do_sample=True, temperature=1e-4))
+    done = {r.uid: r for r in batcher.run_to_completion()}
+    assert done[0].generated == want[0]
+    assert done[1].generated == want[1]
