@@ -51,11 +51,14 @@ class InferenceServer:
 
     # ------------------------------------------------------------------- api
     def submit(self, token_ids: List[int], max_new_tokens: int = 32,
-               eos_token_id: Optional[int] = None) -> Request:
+               eos_token_id: Optional[int] = None, do_sample: bool = False,
+               temperature: float = 1.0, top_k: int = 0,
+               top_p: float = 1.0) -> Request:
         req = Request(uid=next(self._uid),
                       prompt=torch.as_tensor(token_ids, dtype=torch.long),
                       max_new_tokens=max_new_tokens,
-                      eos_token_id=eos_token_id)
+                      eos_token_id=eos_token_id, do_sample=do_sample,
+                      temperature=temperature, top_k=top_k, top_p=top_p)
         ev = threading.Event()
         with self._lock:
             self._events[req.uid] = ev
@@ -65,8 +68,9 @@ class InferenceServer:
 
     def generate(self, token_ids: List[int], max_new_tokens: int = 32,
                  eos_token_id: Optional[int] = None,
-                 timeout: float = 300.0) -> List[int]:
-        req = self.submit(token_ids, max_new_tokens, eos_token_id)
+                 timeout: float = 300.0, **sampling) -> List[int]:
+        req = self.submit(token_ids, max_new_tokens, eos_token_id,
+                          **sampling)
         if not req._event.wait(timeout):
             raise TimeoutError(f"request {req.uid} timed out")
         return req.generated
@@ -82,6 +86,10 @@ def build_app(server: InferenceServer):
         token_ids: List[int]
         max_new_tokens: int = 32
         eos_token_id: Optional[int] = None
+        do_sample: bool = False
+        temperature: float = 1.0
+        top_k: int = 0
+        top_p: float = 1.0
 
     app = FastAPI(title="deepspeed_amd inference")
 
@@ -93,7 +101,10 @@ def build_app(server: InferenceServer):
 
     @app.post("/generate")
     def generate(r: GenRequest):
-        out = server.generate(r.token_ids, r.max_new_tokens, r.eos_token_id)
+        out = server.generate(r.token_ids, r.max_new_tokens, r.eos_token_id,
+                              do_sample=r.do_sample,
+                              temperature=r.temperature,
+                              top_k=r.top_k, top_p=r.top_p)
         return {"generated": out}
 
     return app
