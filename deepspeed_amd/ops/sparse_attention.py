@@ -2,11 +2,16 @@
 Triton block-sparse matmul/softmax with Fixed/BigBird/BSLongformer/
 Variable sparsity configs).
 
-MI355X round-1 implementation: the block LAYOUT machinery (the API users
-configure) with attention computed through SDPA using the layout expanded
-to a dense mask — numerically identical to the reference's kernels, O(S^2)
-compute. A gather-based HIP block kernel that realizes the FLOP savings is
-the planned follow-up; the layouts and module interface are stable.
+MI355X implementation: the block LAYOUT machinery (the API users
+configure) plus a GATHER-based compute path (`block_sparse_attention`)
+that touches only the layout's nonzero blocks — the per-pair [bs,bs] and
+[bs,D] products run as ONE hipBLASLt batched GEMM each on GPU, and the
+cross-block softmax is a segment reduction (index_reduce amax + index_add)
+over the pair axis, so compute and HBM traffic scale with nnz blocks, not
+S^2. When a dense `attn_mask` is composed in, the module falls back to
+SDPA over the expanded mask (numerically identical, dense cost). A fused
+MFMA block kernel (flash-style, per-pair tiles) is the planned follow-up
+behind the same function signature.
 """
 
 import math
@@ -95,6 +100,49 @@ class BigBirdSparsityConfig(SparsityConfig):
         return l
 
 
+@torch.no_grad()
+def _layout_pairs(layout: torch.Tensor):
+    nz = layout.nonzero()                      # [P, 3] = (h, qb, kb)
+    return nz[:, 0], nz[:, 1], nz[:, 2]
+
+
+def block_sparse_attention(q: torch.Tensor, k: torch.Tensor,
+                           v: torch.Tensor, layout: torch.Tensor,
+                           block: int, scale: Optional[float] = None
+                           ) -> torch.Tensor:
+    """Attention restricted to `layout`'s nonzero [block x block] tiles.
+
+    q,k,v: [B, H, S, D]; layout: [H, S/block, S/block] bool. Equals SDPA
+    with the block-expanded boolean mask (rows with at least one live
+    block), but computes only the live tiles."""
+    B, H, S, D = q.shape
+    bs = block
+    nq = S // bs
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    h_idx, qb_idx, kb_idx = _layout_pairs(layout.to(q.device))
+    P = h_idx.numel()
+    qb = q.view(B, H, nq, bs, D)
+    kb = k.view(B, H, nq, bs, D)
+    vb = v.view(B, H, nq, bs, D)
+    qg = qb[:, h_idx, qb_idx].float()                    # [B, P, bs, D]
+    kg = kb[:, h_idx, kb_idx].float()
+    vg = vb[:, h_idx, kb_idx].float()
+    scores = torch.matmul(qg, kg.transpose(-1, -2)) * scale  # [B,P,bs,bs]
+    seg = (h_idx * nq + qb_idx).to(q.device)             # [P] row-group id
+    nseg = H * nq
+    rowmax = scores.amax(dim=-1)                         # [B, P, bs]
+    segmax = torch.full((B, nseg, bs), float("-inf"),
+                        device=q.device).index_reduce_(
+                            1, seg, rowmax, "amax")
+    e = torch.exp(scores - segmax[:, seg].unsqueeze(-1))
+    denom = torch.zeros(B, nseg, bs, device=q.device).index_add_(
+        1, seg, e.sum(dim=-1))
+    num = torch.zeros(B, nseg, bs, D, device=q.device).index_add_(
+        1, seg, torch.matmul(e, vg))
+    out = num / denom.clamp_min(torch.finfo(torch.float32).tiny)[..., None]
+    return out.view(B, H, nq, bs, D).reshape(B, H, S, D).to(q.dtype)
+
+
 class SparseSelfAttention(torch.nn.Module):
     """Applies attention under a block-sparse layout
     (reference sparse_self_attention.py). q,k,v: [B, H, S, D]."""
@@ -103,19 +151,28 @@ class SparseSelfAttention(torch.nn.Module):
         super().__init__()
         self.config = sparsity_config
         self._layouts = {}
+        self._masks = {}
+
+    def _layout(self, seq_len):
+        if seq_len not in self._layouts:
+            self._layouts[seq_len] = self.config.make_layout(seq_len)
+        return self._layouts[seq_len]
 
     def _mask(self, seq_len, device):
-        if seq_len not in self._layouts:
-            layout = self.config.make_layout(seq_len)
+        if seq_len not in self._masks:
+            layout = self._layout(seq_len)
             mask = layout.repeat_interleave(self.config.block, dim=1) \
                          .repeat_interleave(self.config.block, dim=2)
-            self._layouts[seq_len] = mask
-        return self._layouts[seq_len].to(device)
+            self._masks[seq_len] = mask
+        return self._masks[seq_len].to(device)
 
     def forward(self, query, key, value, attn_mask: Optional[torch.Tensor] = None):
         B, H, S, D = query.shape
+        if attn_mask is None:
+            # gather path: compute only the live blocks
+            return block_sparse_attention(query, key, value,
+                                          self._layout(S), self.config.block)
         mask = self._mask(S, query.device).unsqueeze(0)  # [1, H, S, S]
-        if attn_mask is not None:
-            mask = mask & attn_mask.bool()
+        mask = mask & attn_mask.bool()
         return F.scaled_dot_product_attention(query, key, value,
                                               attn_mask=mask)

@@ -75,3 +75,44 @@ def test_ds_report_runs(capsys):
     assert main() == 0
     out = capsys.readouterr().out
     assert "op availability" in out and "fused_adam_flat" in out
+
+
+def test_block_sparse_gather_matches_masked_sdpa():
+    """block_sparse_attention (gather path, only live blocks computed) must
+    equal SDPA over the block-expanded mask, for Fixed (incl. causal),
+    BigBird, and a random layout."""
+    import torch.nn.functional as F
+    from deepspeed_amd.ops.sparse_attention import (
+        BigBirdSparsityConfig, FixedSparsityConfig, block_sparse_attention)
+    torch.manual_seed(0)
+    B, H, S, D, bs = 2, 2, 64, 16, 16
+    q, k, v = (torch.randn(B, H, S, D) for _ in range(3))
+    layouts = [
+        FixedSparsityConfig(H, block=bs, num_local_blocks=2).make_layout(S),
+        FixedSparsityConfig(H, block=bs, num_local_blocks=2,
+                            attention="unidirectional").make_layout(S),
+        BigBirdSparsityConfig(H, block=bs, num_random_blocks=1,
+                              num_sliding_window_blocks=1).make_layout(S),
+        torch.rand(H, S // bs, S // bs) < 0.4,
+    ]
+    for layout in layouts:
+        # every query block must see >= 1 key block (SDPA NaNs otherwise)
+        layout[..., 0] |= layout.sum(-1) == 0
+        got = block_sparse_attention(q, k, v, layout, bs)
+        mask = layout.repeat_interleave(bs, 1).repeat_interleave(bs, 2)
+        want = F.scaled_dot_product_attention(q, k, v,
+                                              attn_mask=mask.unsqueeze(0))
+        torch.testing.assert_close(got, want, atol=2e-5, rtol=1e-4)
+
+
+def test_sparse_self_attention_default_uses_gather_path():
+    from deepspeed_amd.ops.sparse_attention import (FixedSparsityConfig,
+                                                    SparseSelfAttention)
+    torch.manual_seed(1)
+    B, H, S, D = 1, 2, 64, 16
+    q, k, v = (torch.randn(B, H, S, D) for _ in range(3))
+    att = SparseSelfAttention(FixedSparsityConfig(H, block=16,
+                                                  num_local_blocks=2))
+    out_gather = att(q, k, v)
+    out_masked = att(q, k, v, attn_mask=torch.ones(S, S, dtype=torch.bool))
+    torch.testing.assert_close(out_gather, out_masked, atol=2e-5, rtol=1e-4)
