@@ -189,6 +189,9 @@ class Config:
         self.seq_parallel_communication_data_type = g(
             "seq_parallel_communication_data_type", None)
         self.sparse_gradients_enabled = g("sparse_gradients", False)
+        # reference top-level "sparse_attention" block ({"mode": "fixed",
+        # "block": 16, ...}); consumed via build_sparse_attention()
+        self.sparse_attention = g("sparse_attention", None)
         self.zero_allow_untested_optimizer = g("zero_allow_untested_optimizer", True)
 
         self.fp16 = FP16Config(**g("fp16", {}))

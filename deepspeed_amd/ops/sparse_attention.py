@@ -100,6 +100,26 @@ class BigBirdSparsityConfig(SparsityConfig):
         return l
 
 
+def build_sparse_attention(ds_sparse_config: dict,
+                           num_heads: int) -> "SparseSelfAttention":
+    """Build SparseSelfAttention from the ds_config "sparse_attention"
+    block (reference config schema: {"mode": "fixed"|"bigbird"|"dense",
+    "block": 16, mode-specific keys...})."""
+    cfg = dict(ds_sparse_config or {})
+    mode = cfg.pop("mode", "fixed")
+    cls = {"fixed": FixedSparsityConfig,
+           "bigbird": BigBirdSparsityConfig,
+           "dense": DenseSparsityConfig}[mode]
+    import inspect
+    allowed = set(inspect.signature(cls.__init__).parameters) - {"self"}
+    kwargs = {k: v for k, v in cfg.items() if k in allowed}
+    dropped = set(cfg) - set(kwargs)
+    if dropped:
+        from ..utils.logging import log_dist
+        log_dist(f"sparse_attention: ignoring unsupported keys {dropped}")
+    return SparseSelfAttention(cls(num_heads, **kwargs))
+
+
 @torch.no_grad()
 def _layout_pairs(layout: torch.Tensor):
     nz = layout.nonzero()                      # [P, 3] = (h, qb, kb)

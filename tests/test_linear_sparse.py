@@ -116,3 +116,21 @@ def test_sparse_self_attention_default_uses_gather_path():
     out_gather = att(q, k, v)
     out_masked = att(q, k, v, attn_mask=torch.ones(S, S, dtype=torch.bool))
     torch.testing.assert_close(out_gather, out_masked, atol=2e-5, rtol=1e-4)
+
+
+def test_sparse_attention_config_block():
+    """ds_config "sparse_attention" block -> SparseSelfAttention factory
+    (reference top-level config schema)."""
+    from deepspeed_amd.config import Config
+    from deepspeed_amd.ops.sparse_attention import (build_sparse_attention,
+                                                    FixedSparsityConfig)
+    cfg = Config({"train_micro_batch_size_per_gpu": 1,
+                  "sparse_attention": {"mode": "fixed", "block": 16,
+                                       "num_local_blocks": 2,
+                                       "attention": "unidirectional"}})
+    att = build_sparse_attention(cfg.sparse_attention, num_heads=2)
+    assert isinstance(att.config, FixedSparsityConfig)
+    assert att.config.block == 16 and att.config.causal
+    q = torch.randn(1, 2, 64, 8)
+    out = att(q, q, q)
+    assert out.shape == q.shape
