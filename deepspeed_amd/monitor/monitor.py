@@ -90,10 +90,7 @@ class CometMonitor(Monitor):
 
     def write_events(self, events: List[Event]):
         for tag, value, step in events:
-            self.experiment.__internal_api__log_metric__(
-                tag, value, framework="deepspeed_amd", step=step) \
-                if hasattr(self.experiment, "__internal_api__log_metric__") \
-                else self.experiment.log_metric(tag, value, step=step)
+            self.experiment.log_metric(tag, value, step=step)
 
 
 class MonitorMaster(Monitor):
