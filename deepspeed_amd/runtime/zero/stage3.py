@@ -1077,19 +1077,23 @@ class ZeroStage3Optimizer:
                     u.persist = was
                     if u.full is None:
                         continue
-                    if self.modifier_rank is not None and \
-                            self.opt.world_size > 1:
-                        dist.broadcast(u.full, src=self.modifier_rank,
-                                       group=self.opt.dp_group)
-                    if u.full is not u.shard:  # ws=1 aliases them
-                        u.shard.copy_(u.full[self.opt.rank * u.shard_size:
-                                             (self.opt.rank + 1) *
-                                             u.shard_size])
-                    if u.sec_shard is not None:
-                        ss = u.numel // self.opt.hpz_world
-                        u.sec_shard.copy_(
-                            u.full[self.opt.hpz_rank * ss:
-                                   (self.opt.hpz_rank + 1) * ss])
+                    # write back ONLY under modifier_rank (reference
+                    # GatheredParameters semantics) — an unconditional
+                    # copy would round-trip weights through the lossy
+                    # int8 gather when qwZ is on
+                    if self.modifier_rank is not None:
+                        if self.opt.world_size > 1:
+                            dist.broadcast(u.full, src=self.modifier_rank,
+                                           group=self.opt.dp_group)
+                        if u.full is not u.shard:  # ws=1 aliases them
+                            u.shard.copy_(
+                                u.full[self.opt.rank * u.shard_size:
+                                       (self.opt.rank + 1) * u.shard_size])
+                        if u.sec_shard is not None:
+                            ss = u.numel // self.opt.hpz_world
+                            u.sec_shard.copy_(
+                                u.full[self.opt.hpz_rank * ss:
+                                       (self.opt.hpz_rank + 1) * ss])
                     if not u.persist:
                         self.opt._release(u)
             return False
