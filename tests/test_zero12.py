@@ -741,3 +741,34 @@ def _z3_full_zeropp_worker(rank, world):
         ref_model, data, lr, steps, 1, torch.bfloat16)
     for a, b in zip(losses, ref_losses):
         assert abs(a - b) < 1e-1, (losses, ref_losses)
+
+
+def _gathered_readonly_qwz_worker(rank, world):
+    """Read-only GatheredParameters under qwZ must NOT perturb the shards
+    (no lossy int8 round-trip); modifier_rank=0 writes must propagate."""
+    import deepspeed_amd
+    model = TinyNet()
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "zero_optimization": {"stage": 3, "zero_quantized_weights": True,
+                              "zero_quantization_group_size": 64},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-2}},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    opt = engine.optimizer
+    before = [u.shard.clone() for u in opt.units]
+    p = next(engine.module.parameters())
+    with opt.gathered_params(p):           # read-only
+        _ = p.sum()
+    for u, b in zip(opt.units, before):
+        assert torch.equal(u.shard, b), "read-only gather changed a shard"
+    with opt.gathered_params(p, modifier_rank=0):
+        if opt.rank == 0:
+            p.data.fill_(0.125)
+    with opt.gathered_params(p):
+        assert torch.all(p.data == 0.125), "modifier write did not propagate"
+
+
+def test_gathered_params_readonly_qwz_ws2():
+    run_distributed(_gathered_readonly_qwz_worker, world_size=2)
