@@ -150,7 +150,9 @@ def block_sparse_attention(q: torch.Tensor, k: torch.Tensor,
     scores = torch.matmul(qg, kg.transpose(-1, -2)) * scale  # [B,P,bs,bs]
     seg = (h_idx * nq + qb_idx).to(q.device)             # [P] row-group id
     nseg = H * nq
-    rowmax = scores.amax(dim=-1)                         # [B, P, bs]
+    # the softmax shift is gradient-invariant: compute it detached so the
+    # segment amax (whose backward is not needed) stays out of the graph
+    rowmax = scores.detach().amax(dim=-1)                # [B, P, bs]
     segmax = torch.full((B, nseg, bs), float("-inf"),
                         device=q.device).index_reduce_(
                             1, seg, rowmax, "amax")
