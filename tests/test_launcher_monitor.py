@@ -137,3 +137,18 @@ def test_launcher_elastic_restart(tmp_path):
     assert (tmp_path / "a0_r1").read_text() == "29881"
     assert (tmp_path / "a1_r0").read_text() == "29882"
     assert (tmp_path / "a1_r1").exists()
+
+
+def test_launcher_elastic_restarts_exhausted(tmp_path):
+    """A persistently failing worker exhausts max_restarts and the final
+    exit code propagates."""
+    script = tmp_path / "always_fails.py"
+    script.write_text("import os, sys; sys.exit(7)\n")
+    out = subprocess.run(
+        [sys.executable, "-m", "deepspeed_amd.launcher.runner",
+         "--num_gpus", "1", "--max_restarts", "1",
+         "--master_port", "29891", str(script)],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 7
+    assert "elastic restart 1/1" in out.stdout + out.stderr
