@@ -109,6 +109,8 @@ def build_sparse_attention(ds_sparse_config: dict,
     mode = cfg.pop("mode", "fixed")
     cls = {"fixed": FixedSparsityConfig,
            "bigbird": BigBirdSparsityConfig,
+           "bslongformer": BSLongformerSparsityConfig,
+           "variable": VariableSparsityConfig,
            "dense": DenseSparsityConfig}[mode]
     import inspect
     allowed = set(inspect.signature(cls.__init__).parameters) - {"self"}
@@ -163,6 +165,84 @@ def block_sparse_attention(q: torch.Tensor, k: torch.Tensor,
         1, seg, torch.matmul(e, vg))
     out = num / denom.clamp_min(torch.finfo(torch.float32).tiny)[..., None]
     return out.view(B, H, nq, bs, D).reshape(B, H, S, D).to(q.dtype)
+
+
+class BSLongformerSparsityConfig(SparsityConfig):
+    """Longformer pattern (reference BSLongformerSparsityConfig): sliding
+    window + designated global block ROWS/COLUMNS given by block index."""
+
+    def __init__(self, num_heads: int, block: int = 16,
+                 num_sliding_window_blocks: int = 3,
+                 global_block_indices=(0,), global_block_end_indices=None):
+        super().__init__(num_heads, block)
+        self.num_sliding = num_sliding_window_blocks
+        if global_block_end_indices is None:
+            self.globals = [(i, i + 1) for i in global_block_indices]
+        else:
+            assert len(global_block_indices) == len(global_block_end_indices)
+            self.globals = list(zip(global_block_indices,
+                                    global_block_end_indices))
+
+    def make_layout(self, seq_len):
+        l, n = self._empty(seq_len)
+        half = self.num_sliding // 2
+        for i in range(n):
+            for j in range(max(0, i - half), min(n, i + half + 1)):
+                l[:, i, j] = True
+        for lo, hi in self.globals:
+            lo, hi = min(lo, n), min(hi, n)
+            l[:, lo:hi, :] = True   # global rows attend everywhere
+            l[:, :, lo:hi] = True   # everyone attends global columns
+        return l
+
+
+class VariableSparsityConfig(SparsityConfig):
+    """Variable pattern (reference VariableSparsityConfig): per-head-shared
+    layout of local windows of varying sizes + global + optional random
+    blocks; optionally causal."""
+
+    def __init__(self, num_heads: int, block: int = 16,
+                 num_random_blocks: int = 0,
+                 local_window_blocks=(4,),
+                 global_block_indices=(0,), global_block_end_indices=None,
+                 attention: str = "bidirectional", seed: int = 0):
+        super().__init__(num_heads, block)
+        self.num_random_blocks = num_random_blocks
+        self.local_window_blocks = list(local_window_blocks)
+        if global_block_end_indices is None:
+            self.globals = [(i, i + 1) for i in global_block_indices]
+        else:
+            self.globals = list(zip(global_block_indices,
+                                    global_block_end_indices))
+        self.causal = attention == "unidirectional"
+        self.seed = seed
+
+    def make_layout(self, seq_len):
+        l, n = self._empty(seq_len)
+        # tile local windows: first windows use the given sizes, the last
+        # size repeats to cover the sequence (reference semantics)
+        sizes = self.local_window_blocks
+        start, wi = 0, 0
+        while start < n:
+            w = sizes[min(wi, len(sizes) - 1)]
+            end = min(start + w, n)
+            l[:, start:end, start:end] = True
+            start, wi = end, wi + 1
+        for lo, hi in self.globals:
+            lo, hi = min(lo, n), min(hi, n)
+            l[:, lo:hi, :] = True
+            l[:, :, lo:hi] = True
+        if self.num_random_blocks:
+            g = torch.Generator().manual_seed(self.seed)
+            for h in range(self.num_heads):
+                for i in range(n):
+                    for j in torch.randint(0, n, (self.num_random_blocks,),
+                                           generator=g).tolist():
+                        l[h, i, j] = True
+        if self.causal:
+            tri = torch.tril(torch.ones(n, n, dtype=torch.bool))
+            l &= tri
+        return l
 
 
 class SparseSelfAttention(torch.nn.Module):

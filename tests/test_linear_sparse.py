@@ -160,3 +160,35 @@ def test_block_sparse_attention_backward():
     out2.backward(g)
     for a, b in ((q1, q2), (k1, k2), (v1, v2)):
         torch.testing.assert_close(a.grad, b.grad, atol=2e-5, rtol=1e-4)
+
+
+def test_longformer_and_variable_layouts():
+    """BSLongformer and Variable sparsity configs (reference
+    sparsity_config.py): window/global/varying-window semantics, and both
+    run through the gather compute path."""
+    from deepspeed_amd.ops.sparse_attention import (
+        BSLongformerSparsityConfig, VariableSparsityConfig,
+        SparseSelfAttention, block_sparse_attention)
+    import torch.nn.functional as F
+    H, S, bs = 2, 96, 16
+    lf = BSLongformerSparsityConfig(H, block=bs,
+                                    num_sliding_window_blocks=3,
+                                    global_block_indices=(0,))
+    L = lf.make_layout(S)
+    n = S // bs
+    assert L[:, 0, :].all() and L[:, :, 0].all()        # global row+col
+    assert L[0, 3, 2] and L[0, 3, 4] and not L[0, 3, 5]  # window of 3
+    var = VariableSparsityConfig(H, block=bs, local_window_blocks=(1, 2),
+                                 attention="unidirectional")
+    V = var.make_layout(S)
+    assert V[0, 1, 1] and V[0, 1, 2].logical_not()       # window [1,3) causal
+    assert V[0, 2, 1] and not V[0, 2, 3]
+    assert torch.equal(V[0], V[0].tril())                # causal
+    torch.manual_seed(3)
+    q = torch.randn(1, H, S, 8)
+    for layout in (L, V):
+        got = block_sparse_attention(q, q, q, layout, bs)
+        mask = layout.repeat_interleave(bs, 1).repeat_interleave(bs, 2)
+        want = F.scaled_dot_product_attention(q, q, q,
+                                              attn_mask=mask.unsqueeze(0))
+        torch.testing.assert_close(got, want, atol=2e-5, rtol=1e-4)
