@@ -1,31 +1,36 @@
 // Flash-attention forward for MI355X (gfx950 / CDNA4) — hand-written MFMA.
 //
-// Replaces the aotriton SDPA forward on the inference/prefill path
+// Replaces the aotriton SDPA forward on the training/prefill hot path
 // (reference analogue: csrc/transformer/inference softmax+GEMM chain and
-// inference/v2 blocked flash). Structure follows the CDNA4 guide's
-// verified flash ladder (cdna_hip_programming.md §attention).
+// inference/v2 blocked flash). v3 follows the CDNA4 guide's verified
+// 8-wave/QB32/KVBLK64 attention ladder (cdna_hip_programming.md §B).
 //
-// v2 structure (8-wave K/V reuse):
+// v3 structure (fixes v2's 144 B/lane scratch spill and 2-barrier stalls):
 // * 512-thread workgroup = 8 waves x 32 query rows (Q tile = 256 rows);
-//   grid (ceil(S/256), H, B). K and V^T tiles are staged in LDS ONCE per
-//   kv step and consumed by all 8 waves — global K/V traffic drops 8x vs
-//   the naive per-wave version (which measured HBM-bound at 164 TF).
-// * XOR swizzle on the LDS tiles (guide §6 G4: `byte ^= ((row&7)<<4)`)
-//   so the MFMA fragment reads (row-strided ds_read_b128) don't bank
-//   conflict.
+//   grid (ceil(S/256), H, B). KVBLK = 64 rows per step.
+// * K and V^T tiles double-buffered in LDS (64 KB total): ONE
+//   __syncthreads per 64-row kv step instead of two per 32-row step.
+// * async-STAGE split: next tile's global loads are issued into registers
+//   BEFORE the current tile's compute (HBM latency hides under MFMA),
+//   ds_writes land after the compute, before the barrier.
+// * XOR swizzle on both LDS tiles (`chunk ^= row&7`) — bank-conflict-free
+//   row-strided ds_read_b128 fragment reads.
 // * mfma_f32_32x32x16_bf16 with probe-verified layouts
 //   (scripts/mfma_probe.hip, run on gfx950 2026-08-20):
 //     A: row = lane&31, k = reg + 8*(lane>>5)
 //     B: col = lane&31, k = reg + 8*(lane>>5)
 //     C/D: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
-// * swapped S^T = mfma(A=K, B=Q): each lane owns 16 score entries of ONE
-//   query column -> row max/sum is 15 VALU ops + one shfl_xor(32).
-// * P re-layout for PV (C/D -> B fragment) is a register permutation
-//   within the (lane, lane^32) pair: reg' = (kv&3) + 4*(kv>>3),
-//   cross-half values via shfl_xor(32).
-// * KSUB template: number of 32-row kv sub-tiles staged per barrier pair
-//   (KSUB=2 = KVBLK 64, half the barriers — guide ladder step; dbg-only
-//   until GPU-validated, default stays the validated KSUB=1).
+// * swapped S^T = mfma(A=K, B=Q): each lane owns 32 score entries of ONE
+//   query column -> row max/sum is ~31 VALU ops + one shfl_xor(32).
+// * P re-layout for PV (C/D -> B fragment) with COMPILE-TIME register
+//   indices (v2 used half-dependent runtime indices -> scratch): for
+//   pf[ks][reg], the owner half is ho=(reg>>2)&1 and both candidate
+//   source registers rp = (reg&3)+8*(ks&1)+4*{ho,1-ho} are constants;
+//   the cross-half value moves with one shfl_xor(32).
+// * defer-max (RESCALE_THRESHOLD=8): skip the O(64-VALU) rescale pass
+//   when the tile max doesn't raise the running max by >8 (exp stays
+//   bounded by e^8, fp32 accumulators don't care).
+// * s_setprio(1) around the MFMA clusters.
 //
 // Layouts: q,k [B,S,H(kv),128] bf16; vt [B,Hkv,128,S] bf16; o [B,S,H,128].
 // Requires D=128, S % 32 == 0 (wrapper falls back to SDPA otherwise).
@@ -43,6 +48,7 @@ constexpr int NWAVE = 8;  // waves per workgroup
 constexpr int QTILE = QB * NWAVE;
 constexpr int TPB = NWAVE * 64;  // 8 waves x 64 lanes
 constexpr int D = 128;           // head dim
+constexpr int KB = 64;           // kv rows per step
 
 __device__ __forceinline__ short f2bf(float f) {
   union {
@@ -54,20 +60,49 @@ __device__ __forceinline__ short f2bf(float f) {
   return (short)(r >> 16);
 }
 
-// LDS tile helpers: tiles stored as 16-byte chunks with an XOR swizzle on
-// the chunk index so row-strided fragment reads spread across banks.
-// K tile: [KSUB*32][D] bf16 = rows x 16 chunks. V^T: [D][KSUB*32] = 128 x
-// KSUB*4 chunks per row.
+// Swizzled chunk index inside a tile: row-major rows of NC 16-byte chunks,
+// chunk XOR'd with row&7 so row-strided fragment reads spread the banks.
 __device__ __forceinline__ int k_sw(int row, int chunk) {
   return row * 16 + (chunk ^ (row & 7));
 }
-template <int W>
 __device__ __forceinline__ int v_sw(int row, int chunk) {
-  return row * W + (chunk ^ (row & (W - 1)));
+  return row * 8 + (chunk ^ (row & 7));
 }
 
-// VAR ablation (debug): bit0 = K from LDS, bit1 = V from LDS (3 = normal)
-template <bool CAUSAL, int VAR = 3, int KSUB = 1>
+struct StageRegs {
+  bf16x8s k0, k1, v0, v1;
+};
+
+// Issue the global loads for one kv tile (2 K chunks + 2 V^T chunks per
+// thread). Called EARLY so the HBM latency hides under the MFMA phase.
+__device__ __forceinline__ StageRegs stage_load(
+    const short* __restrict__ k, const short* __restrict__ vtp,
+    long long k_base, int HkvD, int kv0, int S, int tid) {
+  StageRegs r;
+  const int krow = tid >> 4, kc = tid & 15;
+  const long long kr0 = min(kv0 + krow, S - 1);
+  const long long kr1 = min(kv0 + krow + 32, S - 1);
+  r.k0 = *(const bf16x8s*)(k + k_base + kr0 * HkvD + kc * 8);
+  r.k1 = *(const bf16x8s*)(k + k_base + kr1 * HkvD + kc * 8);
+  const int vrow = tid >> 3, vc = tid & 7;
+  const int kvc = min(kv0 + vc * 8, S - 8);
+  r.v0 = *(const bf16x8s*)(vtp + (long long)vrow * S + kvc);
+  r.v1 = *(const bf16x8s*)(vtp + (long long)(vrow + 64) * S + kvc);
+  return r;
+}
+
+__device__ __forceinline__ void stage_write(const StageRegs& r,
+                                            lds_chunk* kbuf, lds_chunk* vbuf,
+                                            int tid) {
+  const int krow = tid >> 4, kc = tid & 15;
+  kbuf[k_sw(krow, kc)] = r.k0;
+  kbuf[k_sw(krow + 32, kc)] = r.k1;
+  const int vrow = tid >> 3, vc = tid & 7;
+  vbuf[v_sw(vrow, vc)] = r.v0;
+  vbuf[v_sw(vrow + 64, vc)] = r.v1;
+}
+
+template <bool CAUSAL, bool DEFER = true>
 __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
     const short* __restrict__ q,   // [B, S, H, D]
     const short* __restrict__ k,   // [B, S, Hkv, D]
@@ -76,10 +111,8 @@ __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
     float* __restrict__ lse,       // optional [B, H, S] log-sum-exp
     const int B, const int S, const int H, const int Hkv,
     const float scale) {
-  constexpr int KB = KSUB * 32;   // kv rows staged per barrier pair
-  constexpr int VW = KSUB * 4;    // V^T chunks per row
-  __shared__ lds_chunk kt_lds[KB * 16];
-  __shared__ lds_chunk vt_lds[D * VW];
+  __shared__ lds_chunk kbuf[2][KB * 16];  // K: [64 rows][16 chunks]
+  __shared__ lds_chunk vbuf[2][D * 8];    // V^T: [128 rows][8 chunks]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -111,109 +144,127 @@ __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
 
   const int kv_end_blk = CAUSAL ? min(S, q0b + QTILE) : S;
   const int kv_end_wave = CAUSAL ? min(S, q0w + QB) : S;
+  const int HkvD = Hkv * D;
   const long long k_base = (((long long)b * S) * Hkv + hkv) * D;
   const short* vtp = vt + (((long long)b * Hkv + hkv) * D) * S;
+  const int nt = (kv_end_blk + KB - 1) / KB;
 
-  for (int kv0 = 0; kv0 < kv_end_blk; kv0 += KB) {
-    // ---- cooperative stage: K tile (KB*16 chunks) + V^T (D*VW chunks)
-    __syncthreads();
-    for (int g = tid; g < KB * 16; g += TPB) {
-      const int row = g >> 4, c = g & 15;
-      const int kvr = min(kv0 + row, S - 1);
-      kt_lds[k_sw(row, c)] = *(const bf16x8s*)(
-          k + k_base + (long long)kvr * Hkv * D + c * 8);
-    }
-    for (int g = tid; g < D * VW; g += TPB) {
-      const int vrow = g / VW, vc = g % VW;
-      const int kvc = min(kv0 + vc * 8, S - 8);  // S%8==0 guaranteed
-      vt_lds[v_sw<VW>(vrow, vc)] = *(const bf16x8s*)(
-          vtp + (long long)vrow * S + kvc);
-    }
-    __syncthreads();
+  // ---- prologue: stage tile 0 into buffer 0
+  {
+    StageRegs sr = stage_load(k, vtp, k_base, HkvD, 0, S, tid);
+    stage_write(sr, kbuf[0], vbuf[0], tid);
+  }
+  __syncthreads();
 
-    if (kv0 >= kv_end_wave) continue;  // past my diagonal: barriers only
+  for (int it = 0; it < nt; ++it) {
+    const int kv0 = it * KB;
+    const int cur = it & 1;
 
+    // ---- async stage: issue next tile's loads before compute
+    StageRegs sr;
+    const bool have_next = (it + 1 < nt);
+    if (have_next)
+      sr = stage_load(k, vtp, k_base, HkvD, kv0 + KB, S, tid);
+
+    if (kv0 < kv_end_wave) {
+      const lds_chunk* kt_lds = kbuf[cur];
+      const lds_chunk* vt_lds = vbuf[cur];
+
+      // ---- S^T = K · Q^T for both 32-row kv subtiles
+      f32x16 p2[2];
 #pragma unroll
-    for (int sub = 0; sub < KSUB; ++sub) {
-      const int kv0s = kv0 + sub * 32;
-      if (KSUB > 1 && kv0s >= kv_end_wave) break;  // no barriers inside
-
-      // ---- S^T = K · Q^T : A = K rows (kv), B = Q cols (q)
-      f32x16 st = {};
+      for (int sub = 0; sub < 2; ++sub) {
+        f32x16 st = {};
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int kk = 0; kk < 8; ++kk) {
-        // A slot: row = sub*32+col, d chunk = kk*2 + half (8 bf16 each)
-        bf16x8s kf;
-        if (VAR & 1)
-          kf = kt_lds[k_sw(sub * 32 + col, kk * 2 + half)];
-        else
-          kf = *(const bf16x8s*)(k + k_base +
-                                 (long long)min(kv0s + col, S - 1) * Hkv * D +
-                                 kk * 16 + 8 * half);
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[kk], st, 0, 0, 0);
+        for (int kk = 0; kk < 8; ++kk) {
+          bf16x8s kf = kt_lds[k_sw(sub * 32 + col, kk * 2 + half)];
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[kk], st,
+                                                       0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+        p2[sub] = st;
       }
 
-      // ---- masked scale + online softmax (per-lane: one q, 16 kv entries)
-      float p[16];
+      // ---- masked scale + tile max (lane owns 32 kv entries of one q)
       float mt = -1e30f;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kvl = (r & 3) + 8 * (r >> 2) + 4 * half;
-        float s = st[r] * scale;
-        if ((CAUSAL && kv0s + kvl > qrow) || kv0s + kvl >= S) s = -1e30f;
-        p[r] = s;
-        mt = fmaxf(mt, s);
+      for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvl = kv0 + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          float s = p2[sub][r] * scale;
+          if ((CAUSAL && kvl > qrow) || kvl >= S) s = -1e30f;
+          p2[sub][r] = s;
+          mt = fmaxf(mt, s);
+        }
       }
       mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
-      const float m_new = fmaxf(m_run, mt);
-      const float alpha = __expf(m_run - m_new);
+
+      // ---- online softmax with defer-max
+      bool rescale = true;
+      float m_new;
+      if (DEFER) {
+        rescale = !__all(mt <= m_run + 8.f);
+        m_new = rescale ? fmaxf(m_run, mt) : m_run;
+      } else {
+        m_new = fmaxf(m_run, mt);
+      }
       float lt = 0.f;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        p[r] = __expf(p[r] - m_new);
-        lt += p[r];
+      for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          p2[sub][r] = __expf(p2[sub][r] - m_new);
+          lt += p2[sub][r];
+        }
       }
       lt += __shfl_xor(lt, 32, 64);
-      l_run = l_run * alpha + lt;
+      const float alpha = __expf(m_run - m_new);
+      l_run = (rescale ? l_run * alpha : l_run) + lt;
       m_run = m_new;
 
-      // other half's p values (same q column lives in lane^32)
-      float px[16];
+      // ---- build P B-fragments with compile-time register indices.
+      // pf[ks] covers kv = ks*16 + reg + 8*half; the owner half of that
+      // score is ho=(reg>>2)&1 (independent of the requester's half), at
+      // register rp = (reg&3)+8*(ks&1)+4*h of p2[ks>>1].
+      bf16x8s pf[4];
 #pragma unroll
-      for (int r = 0; r < 16; ++r) px[r] = __shfl_xor(p[r], 32, 64);
-
-      // ---- build P B-fragments: slot (kk2, reg) -> kv = kk2*16+8*half+reg
-      bf16x8s pf[2];
-#pragma unroll
-      for (int kk2 = 0; kk2 < 2; ++kk2) {
+      for (int ks = 0; ks < 4; ++ks) {
 #pragma unroll
         for (int reg = 0; reg < 8; ++reg) {
-          const int kv = kk2 * 16 + 8 * half + reg;
-          const int rp = (kv & 3) + 4 * (kv >> 3);
-          const bool mine = (((kv >> 2) & 1) == half);
-          pf[kk2][reg] = f2bf(mine ? p[rp] : px[rp]);
+          const int ho = (reg >> 2) & 1;      // compile-time
+          const int s_ = ks >> 1;             // compile-time
+          const int rp_own = (reg & 3) + 8 * (ks & 1) + 4 * ho;
+          const int rp_oth = (reg & 3) + 8 * (ks & 1) + 4 * (1 - ho);
+          const float own = p2[s_][rp_own];
+          const float oth = __shfl_xor(p2[s_][rp_oth], 32, 64);
+          pf[ks][reg] = f2bf(half == ho ? own : oth);
         }
       }
 
       // ---- O^T += V^T · P : A = V^T rows (d), k = kv; B = P cols (q)
 #pragma unroll
       for (int dblk = 0; dblk < 4; ++dblk) {
+        if (rescale) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) oacc[dblk][r] *= alpha;
-#pragma unroll
-        for (int kk2 = 0; kk2 < 2; ++kk2) {
-          // A slot: row = dblk*32+col, kv chunk = sub*4 + kk2*2 + half
-          bf16x8s vf;
-          if (VAR & 2)
-            vf = vt_lds[v_sw<VW>(dblk * 32 + col, sub * 4 + kk2 * 2 + half)];
-          else
-            vf = *(const bf16x8s*)(vtp + (long long)(dblk * 32 + col) * S +
-                                   kv0s + kk2 * 16 + 8 * half);
-          oacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              vf, pf[kk2], oacc[dblk], 0, 0, 0);
+          for (int r = 0; r < 16; ++r) oacc[dblk][r] *= alpha;
         }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          bf16x8s vf = vt_lds[v_sw(dblk * 32 + col, ks * 2 + half)];
+          oacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              vf, pf[ks], oacc[dblk], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
       }
     }
+
+    // ---- write next tile into the other buffer; barrier flips it live
+    if (have_next)
+      stage_write(sr, kbuf[cur ^ 1], vbuf[cur ^ 1], tid);
+    __syncthreads();
   }
 
   // ---- epilogue: O[q][d] = O^T / l
@@ -249,37 +300,18 @@ extern "C" void ds_flash_fwd(const void* q, const void* k, const void* vt,
                        (short*)o, (float*)lse, B, S, H, Hkv, scale);
 }
 
-// variants 0-3: K/V LDS ablation at KSUB=1; variant 4: KVBLK=64 (KSUB=2)
+// dbg variants: 0 = defer-max OFF (numerics A/B), everything else = full v3
 extern "C" void ds_flash_fwd_dbg(const void* q, const void* k, const void* vt,
                                  void* o, int B, int S, int H, int Hkv,
                                  float scale, int variant, void* stream) {
   dim3 grid((S + QTILE - 1) / QTILE, H, B);
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
-  switch (variant) {
-    case 0:
-      hipLaunchKernelGGL((flash_fwd_kernel<true, 0>), grid, dim3(TPB), 0, st,
-                         (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, nullptr, B, S, H, Hkv, scale);
-      break;
-    case 1:
-      hipLaunchKernelGGL((flash_fwd_kernel<true, 1>), grid, dim3(TPB), 0, st,
-                         (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, nullptr, B, S, H, Hkv, scale);
-      break;
-    case 2:
-      hipLaunchKernelGGL((flash_fwd_kernel<true, 2>), grid, dim3(TPB), 0, st,
-                         (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, nullptr, B, S, H, Hkv, scale);
-      break;
-    case 4:
-      hipLaunchKernelGGL((flash_fwd_kernel<true, 3, 2>), grid, dim3(TPB), 0,
-                         st, (const short*)q, (const short*)k,
-                         (const short*)vt, (short*)o, nullptr, B, S, H, Hkv,
-                         scale);
-      break;
-    default:
-      hipLaunchKernelGGL((flash_fwd_kernel<true, 3>), grid, dim3(TPB), 0, st,
-                         (const short*)q, (const short*)k, (const short*)vt,
-                         (short*)o, nullptr, B, S, H, Hkv, scale);
-  }
+  if (variant == 0)
+    hipLaunchKernelGGL((flash_fwd_kernel<true, false>), grid, dim3(TPB), 0,
+                       st, (const short*)q, (const short*)k, (const short*)vt,
+                       (short*)o, nullptr, B, S, H, Hkv, scale);
+  else
+    hipLaunchKernelGGL((flash_fwd_kernel<true, true>), grid, dim3(TPB), 0, st,
+                       (const short*)q, (const short*)k, (const short*)vt,
+                       (short*)o, nullptr, B, S, H, Hkv, scale);
 }
