@@ -32,15 +32,35 @@ def _worker(rank, world_size, port, backend, fn, args):
         dist.destroy_process_group()
 
 
-def run_distributed(fn, world_size=2, backend=None, args=(), timeout=120):
-    """Spawn `world_size` processes, each running fn(rank, world_size, *args)."""
+def run_distributed(fn, world_size=2, backend=None, args=(), timeout=300):
+    """Spawn `world_size` processes, each running fn(rank, world_size, *args).
+
+    Joins with a deadline: a hung rank (lost collective, deadlocked
+    barrier) kills the whole group and fails the test instead of hanging
+    the suite forever (the reference's DistributedExec hang detection,
+    tests/unit/common.py:170).
+    """
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     port = _free_port()
     # fork is much faster on the CPU test host; GPU requires spawn
     method = "spawn" if torch.cuda.is_available() else "fork"
-    mp.start_processes(_worker, args=(world_size, port, backend, fn, args),
-                       nprocs=world_size, join=True, start_method=method)
+    ctx = mp.start_processes(_worker,
+                             args=(world_size, port, backend, fn, args),
+                             nprocs=world_size, join=False,
+                             start_method=method)
+    import time
+    deadline = time.monotonic() + timeout
+    while not ctx.join(timeout=max(1.0, deadline - time.monotonic())):
+        if time.monotonic() >= deadline:
+            for p in ctx.processes:
+                if p.is_alive():
+                    p.terminate()
+            for p in ctx.processes:
+                p.join(5)
+            raise RuntimeError(
+                f"distributed test hung: {world_size} ranks did not finish "
+                f"within {timeout}s (backend={backend})")
 
 
 def run_local(fn, backend=None, args=()):
