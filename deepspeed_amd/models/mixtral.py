@@ -143,6 +143,10 @@ class MixtralForCausalLM(nn.Module):
                 module.bias.data.zero_()
         elif isinstance(module, nn.Embedding):
             module.weight.data.normal_(0.0, std)
+        else:
+            from ..moe.experts import FusedExperts
+            if isinstance(module, FusedExperts):
+                module.reset_parameters(std)
 
     def aux_loss(self):
         terms = [layer.l_aux for layer in self.model.layers

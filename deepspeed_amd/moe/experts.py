@@ -1,6 +1,18 @@
-"""Local expert container (reference deepspeed/moe/experts.py).
+"""Local expert containers (reference deepspeed/moe/experts.py).
 
-Each EP rank owns ``num_local_experts`` deep copies of the expert module.
+``Experts``: each EP rank owns ``num_local_experts`` deep copies of the
+expert module, executed as a Python loop — the generic fallback for
+arbitrary expert architectures.
+
+``FusedExperts``: the MI355X grouped-GEMM container (role of the
+reference's CUTLASS grouped GEMM,
+inference/v2/kernels/cutlass_ops/moe_gemm/moe_gemm.cu:175). Expert
+weights live as stacked 3D parameters [E_local, out, in] and all local
+experts run in ONE strided-batched hipBLASLt GEMM per projection
+(torch.bmm) plus the fused gated-activation HIP kernel — 3 launches
+instead of 3*E_local. Used automatically when the expert module is a
+gate/up/down SwiGLU MLP (LlamaMLP-shaped).
+
 Every expert parameter is tagged ``allreduce=False`` + ``group_name`` so the
 ZeRO optimizer reduces it over the expert-data-parallel group instead of
 the full DP group.
@@ -8,7 +20,10 @@ the full DP group.
 
 import copy
 
+import torch
 import torch.nn as nn
+
+from ..ops import swiglu
 
 
 class Experts(nn.Module):
@@ -26,3 +41,58 @@ class Experts(nn.Module):
     def forward(self, x):
         # used only for the single-expert fast path; MOELayer batches itself
         return self.local_experts[0](x)
+
+
+def is_swiglu_mlp(m: nn.Module) -> bool:
+    """True for gate/up/down bias-free MLPs (LlamaMLP & friends)."""
+    return all(
+        isinstance(getattr(m, n, None), nn.Linear)
+        and getattr(m, n).bias is None for n in
+        ("gate_proj", "up_proj", "down_proj"))
+
+
+class FusedExperts(nn.Module):
+    """Grouped expert GEMMs over stacked 3D weights (see module docstring).
+
+    forward consumes the already-dispatched [E_local, N, d_model] batch
+    (capacity-packed rows from the index dispatch) and returns the same
+    shape — MOELayer feeds it exactly what the per-expert loop would see.
+    """
+
+    def __init__(self, template: nn.Module, num_local_experts: int,
+                 expert_group_name=None):
+        super().__init__()
+        d_ff, d_model = template.gate_proj.weight.shape
+        self.num_local_experts = num_local_experts
+        self.d_model, self.d_ff = d_model, d_ff
+        dt = template.gate_proj.weight.dtype
+        e = num_local_experts
+        self.w_gate = nn.Parameter(torch.empty(e, d_ff, d_model, dtype=dt))
+        self.w_up = nn.Parameter(torch.empty(e, d_ff, d_model, dtype=dt))
+        self.w_down = nn.Parameter(torch.empty(e, d_model, d_ff, dtype=dt))
+        self.reset_parameters()
+        # expert 0 keeps the template's weights (parity with deepcopy init)
+        with torch.no_grad():
+            self.w_gate[0].copy_(template.gate_proj.weight)
+            self.w_up[0].copy_(template.up_proj.weight)
+            self.w_down[0].copy_(template.down_proj.weight)
+        for p in self.parameters():
+            p.allreduce = False
+            p.group_name = expert_group_name
+
+    def expert_parameters(self, j):
+        """Per-expert views in LlamaMLP parameter order
+        (gate_proj.weight, up_proj.weight, down_proj.weight) — used by
+        tests/tools that address experts individually."""
+        return [self.w_gate[j], self.w_up[j], self.w_down[j]]
+
+    def reset_parameters(self, std: float = 0.02):
+        for w in (self.w_gate, self.w_up, self.w_down):
+            w.data.normal_(0.0, std)
+
+    def forward(self, x):
+        # x: [E_local, N, d_model]
+        g = torch.bmm(x, self.w_gate.transpose(1, 2))
+        u = torch.bmm(x, self.w_up.transpose(1, 2))
+        h = swiglu(g, u)
+        return torch.bmm(h, self.w_down.transpose(1, 2))

@@ -7,7 +7,7 @@ import torch.nn as nn
 from .. import comm as dist
 from ..parallel import groups
 from ..utils.logging import log_dist
-from .experts import Experts
+from .experts import Experts, FusedExperts, is_swiglu_mlp
 from .sharded_moe import MOELayer, TopKGate
 
 
@@ -34,7 +34,8 @@ class MoE(nn.Module):
                  use_rts: bool = True,
                  use_tutel: bool = False,
                  enable_expert_tensor_parallelism: bool = False,
-                 top2_2nd_expert_sampling: bool = True):
+                 top2_2nd_expert_sampling: bool = True,
+                 use_fused_experts: bool = True):
         super().__init__()
         self.use_residual = use_residual
         self.ep_size = ep_size
@@ -44,11 +45,19 @@ class MoE(nn.Module):
         self.num_local_experts = num_experts // ep_size
         self.expert_group_name = f"ep_size_{ep_size}"
 
-        experts = Experts(expert, self.num_local_experts,
-                          self.expert_group_name)
+        # gate first: FusedExperts' init draws from the global RNG, and the
+        # gate weights must not depend on the expert container choice
         gate = TopKGate(hidden_size, num_experts, k, capacity_factor,
                         eval_capacity_factor, min_capacity, noisy_gate_policy,
                         drop_tokens, use_rts, None, top2_2nd_expert_sampling)
+        if use_fused_experts and is_swiglu_mlp(expert):
+            # grouped-GEMM container: one bmm per projection for all
+            # local experts (MI355X stand-in for CUTLASS moe_gemm)
+            experts = FusedExperts(expert, self.num_local_experts,
+                                   self.expert_group_name)
+        else:
+            experts = Experts(expert, self.num_local_experts,
+                              self.expert_group_name)
         self.deepspeed_moe = MOELayer(gate, experts, self.expert_group_name,
                                       self.ep_size, self.num_local_experts)
         if use_residual:

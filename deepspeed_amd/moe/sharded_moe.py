@@ -225,10 +225,14 @@ class MOELayer(nn.Module):
                                         C, d_model)
         chunks = dispatched.transpose(0, 1).reshape(
             self.num_local_experts, self.ep_size * C, d_model)
-        outs = []
-        for i, expert in enumerate(self.experts.local_experts):
-            outs.append(expert(chunks[i]))
-        expert_out = torch.stack(outs, dim=0)              # [local_E, ep*C, M]
+        from .experts import FusedExperts
+        if isinstance(self.experts, FusedExperts):
+            expert_out = self.experts(chunks)              # [local_E, ep*C, M]
+        else:
+            outs = []
+            for i, expert in enumerate(self.experts.local_experts):
+                outs.append(expert(chunks[i]))
+            expert_out = torch.stack(outs, dim=0)          # [local_E, ep*C, M]
         expert_out = expert_out.reshape(self.num_local_experts, self.ep_size,
                                         C, d_model).transpose(0, 1)
 

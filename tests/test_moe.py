@@ -13,11 +13,19 @@ import torch
 from .common import run_distributed, run_local
 
 
+def _expert_params(experts, j):
+    from deepspeed_amd.moe.experts import FusedExperts
+    if isinstance(experts, FusedExperts):
+        return experts.expert_parameters(j)
+    return list(experts.local_experts[j].parameters())
+
+
 def _seed_experts(moe_layer, ep_rank, num_local):
     """Deterministic per-GLOBAL-expert weights so EP layouts are comparable."""
-    for j, e in enumerate(moe_layer.deepspeed_moe.experts.local_experts):
+    experts = moe_layer.deepspeed_moe.experts
+    for j in range(experts.num_local_experts):
         g = torch.Generator().manual_seed(1000 + ep_rank * num_local + j)
-        for p in e.parameters():
+        for p in _expert_params(experts, j):
             with torch.no_grad():
                 p.copy_(torch.randn(p.shape, generator=g) * 0.05)
 
@@ -71,7 +79,8 @@ def _moe_local_worker(rank, world):
     x = torch.randn(2, 8, h, requires_grad=True)
     out, l_aux, counts = moe(x)
     # single expert, huge capacity: every token routed, gate weight == 1
-    ref = moe.deepspeed_moe.experts.local_experts[0](x)
+    experts = moe.deepspeed_moe.experts
+    ref = experts(x.reshape(1, -1, h)).reshape(x.shape)
     torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
     out.sum().backward()
     assert x.grad is not None and torch.isfinite(x.grad).all()
@@ -100,7 +109,7 @@ def _moe_ep_worker(rank, world):
     moe_local = build(1)
     for j in range(E):
         g = torch.Generator().manual_seed(1000 + j)
-        for p in moe_local.deepspeed_moe.experts.local_experts[j].parameters():
+        for p in _expert_params(moe_local.deepspeed_moe.experts, j):
             with torch.no_grad():
                 p.copy_(torch.randn(p.shape, generator=g) * 0.05)
 
