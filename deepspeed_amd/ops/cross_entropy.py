@@ -20,7 +20,9 @@ class _FusedCE(torch.autograd.Function):
         loss, lse = ext.ce_fwd(logits, labels, ignore_index)
         ctx.save_for_backward(logits, labels, lse)
         ctx.ignore_index = ignore_index
-        return loss.sum(), (labels != ignore_index).sum().float()
+        count = (labels != ignore_index).sum().float()
+        ctx.mark_non_differentiable(count)
+        return loss.sum(), count
 
     @staticmethod
     def backward(ctx, gsum, _gcount):

@@ -42,6 +42,7 @@ namespace {
 typedef __attribute__((ext_vector_type(8))) short bf16x8s;
 typedef __attribute__((ext_vector_type(8))) short lds_chunk;  // 16 B
 typedef float f32x16 __attribute__((ext_vector_type(16)));
+typedef __attribute__((ext_vector_type(2))) int i32x2;
 
 constexpr int QB = 32;    // query rows per wave
 constexpr int NWAVE = 8;  // waves per workgroup
@@ -186,17 +187,33 @@ __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
         p2[sub] = st;
       }
 
-      // ---- masked scale + tile max (lane owns 32 kv entries of one q)
+      // ---- masked scale + tile max (lane owns 32 kv entries of one q).
+      // Interior tiles (fully below the causal diagonal for every q row
+      // of this wave, fully inside S) skip the 64 mask compares.
       float mt = -1e30f;
+      const bool interior = kv0 + KB <= (CAUSAL ? q0w + 1 : S) &&
+                            kv0 + KB <= S;
+      if (interior) {
 #pragma unroll
-      for (int sub = 0; sub < 2; ++sub) {
+        for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int kvl = kv0 + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-          float s = p2[sub][r] * scale;
-          if ((CAUSAL && kvl > qrow) || kvl >= S) s = -1e30f;
-          p2[sub][r] = s;
-          mt = fmaxf(mt, s);
+          for (int r = 0; r < 16; ++r) {
+            const float s = p2[sub][r] * scale;
+            p2[sub][r] = s;
+            mt = fmaxf(mt, s);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kvl = kv0 + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+            float s = p2[sub][r] * scale;
+            if ((CAUSAL && kvl > qrow) || kvl >= S) s = -1e30f;
+            p2[sub][r] = s;
+            mt = fmaxf(mt, s);
+          }
         }
       }
       mt = fmaxf(mt, __shfl_xor(mt, 32, 64));
@@ -227,19 +244,25 @@ __global__ __launch_bounds__(TPB) void flash_fwd_kernel(
       // ---- build P B-fragments with compile-time register indices.
       // pf[ks] covers kv = ks*16 + reg + 8*half; the owner half of that
       // score is ho=(reg>>2)&1 (independent of the requester's half), at
-      // register rp = (reg&3)+8*(ks&1)+4*h of p2[ks>>1].
+      // register rp = (reg&3)+8*(ks&1)+4*h of p2[ks>>1]. One
+      // permlane32_swap(a=low-half need, b=high-half need) delivers BOTH
+      // fragment slots (reg=rr at r0, reg=rr+4 at r1) — 16 VALU-pipe
+      // permlanes per tile instead of 32 LDS-pipe ds_bpermutes.
       bf16x8s pf[4];
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
 #pragma unroll
-        for (int reg = 0; reg < 8; ++reg) {
-          const int ho = (reg >> 2) & 1;      // compile-time
-          const int s_ = ks >> 1;             // compile-time
-          const int rp_own = (reg & 3) + 8 * (ks & 1) + 4 * ho;
-          const int rp_oth = (reg & 3) + 8 * (ks & 1) + 4 * (1 - ho);
-          const float own = p2[s_][rp_own];
-          const float oth = __shfl_xor(p2[s_][rp_oth], 32, 64);
-          pf[ks][reg] = f2bf(half == ho ? own : oth);
+        for (int rr = 0; rr < 4; ++rr) {
+          const int s_ = ks >> 1;                   // compile-time
+          const int rpA = rr + 8 * (ks & 1);        // h=0 requester's need
+          const int rpB = rpA + 4;                  // h=1 requester's need
+          // inline asm: the __builtin_amdgcn_permlane32_swap intrinsic is
+          // wrongly CSE'd across calls with different args by this LLVM
+          // (16 calls collapse to 2 — see /tmp repro pl3/pl6, 2026-08-20)
+          float a = p2[s_][rpA], b = p2[s_][rpB];
+          asm("v_permlane32_swap_b32 %0, %1" : "+v"(a), "+v"(b));
+          pf[ks][rr] = f2bf(a);
+          pf[ks][rr + 4] = f2bf(b);
         }
       }
 
