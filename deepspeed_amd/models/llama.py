@@ -102,11 +102,17 @@ class LlamaAttention(nn.Module):
             return self.o_proj(o.reshape(B, S, -1))
 
         if kv_cache is None and attention_fn is None:
-            from ..ops.attention import flash_attn_available, flash_attn_fwd
+            from ..ops.attention import (flash_attn_available, flash_attn_fwd,
+                                         flash_attn_func,
+                                         flash_train_available)
             if flash_attn_available(q, k):
                 # hand-written CDNA4 MFMA flash kernel, BSHD layout — skips
                 # the transpose entirely (inference/no-grad prefill path)
                 o = flash_attn_fwd(q, k, v, causal=True)
+                return self.o_proj(o.reshape(B, S, -1))
+            if torch.is_grad_enabled() and flash_train_available(q, k):
+                # differentiable hand-written path (fwd + dk/dv/dq kernels)
+                o = flash_attn_func(q, k, v, causal=True)
                 return self.o_proj(o.reshape(B, S, -1))
 
         # [B, H, S, D] for SDPA

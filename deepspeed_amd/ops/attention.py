@@ -15,18 +15,30 @@ import torch
 from ._loader import get_ext
 
 
-def flash_attn_available(q: torch.Tensor, k: torch.Tensor) -> bool:
-    # Numerics-verified but not yet past aotriton SDPA's throughput
-    # (130 vs 400 TF — the guide ladder's KVBLK=64/async-stage/defer-max
-    # steps are pending), so the model hot path keeps SDPA unless opted in.
-    import os
-    if os.environ.get("DS_AMD_FLASH") != "1":
-        return False
+def _eligible(q: torch.Tensor, k: torch.Tensor) -> bool:
     return (get_ext() is not None and q.is_cuda
             and q.dtype == torch.bfloat16
             and q.size(-1) == 128 and q.size(1) % 32 == 0
-            and q.size(1) == k.size(1)
-            and not torch.is_grad_enabled())
+            and q.size(1) == k.size(1))
+
+
+def flash_attn_available(q: torch.Tensor, k: torch.Tensor) -> bool:
+    """Inference/prefill forward: ON by default (v3 kernel beats aotriton
+    SDPA on the bench shape, 417 vs <=402 TF). DS_AMD_FLASH=0 opts out."""
+    import os
+    if os.environ.get("DS_AMD_FLASH", "1") == "0":
+        return False
+    return _eligible(q, k) and not torch.is_grad_enabled()
+
+
+def flash_train_available(q: torch.Tensor, k: torch.Tensor) -> bool:
+    """Training fwd+bwd path (FlashAttnFunc). Gated by DS_AMD_FLASH_TRAIN
+    until the v2 backward beats the aotriton backward on the bench shape
+    (flip the default then)."""
+    import os
+    if os.environ.get("DS_AMD_FLASH_TRAIN", "0") != "1":
+        return False
+    return _eligible(q, k)
 
 
 def flash_attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,

@@ -1,38 +1,51 @@
 // Flash-attention backward for MI355X (gfx950 / CDNA4) — hand-written MFMA.
 //
-// Implements docs/flash_bwd_design.md, tile-for-tile identical to the
-// CPU-validated blueprint ops/flash_bwd_ref.py (which matches autograd).
-// Status: compiled + index-math simulated; GPU numerics validation is the
-// first round-2 task — reachable only through the _dbg bindings until then.
+// v2: 8-wave workgroups with LDS-shared operand tiles. v1 ran one wave per
+// 32-row tile and re-read Q/dO (or K/V) from global for every tile pair —
+// ~30 flops/byte, HBM-bound at ~110 TF. v2 blocks 256 rows per workgroup
+// (8 waves x 32) and stages the per-iteration operand tiles in
+// double-buffered LDS shared by all 8 waves: ~260 flops/byte, compute-
+// bound. Same math as v1 (tile-for-tile the CPU-validated blueprint
+// ops/flash_bwd_ref.py; GPU-numerics-validated vs torch autograd).
 //
 // Layouts (ALL [B, H(kv), S, D] = BHSD, contiguous):
 //   q, o, do: [B, H, S, D]      k, v: [B, Hkv, S, D]
 //   qt, kt, dot: transposed copies [B, H(kv), D, S] (wrapper-made) so the
-//   dK/dV/dQ MFMAs' B-operands (k-dim = q or kv) read contiguously.
-//   lse, delta: [B, H, S] fp32.   dq: [B,H,S,D] fp32? -> bf16 out.
-//   dk, dv: [B, Hkv, S, D] bf16 (query-head group summed in registers).
+//   dK/dV/dQ MFMAs' B-operands (k-dim = q or kv) stage contiguously.
+//   lse, delta: [B, H, S] fp32.  dq/dk/dv bf16 outputs.
 //
 // MFMA fragment maps (probe-verified, scripts/mfma_probe.hip):
 //   A: row=lane&31, k=reg+8*(lane>>5) · B: col=lane&31, same k
 //   C/D: col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)
 //
-// Grid: kernel1 (dk/dv): (S/32, Hkv, B), 64 threads (one wave per kv tile;
-// the query-head group G=H/Hkv is an inner loop so dk/dv accumulate in
-// registers with no atomics). kernel2 (dq): (S/32, H, B).
+// dkdv kernel: grid (S/256, Hkv, B). Each wave owns 32 kv rows (K/V
+// A-fragments preloaded in registers; dk/dv accumulate in registers across
+// the whole q loop — the GQA head group G=H/Hkv is an inner loop, no
+// atomics). Per (g, q-tile) iteration the block stages Q[32][128],
+// dO[32][128], qt[128][32], dot[128][32] (32 KB) double-buffered with the
+// async issue-early/write-late split and ONE barrier per iteration.
+//
+// dq kernel: grid (S/256, H, B). Each wave owns 32 q rows (Q/dO
+// B-fragments + dq accumulator in registers); stages K[32][128],
+// V[32][128], kt[128][32] per kv tile.
 //
 // The C/D -> A-operand change of axis for P^T / dS^T goes through a
-// 32x32 bf16 LDS tile: store via the C/D map, read back as A fragments
-// (row=lane&31, 8 consecutive k) — 16-byte aligned ds reads.
+// per-wave 32x32 bf16 LDS scratch (XOR-swizzled 16B chunks), no barrier
+// needed (same-wave LDS RAW is ordered by lgkmcnt).
 
 #include "ds_kernels.h"
 
 namespace {
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8s;
+typedef __attribute__((ext_vector_type(8))) short lds_chunk;  // 16 B
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
-constexpr int T = 32;   // tile rows
-constexpr int D = 128;  // head dim
+constexpr int T = 32;     // tile rows
+constexpr int D = 128;    // head dim
+constexpr int NW = 8;     // waves per workgroup
+constexpr int MT = T * NW;  // macro tile (256 rows)
+constexpr int TPB = NW * 64;
 
 __device__ __forceinline__ short f2bf(float f) {
   union {
@@ -49,16 +62,64 @@ __device__ __forceinline__ int cd_row(int r, int half) {
   return (r & 3) + 8 * (r >> 2) + 4 * half;
 }
 
-// store a 32x32 fp32 tile (held as C/D fragments, val[16]) into LDS bf16
-// [row][col] row-major; then A-frags read ldsP[lane&31][kk*16+8*half .. +8]
-__device__ __forceinline__ void cd_to_lds(const float* val, short* lds,
-                                          int col, int half) {
-#pragma unroll
-  for (int r = 0; r < 16; ++r) lds[cd_row(r, half) * T + col] = f2bf(val[r]);
+// [32][128] tiles: 16 chunks/row, XOR over row&7 (fwd-verified pattern)
+__device__ __forceinline__ int sw16(int row, int chunk) {
+  return row * 16 + (chunk ^ (row & 7));
+}
+// [128][32] tiles and 32x32 scratch: 4 chunks/row, XOR over row&3
+__device__ __forceinline__ int sw4(int row, int chunk) {
+  return row * 4 + (chunk ^ (row & 3));
+}
+
+// scalar bf16 store into a sw4-swizzled [32][32] scratch
+__device__ __forceinline__ void scr_store(short* scr, int row, int col,
+                                          short v) {
+  const int chunk = (col >> 3) ^ (row & 3);
+  scr[row * 32 + chunk * 8 + (col & 7)] = v;
+}
+// 8-element read at (row, k0=kk*16+8*half) from the same scratch
+__device__ __forceinline__ bf16x8s scr_read(const short* scr, int row,
+                                            int kk, int half) {
+  const int chunk = (2 * kk + half) ^ (row & 3);
+  return *(const bf16x8s*)(scr + row * 32 + chunk * 8);
+}
+
+// ---------------------------------------------------------------- dk/dv
+
+struct DkdvStage {
+  bf16x8s qv, dov, qtv, dotv;
+};
+
+__device__ __forceinline__ DkdvStage dkdv_load(
+    const short* __restrict__ q, const short* __restrict__ dout,
+    const short* __restrict__ qt, const short* __restrict__ dot,
+    long long qbase, long long tbase, int qs, int S, int tid) {
+  DkdvStage r;
+  const int row = tid >> 4, c = tid & 15;      // [32][128] tiles
+  const long long qoff = qbase + (long long)(qs + row) * D + c * 8;
+  r.qv = *(const bf16x8s*)(q + qoff);
+  r.dov = *(const bf16x8s*)(dout + qoff);
+  const int vr = tid >> 2, vc = tid & 3;       // [128][32] tiles
+  const long long toff = tbase + (long long)vr * S + qs + vc * 8;
+  r.qtv = *(const bf16x8s*)(qt + toff);
+  r.dotv = *(const bf16x8s*)(dot + toff);
+  return r;
+}
+
+__device__ __forceinline__ void dkdv_write(const DkdvStage& r,
+                                           lds_chunk* qb, lds_chunk* dob,
+                                           lds_chunk* qtb, lds_chunk* dotb,
+                                           int tid) {
+  const int row = tid >> 4, c = tid & 15;
+  qb[sw16(row, c)] = r.qv;
+  dob[sw16(row, c)] = r.dov;
+  const int vr = tid >> 2, vc = tid & 3;
+  qtb[sw4(vr, vc)] = r.qtv;
+  dotb[sw4(vr, vc)] = r.dotv;
 }
 
 template <bool CAUSAL>
-__global__ __launch_bounds__(64) void flash_bwd_dkdv_kernel(
+__global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const short* __restrict__ qt, const short* __restrict__ dot,
@@ -66,101 +127,135 @@ __global__ __launch_bounds__(64) void flash_bwd_dkdv_kernel(
     short* __restrict__ dk, short* __restrict__ dv,
     const int B, const int S, const int H, const int Hkv,
     const float scale) {
-  __shared__ __align__(16) short ldsP[T * T];
-  __shared__ __align__(16) short ldsD[T * T];
+  __shared__ lds_chunk qbuf[2][T * 16];
+  __shared__ lds_chunk dobuf[2][T * 16];
+  __shared__ lds_chunk qtbuf[2][D * 4];
+  __shared__ lds_chunk dotbuf[2][D * 4];
+  __shared__ short pscr[NW][T * T];
+  __shared__ short dscr[NW][T * T];
 
-  const int lane = threadIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
   const int col = lane & 31;
   const int half = lane >> 5;
-  const int kv0 = blockIdx.x * T;
+  const int kv0b = blockIdx.x * MT;
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
   const int G = H / Hkv;
-  const int kvrow = kv0 + col;
+  const int my_kv0 = kv0b + wid * T;
+  const int kvrow = my_kv0 + col;
+  const int kvload = min(kvrow, S - 1);
 
-  // ---- preload K and V A-fragments for this kv tile (row = kvrow)
-  bf16x8s kf[8], vf[8];
-  {
-    const long long base =
-        (((long long)b * Hkv + hkv) * S + kvrow) * D + 8 * half;
-#pragma unroll
-    for (int kk = 0; kk < 8; ++kk) {
-      kf[kk] = *(const bf16x8s*)(k + base + kk * 16);
-      vf[kk] = *(const bf16x8s*)(v + base + kk * 16);
-    }
-  }
+  // K/V A-fragments are re-read from global per q-tile (row fixed per
+  // lane -> L1-resident after the first tile). Preloading both in
+  // registers (64 VGPRs) pushed the kernel to 268 B/lane scratch spill.
+  const long long kvbase =
+      (((long long)b * Hkv + hkv) * S + kvload) * D + 8 * half;
 
   f32x16 dvacc[4] = {};  // [kv rows x d cols], col=lane&31 = d_local
   f32x16 dkacc[4] = {};
 
-  for (int g = 0; g < G; ++g) {
+  const int q_start = CAUSAL ? kv0b : 0;  // 32-aligned (S % 32 == 0)
+  const int nq = (S - q_start + T - 1) / T;
+  const int total = G * nq;
+
+  // iteration -> (g, qs, h) helpers
+  auto qb_of = [&](int it, int& g, int& qs, long long& qbase,
+                   long long& tbase, long long& sbase) {
+    g = it / nq;
+    qs = q_start + (it % nq) * T;
     const int h = hkv * G + g;
-    const long long qbase = (((long long)b * H + h) * S) * D;
-    const long long tbase = (((long long)b * H + h) * D) * S;
-    const long long sbase = ((long long)b * H + h) * S;
+    qbase = (((long long)b * H + h) * S) * D;
+    tbase = (((long long)b * H + h) * D) * S;
+    sbase = ((long long)b * H + h) * S;
+  };
 
-    const int q_start = CAUSAL ? kv0 : 0;
-    for (int qs = q_start; qs < S; qs += T) {
-      const int qrow = qs + col;
-      // B-frags of Q and dO for this q tile (col = qrow, k slots along d)
-      bf16x8s qf[8], dof[8];
-#pragma unroll
-      for (int kk = 0; kk < 8; ++kk) {
-        qf[kk] = *(const bf16x8s*)(q + qbase + (long long)qrow * D +
-                                   kk * 16 + 8 * half);
-        dof[kk] = *(const bf16x8s*)(dout + qbase + (long long)qrow * D +
-                                    kk * 16 + 8 * half);
-      }
+  {
+    int g, qs;
+    long long qbase, tbase, sbase;
+    qb_of(0, g, qs, qbase, tbase, sbase);
+    DkdvStage sr = dkdv_load(q, dout, qt, dot, qbase, tbase, qs, S, tid);
+    dkdv_write(sr, qbuf[0], dobuf[0], qtbuf[0], dotbuf[0], tid);
+  }
+  __syncthreads();
 
-      // S^T = K Q^T ; dP^T = V dO^T
+  short* pw = pscr[wid];
+  short* dw = dscr[wid];
+
+  for (int it = 0; it < total; ++it) {
+    const int cur = it & 1;
+    int g, qs;
+    long long qbase, tbase, sbase;
+    qb_of(it, g, qs, qbase, tbase, sbase);
+
+    DkdvStage sr;
+    const bool have_next = (it + 1 < total);
+    if (have_next) {
+      int g2, qs2;
+      long long qb2, tb2, sb2;
+      qb_of(it + 1, g2, qs2, qb2, tb2, sb2);
+      sr = dkdv_load(q, dout, qt, dot, qb2, tb2, qs2, S, tid);
+    }
+
+    if (!CAUSAL || qs + T - 1 >= my_kv0) {
+      const lds_chunk* qb = qbuf[cur];
+      const lds_chunk* dob = dobuf[cur];
+      const lds_chunk* qtb = qtbuf[cur];
+      const lds_chunk* dotb = dotbuf[cur];
+
+      // ---- S^T = K Q^T ; dP^T = V dO^T  (B-frags from LDS, col = q)
       f32x16 st = {}, dpt = {};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kk = 0; kk < 8; ++kk) {
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[kk], qf[kk], st,
-                                                     0, 0, 0);
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[kk], dof[kk], dpt,
+        const bf16x8s kfk = *(const bf16x8s*)(k + kvbase + kk * 16);
+        const bf16x8s vfk = *(const bf16x8s*)(v + kvbase + kk * 16);
+        const bf16x8s qB = qb[sw16(col, kk * 2 + half)];
+        const bf16x8s doB = dob[sw16(col, kk * 2 + half)];
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfk, qB, st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfk, doB, dpt,
                                                       0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
 
+      const int qrow = qs + col;
       const float l = lse[sbase + qrow];
       const float dl = delta[sbase + qrow];
-      float pt[16], dst[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int kvl = cd_row(r, half);
-        float s = st[r] * scale;
-        const bool dead = CAUSAL && (kv0 + kvl > qrow);
-        pt[r] = dead ? 0.f : __expf(s - l);
-        dst[r] = pt[r] * (dpt[r] - dl) * scale;
+        const int kvl = cd_row(r, half);  // kv row within my wave tile
+        const bool dead = (CAUSAL && (my_kv0 + kvl > qrow)) ||
+                          (my_kv0 + kvl >= S);
+        const float pt = dead ? 0.f : __expf(st[r] * scale - l);
+        const float dst = pt * (dpt[r] - dl) * scale;
+        scr_store(pw, kvl, col, f2bf(pt));   // [kv][q]
+        scr_store(dw, kvl, col, f2bf(dst));
       }
 
-      // transpose P^T and dS^T through LDS into A-operand layout
-      __syncthreads();
-      cd_to_lds(pt, ldsP, col, half);
-      cd_to_lds(dst, ldsD, col, half);
-      __syncthreads();
-
-      // dV[kv][d] += P^T(k=q) dO^T-asB ; dK[kv][d] += dS^T(k=q) Q^T-asB
+      // ---- dV[kv][d] += P^T(k=q) dO-asB ; dK[kv][d] += dS^T(k=q) Q-asB
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int dblk = 0; dblk < 4; ++dblk) {
-        const long long trow = (long long)(dblk * 32 + col) * S + qs;
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-          bf16x8s pA = *(const bf16x8s*)(ldsP + (lane & 31) * T + kk * 16 +
-                                         8 * half);
-          bf16x8s dA = *(const bf16x8s*)(ldsD + (lane & 31) * T + kk * 16 +
-                                         8 * half);
-          bf16x8s doB = *(const bf16x8s*)(dot + tbase + trow + kk * 16 +
-                                          8 * half);
-          bf16x8s qB = *(const bf16x8s*)(qt + tbase + trow + kk * 16 +
-                                         8 * half);
+          const bf16x8s pA = scr_read(pw, col, kk, half);
+          const bf16x8s dA = scr_read(dw, col, kk, half);
+          const bf16x8s doB = dotb[sw4(dblk * 32 + col, kk * 2 + half)];
+          const bf16x8s qB2 = qtb[sw4(dblk * 32 + col, kk * 2 + half)];
           dvacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               pA, doB, dvacc[dblk], 0, 0, 0);
           dkacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              dA, qB, dkacc[dblk], 0, 0, 0);
+              dA, qB2, dkacc[dblk], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
+
+    if (have_next)
+      dkdv_write(sr, qbuf[cur ^ 1], dobuf[cur ^ 1], qtbuf[cur ^ 1],
+                 dotbuf[cur ^ 1], tid);
+    __syncthreads();
   }
 
   // ---- write dk/dv (C/D: col = d_local, row = kv via reg map)
@@ -170,7 +265,8 @@ __global__ __launch_bounds__(64) void flash_bwd_dkdv_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kvl = cd_row(r, half);
-      const long long off = obase + (long long)(kv0 + kvl) * D +
+      if (my_kv0 + kvl >= S) continue;
+      const long long off = obase + (long long)(my_kv0 + kvl) * D +
                             dblk * 32 + col;
       dv[off] = f2bf(dvacc[dblk][r]);
       dk[off] = f2bf(dkacc[dblk][r]);
@@ -178,8 +274,41 @@ __global__ __launch_bounds__(64) void flash_bwd_dkdv_kernel(
   }
 }
 
+// ------------------------------------------------------------------- dq
+
+struct DqStage {
+  bf16x8s kv_, vv, ktv;
+};
+
+__device__ __forceinline__ DqStage dq_load(const short* __restrict__ k,
+                                           const short* __restrict__ v,
+                                           const short* __restrict__ kt,
+                                           long long kbase, long long ktbase,
+                                           int kv0, int S, int tid) {
+  DqStage r;
+  const int row = tid >> 4, c = tid & 15;
+  const long long koff = kbase + (long long)min(kv0 + row, S - 1) * D + c * 8;
+  r.kv_ = *(const bf16x8s*)(k + koff);
+  r.vv = *(const bf16x8s*)(v + koff);
+  const int vr = tid >> 2, vc = tid & 3;
+  const long long toff = ktbase + (long long)vr * S +
+                         min(kv0 + vc * 8, S - 8);
+  r.ktv = *(const bf16x8s*)(kt + toff);
+  return r;
+}
+
+__device__ __forceinline__ void dq_write(const DqStage& r, lds_chunk* kb,
+                                         lds_chunk* vb, lds_chunk* ktb,
+                                         int tid) {
+  const int row = tid >> 4, c = tid & 15;
+  kb[sw16(row, c)] = r.kv_;
+  vb[sw16(row, c)] = r.vv;
+  const int vr = tid >> 2, vc = tid & 3;
+  ktb[sw4(vr, vc)] = r.ktv;
+}
+
 template <bool CAUSAL>
-__global__ __launch_bounds__(64) void flash_bwd_dq_kernel(
+__global__ __launch_bounds__(TPB) void flash_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const short* __restrict__ kt,
@@ -187,94 +316,120 @@ __global__ __launch_bounds__(64) void flash_bwd_dq_kernel(
     short* __restrict__ dq,
     const int B, const int S, const int H, const int Hkv,
     const float scale) {
-  __shared__ __align__(16) short ldsD[T * T];
+  __shared__ lds_chunk kbuf[2][T * 16];
+  __shared__ lds_chunk vbuf[2][T * 16];
+  __shared__ lds_chunk ktbuf[2][D * 4];
+  __shared__ short dscr[NW][T * T];
 
-  const int lane = threadIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
   const int col = lane & 31;
   const int half = lane >> 5;
-  const int qs = blockIdx.x * T;
+  const int q0b = blockIdx.x * MT;
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int hkv = h / (H / Hkv);
-  const int qrow = qs + col;
+  const int q0w = q0b + wid * T;
+  const int qrow = q0w + col;
+  const int qload = min(qrow, S - 1);
 
   const long long qbase = (((long long)b * H + h) * S) * D;
   const long long kbase = (((long long)b * Hkv + hkv) * S) * D;
   const long long ktbase = (((long long)b * Hkv + hkv) * D) * S;
   const long long sbase = ((long long)b * H + h) * S;
 
-  // B-frags of Q and dO for my q tile
+  // ---- B-frags of Q and dO for my q rows
   bf16x8s qf[8], dof[8];
 #pragma unroll
   for (int kk = 0; kk < 8; ++kk) {
-    qf[kk] = *(const bf16x8s*)(q + qbase + (long long)qrow * D + kk * 16 +
+    qf[kk] = *(const bf16x8s*)(q + qbase + (long long)qload * D + kk * 16 +
                                8 * half);
-    dof[kk] = *(const bf16x8s*)(dout + qbase + (long long)qrow * D +
+    dof[kk] = *(const bf16x8s*)(dout + qbase + (long long)qload * D +
                                 kk * 16 + 8 * half);
   }
-  const float l = lse[sbase + qrow];
-  const float dl = delta[sbase + qrow];
+  const float l = lse[sbase + qload];
+  const float dl = delta[sbase + qload];
 
   f32x16 dqacc[4] = {};  // rows=q, cols=d (col=lane&31=d_local)
 
-  const int kv_end = CAUSAL ? min(S, qs + T) : S;
-  for (int kv0 = 0; kv0 < kv_end; kv0 += T) {
-    // A-frags of K and V for this kv tile (row = kv0+col)
-    bf16x8s kf[8], vf[8];
-    const long long base = kbase + (long long)(kv0 + col) * D + 8 * half;
-#pragma unroll
-    for (int kk = 0; kk < 8; ++kk) {
-      kf[kk] = *(const bf16x8s*)(k + base + kk * 16);
-      vf[kk] = *(const bf16x8s*)(v + base + kk * 16);
-    }
-    f32x16 st = {}, dpt = {};
-#pragma unroll
-    for (int kk = 0; kk < 8; ++kk) {
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[kk], qf[kk], st,
-                                                   0, 0, 0);
-      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[kk], dof[kk], dpt,
-                                                    0, 0, 0);
-    }
-    float dst[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int kvl = cd_row(r, half);
-      float s = st[r] * scale;
-      const bool dead = CAUSAL && (kv0 + kvl > qrow);
-      const float pt = dead ? 0.f : __expf(s - l);
-      dst[r] = pt * (dpt[r] - dl) * scale;
-    }
+  const int kv_end_blk = CAUSAL ? min(S, q0b + MT) : S;
+  const int kv_end_wave = CAUSAL ? min(S, q0w + T) : S;
+  const int nt = (kv_end_blk + T - 1) / T;
 
-    // dS^T (rows kv, cols q) -> LDS -> read as dS A-frags (rows q, k=kv):
-    // store transposed: lds[q][kv] = dst  (q = col, kv = cd_row)
-    __syncthreads();
-#pragma unroll
-    for (int r = 0; r < 16; ++r)
-      ldsD[col * T + cd_row(r, half)] = f2bf(dst[r]);
-    __syncthreads();
+  {
+    DqStage sr = dq_load(k, v, kt, kbase, ktbase, 0, S, tid);
+    dq_write(sr, kbuf[0], vbuf[0], ktbuf[0], tid);
+  }
+  __syncthreads();
 
-    // dQ[q][d] += dS(row=q, k=kv) · K^T-asB(k=kv, col=d)
+  short* dw = dscr[wid];
+
+  for (int it = 0; it < nt; ++it) {
+    const int kv0 = it * T;
+    const int cur = it & 1;
+    DqStage sr;
+    const bool have_next = (it + 1 < nt);
+    if (have_next)
+      sr = dq_load(k, v, kt, kbase, ktbase, kv0 + T, S, tid);
+
+    if (kv0 < kv_end_wave) {
+      const lds_chunk* kb = kbuf[cur];
+      const lds_chunk* vb = vbuf[cur];
+      const lds_chunk* ktb = ktbuf[cur];
+
+      f32x16 st = {}, dpt = {};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int dblk = 0; dblk < 4; ++dblk) {
-      const long long trow = ktbase + (long long)(dblk * 32 + col) * S + kv0;
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-        bf16x8s dA = *(const bf16x8s*)(ldsD + (lane & 31) * T + kk * 16 +
-                                       8 * half);
-        bf16x8s kB = *(const bf16x8s*)(kt + trow + kk * 16 + 8 * half);
-        dqacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            dA, kB, dqacc[dblk], 0, 0, 0);
+      for (int kk = 0; kk < 8; ++kk) {
+        const bf16x8s kA = kb[sw16(col, kk * 2 + half)];
+        const bf16x8s vA = vb[sw16(col, kk * 2 + half)];
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kA, qf[kk], st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vA, dof[kk], dpt,
+                                                      0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
+
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kvl = cd_row(r, half);
+        const bool dead = (CAUSAL && (kv0 + kvl > qrow)) ||
+                          (kv0 + kvl >= S);
+        const float pt = dead ? 0.f : __expf(st[r] * scale - l);
+        const float dst = pt * (dpt[r] - dl) * scale;
+        // store TRANSPOSED: [q][kv] so dS reads as A-frags (rows q)
+        scr_store(dw, col, kvl, f2bf(dst));
+      }
+
+      // ---- dQ[q][d] += dS(row=q, k=kv) · K^T-asB(k=kv, col=d)
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dblk = 0; dblk < 4; ++dblk) {
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const bf16x8s dA = scr_read(dw, col, kk, half);
+          const bf16x8s kB = ktb[sw4(dblk * 32 + col, kk * 2 + half)];
+          dqacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              dA, kB, dqacc[dblk], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
+
+    if (have_next)
+      dq_write(sr, kbuf[cur ^ 1], vbuf[cur ^ 1], ktbuf[cur ^ 1], tid);
+    __syncthreads();
   }
 
-  // write dq (rows=q via reg map, col=d_local)
+  // ---- write dq (rows=q via reg map, col=d_local)
+  if (q0w >= S) return;
 #pragma unroll
   for (int dblk = 0; dblk < 4; ++dblk) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int ql = cd_row(r, half);
-      dq[qbase + (long long)(qs + ql) * D + dblk * 32 + col] =
+      if (q0w + ql >= S) continue;
+      dq[qbase + (long long)(q0w + ql) * D + dblk * 32 + col] =
           f2bf(dqacc[dblk][r]);
     }
   }
@@ -289,25 +444,25 @@ extern "C" void ds_flash_bwd(const void* q, const void* k, const void* v,
                              void* dk, void* dv, int B, int S, int H,
                              int Hkv, float scale, int causal, void* stream) {
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
-  dim3 g1((S + T - 1) / T, Hkv, B);
-  dim3 g2((S + T - 1) / T, H, B);
+  dim3 g1((S + MT - 1) / MT, Hkv, B);
+  dim3 g2((S + MT - 1) / MT, H, B);
   if (causal) {
-    hipLaunchKernelGGL((flash_bwd_dkdv_kernel<true>), g1, dim3(64), 0, st,
+    hipLaunchKernelGGL((flash_bwd_dkdv_kernel<true>), g1, dim3(TPB), 0, st,
                        (const short*)q, (const short*)k, (const short*)v,
                        (const short*)dout, (const short*)qt,
                        (const short*)dot, lse, delta, (short*)dk, (short*)dv,
                        B, S, H, Hkv, scale);
-    hipLaunchKernelGGL((flash_bwd_dq_kernel<true>), g2, dim3(64), 0, st,
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<true>), g2, dim3(TPB), 0, st,
                        (const short*)q, (const short*)k, (const short*)v,
                        (const short*)dout, (const short*)kt, lse, delta,
                        (short*)dq, B, S, H, Hkv, scale);
   } else {
-    hipLaunchKernelGGL((flash_bwd_dkdv_kernel<false>), g1, dim3(64), 0, st,
+    hipLaunchKernelGGL((flash_bwd_dkdv_kernel<false>), g1, dim3(TPB), 0, st,
                        (const short*)q, (const short*)k, (const short*)v,
                        (const short*)dout, (const short*)qt,
                        (const short*)dot, lse, delta, (short*)dk, (short*)dv,
                        B, S, H, Hkv, scale);
-    hipLaunchKernelGGL((flash_bwd_dq_kernel<false>), g2, dim3(64), 0, st,
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<false>), g2, dim3(TPB), 0, st,
                        (const short*)q, (const short*)k, (const short*)v,
                        (const short*)dout, (const short*)kt, lse, delta,
                        (short*)dq, B, S, H, Hkv, scale);
