@@ -109,6 +109,9 @@ class RMSNorm(torch.nn.Module):
             torch.ones(hidden_size, dtype=dtype, device=device))
         self.eps = eps
 
+    def reset_parameters(self):
+        torch.nn.init.ones_(self.weight)
+
     def forward(self, x):
         return _RMSNormFn.apply(x, self.weight, self.eps)
 
@@ -124,6 +127,11 @@ class FusedLayerNorm(torch.nn.Module):
         self.bias = torch.nn.Parameter(
             torch.zeros(hidden_size, dtype=dtype, device=device)) if bias else None
         self.eps = eps
+
+    def reset_parameters(self):
+        torch.nn.init.ones_(self.weight)
+        if self.bias is not None:
+            torch.nn.init.zeros_(self.bias)
 
     def forward(self, x):
         return _LayerNormFn.apply(x, self.weight, self.bias, self.eps)

@@ -135,8 +135,11 @@ class Engine(torch.nn.Module):
             self._broadcast_model()
         else:
             # ZeRO-3: params may already be partitioned (zero.Init); the
-            # stage-3 optimizer handles placement and bcast.
-            if not dont_change_device:
+            # stage-3 optimizer handles placement and bcast. Meta-init'd
+            # modules (zero.Init(remote_device="meta")) must not be moved —
+            # stage 3 materializes them unit by unit at partition time.
+            if not dont_change_device and \
+                    not any(p.is_meta for p in self.module.parameters()):
                 self.module.to(self.device)
 
     def _broadcast_model(self):

@@ -34,13 +34,33 @@ def register_stage3(opt):
 
 
 class Init:
+    """Context for constructing models under ZeRO-3.
+
+    Two modes:
+    * default: registered floating params are cast to ``dtype`` the moment
+      they register (keeps the transient host copy small; partitioning
+      happens at engine construction — the 288 GB HBM design stance).
+    * ``remote_device="meta"``: params are registered on the META device —
+      NO weight memory is allocated anywhere at construction, so models
+      whose full weights exceed device+host memory can still be built.
+      The stage-3 optimizer materializes each module unit transiently at
+      partition time (``reset_parameters()`` per module, rank-0 broadcast)
+      and keeps only the local shard. Reference analogue:
+      partition_parameters.py Init :824 (construction-time sharding) +
+      remote_device handling :1551.
+    """
+
     def __init__(self, module=None, config=None, dtype=None, enabled=True,
-                 **_ignored):
+                 remote_device=None, meta_device=False, **_ignored):
         self.dtype = dtype or torch.bfloat16
         self.enabled = enabled
+        self.meta = meta_device or remote_device == "meta"
         self._orig_init = None
         if module is not None:  # eager form: cast an existing module
-            module.to(self.dtype)
+            if self.meta:
+                module.to_empty(device="meta")
+            else:
+                module.to(self.dtype)
 
     def __enter__(self):
         if not self.enabled:
@@ -53,18 +73,32 @@ class Init:
         self._orig_register = nn.Module.register_parameter
         target_dtype = self.dtype
 
+        meta = self.meta
+
         @functools.wraps(self._orig_setattr)
         def wrapped_setattr(mod, name, value):
             if isinstance(value, nn.Parameter) and value.is_floating_point() \
                     and not value.is_meta:
-                value.data = value.data.to(target_dtype)
+                if meta:
+                    value = nn.Parameter(
+                        torch.empty(value.shape, dtype=target_dtype,
+                                    device="meta"),
+                        requires_grad=value.requires_grad)
+                else:
+                    value.data = value.data.to(target_dtype)
             self._orig_setattr(mod, name, value)
 
         @functools.wraps(self._orig_register)
         def wrapped_register(mod, name, param):
             if isinstance(param, nn.Parameter) and \
                     param.is_floating_point() and not param.is_meta:
-                param.data = param.data.to(target_dtype)
+                if meta:
+                    param = nn.Parameter(
+                        torch.empty(param.shape, dtype=target_dtype,
+                                    device="meta"),
+                        requires_grad=param.requires_grad)
+                else:
+                    param.data = param.data.to(target_dtype)
             self._orig_register(mod, name, param)
 
         nn.Module.__setattr__ = wrapped_setattr

@@ -312,7 +312,31 @@ class ZeroStage3Optimizer:
         dtype = self._config_dtype or self.units[0].params[0].dtype
         self._dtype = dtype
         self._empty = torch.empty(0, dtype=dtype, device=dev)
+        warned_no_reset = False
         for u in self.units:
+            if any(p.data.is_meta for p in u.params):
+                # zero.Init(remote_device="meta"): the unit was never
+                # allocated anywhere. Materialize it transiently on the
+                # device, run the module's own reset_parameters (rank-0's
+                # draw wins via the broadcast below), shard, free — peak
+                # memory is ONE module unit, so models larger than
+                # device+host memory can be constructed and trained.
+                for p in u.params:
+                    # set_data cannot cross meta->real; swap_tensors keeps
+                    # the Parameter object identity (optimizer groups, unit
+                    # lists) while materializing storage
+                    new = torch.empty(p.shape, dtype=p.dtype, device=dev)
+                    torch.utils.swap_tensors(p, torch.nn.Parameter(
+                        new, requires_grad=p.requires_grad))
+                if hasattr(u.module, "reset_parameters"):
+                    u.module.reset_parameters()
+                elif not warned_no_reset:
+                    from ...utils.logging import logger
+                    logger.warning(
+                        f"meta-init module '{u.name}' has no "
+                        "reset_parameters(); its weights start ZEROED — "
+                        "load a checkpoint or pass an init_fn")
+                    warned_no_reset = True
             full = torch.empty(u.numel, dtype=dtype, device=dev)
             for p, off in zip(u.params, u.offsets):
                 full[off:off + p.numel()].copy_(p.data.reshape(-1).to(dev, dtype))
@@ -1189,6 +1213,7 @@ class ZeroStage3Optimizer:
             "stage": 3,
             "world_size": self.world_size,
             "rank": self.rank,
+            "dtype": str(self._dtype),
             "loss_scaler": self.loss_scaler.state_dict(),
             "fp32_flat_groups": [m.data if m.numel() else m
                                  for m in self.group_masters],
@@ -1268,6 +1293,12 @@ class ZeroStage3Optimizer:
     def load_state_dict(self, sd, load_optimizer_states=True):
         assert sd["world_size"] == self.world_size, \
             "ZeRO-3 checkpoint reshaping requires the universal checkpoint path"
+        ck_dtype = sd.get("dtype")
+        if ck_dtype is not None and ck_dtype != str(self._dtype):
+            raise ValueError(
+                f"ZeRO-3 checkpoint was saved by a {ck_dtype} engine but "
+                f"this engine runs {self._dtype}: the flat master layouts "
+                "differ — convert through the universal checkpoint instead")
         self.loss_scaler.load_state_dict(sd["loss_scaler"])
         if self.nvme_offload:
             for gi, flat in enumerate(sd["fp32_flat_groups"]):
