@@ -118,7 +118,9 @@ __device__ __forceinline__ void dkdv_write(const DkdvStage& r,
   dotb[sw4(vr, vc)] = r.dotv;
 }
 
-template <bool CAUSAL>
+// VAR ablation (probe-only): bit0 = skip compute, bit1 = skip scr+outer
+// mfma (accumulate st/dpt junk to stay live), bit2 = skip staging
+template <bool CAUSAL, int VAR = 0>
 __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -190,7 +192,7 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     qb_of(it, g, qs, qbase, tbase, sbase);
 
     DkdvStage sr;
-    const bool have_next = (it + 1 < total);
+    const bool have_next = (it + 1 < total) && !(VAR & 4);
     if (have_next) {
       int g2, qs2;
       long long qb2, tb2, sb2;
@@ -198,7 +200,7 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
       sr = dkdv_load(q, dout, qt, dot, qb2, tb2, qs2, S, tid);
     }
 
-    if (!CAUSAL || qs + T - 1 >= my_kv0) {
+    if (!(VAR & 1) && (!CAUSAL || qs + T - 1 >= my_kv0)) {
       const lds_chunk* qb = qbuf[cur];
       const lds_chunk* dob = dobuf[cur];
       const lds_chunk* qtb = qtbuf[cur];
@@ -222,34 +224,42 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
       const int qrow = qs + col;
       const float l = lse[sbase + qrow];
       const float dl = delta[sbase + qrow];
+      if (VAR & 2) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kvl = cd_row(r, half);  // kv row within my wave tile
-        const bool dead = (CAUSAL && (my_kv0 + kvl > qrow)) ||
-                          (my_kv0 + kvl >= S);
-        const float pt = dead ? 0.f : __expf(st[r] * scale - l);
-        const float dst = pt * (dpt[r] - dl) * scale;
-        scr_store(pw, kvl, col, f2bf(pt));   // [kv][q]
-        scr_store(dw, kvl, col, f2bf(dst));
-      }
-
-      // ---- dV[kv][d] += P^T(k=q) dO-asB ; dK[kv][d] += dS^T(k=q) Q-asB
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int dblk = 0; dblk < 4; ++dblk) {
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk) {
-          const bf16x8s pA = scr_read(pw, col, kk, half);
-          const bf16x8s dA = scr_read(dw, col, kk, half);
-          const bf16x8s doB = dotb[sw4(dblk * 32 + col, kk * 2 + half)];
-          const bf16x8s qB2 = qtb[sw4(dblk * 32 + col, kk * 2 + half)];
-          dvacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pA, doB, dvacc[dblk], 0, 0, 0);
-          dkacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              dA, qB2, dkacc[dblk], 0, 0, 0);
+        for (int r = 0; r < 16; ++r) {  // keep st/dpt live, skip the rest
+          dvacc[0][r] += st[r] * l;
+          dkacc[0][r] += dpt[r] * dl;
         }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvl = cd_row(r, half);  // kv row within my wave tile
+          const bool dead = (CAUSAL && (my_kv0 + kvl > qrow)) ||
+                            (my_kv0 + kvl >= S);
+          const float pt = dead ? 0.f : __expf(st[r] * scale - l);
+          const float dst = pt * (dpt[r] - dl) * scale;
+          scr_store(pw, kvl, col, f2bf(pt));   // [kv][q]
+          scr_store(dw, kvl, col, f2bf(dst));
+        }
+
+        // ---- dV[kv][d] += P^T(k=q) dO-asB ; dK[kv][d] += dS^T(k=q) Q-asB
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int dblk = 0; dblk < 4; ++dblk) {
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            const bf16x8s pA = scr_read(pw, col, kk, half);
+            const bf16x8s dA = scr_read(dw, col, kk, half);
+            const bf16x8s doB = dotb[sw4(dblk * 32 + col, kk * 2 + half)];
+            const bf16x8s qB2 = qtb[sw4(dblk * 32 + col, kk * 2 + half)];
+            dvacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pA, doB, dvacc[dblk], 0, 0, 0);
+            dkacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                dA, qB2, dkacc[dblk], 0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
 
     if (have_next)
@@ -435,7 +445,31 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dq_kernel(
   }
 }
 
-}  // namespace
+// probe-only ablation entry (scripts/bwd_ablate.hip)
+extern "C" void ds_flash_bwd_dkdv_dbg(const void* q, const void* k,
+                                      const void* v, const void* dout,
+                                      const void* qt, const void* dot,
+                                      const float* lse, const float* delta,
+                                      void* dk, void* dv, int B, int S,
+                                      int H, int Hkv, float scale,
+                                      int variant, void* stream) {
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  dim3 g1((S + MT - 1) / MT, Hkv, B);
+#define L(V) hipLaunchKernelGGL((flash_bwd_dkdv_kernel<true, V>), g1, \
+      dim3(TPB), 0, st, (const short*)q, (const short*)k, (const short*)v, \
+      (const short*)dout, (const short*)qt, (const short*)dot, lse, delta, \
+      (short*)dk, (short*)dv, B, S, H, Hkv, scale)
+  switch (variant) {
+    case 1: L(1); break;
+    case 2: L(2); break;
+    case 4: L(4); break;
+    case 5: L(5); break;
+    default: L(0);
+  }
+#undef L
+}
+
+}  // namespace probe internal end marker (kernels above)
 
 extern "C" void ds_flash_bwd(const void* q, const void* k, const void* v,
                              const void* dout, const void* qt,
