@@ -103,3 +103,19 @@ def test_universal_checkpoint_dp4_to_dp2(tmp_path):
     from deepspeed_amd.checkpoint import ds_to_universal
     ds_to_universal(tmp, tag="step2")
     run_distributed(_resume_worker, world_size=2, args=(tmp, 1))
+
+
+import pytest
+
+
+@pytest.mark.parametrize("stage", [1, 2, 3])
+@pytest.mark.parametrize("save_ws,load_ws", [(4, 2), (2, 4)])
+def test_universal_reshape_matrix(tmp_path, stage, save_ws, load_ws):
+    """Cross-world-size reshaping matrix, shrink AND grow, for every ZeRO
+    stage including optimizer state (reference DistributedFixture pattern,
+    tests/unit/common.py:354 + tests/unit/checkpoint/)."""
+    tmp = str(tmp_path)
+    run_distributed(_save_worker, world_size=save_ws, args=(tmp, stage))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="step2")
+    run_distributed(_resume_worker, world_size=load_ws, args=(tmp, stage))
