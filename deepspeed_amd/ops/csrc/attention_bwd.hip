@@ -86,36 +86,31 @@ __device__ __forceinline__ bf16x8s scr_read(const short* scr, int row,
 
 // ---------------------------------------------------------------- dk/dv
 
-struct DkdvStage {
-  bf16x8s qv, dov, qtv, dotv;
-};
-
-__device__ __forceinline__ DkdvStage dkdv_load(
+// Staging via global_load_lds: the LDS destination is linear
+// (wave-uniform base + lane*16, the hardware contract), and the SWIZZLE is
+// folded into the per-lane GLOBAL source address (inverse permutation) —
+// zero staging registers, fully async (vmcnt-counted, drained by the
+// barrier). Each thread issues one 16 B chunk per tile kind.
+__device__ __forceinline__ void dkdv_stage(
     const short* __restrict__ q, const short* __restrict__ dout,
     const short* __restrict__ qt, const short* __restrict__ dot,
+    lds_chunk* qb, lds_chunk* dob, lds_chunk* qtb, lds_chunk* dotb,
     long long qbase, long long tbase, int qs, int S, int tid) {
-  DkdvStage r;
-  const int row = tid >> 4, c = tid & 15;      // [32][128] tiles
+  const int j = tid;              // linear LDS chunk slot 0..511
+  // [32][128] tiles (16 chunks/row): slot j holds (row=j>>4, c=(j&15)^(row&7))
+  const int row = j >> 4, c = (j & 15) ^ (row & 7);
   const long long qoff = qbase + (long long)(qs + row) * D + c * 8;
-  r.qv = *(const bf16x8s*)(q + qoff);
-  r.dov = *(const bf16x8s*)(dout + qoff);
-  const int vr = tid >> 2, vc = tid & 3;       // [128][32] tiles
+  __builtin_amdgcn_global_load_lds((const unsigned int*)(q + qoff),
+                                   (unsigned int*)(qb + j), 16, 0, 0);
+  __builtin_amdgcn_global_load_lds((const unsigned int*)(dout + qoff),
+                                   (unsigned int*)(dob + j), 16, 0, 0);
+  // [128][32] tiles (4 chunks/row): slot j holds (vr=j>>2, vc=(j&3)^(vr&3))
+  const int vr = j >> 2, vc = (j & 3) ^ (vr & 3);
   const long long toff = tbase + (long long)vr * S + qs + vc * 8;
-  r.qtv = *(const bf16x8s*)(qt + toff);
-  r.dotv = *(const bf16x8s*)(dot + toff);
-  return r;
-}
-
-__device__ __forceinline__ void dkdv_write(const DkdvStage& r,
-                                           lds_chunk* qb, lds_chunk* dob,
-                                           lds_chunk* qtb, lds_chunk* dotb,
-                                           int tid) {
-  const int row = tid >> 4, c = tid & 15;
-  qb[sw16(row, c)] = r.qv;
-  dob[sw16(row, c)] = r.dov;
-  const int vr = tid >> 2, vc = tid & 3;
-  qtb[sw4(vr, vc)] = r.qtv;
-  dotb[sw4(vr, vc)] = r.dotv;
+  __builtin_amdgcn_global_load_lds((const unsigned int*)(qt + toff),
+                                   (unsigned int*)(qtb + j), 16, 0, 0);
+  __builtin_amdgcn_global_load_lds((const unsigned int*)(dot + toff),
+                                   (unsigned int*)(dotb + j), 16, 0, 0);
 }
 
 // VAR ablation (probe-only): bit0 = skip compute, bit1 = skip scr+outer
@@ -149,11 +144,17 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
   const int kvrow = my_kv0 + col;
   const int kvload = min(kvrow, S - 1);
 
-  // K/V A-fragments are re-read from global per q-tile (row fixed per
-  // lane -> L1-resident after the first tile). Preloading both in
-  // registers (64 VGPRs) pushed the kernel to 268 B/lane scratch spill.
+  // K A-fragments preload in registers; V fragments are issued at the top
+  // of each iteration and consumed AFTER the 8 st-MFMAs (their HBM/L1
+  // latency hides under the QK cluster). The v2.0 design re-read BOTH from
+  // global inside the mfma loop — the dbg ablation (r2_call6) showed that
+  // serial load->mfma chain was 13.3 of 17 ms.
   const long long kvbase =
       (((long long)b * Hkv + hkv) * S + kvload) * D + 8 * half;
+  bf16x8s kf[8];
+#pragma unroll
+  for (int kk = 0; kk < 8; ++kk)
+    kf[kk] = *(const bf16x8s*)(k + kvbase + kk * 16);
 
   f32x16 dvacc[4] = {};  // [kv rows x d cols], col=lane&31 = d_local
   f32x16 dkacc[4] = {};
@@ -177,8 +178,8 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     int g, qs;
     long long qbase, tbase, sbase;
     qb_of(0, g, qs, qbase, tbase, sbase);
-    DkdvStage sr = dkdv_load(q, dout, qt, dot, qbase, tbase, qs, S, tid);
-    dkdv_write(sr, qbuf[0], dobuf[0], qtbuf[0], dotbuf[0], tid);
+    dkdv_stage(q, dout, qt, dot, qbuf[0], dobuf[0], qtbuf[0], dotbuf[0],
+               qbase, tbase, qs, S, tid);
   }
   __syncthreads();
 
@@ -191,13 +192,13 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     long long qbase, tbase, sbase;
     qb_of(it, g, qs, qbase, tbase, sbase);
 
-    DkdvStage sr;
     const bool have_next = (it + 1 < total) && !(VAR & 4);
     if (have_next) {
       int g2, qs2;
       long long qb2, tb2, sb2;
       qb_of(it + 1, g2, qs2, qb2, tb2, sb2);
-      sr = dkdv_load(q, dout, qt, dot, qb2, tb2, qs2, S, tid);
+      dkdv_stage(q, dout, qt, dot, qbuf[cur ^ 1], dobuf[cur ^ 1],
+                 qtbuf[cur ^ 1], dotbuf[cur ^ 1], qb2, tb2, qs2, S, tid);
     }
 
     if (!(VAR & 1) && (!CAUSAL || qs + T - 1 >= my_kv0)) {
@@ -207,16 +208,23 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
       const lds_chunk* dotb = dotbuf[cur];
 
       // ---- S^T = K Q^T ; dP^T = V dO^T  (B-frags from LDS, col = q)
+      // issue V loads first: 8 independent global loads, then the st
+      // cluster covers their latency before dpt consumes them
+      bf16x8s vfk[8];
+#pragma unroll
+      for (int kk = 0; kk < 8; ++kk)
+        vfk[kk] = *(const bf16x8s*)(v + kvbase + kk * 16);
       f32x16 st = {}, dpt = {};
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kk = 0; kk < 8; ++kk) {
-        const bf16x8s kfk = *(const bf16x8s*)(k + kvbase + kk * 16);
-        const bf16x8s vfk = *(const bf16x8s*)(v + kvbase + kk * 16);
         const bf16x8s qB = qb[sw16(col, kk * 2 + half)];
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[kk], qB, st, 0, 0, 0);
+      }
+#pragma unroll
+      for (int kk = 0; kk < 8; ++kk) {
         const bf16x8s doB = dob[sw16(col, kk * 2 + half)];
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfk, qB, st, 0, 0, 0);
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfk, doB, dpt,
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfk[kk], doB, dpt,
                                                       0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
@@ -262,9 +270,6 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
       }
     }
 
-    if (have_next)
-      dkdv_write(sr, qbuf[cur ^ 1], dobuf[cur ^ 1], qtbuf[cur ^ 1],
-                 dotbuf[cur ^ 1], tid);
     __syncthreads();
   }
 

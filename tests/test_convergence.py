@@ -1,0 +1,82 @@
+"""Convergence parity: the engine's real loss curve must track plain torch
+DDP-style training (same data, same init, fp32 AdamW reference) within
+dtype tolerance over a meaningful number of steps — the reference's core
+correctness pattern (tests/unit/runtime/zero/test_zero.py)."""
+
+import copy
+
+import pytest
+import torch
+
+from .common import run_distributed
+
+STEPS = 20
+
+
+def _curve_worker(rank, world, stage):
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_mini
+
+    cfg = llama_mini()
+    torch.manual_seed(5)
+    model = LlamaForCausalLM(cfg)
+    ref = copy.deepcopy(model)
+
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "bf16": {"enabled": True},
+        "zero_optimization": {"stage": stage, "overlap_comm": True},
+        "optimizer": {"type": "AdamW", "params": {"lr": 3e-4}}})
+
+    # fp32 DDP reference: grads averaged over all ranks' batches
+    opt_ref = torch.optim.AdamW(ref.parameters(), lr=3e-4)
+
+    g = torch.Generator().manual_seed(123)
+    # fixed batches -> the loss curve must actually descend (memorization)
+    batches = [torch.randint(0, cfg.vocab_size, (2, 64), generator=g)
+               for _ in range(world)]
+    eng_curve, ref_curve = [], []
+    for _ in range(STEPS):
+        ids = batches[rank]
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+        # whole-job mean loss for the curve
+        t = loss.detach().clone().float()
+        deepspeed_amd.comm.all_reduce(t)
+        eng_curve.append(t.item() / world)
+
+        tot = 0.0
+        for b in batches:
+            l2 = ref(b, labels=b)
+            (l2 / world).backward()
+            tot += l2.item() / world
+        opt_ref.step()
+        opt_ref.zero_grad()
+        ref_curve.append(tot)
+
+    eng = torch.tensor(eng_curve)
+    reft = torch.tensor(ref_curve)
+    # the curves track point by point while the loss is in a meaningful
+    # range (the memorization tail -> 0 amplifies bf16-vs-fp32 relative
+    # noise, so the tail is held only to convergence, not parity)
+    assert torch.allclose(eng[:8], reft[:8], rtol=0.05, atol=0.05), \
+        (eng_curve, ref_curve)
+    # and training actually converges (fixed batches memorize fast)
+    assert eng_curve[-1] < 0.2 and ref_curve[-1] < 0.2, \
+        (eng_curve, ref_curve)
+
+
+@pytest.mark.parametrize("stage", [2, 3])
+def test_convergence_parity_ws2(stage):
+    run_distributed(_curve_worker, world_size=2, args=(stage,), timeout=600)
+
+
+def test_convergence_parity_ws8():
+    run_distributed(_curve_worker, world_size=8, args=(2,), timeout=900)
+
+
+@pytest.mark.gpu
+def test_convergence_parity_gpu():
+    """Single-rank GPU curve vs fp32 torch (HIP kernels on the hot path)."""
+    run_distributed(_curve_worker, world_size=1, args=(3,), timeout=600)
