@@ -113,8 +113,16 @@ __device__ __forceinline__ void dkdv_stage(
                                    (unsigned int*)(dotb + j), 16, 0, 0);
 }
 
-// VAR ablation (probe-only): bit0 = skip compute, bit1 = skip scr+outer
-// mfma (accumulate st/dpt junk to stay live), bit2 = skip staging
+// dkdv v3 geometry: KV macro-tile = 128 rows = 4 wave-PAIRS x 32 rows.
+// Both waves of a pair cover the same 32 kv rows (K/V staged once per
+// pair in LDS, read as A-fragments like the dq kernel) and split the d
+// dimension: wave pm=0 accumulates dblk 0-1, pm=1 dblk 2-3 — so the
+// dk/dv accumulators are 64 VGPRs per wave instead of 128 (the
+// register-resident variants spilled 100-268 B/lane and the per-
+// iteration global A-fragment loads serialized against the MFMAs,
+// r2_call6 ablation).
+constexpr int MTK = 128;  // kv rows per workgroup
+
 template <bool CAUSAL, int VAR = 0>
 __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
@@ -124,49 +132,59 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     short* __restrict__ dk, short* __restrict__ dv,
     const int B, const int S, const int H, const int Hkv,
     const float scale) {
-  __shared__ lds_chunk qbuf[2][T * 16];
-  __shared__ lds_chunk dobuf[2][T * 16];
-  __shared__ lds_chunk qtbuf[2][D * 4];
-  __shared__ lds_chunk dotbuf[2][D * 4];
-  __shared__ short pscr[NW][T * T];
-  __shared__ short dscr[NW][T * T];
+  __shared__ lds_chunk kscr[4][T * 16];  // K [32][128] per pair (32 KB)
+  __shared__ lds_chunk vscr[4][T * 16];  // V (32 KB)
+  __shared__ lds_chunk qbuf[T * 16];     // single-buffered q tiles (32 KB)
+  __shared__ lds_chunk dobuf[T * 16];
+  __shared__ lds_chunk qtbuf[D * 4];
+  __shared__ lds_chunk dotbuf[D * 4];
+  __shared__ short pscr[NW][T * T];      // per-wave transpose scratch
+  __shared__ short dscr[NW][T * T];      // (32 KB)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
+  const int pair = wid >> 1;
+  const int pm = wid & 1;                // my d half: dblk = pm*2 + {0,1}
   const int col = lane & 31;
   const int half = lane >> 5;
-  const int kv0b = blockIdx.x * MT;
+  const int kv0b = blockIdx.x * MTK;
   const int hkv = blockIdx.y;
   const int b = blockIdx.z;
   const int G = H / Hkv;
-  const int my_kv0 = kv0b + wid * T;
+  const int my_kv0 = kv0b + pair * T;
   const int kvrow = my_kv0 + col;
-  const int kvload = min(kvrow, S - 1);
 
-  // K A-fragments preload in registers; V fragments are issued at the top
-  // of each iteration and consumed AFTER the 8 st-MFMAs (their HBM/L1
-  // latency hides under the QK cluster). The v2.0 design re-read BOTH from
-  // global inside the mfma loop — the dbg ablation (r2_call6) showed that
-  // serial load->mfma chain was 13.3 of 17 ms.
-  const long long kvbase =
-      (((long long)b * Hkv + hkv) * S + kvload) * D + 8 * half;
-  bf16x8s kf[8];
+  // ---- stage K/V for all 4 pairs (linear dst, pre-swizzled src)
+  {
+    const long long kb0 = (((long long)b * Hkv + hkv) * S) * D;
+    const int HkvD = Hkv * D;
 #pragma unroll
-  for (int kk = 0; kk < 8; ++kk)
-    kf[kk] = *(const bf16x8s*)(k + kvbase + kk * 16);
+    for (int t = 0; t < 4; ++t) {
+      const int j = tid + t * TPB;       // 0..2047
+      const int p = j >> 9, jj = j & 511;
+      const int row = jj >> 4, c = (jj & 15) ^ (row & 7);
+      const long long off =
+          kb0 + (long long)min(kv0b + p * T + row, S - 1) * HkvD + c * 8;
+      __builtin_amdgcn_global_load_lds((const unsigned int*)(k + off),
+                                       (unsigned int*)(&kscr[0][0] + j),
+                                       16, 0, 0);
+      __builtin_amdgcn_global_load_lds((const unsigned int*)(v + off),
+                                       (unsigned int*)(&vscr[0][0] + j),
+                                       16, 0, 0);
+    }
+  }
 
-  f32x16 dvacc[4] = {};  // [kv rows x d cols], col=lane&31 = d_local
-  f32x16 dkacc[4] = {};
+  f32x16 dvacc[2] = {};  // my two 32-wide d blocks
+  f32x16 dkacc[2] = {};
 
   const int q_start = CAUSAL ? kv0b : 0;  // 32-aligned (S % 32 == 0)
   const int nq = (S - q_start + T - 1) / T;
   const int total = G * nq;
 
-  // iteration -> (g, qs, h) helpers
-  auto qb_of = [&](int it, int& g, int& qs, long long& qbase,
-                   long long& tbase, long long& sbase) {
-    g = it / nq;
+  auto qb_of = [&](int it, int& qs, long long& qbase, long long& tbase,
+                   long long& sbase) {
+    const int g = it / nq;
     qs = q_start + (it % nq) * T;
     const int h = hkv * G + g;
     qbase = (((long long)b * H + h) * S) * D;
@@ -175,57 +193,39 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
   };
 
   {
-    int g, qs;
+    int qs;
     long long qbase, tbase, sbase;
-    qb_of(0, g, qs, qbase, tbase, sbase);
-    dkdv_stage(q, dout, qt, dot, qbuf[0], dobuf[0], qtbuf[0], dotbuf[0],
+    qb_of(0, qs, qbase, tbase, sbase);
+    dkdv_stage(q, dout, qt, dot, qbuf, dobuf, qtbuf, dotbuf,
                qbase, tbase, qs, S, tid);
   }
   __syncthreads();
 
   short* pw = pscr[wid];
   short* dw = dscr[wid];
+  const lds_chunk* kp = kscr[pair];
+  const lds_chunk* vp = vscr[pair];
 
   for (int it = 0; it < total; ++it) {
-    const int cur = it & 1;
-    int g, qs;
+    int qs;
     long long qbase, tbase, sbase;
-    qb_of(it, g, qs, qbase, tbase, sbase);
-
-    const bool have_next = (it + 1 < total) && !(VAR & 4);
-    if (have_next) {
-      int g2, qs2;
-      long long qb2, tb2, sb2;
-      qb_of(it + 1, g2, qs2, qb2, tb2, sb2);
-      dkdv_stage(q, dout, qt, dot, qbuf[cur ^ 1], dobuf[cur ^ 1],
-                 qtbuf[cur ^ 1], dotbuf[cur ^ 1], qb2, tb2, qs2, S, tid);
-    }
+    qb_of(it, qs, qbase, tbase, sbase);
 
     if (!(VAR & 1) && (!CAUSAL || qs + T - 1 >= my_kv0)) {
-      const lds_chunk* qb = qbuf[cur];
-      const lds_chunk* dob = dobuf[cur];
-      const lds_chunk* qtb = qtbuf[cur];
-      const lds_chunk* dotb = dotbuf[cur];
-
-      // ---- S^T = K Q^T ; dP^T = V dO^T  (B-frags from LDS, col = q)
-      // issue V loads first: 8 independent global loads, then the st
-      // cluster covers their latency before dpt consumes them
-      bf16x8s vfk[8];
-#pragma unroll
-      for (int kk = 0; kk < 8; ++kk)
-        vfk[kk] = *(const bf16x8s*)(v + kvbase + kk * 16);
+      // ---- S^T = K Q^T ; dP^T = V dO^T  (A from pair LDS, B from tiles)
       f32x16 st = {}, dpt = {};
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kk = 0; kk < 8; ++kk) {
-        const bf16x8s qB = qb[sw16(col, kk * 2 + half)];
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf[kk], qB, st, 0, 0, 0);
+        const bf16x8s kA = kp[sw16(col, kk * 2 + half)];
+        const bf16x8s qB = qbuf[sw16(col, kk * 2 + half)];
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kA, qB, st, 0, 0, 0);
       }
 #pragma unroll
       for (int kk = 0; kk < 8; ++kk) {
-        const bf16x8s doB = dob[sw16(col, kk * 2 + half)];
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfk[kk], doB, dpt,
-                                                      0, 0, 0);
+        const bf16x8s vA = vp[sw16(col, kk * 2 + half)];
+        const bf16x8s doB = dobuf[sw16(col, kk * 2 + half)];
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vA, doB, dpt, 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
 
@@ -241,7 +241,7 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int kvl = cd_row(r, half);  // kv row within my wave tile
+          const int kvl = cd_row(r, half);  // kv row within my pair tile
           const bool dead = (CAUSAL && (my_kv0 + kvl > qrow)) ||
                             (my_kv0 + kvl >= S);
           const float pt = dead ? 0.f : __expf(st[r] * scale - l);
@@ -253,41 +253,53 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
         // ---- dV[kv][d] += P^T(k=q) dO-asB ; dK[kv][d] += dS^T(k=q) Q-asB
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int dblk = 0; dblk < 4; ++dblk) {
+        for (int db = 0; db < 2; ++db) {
+          const int dblk = pm * 2 + db;
 #pragma unroll
           for (int kk = 0; kk < 2; ++kk) {
             const bf16x8s pA = scr_read(pw, col, kk, half);
             const bf16x8s dA = scr_read(dw, col, kk, half);
-            const bf16x8s doB = dotb[sw4(dblk * 32 + col, kk * 2 + half)];
-            const bf16x8s qB2 = qtb[sw4(dblk * 32 + col, kk * 2 + half)];
-            dvacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                pA, doB, dvacc[dblk], 0, 0, 0);
-            dkacc[dblk] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                dA, qB2, dkacc[dblk], 0, 0, 0);
+            const bf16x8s doB = dotbuf[sw4(dblk * 32 + col, kk * 2 + half)];
+            const bf16x8s qB2 = qtbuf[sw4(dblk * 32 + col, kk * 2 + half)];
+            dvacc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pA, doB, dvacc[db], 0, 0, 0);
+            dkacc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                dA, qB2, dkacc[db], 0, 0, 0);
           }
         }
         __builtin_amdgcn_s_setprio(0);
       }
     }
 
-    __syncthreads();
+    __syncthreads();  // everyone done reading the tile buffers
+    const bool have_next = (it + 1 < total) && !(VAR & 4);
+    if (have_next) {
+      int qs2;
+      long long qb2, tb2, sb2;
+      qb_of(it + 1, qs2, qb2, tb2, sb2);
+      dkdv_stage(q, dout, qt, dot, qbuf, dobuf, qtbuf, dotbuf,
+                 qb2, tb2, qs2, S, tid);
+    }
+    __syncthreads();  // loads drained (barrier waits vmcnt)
   }
 
-  // ---- write dk/dv (C/D: col = d_local, row = kv via reg map)
+  // ---- write dk/dv for my 2 d blocks (col = d_local, row = kv via map)
   const long long obase = (((long long)b * Hkv + hkv) * S) * D;
 #pragma unroll
-  for (int dblk = 0; dblk < 4; ++dblk) {
+  for (int db = 0; db < 2; ++db) {
+    const int dblk = pm * 2 + db;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kvl = cd_row(r, half);
       if (my_kv0 + kvl >= S) continue;
       const long long off = obase + (long long)(my_kv0 + kvl) * D +
                             dblk * 32 + col;
-      dv[off] = f2bf(dvacc[dblk][r]);
-      dk[off] = f2bf(dkacc[dblk][r]);
+      dv[off] = f2bf(dvacc[db][r]);
+      dk[off] = f2bf(dkacc[db][r]);
     }
   }
 }
+
 
 // ------------------------------------------------------------------- dq
 
@@ -334,7 +346,7 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dq_kernel(
   __shared__ lds_chunk kbuf[2][T * 16];
   __shared__ lds_chunk vbuf[2][T * 16];
   __shared__ lds_chunk ktbuf[2][D * 4];
-  __shared__ short dscr[NW][T * T];
+  __shared__ short dscr2[NW][T * T];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -378,7 +390,7 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dq_kernel(
   }
   __syncthreads();
 
-  short* dw = dscr[wid];
+  short* dw = dscr2[wid];
 
   for (int it = 0; it < nt; ++it) {
     const int kv0 = it * T;
@@ -483,7 +495,7 @@ extern "C" void ds_flash_bwd(const void* q, const void* k, const void* v,
                              void* dk, void* dv, int B, int S, int H,
                              int Hkv, float scale, int causal, void* stream) {
   hipStream_t st = reinterpret_cast<hipStream_t>(stream);
-  dim3 g1((S + MT - 1) / MT, Hkv, B);
+  dim3 g1((S + MTK - 1) / MTK, Hkv, B);
   dim3 g2((S + MT - 1) / MT, H, B);
   if (causal) {
     hipLaunchKernelGGL((flash_bwd_dkdv_kernel<true>), g1, dim3(TPB), 0, st,
