@@ -31,6 +31,7 @@ from .fp16.loss_scaler import create_loss_scaler, LossScaler
 from .utils import (DummyOptim, get_global_norm_of_tensors,
                     clip_tensors_by_global_norm, see_memory_usage)
 from .zero.stage12 import ZeroStage12Optimizer
+from ..utils.nvtx import instrument_w_nvtx
 
 MEMORY_OPT_ALLREDUCE_SIZE = 500_000_000
 
@@ -314,6 +315,7 @@ class Engine(torch.nn.Module):
 
     # ------------------------------------------------------------ hot path
 
+    @instrument_w_nvtx
     def forward(self, *inputs, **kwargs):
         if self.wall_clock_breakdown:
             self.timers(FORWARD_GLOBAL_TIMER).start()
@@ -337,6 +339,7 @@ class Engine(torch.nn.Module):
             self.timers(FORWARD_GLOBAL_TIMER).stop()
         return loss
 
+    @instrument_w_nvtx
     def backward(self, loss, retain_graph=False, scale_wrt_gas=True):
         boundary = (self.micro_steps + 1) % self.gradient_accumulation_steps == 0 \
             and not self.inside_no_sync
@@ -421,6 +424,7 @@ class Engine(torch.nn.Module):
                 flush()
         flush()
 
+    @instrument_w_nvtx
     def step(self, lr_kwargs=None):
         if not self._is_gradient_accumulation_boundary:
             return

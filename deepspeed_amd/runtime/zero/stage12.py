@@ -37,6 +37,7 @@ from ... import comm as dist
 from ...utils.logging import log_dist
 from ..fp16.loss_scaler import LossScalerBase, LossScaler
 from ..utils import ALIGNMENT
+from ...utils.nvtx import instrument_w_nvtx
 
 
 class _Bucket:
@@ -332,12 +333,14 @@ class ZeroStage12Optimizer:
         else:
             self.loss_scaler.backward(loss.float(), retain_graph=retain_graph)
 
+    @instrument_w_nvtx
     def reduce_gradients(self):
         """Called by the engine at the gradient-accumulation boundary after the
         last micro-backward: flush buckets whose hooks fired pre-boundary."""
         self._finish_reductions()
 
     @torch.no_grad()
+    @instrument_w_nvtx
     def step(self, closure=None):
         assert closure is None, "closure not supported"
         self._finish_reductions()

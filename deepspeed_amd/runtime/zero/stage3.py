@@ -46,6 +46,7 @@ from ... import comm as dist
 from ...utils.logging import log_dist
 from ..fp16.loss_scaler import LossScalerBase, LossScaler
 from ..utils import ALIGNMENT
+from ...utils.nvtx import instrument_w_nvtx
 
 FREE, INFLIGHT, AVAILABLE = 0, 1, 2
 
@@ -444,6 +445,7 @@ class ZeroStage3Optimizer:
             g //= 2
         return g
 
+    @instrument_w_nvtx
     def _launch_gather(self, u: _Unit):
         if u.status != FREE:
             return
@@ -618,6 +620,7 @@ class ZeroStage3Optimizer:
 
     # -------------------------------------------------------------- reduction
 
+    @instrument_w_nvtx
     def _reduce_unit(self, u: _Unit):
         """reduce-scatter one unit's grads; accumulate shard into fp32."""
         if u.grad_full is None:
@@ -707,6 +710,7 @@ class ZeroStage3Optimizer:
             if i < len(t):
                 self._bwd_cursor = i + 1
 
+    @instrument_w_nvtx
     def _prefetch(self, trace, cursor):
         budget = self.prefetch_bucket_size
         i = cursor
@@ -768,6 +772,7 @@ class ZeroStage3Optimizer:
         self._drain_inflight_rs(limit=0)
 
     @torch.no_grad()
+    @instrument_w_nvtx
     def step(self, closure=None):
         assert closure is None, "closure not supported"
         self._drain_inflight_rs(limit=0)
