@@ -12,6 +12,16 @@ extern "C" void ds_fused_adam_flat(float* p, const void* g, int grad_dtype,
                                    float lr, float beta1, float beta2, float eps,
                                    float weight_decay, int step, float inv_scale,
                                    int adamw, void* stream);
+extern "C" void ds_fused_lion(float* p, const void* g, int grad_dtype,
+                              float* m, void* p16, long long n, float lr,
+                              float beta1, float beta2, float weight_decay,
+                              float inv_scale, void* stream);
+extern "C" void ds_fused_lamb(float* p, const void* g, int grad_dtype,
+                              float* m, float* v, float* u, float* norms2,
+                              void* p16, long long n, float lr, float beta1,
+                              float beta2, float eps, float weight_decay,
+                              int step, float max_coeff, float min_coeff,
+                              float inv_scale, void* stream);
 extern "C" void ds_norm_fwd(const void* x, const void* w, const void* b,
                             void* y, float* invrms, float* mean, int rows,
                             int H, float eps, int ln, int dtype, void* stream);
@@ -107,6 +117,41 @@ void fused_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                      p.numel(), (float)lr, (float)beta1, (float)beta2,
                      (float)eps, (float)weight_decay, (int)step,
                      (float)inv_scale, adamw ? 1 : 0, cur_stream());
+}
+
+void fused_lion(at::Tensor p, at::Tensor g, at::Tensor m,
+                c10::optional<at::Tensor> p16, double lr, double beta1,
+                double beta2, double weight_decay, double inv_scale) {
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous(),
+              "fused_lion: tensors must be contiguous");
+  TORCH_CHECK(p.scalar_type() == at::kFloat && m.scalar_type() == at::kFloat,
+              "p/m must be fp32");
+  void* p16_ptr = nullptr;
+  if (p16.has_value() && p16->defined()) p16_ptr = p16->data_ptr();
+  ds_fused_lion(p.data_ptr<float>(), g.data_ptr(), dtype_code(g),
+                m.data_ptr<float>(), p16_ptr, p.numel(), (float)lr,
+                (float)beta1, (float)beta2, (float)weight_decay,
+                (float)inv_scale, cur_stream());
+}
+
+void fused_lamb(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                at::Tensor u, at::Tensor norms2,
+                c10::optional<at::Tensor> p16, double lr, double beta1,
+                double beta2, double eps, double weight_decay, int64_t step,
+                double max_coeff, double min_coeff, double inv_scale) {
+  TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous() &&
+              v.is_contiguous() && u.is_contiguous(),
+              "fused_lamb: tensors must be contiguous");
+  TORCH_CHECK(norms2.numel() == 2 && norms2.scalar_type() == at::kFloat,
+              "norms2 must be fp32[2]");
+  void* p16_ptr = nullptr;
+  if (p16.has_value() && p16->defined()) p16_ptr = p16->data_ptr();
+  ds_fused_lamb(p.data_ptr<float>(), g.data_ptr(), dtype_code(g),
+                m.data_ptr<float>(), v.data_ptr<float>(), u.data_ptr<float>(),
+                norms2.data_ptr<float>(), p16_ptr, p.numel(), (float)lr,
+                (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                (int)step, (float)max_coeff, (float)min_coeff,
+                (float)inv_scale, cur_stream());
 }
 
 void cpu_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
@@ -319,6 +364,8 @@ class AioHandle {
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_lion", &fused_lion, "fused Lion step (GPU)");
+  m.def("fused_lamb", &fused_lamb, "fused LAMB step (GPU, 2-phase)");
   m.def("fused_adam_flat", &fused_adam_flat,
         "Fused Adam/AdamW on flat fp32 master + 16-bit grad shard");
   m.def("cpu_adam_flat", &cpu_adam_flat,

@@ -1,13 +1,15 @@
 """LAMB optimizer (reference: deepspeed/ops/lamb/fused_lamb.py over
 csrc/lamb/fused_lamb_cuda_kernel.cu).
 
-Layerwise trust-ratio Adam for very large batch sizes. The per-parameter
-norm reductions + elementwise update run as a handful of torch kernels on
-ROCm; the flat-shard fused-kernel treatment (like adam.hip) is only worth
-it if LAMB becomes a bench path.
+Layerwise trust-ratio Adam for very large batch sizes. On GPU fp32 params the
+step runs as the hand-written 2-phase HIP kernel (csrc/optim.hip: Adam
+direction with fused squared-norm reductions, then trust-ratio apply);
+other dtypes/devices fall back to the torch composition below.
 """
 
 import torch
+
+from ._loader import get_ext
 
 
 class FusedLamb(torch.optim.Optimizer):
@@ -41,6 +43,24 @@ class FusedLamb(torch.optim.Optimizer):
                                                            dtype=torch.float32)
                 state["step"] += 1
                 m, v = state["exp_avg"], state["exp_avg_sq"]
+                ext = get_ext()
+                if (ext is not None and p.is_cuda
+                        and p.dtype == torch.float32
+                        and group["bias_correction"]):
+                    # hand-written 2-phase HIP kernel (csrc/optim.hip):
+                    # Adam direction + fused norm reductions, then the
+                    # trust-ratio apply — no host sync for the norms
+                    if "u_buf" not in state:
+                        state["u_buf"] = torch.empty_like(p)
+                        state["norms2"] = torch.zeros(
+                            2, dtype=torch.float32, device=p.device)
+                    ext.fused_lamb(p.data, p.grad.contiguous(), m, v,
+                                   state["u_buf"], state["norms2"], None,
+                                   group["lr"], beta1, beta2, group["eps"],
+                                   group["weight_decay"], state["step"],
+                                   group["max_coeff"], group["min_coeff"],
+                                   1.0)
+                    continue
                 m.mul_(beta1).add_(g, alpha=1 - beta1)
                 v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
                 if group["bias_correction"]:

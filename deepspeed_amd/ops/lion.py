@@ -8,6 +8,8 @@ a dedicated HIP kernel is a later optimization.
 
 import torch
 
+from ._loader import get_ext
+
 
 class Lion(torch.optim.Optimizer):
     """p -= lr * (sign(b1*m + (1-b1)*g) + wd*p);  m = b2*m + (1-b2)*g"""
@@ -34,6 +36,14 @@ class Lion(torch.optim.Optimizer):
                 if len(state) == 0:
                     state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
                 m = state["exp_avg"]
+                ext = get_ext()
+                if (ext is not None and p.is_cuda
+                        and p.dtype == torch.float32):
+                    # fused HIP kernel (csrc/optim.hip); update math is
+                    # identical to the torch path below
+                    ext.fused_lion(p.data, p.grad.contiguous(), m, None,
+                                   lr, beta1, beta2, wd, 1.0)
+                    continue
                 if wd != 0:
                     p.mul_(1 - lr * wd)
                 update = m.mul(beta1).add_(g, alpha=1 - beta1).sign_()
