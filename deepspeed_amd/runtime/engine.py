@@ -205,6 +205,15 @@ class Engine(torch.nn.Module):
             return FusedLamb(model_parameters, **params)
         if name == "adagrad":
             return torch.optim.Adagrad(model_parameters, **params)
+        if name in ("onebitadam", "onebit_adam"):
+            from .fp16.onebit import OnebitAdam
+            return OnebitAdam(model_parameters, **params)
+        if name in ("onebitlamb", "onebit_lamb"):
+            from .fp16.onebit import OnebitLamb
+            return OnebitLamb(model_parameters, **params)
+        if name in ("zerooneadam", "zero_one_adam"):
+            from .fp16.onebit import ZeroOneAdam
+            return ZeroOneAdam(model_parameters, **params)
         raise ValueError(f"unsupported optimizer type {cfg.type}")
 
     def _configure_optimizer(self, model_parameters):

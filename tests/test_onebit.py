@@ -111,3 +111,31 @@ def _zoadam_worker(rank, world):
 
 def test_zero_one_adam_trains():
     run_distributed(_zoadam_worker, world_size=2)
+
+
+def _onebit_lamb_worker(rank, world):
+    from deepspeed_amd.runtime.fp16.onebit import OnebitLamb
+    torch.manual_seed(9)
+    model = torch.nn.Linear(16, 1)
+    opt = OnebitLamb(model.parameters(), lr=0.2, freeze_step=5)
+    torch.manual_seed(300)  # same data on both ranks
+    X = torch.randn(64, 16)
+    y = X @ torch.randn(16, 1)
+    losses = []
+    for _ in range(60):
+        loss = torch.nn.functional.mse_loss(model(X), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert opt.lamb_freeze_key
+    assert losses[-1] < 0.3 * losses[0], losses[::8]
+    import torch.distributed as td
+    wt = model.weight.detach().clone()
+    peers = [torch.empty_like(wt) for _ in range(world)]
+    td.all_gather(peers, wt)
+    assert torch.equal(peers[0], peers[1])
+
+
+def test_onebit_lamb_trains():
+    run_distributed(_onebit_lamb_worker, world_size=2)
