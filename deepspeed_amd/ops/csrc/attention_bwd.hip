@@ -155,17 +155,18 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
   const int my_kv0 = kv0b + pair * T;
   const int kvrow = my_kv0 + col;
 
-  // ---- stage K/V for all 4 pairs (linear dst, pre-swizzled src)
+  // ---- stage K/V for all 4 pairs (linear dst, pre-swizzled src).
+  // k/v are BHSD here ([B,Hkv,S,D]) — kv-row stride is D, unlike the
+  // forward kernel's [B,S,Hkv,D] layout.
   {
     const long long kb0 = (((long long)b * Hkv + hkv) * S) * D;
-    const int HkvD = Hkv * D;
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const int j = tid + t * TPB;       // 0..2047
       const int p = j >> 9, jj = j & 511;
       const int row = jj >> 4, c = (jj & 15) ^ (row & 7);
       const long long off =
-          kb0 + (long long)min(kv0b + p * T + row, S - 1) * HkvD + c * 8;
+          kb0 + (long long)min(kv0b + p * T + row, S - 1) * D + c * 8;
       __builtin_amdgcn_global_load_lds((const unsigned int*)(k + off),
                                        (unsigned int*)(&kscr[0][0] + j),
                                        16, 0, 0);

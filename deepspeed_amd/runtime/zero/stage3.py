@@ -370,6 +370,10 @@ class ZeroStage3Optimizer:
             del full
         if accel.available():
             torch.cuda.empty_cache()
+        # safe mode: unit count/sizes must agree before any collective
+        from ...utils.safe_mode import checked
+        checked([u.numel for u in self.units], group=self.dp_group,
+                what="ZeRO-3 unit sizes")
         # persistent units stay materialized
         for u in self.units:
             if u.persist:
@@ -732,6 +736,10 @@ class ZeroStage3Optimizer:
                     dedup.append(idx)
             self._trace = dedup
             self._rtrace = list(reversed(dedup))
+            # safe mode: a divergent fetch order across ranks means the
+            # next step's coalesced gathers would deadlock — assert now
+            from ...utils.safe_mode import checked
+            checked(self._trace, group=self.dp_group, what="ZeRO-3 fetch trace")
         elif self._trace_complete and self._trace_misses > \
                 max(4, len(self._trace) // 4):
             # trace invalidation (reference partitioned_param_coordinator

@@ -509,3 +509,56 @@ def test_evoformer_attention_reference_math():
     out_nb = DS4Sci_EvoformerAttention(Q, K, V)
     out_b1 = DS4Sci_EvoformerAttention(Q, K, V, [bias1, None])
     assert not torch.allclose(out_nb, out_b1)
+
+
+def _safe_mode_worker(rank, world):
+    from deepspeed_amd.utils.safe_mode import (
+        assert_ints_same_as_other_ranks, enable_safe_mode, checked)
+    # identical sequences pass
+    assert_ints_same_as_other_ranks([1, 2, 3], what="ok-case")
+    # divergent sequences raise on every rank, naming the position
+    import pytest
+    with pytest.raises(RuntimeError, match="diverges|length"):
+        assert_ints_same_as_other_ranks([1, 2, 3 + rank], what="bad-case")
+    # checked() is a no-op until enabled
+    checked([rank], what="gated")
+    enable_safe_mode(True)
+    try:
+        with pytest.raises(RuntimeError):
+            checked([rank], what="gated-on")
+    finally:
+        enable_safe_mode(False)
+
+
+def test_safe_mode_cross_rank_asserts():
+    from .common import run_distributed
+    run_distributed(_safe_mode_worker, world_size=2)
+
+
+def test_safe_mode_zero3_trace():
+    """ZeRO-3 end-to-end with safe mode on: consistent ranks pass."""
+    import os
+
+    def worker(rank, world):
+        os.environ["DS_AMD_SAFE_MODE"] = "1"
+        import importlib
+        from deepspeed_amd.utils import safe_mode
+        importlib.reload(safe_mode)
+        import deepspeed_amd
+        import torch
+        from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+        cfg = llama_tiny()
+        torch.manual_seed(4)
+        model = LlamaForCausalLM(cfg)
+        engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+            "train_micro_batch_size_per_gpu": 2,
+            "zero_optimization": {"stage": 3},
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+        for _ in range(2):
+            ids = torch.randint(0, cfg.vocab_size, (2, 16))
+            loss = engine(ids, labels=ids)
+            engine.backward(loss)
+            engine.step()
+
+    from .common import run_distributed
+    run_distributed(worker, world_size=2)
