@@ -96,6 +96,16 @@ class InferenceEngine(nn.Module):
         input_ids: [B, S] prompt. Returns [B, S + new] including the prompt.
         """
         layers, kv_heads, head_dim, max_seq = self._model_geometry()
+        if self.config.enable_cuda_graph and self.device.type == "cuda":
+            from .graph import graph_generate
+            return graph_generate(
+                self.module, input_ids.to(self.device), n_layers=layers,
+                kv_heads=kv_heads, head_dim=head_dim, max_seq=max_seq,
+                dtype=(self.config.dtype if self.config.dtype
+                       != torch.float32 else torch.float32),
+                max_new_tokens=max_new_tokens, do_sample=do_sample,
+                temperature=temperature, top_k=top_k, top_p=top_p,
+                eos_token_id=eos_token_id)
         return kv_generate(self.module, input_ids.to(self.device),
                            n_layers=layers, kv_heads=kv_heads,
                            head_dim=head_dim, max_seq=max_seq,
