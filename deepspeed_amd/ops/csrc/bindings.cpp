@@ -49,6 +49,9 @@ extern "C" void ds_groupwise_quant(const void* x, int dtype, void* q,
 extern "C" void ds_groupwise_dequant(const void* q, const float* scales,
                                      void* out, int dtype, long long n,
                                      int group_size, int bits, void* stream);
+extern "C" void ds_fp_quantize(const void* x, int dtype, void* out,
+                               float* scales, long long n, int group_size,
+                               int bits, int dequant, void* stream);
 extern "C" void* ds_aio_create(long long block_size, int n_threads);
 extern "C" void ds_aio_destroy(void* h);
 extern "C" int ds_aio_pwrite(void* h, const void* data, long long nbytes,
@@ -152,6 +155,35 @@ void fused_lamb(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                 (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
                 (int)step, (float)max_coeff, (float)min_coeff,
                 (float)inv_scale, cur_stream());
+}
+
+std::vector<at::Tensor> fp_quantize(at::Tensor x, int64_t group_size,
+                                    int64_t bits) {
+  TORCH_CHECK(x.is_contiguous() && x.is_cuda(), "fp_quantize: cuda contiguous");
+  TORCH_CHECK(bits == 4 || bits == 6 || bits == 8 || bits == 12,
+              "bits must be 4/6/8/12");
+  TORCH_CHECK(group_size % 8 == 0, "group_size % 8 == 0");
+  const long long n = x.numel();
+  const long long groups = (n + group_size - 1) / group_size;
+  const int vper3 = 24 / (int)bits;
+  const long long n3 = (n + vper3 - 1) / vper3;
+  auto out = at::empty({n3 * 3}, x.options().dtype(at::kByte));
+  auto scales = at::empty({groups}, x.options().dtype(at::kFloat));
+  ds_fp_quantize(x.data_ptr(), dtype_code(x), out.data_ptr(),
+                 scales.data_ptr<float>(), n, (int)group_size, (int)bits, 0,
+                 cur_stream());
+  return {out, scales};
+}
+
+at::Tensor fp_dequantize(at::Tensor q, at::Tensor scales, int64_t numel,
+                         int64_t group_size, int64_t bits,
+                         at::ScalarType out_dtype) {
+  TORCH_CHECK(q.is_contiguous() && q.is_cuda(), "fp_dequantize: cuda");
+  auto out = at::empty({numel}, q.options().dtype(out_dtype));
+  ds_fp_quantize(q.data_ptr(), dtype_code(out), out.data_ptr(),
+                 scales.data_ptr<float>(), numel, (int)group_size, (int)bits,
+                 1, cur_stream());
+  return out;
 }
 
 void cpu_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
@@ -364,6 +396,8 @@ class AioHandle {
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fp_quantize", &fp_quantize, "groupwise FP4/6/8/12 quantize");
+  m.def("fp_dequantize", &fp_dequantize, "groupwise FP dequantize");
   m.def("fused_lion", &fused_lion, "fused Lion step (GPU)");
   m.def("fused_lamb", &fused_lamb, "fused LAMB step (GPU, 2-phase)");
   m.def("fused_adam_flat", &fused_adam_flat,

@@ -108,3 +108,33 @@ def test_fp8_quantizer_class():
     y = fq.dequantize(q, s)
     assert y.shape == x.shape and y.dtype == torch.bfloat16
     assert (y.float() - x.float()).abs().mean() < 0.05
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("bits", [4, 6, 8, 12])
+def test_fp_quantizer_roundtrip_gpu(bits):
+    """HIP FP4/6/8/12 kernel vs the bit-accurate torch emulation."""
+    from deepspeed_amd.ops.fp_quantizer import (fp_quantize, fp_dequantize,
+                                                fp_emulate_reference)
+    torch.manual_seed(0)
+    x = torch.randn(4096 + 56, device="cuda") * 3
+    q, scales = fp_quantize(x, bits=bits, group_size=128)
+    got = fp_dequantize(q, scales, x.numel(), bits=bits, group_size=128,
+                        out_dtype=torch.float32)
+    want = fp_emulate_reference(x, bits=bits, group_size=128)
+    torch.testing.assert_close(got.view(-1), want.view(-1),
+                               rtol=1e-6, atol=1e-6)
+    # compression: FP6 must actually pack 4 values into 3 bytes
+    if bits == 6:
+        assert q.numel() <= (x.numel() * 6 + 23) // 24 * 3 + 3
+
+
+def test_fp_emulation_properties_cpu():
+    """The emulation itself: max error bounds per format."""
+    from deepspeed_amd.ops.fp_quantizer import fp_emulate_reference
+    torch.manual_seed(1)
+    x = torch.randn(2048) * 5
+    for bits, tol in ((4, 0.3), (6, 0.15), (8, 0.07), (12, 0.04)):
+        y = fp_emulate_reference(x, bits=bits, group_size=256)
+        rel = ((x - y).abs() / x.abs().clamp_min(1e-3)).median()
+        assert rel < tol, (bits, rel)
