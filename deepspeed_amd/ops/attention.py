@@ -41,6 +41,27 @@ def flash_train_available(q: torch.Tensor, k: torch.Tensor) -> bool:
     return _eligible(q, k)
 
 
+def _vt_from_bshd(v):
+    """[B, S, Hkv, D] -> [B, Hkv, D, S] via the tiled transpose kernel."""
+    ext = get_ext()
+    B, S, Hk, D = v.shape
+    if ext is None or not v.is_cuda or v.dtype != torch.bfloat16:
+        return v.permute(0, 2, 3, 1).contiguous()
+    out = ext.transpose_bf16(v.contiguous(), B * Hk, S, D, Hk * D, Hk, D,
+                             S * Hk * D)
+    return out.view(B, Hk, D, S)
+
+
+def _t_last2_bhsd(x):
+    """[B, H, S, D] -> [B, H, D, S] via the tiled transpose kernel."""
+    ext = get_ext()
+    B, H, S, D = x.shape
+    if ext is None or not x.is_cuda or x.dtype != torch.bfloat16:
+        return x.transpose(-1, -2).contiguous()
+    out = ext.transpose_bf16(x.contiguous(), B * H, S, D, D, 1, 0, S * D)
+    return out.view(B, H, D, S)
+
+
 def flash_attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                    causal: bool = True,
                    scale: Optional[float] = None) -> torch.Tensor:
@@ -48,7 +69,7 @@ def flash_attn_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     only (no autograd)."""
     ext = get_ext()
     scale = scale if scale is not None else 1.0 / math.sqrt(q.size(-1))
-    vt = v.permute(0, 2, 3, 1).contiguous()  # [B, Hkv, D, S]
+    vt = _vt_from_bshd(v)  # [B, Hkv, D, S]
     return ext.flash_attn_fwd(q.contiguous(), k.contiguous(), vt, scale,
                               causal)
 
@@ -72,9 +93,9 @@ def flash_attn_bwd(q, k, v, o, do, lse, causal=True, scale=None):
     ext = get_ext()
     scale = scale if scale is not None else 1.0 / math.sqrt(q.size(-1))
     delta = (do.float() * o.float()).sum(-1)              # [B,H,S]
-    qt = q.transpose(-1, -2).contiguous()
-    kt = k.transpose(-1, -2).contiguous()
-    dot = do.transpose(-1, -2).contiguous()
+    qt = _t_last2_bhsd(q)
+    kt = _t_last2_bhsd(k)
+    dot = _t_last2_bhsd(do)
     return ext.flash_attn_bwd(q.contiguous(), k.contiguous(), v.contiguous(),
                               do.contiguous(), qt, kt, dot,
                               lse.contiguous().float(), delta.contiguous(),
@@ -89,7 +110,7 @@ class FlashAttnFunc(torch.autograd.Function):
     def forward(ctx, q, k, v, causal, scale):
         ext = get_ext()
         scale = scale if scale is not None else 1.0 / math.sqrt(q.size(-1))
-        vt = v.permute(0, 2, 3, 1).contiguous()
+        vt = _vt_from_bshd(v)
         o, lse = ext.flash_attn_fwd_lse(q.contiguous(), k.contiguous(), vt,
                                         scale, causal)
         ctx.save_for_backward(q, k, v, o, lse)

@@ -52,6 +52,10 @@ extern "C" void ds_groupwise_dequant(const void* q, const float* scales,
 extern "C" void ds_fp_quantize(const void* x, int dtype, void* out,
                                float* scales, long long n, int group_size,
                                int bits, int dequant, void* stream);
+extern "C" void ds_transpose_bf16(const void* src, void* dst, int n_batch,
+                                  int R, int C, long long row_stride,
+                                  int inner, long long inner_stride,
+                                  long long outer_stride, void* stream);
 extern "C" void* ds_aio_create(long long block_size, int n_threads);
 extern "C" void ds_aio_destroy(void* h);
 extern "C" int ds_aio_pwrite(void* h, const void* data, long long nbytes,
@@ -184,6 +188,18 @@ at::Tensor fp_dequantize(at::Tensor q, at::Tensor scales, int64_t numel,
                  scales.data_ptr<float>(), numel, (int)group_size, (int)bits,
                  1, cur_stream());
   return out;
+}
+
+at::Tensor transpose_bf16(at::Tensor src, int64_t n_batch, int64_t R,
+                          int64_t C, int64_t row_stride, int64_t inner,
+                          int64_t inner_stride, int64_t outer_stride) {
+  TORCH_CHECK(src.is_cuda() && src.scalar_type() == at::kBFloat16,
+              "transpose_bf16: cuda bf16");
+  auto dst = at::empty({n_batch, C, R}, src.options());
+  ds_transpose_bf16(src.data_ptr(), dst.data_ptr(), (int)n_batch, (int)R,
+                    (int)C, row_stride, (int)inner, inner_stride,
+                    outer_stride, cur_stream());
+  return dst;
 }
 
 void cpu_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
@@ -396,6 +412,7 @@ class AioHandle {
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("transpose_bf16", &transpose_bf16, "tiled bf16 batched transpose");
   m.def("fp_quantize", &fp_quantize, "groupwise FP4/6/8/12 quantize");
   m.def("fp_dequantize", &fp_dequantize, "groupwise FP dequantize");
   m.def("fused_lion", &fused_lion, "fused Lion step (GPU)");

@@ -76,8 +76,19 @@ print(f"SDPA fwd+bwd:   {t_fb*1e3:7.2f} ms  ({fwd_flops*3.5/t_fb/1e12:6.0f} TF e
 qh, kh, vh = (x.transpose(1, 2).contiguous() for x in (q, k, v))
 oh = o.transpose(1, 2).contiguous()
 doh = do_b.contiguous()
-from deepspeed_amd.ops.attention import flash_attn_bwd
+from deepspeed_amd.ops.attention import flash_attn_bwd, _t_last2_bhsd
 t = time_fn(lambda: flash_attn_bwd(qh, kh, vh, oh, doh, lse, True), n=5)
 bwd_flops = fwd_flops * 2.5
 print(f"our bwd (wrap): {t*1e3:7.2f} ms  {bwd_flops/t/1e12:6.0f} TF")
+
+# raw kernel split (pre-transposed inputs)
+delta = (doh.float() * oh.float()).sum(-1)
+qt2, kt2, dot2 = _t_last2_bhsd(qh), _t_last2_bhsd(kh), _t_last2_bhsd(doh)
+t = time_fn(lambda: _C.flash_attn_bwd(qh, kh, vh, doh, qt2, kt2, dot2,
+                                      lse.float(), delta, scale, True), n=5)
+print(f"bwd kernels:    {t*1e3:7.2f} ms  {bwd_flops/t/1e12:6.0f} TF")
+t = time_fn(lambda: _t_last2_bhsd(qh))
+print(f"transpose k:    {t*1e3:7.2f} ms  ({qh.numel()*4/t/1e12:5.2f} TB/s)")
+t = time_fn(lambda: qh.transpose(-1, -2).contiguous())
+print(f"transpose torch:{t*1e3:7.2f} ms  ({qh.numel()*4/t/1e12:5.2f} TB/s)")
 print("done")

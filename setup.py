@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "adam.hip"),
         os.path.join(CSRC, "optim.hip"),
         os.path.join(CSRC, "fp_quant.hip"),
+        os.path.join(CSRC, "transpose.hip"),
         os.path.join(CSRC, "quantize.hip"),
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "attention_bwd.hip"),

@@ -399,3 +399,18 @@ def test_fused_lamb_gpu_matches_torch():
         og.step()
         oc.step()
     torch.testing.assert_close(p_gpu.cpu(), p_cpu, rtol=1e-4, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_transpose_bf16_kernel():
+    """Tiled transpose kernel vs torch permute for both wrapper layouts."""
+    from deepspeed_amd.ops.attention import _vt_from_bshd, _t_last2_bhsd
+    torch.manual_seed(0)
+    v = torch.randn(2, 160, 3, 128, device="cuda", dtype=torch.bfloat16)
+    got = _vt_from_bshd(v)
+    want = v.permute(0, 2, 3, 1).contiguous()
+    assert torch.equal(got, want)
+    x = torch.randn(2, 4, 256, 128, device="cuda", dtype=torch.bfloat16)
+    got = _t_last2_bhsd(x)
+    want = x.transpose(-1, -2).contiguous()
+    assert torch.equal(got, want)
