@@ -117,7 +117,7 @@ def _adam_hyperparams(optimizer, group):
 
 
 def fused_adam_step(optimizer, group, master_param, grad_shard, combined_scale,
-                    segments=None) -> bool:
+                    segments=None, bump_step=True) -> bool:
     """ZeRO flat-shard step: fused HIP kernel over the group's fp32 master
     shard with 16-bit grads consumed in place.
 
@@ -139,7 +139,8 @@ def fused_adam_step(optimizer, group, master_param, grad_shard, combined_scale,
         state["step"] = 0
         state["exp_avg"] = torch.zeros_like(master_param, dtype=torch.float32)
         state["exp_avg_sq"] = torch.zeros_like(master_param, dtype=torch.float32)
-    state["step"] += 1
+    if bump_step:
+        state["step"] += 1
     scale = float(combined_scale) if not torch.is_tensor(combined_scale) \
         else float(combined_scale.item())
     inv_scale = 1.0 / scale

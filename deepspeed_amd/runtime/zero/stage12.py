@@ -121,8 +121,13 @@ class ZeroStage12Optimizer:
         self._register_hooks()
 
         self.fused_adam_fn = None
-        if fused_adam and not cpu_offload:
+        if fused_adam:
             self.fused_adam_fn = self._try_fused_adam()
+        # pipelined offload bookkeeping: per-bucket D2H completion events
+        # and a device-side grad-norm accumulator (so step() never blocks
+        # on the full D2H drain — reference stage_1_and_2.py:1193 overlap)
+        self._bucket_d2h_ev = {}
+        self._dev_norm_sq = None
 
         log_dist(f"ZeRO stage {stage}: world={self.world_size} "
                  f"buckets={len(self.buckets)} "
@@ -311,18 +316,37 @@ class ZeroStage12Optimizer:
                     bucket.grad_flat.div_(scale)
                     src = recv  # view into grad_flat, now averaged
                     if self.cpu_offload:
+                        self._accum_dev_norm(src)
                         shard_dst.copy_(src, non_blocking=True)
+                        self._record_d2h(bucket)
                     else:
                         shard_dst.copy_(src)
                 elif self.cpu_offload:
-                    shard_dst.copy_(recv.div(scale), non_blocking=True)
+                    src = recv.div(scale)
+                    self._accum_dev_norm(src)
+                    shard_dst.copy_(src, non_blocking=True)
+                    self._record_d2h(bucket)
                 else:
                     torch.div(recv, scale, out=shard_dst)
         if self.overlap_comm:
             accel.current_stream().wait_stream(self._comm_stream)
-        if self.cpu_offload:
-            accel.synchronize()
         self._inflight.clear()
+
+    def _accum_dev_norm(self, t):
+        """Squared-norm of a grad shard, accumulated ON DEVICE while the
+        D2H copy streams — step() then never waits for host grads just to
+        decide overflow/clipping."""
+        if t.numel() == 0 or not t.is_cuda:
+            return
+        nsq = t.float().pow(2).sum()
+        self._dev_norm_sq = nsq if self._dev_norm_sq is None \
+            else self._dev_norm_sq + nsq
+
+    def _record_d2h(self, bucket):
+        if accel.available():
+            ev = self._bucket_d2h_ev.get(bucket.index) or accel.event()
+            ev.record(accel.current_stream())
+            self._bucket_d2h_ev[bucket.index] = ev
 
     # ------------------------------------------------------------------- api
 
@@ -349,7 +373,10 @@ class ZeroStage12Optimizer:
         scale = self.loss_scaler.loss_scale
         owned = [g for g in self.group_owned_grads if g.numel() > 0]
         norm_sq_dev = None
-        if owned:
+        if self.cpu_offload and self._dev_norm_sq is not None:
+            norm_sq_dev = self._dev_norm_sq
+            self._dev_norm_sq = None
+        elif owned:
             norms = torch._foreach_norm(owned, 2.0)
             norm_sq_dev = torch.stack([n.float() for n in norms]).pow(2).sum()
             if dist.is_initialized() and self.world_size > 1:
@@ -396,6 +423,7 @@ class ZeroStage12Optimizer:
                 continue
             grads = self.group_owned_grads[gi]
             segments = []
+            seg_buckets = []
             for b in self.buckets:
                 if b.group_idx != gi:
                     continue
@@ -406,10 +434,26 @@ class ZeroStage12Optimizer:
                 else:
                     wrote_params = False
                 segments.append((b.master_offset, b.shard_size, out16))
-            ok = self.fused_adam_fn(self.optimizer, group, master, grads,
-                                    combined_scale, segments=segments)
-            if not ok:
-                return False, False
+                seg_buckets.append(b)
+            if self.cpu_offload:
+                # pipelined: wait only THIS bucket's D2H, step it on the
+                # host (AVX cpu_adam_flat + pinned bf16 async H2D), move on
+                first = True
+                for seg, b in zip(segments, seg_buckets):
+                    ev = self._bucket_d2h_ev.get(b.index)
+                    if ev is not None:
+                        ev.synchronize()
+                    ok = self.fused_adam_fn(self.optimizer, group, master,
+                                            grads, combined_scale,
+                                            segments=[seg], bump_step=first)
+                    if not ok:
+                        return False, False
+                    first = False
+            else:
+                ok = self.fused_adam_fn(self.optimizer, group, master, grads,
+                                        combined_scale, segments=segments)
+                if not ok:
+                    return False, False
         return True, wrote_params
 
     def _torch_step(self, combined_scale):
