@@ -47,6 +47,12 @@ class InferenceEngine(nn.Module):
         self._kv = None
         self._graph = None
 
+        if config.replace_with_kernel_inject:
+            from ..module_inject import (replace_transformer_layer,
+                                         HFInjectionPolicy)
+            self.injection_policy = HFInjectionPolicy()
+            replace_transformer_layer(self.module, self.injection_policy)
+
         tp = config.tp_size
         if tp > 1:
             if not dist.is_initialized():
