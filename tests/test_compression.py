@@ -71,3 +71,61 @@ def test_quant_act_ema():
     # beyond it clamp (standard QAT behavior)
     inside = x.abs() <= qa.range.item()
     assert (y - x)[inside].abs().max() < 0.1
+
+
+def test_kd_loss_blend_and_intermediate():
+    from deepspeed_amd.compression import KDLoss
+    torch.manual_seed(0)
+    s = torch.randn(4, 16, requires_grad=True)
+    t = torch.randn(4, 16)
+    hard = torch.tensor(2.0)
+    kd = KDLoss(temperature=2.0, alpha=0.5, beta=0.1)
+    loss = kd(s, t, hard_loss=hard,
+              student_states=[torch.randn(4, 8, requires_grad=True)],
+              teacher_states=[torch.randn(4, 8)])
+    assert loss.requires_grad and torch.isfinite(loss)
+    loss.backward()
+    assert s.grad is not None
+    # identical logits at alpha=0 -> pure soft loss ~ 0
+    kd0 = KDLoss(alpha=0.0)
+    z = torch.randn(4, 16)
+    assert kd0(z, z.clone()).abs() < 1e-6
+
+
+def test_build_reduced_student():
+    from deepspeed_amd.compression import build_reduced_student
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    cfg = llama_tiny()
+    torch.manual_seed(1)
+    teacher = LlamaForCausalLM(cfg)
+    student = build_reduced_student(teacher, keep_layers=[0])
+    assert len(student.model.layers) == 1
+    assert len(teacher.model.layers) == cfg.num_layers  # teacher untouched
+    # kept layer's weights are the teacher's layer-0 weights
+    tw = teacher.model.layers[0].self_attn.q_proj.weight
+    sw = student.model.layers[0].self_attn.q_proj.weight
+    assert torch.equal(tw, sw)
+    ids = torch.randint(0, cfg.vocab_size, (2, 8))
+    assert torch.isfinite(student(ids, labels=ids))
+
+
+def test_compression_scheduler_offsets():
+    from deepspeed_amd.compression import (CompressionScheduler,
+                                           LinearLayer_Compress)
+    lin = LinearLayer_Compress(8, 8)
+    sched = CompressionScheduler()
+    sched.register(lin, "weight_quantization", offset=5, end=10, bits=8)
+    sched.register(lin, "sparse_pruning", offset=7, ratio=0.5)
+    sched.step(0)
+    assert lin.weight_quant_bits == 0 and lin.sparse_mask.numel() == 0
+    sched.step(5)
+    assert lin.weight_quant_bits == 8
+    sched.step(7)
+    assert lin.sparse_mask.numel() > 0
+    sched.step(10)  # freeze point: quantization baked into the weight
+    sd = sched.state_dict()
+    s2 = CompressionScheduler()
+    s2.register(lin, "weight_quantization", offset=5, end=10, bits=8)
+    s2.register(lin, "sparse_pruning", offset=7, ratio=0.5)
+    s2.load_state_dict(sd)
+    assert s2._done == sched._done
