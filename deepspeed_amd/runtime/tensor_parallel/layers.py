@@ -49,6 +49,36 @@ class _ReduceFromTP(torch.autograd.Function):
         return grad, None
 
 
+class _ColumnParallelFn(torch.autograd.Function):
+    """Fused column-parallel linear with the Domino/Megatron backward
+    overlap (reference runtime/domino/transformer.py): the dx all-reduce
+    launches ASYNC and the weight/bias gradient GEMMs compute under it —
+    the collective rides xGMI while MFMA does the dW math."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, group):
+        ctx.save_for_backward(x, weight)
+        ctx.group = group
+        ctx.has_bias = bias is not None
+        return nn.functional.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, grad):
+        x, weight = ctx.saved_tensors
+        grad = grad.contiguous()
+        dx = grad @ weight
+        handle = None
+        if dist.get_world_size(ctx.group) > 1:
+            handle = dist.all_reduce(dx, group=ctx.group, async_op=True)
+        g2 = grad.reshape(-1, grad.shape[-1])
+        x2 = x.reshape(-1, x.shape[-1])
+        dw = g2.t() @ x2                       # overlaps the dx all-reduce
+        db = g2.sum(0) if ctx.has_bias else None
+        if handle is not None:
+            handle.wait()
+        return dx, dw, db, None
+
+
 class ColumnParallelLinear(nn.Module):
     def __init__(self, weight_shard: torch.Tensor, bias_shard, group):
         super().__init__()
@@ -60,8 +90,7 @@ class ColumnParallelLinear(nn.Module):
             self.bias.tensor_model_parallel = True
 
     def forward(self, x):
-        x = _CopyToTP.apply(x, self.group)
-        return nn.functional.linear(x, self.weight, self.bias)
+        return _ColumnParallelFn.apply(x, self.weight, self.bias, self.group)
 
 
 class RowParallelLinear(nn.Module):
