@@ -25,6 +25,24 @@ class ConfigModel(BaseModel):
     model_config = {"extra": "forbid", "validate_assignment": True}
 
 
+class CurriculumLearningConfig(ConfigModel):
+    enabled: bool = False
+    curriculum_type: str = "fixed_linear"
+    min_difficulty: int = 1
+    max_difficulty: int = 10
+    schedule_config: dict = {}
+    # difficulty source: "seqlen" derives per-sample difficulty from the
+    # sample's length; "index" loads the DataAnalyzer's metric index file
+    difficulty_metric: str = "seqlen"
+    metric_path: Optional[str] = None
+
+
+class DataEfficiencyConfig(ConfigModel):
+    enabled: bool = False
+    seed: int = 1234
+    curriculum_learning: CurriculumLearningConfig = CurriculumLearningConfig()
+
+
 class FP16Config(ConfigModel):
     enabled: bool = False
     loss_scale: float = 0.0  # 0 => dynamic
@@ -217,6 +235,8 @@ class Config:
         self.pipeline = PipelineConfig(**g("pipeline", {}))
         self.tensor_parallel = TensorParallelConfig(**g("tensor_parallel", {}))
         self.aio = AIOConfig(**g("aio", {}))
+        self.data_efficiency = DataEfficiencyConfig(
+            **g("data_efficiency", {}))
 
         self.data_types_grad_accum_dtype = (
             g("data_types", {}).get("grad_accum_dtype", None))

@@ -463,6 +463,9 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dq_kernel(
   }
 }
 
+
+}  // namespace
+
 // probe-only ablation entry (scripts/bwd_ablate.hip)
 extern "C" void ds_flash_bwd_dkdv_dbg(const void* q, const void* k,
                                       const void* v, const void* dout,
@@ -486,8 +489,6 @@ extern "C" void ds_flash_bwd_dkdv_dbg(const void* q, const void* k,
   }
 #undef L
 }
-
-}  // namespace probe internal end marker (kernels above)
 
 extern "C" void ds_flash_bwd(const void* q, const void* k, const void* v,
                              const void* dout, const void* qt,

@@ -52,6 +52,13 @@ extern "C" void ds_groupwise_dequant(const void* q, const float* scales,
 extern "C" void ds_fp_quantize(const void* x, int dtype, void* out,
                                float* scales, long long n, int group_size,
                                int bits, int dequant, void* stream);
+extern "C" void ds_flash_bwd_dkdv_dbg(const void* q, const void* k,
+                                      const void* v, const void* dout,
+                                      const void* qt, const void* dot,
+                                      const float* lse, const float* delta,
+                                      void* dk, void* dv, int B, int S,
+                                      int H, int Hkv, float scale,
+                                      int variant, void* stream);
 extern "C" void ds_transpose_bf16(const void* src, void* dst, int n_batch,
                                   int R, int C, long long row_stride,
                                   int inner, long long inner_stride,
@@ -412,6 +419,20 @@ class AioHandle {
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("flash_bwd_dkdv_dbg",
+        [](at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout,
+           at::Tensor qt, at::Tensor dot, at::Tensor lse, at::Tensor delta,
+           at::Tensor dk, at::Tensor dv, double scale, int64_t variant) {
+          const int B = q.size(0), H = q.size(1), S = q.size(2);
+          const int Hkv = k.size(1);
+          ds_flash_bwd_dkdv_dbg(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                                dout.data_ptr(), qt.data_ptr(),
+                                dot.data_ptr(), lse.data_ptr<float>(),
+                                delta.data_ptr<float>(), dk.data_ptr(),
+                                dv.data_ptr(), B, S, H, Hkv, (float)scale,
+                                (int)variant, cur_stream());
+        },
+        "dkdv-only launcher (perf diagnosis)");
   m.def("transpose_bf16", &transpose_bf16, "tiled bf16 batched transpose");
   m.def("fp_quantize", &fp_quantize, "groupwise FP4/6/8/12 quantize");
   m.def("fp_dequantize", &fp_dequantize, "groupwise FP dequantize");

@@ -91,3 +91,26 @@ class DataAnalyzer:
 def load_index_to_metric(save_path: str, metric: str) -> List[float]:
     with open(os.path.join(save_path, f"{metric}_index_to_metric.json")) as f:
         return json.load(f)
+
+
+def _map_worker(args):
+    (dataset, names, fns, save_path, wid, nw) = args
+    DataAnalyzer(dataset, names, fns, save_path, wid, nw).run_map()
+
+
+def run_analysis_parallel(dataset, metric_names=None, metric_functions=None,
+                          save_path="./data_analysis", num_workers=4):
+    """Map-reduce over `num_workers` PROCESSES (reference DataAnalyzer's
+    multi-worker map, data_analyzer.py:22 run_map_reduce): each worker
+    writes its metric shard, then one reduce merges + builds the
+    index_to_metric / metric_to_sample files."""
+    import multiprocessing as mp
+    names = metric_names or ["seqlen"]
+    fns = metric_functions or [metric_seqlen]
+    ctx = mp.get_context("fork")
+    with ctx.Pool(num_workers) as pool:
+        pool.map(_map_worker,
+                 [(dataset, names, fns, save_path, w, num_workers)
+                  for w in range(num_workers)])
+    return DataAnalyzer(dataset, names, fns, save_path, 0,
+                        num_workers).run_reduce()

@@ -580,7 +580,48 @@ class Engine(torch.nn.Module):
 
     # ----------------------------------------------------------- data loader
 
-    def deepspeed_io(self, dataset, batch_size=None, num_workers=0, collate_fn=None):
+    def deepspeed_io(self, dataset, batch_size=None, num_workers=0,
+                     collate_fn=None, difficulties=None):
+        """Build the training dataloader. With
+        ``data_efficiency.curriculum_learning`` enabled, samples batch
+        indices through the curriculum-aware DeepSpeedDataSampler
+        (difficulty = sample length by default, or the DataAnalyzer index
+        named by ``metric_path``) — reference engine.deepspeed_io:1831 +
+        data_pipeline/data_sampler.py."""
+        de = self.config.data_efficiency
+        cl = de.curriculum_learning
+        if de.enabled and cl.enabled:
+            from .data_pipeline.curriculum_scheduler import \
+                CurriculumScheduler
+            from .data_pipeline.data_sampler import DeepSpeedDataSampler
+            if difficulties is None:
+                if cl.metric_path:
+                    from .data_pipeline.data_analyzer import \
+                        load_index_to_metric
+                    difficulties = load_index_to_metric(
+                        cl.metric_path, cl.difficulty_metric)
+                else:
+                    difficulties = [
+                        len(dataset[i][0]) if isinstance(dataset[i],
+                                                         (tuple, list))
+                        else len(dataset[i]) for i in range(len(dataset))]
+            sched = CurriculumScheduler({
+                "curriculum_type": cl.curriculum_type,
+                "min_difficulty": cl.min_difficulty,
+                "max_difficulty": cl.max_difficulty,
+                "schedule_config": dict(cl.schedule_config)})
+            self.curriculum_scheduler = sched
+            sampler = DeepSpeedDataSampler(
+                difficulties, sched,
+                batch_size=batch_size or
+                self.train_micro_batch_size_per_gpu,
+                dp_rank=dist.get_rank(self.dp_group),
+                dp_size=self.dp_world_size, seed=de.seed)
+            self.curriculum_sampler = sampler
+            from torch.utils.data import DataLoader
+            return DataLoader(dataset, batch_sampler=sampler,
+                              num_workers=num_workers,
+                              collate_fn=collate_fn)
         from .dataloader import build_dataloader
         return build_dataloader(
             dataset,

@@ -87,6 +87,16 @@ qt2, kt2, dot2 = _t_last2_bhsd(qh), _t_last2_bhsd(kh), _t_last2_bhsd(doh)
 t = time_fn(lambda: _C.flash_attn_bwd(qh, kh, vh, doh, qt2, kt2, dot2,
                                       lse.float(), delta, scale, True), n=5)
 print(f"bwd kernels:    {t*1e3:7.2f} ms  {bwd_flops/t/1e12:6.0f} TF")
+dk_o = torch.empty_like(kh)
+dv_o = torch.empty_like(vh)
+t = time_fn(lambda: _C.flash_bwd_dkdv_dbg(qh, kh, vh, doh, qt2, dot2,
+                                          lse.float(), delta, dk_o, dv_o,
+                                          scale, 0), n=10)
+print(f"dkdv only:      {t*1e3:7.2f} ms")
+t = time_fn(lambda: _C.flash_bwd_dkdv_dbg(qh, kh, vh, doh, qt2, dot2,
+                                          lse.float(), delta, dk_o, dv_o,
+                                          scale, 0), n=30)
+print(f"dkdv only n30:  {t*1e3:7.2f} ms")
 t = time_fn(lambda: _t_last2_bhsd(qh))
 print(f"transpose k:    {t*1e3:7.2f} ms  ({qh.numel()*4/t/1e12:5.2f} TB/s)")
 t = time_fn(lambda: qh.transpose(-1, -2).contiguous())

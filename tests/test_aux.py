@@ -562,3 +562,54 @@ def test_safe_mode_zero3_trace():
 
     from .common import run_distributed
     run_distributed(worker, world_size=2)
+
+
+def test_data_analyzer_parallel_map_reduce(tmp_path):
+    from deepspeed_amd.runtime.data_pipeline.data_analyzer import (
+        run_analysis_parallel, load_index_to_metric)
+    data = [list(range(3 + i % 7)) for i in range(40)]
+    out = run_analysis_parallel(data, save_path=str(tmp_path), num_workers=3)
+    assert out["seqlen"] == [len(s) for s in data]
+    assert load_index_to_metric(str(tmp_path), "seqlen") == out["seqlen"]
+
+
+def test_curriculum_dataloader_from_config():
+    """data_efficiency.curriculum_learning drives the engine dataloader:
+    early batches contain only short samples, longer appear as the
+    difficulty schedule opens up."""
+    from .common import run_local
+
+    def worker(rank, world):
+        import torch
+        import deepspeed_amd
+
+        class ToyDS(torch.utils.data.Dataset):
+            def __len__(self):
+                return 64
+
+            def __getitem__(self, i):
+                n = 4 + (i % 8)
+                return torch.full((n,), i, dtype=torch.long), n
+
+        model = torch.nn.Linear(4, 4)
+        engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+            "train_micro_batch_size_per_gpu": 4,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "data_efficiency": {
+                "enabled": True,
+                "curriculum_learning": {
+                    "enabled": True, "curriculum_type": "fixed_linear",
+                    "min_difficulty": 5, "max_difficulty": 12,
+                    "schedule_config": {"total_curriculum_step": 8,
+                                        "difficulty_step": 1}}}})
+        dl = engine.deepspeed_io(ToyDS(), collate_fn=lambda b: b)
+        lens_per_batch = []
+        for batch in dl:
+            lens_per_batch.append(max(n for _, n in batch))
+            if len(lens_per_batch) >= 8:
+                break
+        assert lens_per_batch[0] <= 5, lens_per_batch
+        assert max(lens_per_batch) > 5, lens_per_batch
+        assert engine.curriculum_scheduler is not None
+
+    run_local(worker)
