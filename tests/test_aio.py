@@ -61,3 +61,16 @@ def test_tensor_swapper(tmp_path):
     assert a2.dtype == torch.float32 and b2.dtype == torch.float64
     sw.remove("a")
     assert not os.path.exists(str(tmp_path / "a.swp"))
+
+
+def test_nvme_tune_sweep(tmp_path):
+    """ds_nvme_tune equivalent: sweep produces per-combo bandwidths and a
+    valid suggested aio config block."""
+    from deepspeed_amd.utils.nvme_tune import tune
+    results, cfg = tune(path=str(tmp_path), size_mb=4,
+                        block_kbs=(128, 512), thread_counts=(1, 2),
+                        verbose=False)
+    assert len(results) == 4
+    assert all(r["write_GBps"] > 0 and r["read_GBps"] > 0 for r in results)
+    assert cfg["aio"]["block_size"] in (128 * 1024, 512 * 1024)
+    assert cfg["aio"]["thread_count"] in (1, 2)
