@@ -19,12 +19,12 @@ def _bench_combo(ext, path, data, block_kb, threads):
     p = os.path.join(path, f"nvme_tune_{block_kb}_{threads}.bin")
     try:
         t0 = time.perf_counter()
-        h.pwrite(data, p)
+        h.async_pwrite(data, p)
         h.wait()
         tw = time.perf_counter() - t0
         out = torch.empty_like(data)
         t0 = time.perf_counter()
-        h.pread(out, p)
+        h.async_pread(out, p)
         h.wait()
         tr = time.perf_counter() - t0
         n = data.numel()
