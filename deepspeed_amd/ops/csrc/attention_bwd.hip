@@ -70,6 +70,10 @@ __device__ __forceinline__ int sw16(int row, int chunk) {
 __device__ __forceinline__ int sw4(int row, int chunk) {
   return row * 4 + (chunk ^ (row & 3));
 }
+// [128][64] tiles: 8 chunks/row, XOR over row&7
+__device__ __forceinline__ int sw8(int row, int chunk) {
+  return row * 8 + (chunk ^ (row & 7));
+}
 
 // scalar bf16 store into a sw4-swizzled [32][32] scratch
 __device__ __forceinline__ void scr_store(short* scr, int row, int col,
@@ -96,21 +100,26 @@ __device__ __forceinline__ void dkdv_stage(
     const short* __restrict__ qt, const short* __restrict__ dot,
     lds_chunk* qb, lds_chunk* dob, lds_chunk* qtb, lds_chunk* dotb,
     long long qbase, long long tbase, int qs, int S, int tid) {
-  const int j = tid;              // linear LDS chunk slot 0..511
-  // [32][128] tiles (16 chunks/row): slot j holds (row=j>>4, c=(j&15)^(row&7))
-  const int row = j >> 4, c = (j & 15) ^ (row & 7);
-  const long long qoff = qbase + (long long)(qs + row) * D + c * 8;
-  __builtin_amdgcn_global_load_lds((const unsigned int*)(q + qoff),
-                                   (unsigned int*)(qb + j), 16, 0, 0);
-  __builtin_amdgcn_global_load_lds((const unsigned int*)(dout + qoff),
-                                   (unsigned int*)(dob + j), 16, 0, 0);
-  // [128][32] tiles (4 chunks/row): slot j holds (vr=j>>2, vc=(j&3)^(vr&3))
-  const int vr = j >> 2, vc = (j & 3) ^ (vr & 3);
-  const long long toff = tbase + (long long)vr * S + qs + vc * 8;
-  __builtin_amdgcn_global_load_lds((const unsigned int*)(qt + toff),
-                                   (unsigned int*)(qtb + j), 16, 0, 0);
-  __builtin_amdgcn_global_load_lds((const unsigned int*)(dot + toff),
-                                   (unsigned int*)(dotb + j), 16, 0, 0);
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    const int j = tid + t * TPB;    // linear LDS chunk slot 0..1023
+    // [64][128] tiles (16 chunks/row): slot j = (row=j>>4, c=(j&15)^(row&7))
+    const int row = j >> 4, c = (j & 15) ^ (row & 7);
+    const long long qoff =
+        qbase + (long long)min(qs + row, S - 1) * D + c * 8;
+    __builtin_amdgcn_global_load_lds((const unsigned int*)(q + qoff),
+                                     (unsigned int*)(qb + j), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((const unsigned int*)(dout + qoff),
+                                     (unsigned int*)(dob + j), 16, 0, 0);
+    // [128][64] tiles (8 chunks/row): slot j = (vr=j>>3, vc=(j&7)^(vr&7))
+    const int vr = j >> 3, vc = (j & 7) ^ (vr & 7);
+    const long long toff = tbase + (long long)vr * S +
+                           min(qs + vc * 8, S - 8);
+    __builtin_amdgcn_global_load_lds((const unsigned int*)(qt + toff),
+                                     (unsigned int*)(qtb + j), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((const unsigned int*)(dot + toff),
+                                     (unsigned int*)(dotb + j), 16, 0, 0);
+  }
 }
 
 // dkdv v3 geometry: KV macro-tile = 128 rows = 4 wave-PAIRS x 32 rows.
@@ -132,14 +141,14 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     short* __restrict__ dk, short* __restrict__ dv,
     const int B, const int S, const int H, const int Hkv,
     const float scale) {
-  __shared__ lds_chunk kscr[4][T * 16];  // K [32][128] per pair (32 KB)
-  __shared__ lds_chunk vscr[4][T * 16];  // V (32 KB)
-  __shared__ lds_chunk qbuf[T * 16];     // single-buffered q tiles (32 KB)
-  __shared__ lds_chunk dobuf[T * 16];
-  __shared__ lds_chunk qtbuf[D * 4];
-  __shared__ lds_chunk dotbuf[D * 4];
-  __shared__ short pscr[NW][T * T];      // per-wave transpose scratch
-  __shared__ short dscr[NW][T * T];      // (32 KB)
+  __shared__ lds_chunk kscr[4][T * 16];   // K [32][128] per pair (32 KB)
+  __shared__ lds_chunk vscr[4][T * 16];   // V (32 KB)
+  __shared__ lds_chunk qbuf[2 * T * 16];  // single-buffered 64-row q tiles
+  __shared__ lds_chunk dobuf[2 * T * 16];
+  __shared__ lds_chunk qtbuf[D * 8];      // [128][64]
+  __shared__ lds_chunk dotbuf[D * 8];     // (64 KB of tiles total)
+  __shared__ short pscr[NW][T * T];       // per-wave scratch, REUSED for
+                                          // pt then dst per subtile (16 KB)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -179,14 +188,15 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
   f32x16 dvacc[2] = {};  // my two 32-wide d blocks
   f32x16 dkacc[2] = {};
 
-  const int q_start = CAUSAL ? kv0b : 0;  // 32-aligned (S % 32 == 0)
-  const int nq = (S - q_start + T - 1) / T;
+  const int QT = 2 * T;                    // 64 q rows per iteration
+  const int q_start = CAUSAL ? kv0b : 0;   // 64-aligned (kv0b % 128 == 0)
+  const int nq = (S - q_start + QT - 1) / QT;
   const int total = G * nq;
 
   auto qb_of = [&](int it, int& qs, long long& qbase, long long& tbase,
                    long long& sbase) {
     const int g = it / nq;
-    qs = q_start + (it % nq) * T;
+    qs = q_start + (it % nq) * (2 * T);
     const int h = hkv * G + g;
     qbase = (((long long)b * H + h) * S) * D;
     tbase = (((long long)b * H + h) * D) * S;
@@ -203,7 +213,6 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
   __syncthreads();
 
   short* pw = pscr[wid];
-  short* dw = dscr[wid];
   const lds_chunk* kp = kscr[pair];
   const lds_chunk* vp = vscr[pair];
 
@@ -212,46 +221,51 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
     long long qbase, tbase, sbase;
     qb_of(it, qs, qbase, tbase, sbase);
 
-    if (!(VAR & 1) && (!CAUSAL || qs + T - 1 >= my_kv0)) {
-      // ---- S^T = K Q^T ; dP^T = V dO^T  (A from pair LDS, B from tiles)
-      f32x16 st = {}, dpt = {};
-      __builtin_amdgcn_s_setprio(1);
+    if (!(VAR & 1)) {
 #pragma unroll
-      for (int kk = 0; kk < 8; ++kk) {
-        const bf16x8s kA = kp[sw16(col, kk * 2 + half)];
-        const bf16x8s qB = qbuf[sw16(col, kk * 2 + half)];
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kA, qB, st, 0, 0, 0);
-      }
+      for (int sub = 0; sub < 2; ++sub) {       // two 32-row q subtiles
+        const int qs2 = qs + sub * T;
+        if (CAUSAL && qs2 + T - 1 < my_kv0) continue;
+        // ---- S^T = K Q^T ; dP^T = V dO^T (A from pair LDS, B from tiles)
+        f32x16 st = {}, dpt = {};
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int kk = 0; kk < 8; ++kk) {
-        const bf16x8s vA = vp[sw16(col, kk * 2 + half)];
-        const bf16x8s doB = dobuf[sw16(col, kk * 2 + half)];
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vA, doB, dpt, 0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
-
-      const int qrow = qs + col;
-      const float l = lse[sbase + qrow];
-      const float dl = delta[sbase + qrow];
-      if (VAR & 2) {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {  // keep st/dpt live, skip the rest
-          dvacc[0][r] += st[r] * l;
-          dkacc[0][r] += dpt[r] * dl;
+        for (int kk = 0; kk < 8; ++kk) {
+          const bf16x8s kA = kp[sw16(col, kk * 2 + half)];
+          const bf16x8s qB = qbuf[sw16(sub * T + col, kk * 2 + half)];
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kA, qB, st, 0, 0, 0);
         }
-      } else {
+#pragma unroll
+        for (int kk = 0; kk < 8; ++kk) {
+          const bf16x8s vA = vp[sw16(col, kk * 2 + half)];
+          const bf16x8s doB = dobuf[sw16(sub * T + col, kk * 2 + half)];
+          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vA, doB, dpt,
+                                                        0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+
+        const int qrow = qs2 + col;
+        const float l = lse[sbase + min(qrow, S - 1)];
+        const float dl = delta[sbase + min(qrow, S - 1)];
+        if (VAR & 2) {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {  // keep st/dpt live, skip the rest
+            dvacc[0][r] += st[r] * l;
+            dkacc[0][r] += dpt[r] * dl;
+          }
+          continue;
+        }
+        // pt pass: write P^T into the (single, reused) scratch
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int kvl = cd_row(r, half);  // kv row within my pair tile
           const bool dead = (CAUSAL && (my_kv0 + kvl > qrow)) ||
-                            (my_kv0 + kvl >= S);
+                            (my_kv0 + kvl >= S) || (qrow >= S);
           const float pt = dead ? 0.f : __expf(st[r] * scale - l);
-          const float dst = pt * (dpt[r] - dl) * scale;
-          scr_store(pw, kvl, col, f2bf(pt));   // [kv][q]
-          scr_store(dw, kvl, col, f2bf(dst));
+          st[r] = pt;                       // keep pt for the dst pass
+          scr_store(pw, kvl, col, f2bf(pt));
         }
-
-        // ---- dV[kv][d] += P^T(k=q) dO-asB ; dK[kv][d] += dS^T(k=q) Q-asB
+        // ---- dV[kv][d] += P^T(k=q) dO-asB
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int db = 0; db < 2; ++db) {
@@ -259,11 +273,30 @@ __global__ __launch_bounds__(TPB) void flash_bwd_dkdv_kernel(
 #pragma unroll
           for (int kk = 0; kk < 2; ++kk) {
             const bf16x8s pA = scr_read(pw, col, kk, half);
-            const bf16x8s dA = scr_read(dw, col, kk, half);
-            const bf16x8s doB = dotbuf[sw4(dblk * 32 + col, kk * 2 + half)];
-            const bf16x8s qB2 = qtbuf[sw4(dblk * 32 + col, kk * 2 + half)];
+            const bf16x8s doB =
+                dotbuf[sw8(dblk * 32 + col, sub * 4 + kk * 2 + half)];
             dvacc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 pA, doB, dvacc[db], 0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        // dst pass: overwrite the scratch (same-wave LDS RAW is ordered)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvl = cd_row(r, half);
+          const float dst = st[r] * (dpt[r] - dl) * scale;
+          scr_store(pw, kvl, col, f2bf(dst));
+        }
+        // ---- dK[kv][d] += dS^T(k=q) Q-asB
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int db = 0; db < 2; ++db) {
+          const int dblk = pm * 2 + db;
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk) {
+            const bf16x8s dA = scr_read(pw, col, kk, half);
+            const bf16x8s qB2 =
+                qtbuf[sw8(dblk * 32 + col, sub * 4 + kk * 2 + half)];
             dkacc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 dA, qB2, dkacc[db], 0, 0, 0);
           }
