@@ -63,6 +63,20 @@ extern "C" void ds_transpose_bf16(const void* src, void* dst, int n_batch,
                                   int R, int C, long long row_stride,
                                   int inner, long long inner_stride,
                                   long long outer_stride, void* stream);
+extern "C" void ds_fused_softmax(const void* x, const void* mask, void* y,
+                                 const float* alibi_slopes, long long rows,
+                                 int n, int heads, int sq, int mask_stride,
+                                 float scale, int causal, int dtype,
+                                 void* stream);
+extern "C" void ds_fused_dropout(const void* x, const void* bias,
+                                 const void* residual, void* y,
+                                 unsigned char* mask_out, long long n,
+                                 int cols, float ratio,
+                                 unsigned long long seed, int dtype,
+                                 void* stream);
+extern "C" void ds_dropout_bwd(const void* dy, const unsigned char* mask,
+                               void* dx, long long n, float ratio, int dtype,
+                               void* stream);
 extern "C" void* ds_aio_create(long long block_size, int n_threads);
 extern "C" void ds_aio_destroy(void* h);
 extern "C" int ds_aio_pwrite(void* h, const void* data, long long nbytes,
@@ -207,6 +221,55 @@ at::Tensor transpose_bf16(at::Tensor src, int64_t n_batch, int64_t R,
                     (int)C, row_stride, (int)inner, inner_stride,
                     outer_stride, cur_stream());
   return dst;
+}
+
+at::Tensor fused_softmax(at::Tensor x, c10::optional<at::Tensor> mask,
+                         c10::optional<at::Tensor> alibi, int64_t heads,
+                         int64_t sq, double scale, bool causal) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "fused_softmax: cuda");
+  const int n = x.size(-1);
+  const long long rows = x.numel() / n;
+  auto y = at::empty_like(x);
+  const void* mp = nullptr;
+  int mstride = 0;
+  if (mask.has_value() && mask->defined()) {
+    TORCH_CHECK(mask->scalar_type() == x.scalar_type(), "mask dtype");
+    mp = mask->data_ptr();
+    mstride = mask->numel() / mask->size(0);
+  }
+  const float* sl = nullptr;
+  if (alibi.has_value() && alibi->defined())
+    sl = alibi->data_ptr<float>();
+  ds_fused_softmax(x.data_ptr(), mp, y.data_ptr(), sl, rows, n, (int)heads,
+                   (int)sq, mstride, (float)scale, causal ? 1 : 0,
+                   dtype_code(x), cur_stream());
+  return y;
+}
+
+std::vector<at::Tensor> fused_dropout(at::Tensor x,
+                                      c10::optional<at::Tensor> bias,
+                                      c10::optional<at::Tensor> residual,
+                                      double ratio, int64_t seed) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "fused_dropout: cuda");
+  auto y = at::empty_like(x);
+  auto m = at::empty({x.numel()}, x.options().dtype(at::kByte));
+  const void* bp = bias.has_value() && bias->defined() ? bias->data_ptr()
+                                                       : nullptr;
+  const void* rp = residual.has_value() && residual->defined()
+                       ? residual->data_ptr() : nullptr;
+  ds_fused_dropout(x.data_ptr(), bp, rp, y.data_ptr(),
+                   m.data_ptr<unsigned char>(), x.numel(),
+                   (int)x.size(-1), (float)ratio,
+                   (unsigned long long)seed, dtype_code(x), cur_stream());
+  return {y, m};
+}
+
+at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double ratio) {
+  auto dx = at::empty_like(dy);
+  ds_dropout_bwd(dy.contiguous().data_ptr(),
+                 mask.data_ptr<unsigned char>(), dx.data_ptr(), dy.numel(),
+                 (float)ratio, dtype_code(dy), cur_stream());
+  return dx;
 }
 
 void cpu_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
@@ -433,6 +496,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                                 (int)variant, cur_stream());
         },
         "dkdv-only launcher (perf diagnosis)");
+  m.def("fused_softmax", &fused_softmax,
+        "fused masked/alibi/causal softmax");
+  m.def("fused_dropout", &fused_dropout, "fused bias+dropout(+residual)");
+  m.def("dropout_bwd", &dropout_bwd, "dropout backward from saved mask");
   m.def("transpose_bf16", &transpose_bf16, "tiled bf16 batched transpose");
   m.def("fp_quantize", &fp_quantize, "groupwise FP4/6/8/12 quantize");
   m.def("fp_dequantize", &fp_dequantize, "groupwise FP dequantize");

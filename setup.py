@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "optim.hip"),
         os.path.join(CSRC, "fp_quant.hip"),
         os.path.join(CSRC, "transpose.hip"),
+        os.path.join(CSRC, "softmax_dropout.hip"),
         os.path.join(CSRC, "quantize.hip"),
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "attention_bwd.hip"),
