@@ -613,3 +613,34 @@ def test_curriculum_dataloader_from_config():
         assert engine.curriculum_scheduler is not None
 
     run_local(worker)
+
+
+def test_contiguous_allocator_defrag():
+    import torch
+    from deepspeed_amd.runtime.zero.contiguous_allocator import (
+        ContiguousAllocator)
+    a = ContiguousAllocator(10240, dtype=torch.float32, device="cpu")
+    hs = [a.allocate(1500) for _ in range(6)]
+    for i, h in enumerate(hs):
+        h.tensor.fill_(float(i))
+    # free alternating -> fragmented: 3 live x 1500, but gaps of 1536
+    for h in hs[::2]:
+        h.release()
+    assert a.largest_free_block() < 4096
+    # a 4000-elem allocation only fits after defragmentation
+    big = a.allocate(4000)          # triggers defragment() internally
+    big.tensor.fill_(9.0)
+    # survivors kept their contents through the migration
+    for i, h in zip((1, 3, 5), hs[1::2]):
+        assert torch.all(h.tensor == float(i)), i
+    assert torch.all(big.tensor == 9.0)
+    # overlapping down-move correctness: one huge survivor shifted by less
+    # than its own length
+    b = ContiguousAllocator(8192, dtype=torch.float32, device="cpu")
+    small = b.allocate(512)
+    payload = b.allocate(6000)
+    payload.tensor.copy_(torch.arange(6000, dtype=torch.float32))
+    small.release()
+    b.defragment()
+    assert torch.equal(payload.tensor,
+                       torch.arange(6000, dtype=torch.float32))
