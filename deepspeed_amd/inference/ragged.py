@@ -383,6 +383,36 @@ class PagedKVCache:
         vb = vb.permute(0, 2, 1, 3, 4).reshape(n, Hkv, nblk * bs, D)
         return kb[:, :, :self._L], vb[:, :, :self._L]
 
+    def decode_attention(self, layer_idx: int, q, k, v, scale=None):
+        """Single-token decode THROUGH the paged pool: writes the new
+        token's k/v into its block and runs the split-S flash-decode HIP
+        kernel over the block table — the padded [n, Hkv, L, D] views are
+        never materialized (reference blocked_flash path). q/k/v are
+        [n, 1, H(kv), D] (BSHD step slices); returns [n, 1, H, D] or None
+        when ineligible (caller falls back to update() + SDPA)."""
+        from ..ops.paged_attention import (paged_decode_attention,
+                                           paged_decode_available)
+        if self._q_len != 1 or not q.is_cuda or q.dtype != torch.bfloat16:
+            return None
+        kp = self.k[layer_idx]
+        if not paged_decode_available(q.reshape(q.size(0), -1, q.size(-1)),
+                                      kp):
+            return None
+        bs = self.block_size
+        for i, s_ in enumerate(self._slots):
+            p = int(self._starts[i])
+            blk = self.block_table[s_][p // bs]
+            self.k[layer_idx][blk, :, p % bs] = k[i, 0]
+            self.v[layer_idx][blk, :, p % bs] = v[i, 0]
+        nblk = (int(self._starts.max()) + 1 + bs - 1) // bs
+        table = torch.tensor(
+            [(self.block_table[s_] + [0] * nblk)[:nblk]
+             for s_ in self._slots], dtype=torch.int32, device=q.device)
+        lens = (self._starts + 1).to(q.device, torch.int32)
+        o = paged_decode_attention(q[:, 0], kp, self.v[layer_idx], table,
+                                   lens, scale=scale)
+        return o.unsqueeze(1)
+
     def end_step(self):
         idx = torch.as_tensor(self._slots, device=self.lens.device)
         if self._packed:

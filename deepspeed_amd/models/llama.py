@@ -115,6 +115,16 @@ class LlamaAttention(nn.Module):
                 o = flash_attn_func(q, k, v, causal=True)
                 return self.o_proj(o.reshape(B, S, -1))
 
+        if (S == 1 and kv_cache is not None
+                and hasattr(kv_cache, "decode_attention")
+                and attention_fn is None):
+            # paged flash-decode kernel over the block table (no padded
+            # KV materialization); None -> ineligible, normal path below
+            o = kv_cache.decode_attention(self.layer_idx, q, k, v)
+            if o is not None:
+                kv_cache.last_decoded = True
+                return self.o_proj(o.reshape(B, S, -1))
+
         # [B, H, S, D] for SDPA
         q = q.transpose(1, 2)
         k = k.transpose(1, 2)

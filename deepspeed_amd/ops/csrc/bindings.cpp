@@ -59,6 +59,12 @@ extern "C" void ds_flash_bwd_dkdv_dbg(const void* q, const void* k,
                                       void* dk, void* dv, int B, int S,
                                       int H, int Hkv, float scale,
                                       int variant, void* stream);
+extern "C" void ds_paged_decode(const void* q, const void* kpool,
+                                const void* vpool, const int* block_table,
+                                const int* lens, float* part, float* part_ml,
+                                void* o, int n, int H, int Hkv, int BS,
+                                int max_blocks, int splits, float scale,
+                                void* stream);
 extern "C" void ds_transpose_bf16(const void* src, void* dst, int n_batch,
                                   int R, int C, long long row_stride,
                                   int inner, long long inner_stride,
@@ -270,6 +276,33 @@ at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double ratio) {
                  mask.data_ptr<unsigned char>(), dx.data_ptr(), dy.numel(),
                  (float)ratio, dtype_code(dy), cur_stream());
   return dx;
+}
+
+at::Tensor paged_decode(at::Tensor q, at::Tensor kpool, at::Tensor vpool,
+                        at::Tensor block_table, at::Tensor lens,
+                        int64_t splits, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+              q.is_contiguous() && kpool.is_contiguous() &&
+              vpool.is_contiguous(), "paged_decode: bf16 cuda contiguous");
+  TORCH_CHECK(block_table.scalar_type() == at::kInt &&
+              lens.scalar_type() == at::kInt, "table/lens must be int32");
+  const int n = q.size(0), H = q.size(1);
+  const int Hkv = kpool.size(1), BS = kpool.size(2);
+  TORCH_CHECK(q.size(2) == 128 && kpool.size(3) == 128, "D must be 128");
+  const int G = H / Hkv;
+  TORCH_CHECK(G == 1 || G == 2 || G == 4 || G == 8, "H/Hkv must be 1/2/4/8");
+  const int mb = block_table.size(1);
+  auto part = at::empty({(long)n, (long)Hkv, splits, (long)G, 128L},
+                        q.options().dtype(at::kFloat));
+  auto part_ml = at::empty({(long)n, (long)Hkv, splits, (long)G, 2L},
+                           q.options().dtype(at::kFloat));
+  auto o = at::empty_like(q);
+  ds_paged_decode(q.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+                  block_table.data_ptr<int>(), lens.data_ptr<int>(),
+                  part.data_ptr<float>(), part_ml.data_ptr<float>(),
+                  o.data_ptr(), n, H, Hkv, BS, mb, (int)splits, (float)scale,
+                  cur_stream());
+  return o;
 }
 
 void cpu_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
@@ -496,6 +529,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                                 (int)variant, cur_stream());
         },
         "dkdv-only launcher (perf diagnosis)");
+  m.def("paged_decode", &paged_decode,
+        "split-S flash-decode over a paged KV block table");
   m.def("fused_softmax", &fused_softmax,
         "fused masked/alibi/causal softmax");
   m.def("fused_dropout", &fused_dropout, "fused bias+dropout(+residual)");

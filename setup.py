@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "fp_quant.hip"),
         os.path.join(CSRC, "transpose.hip"),
         os.path.join(CSRC, "softmax_dropout.hip"),
+        os.path.join(CSRC, "paged_decode.hip"),
         os.path.join(CSRC, "quantize.hip"),
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "attention_bwd.hip"),

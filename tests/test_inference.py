@@ -396,3 +396,33 @@ def test_graph_generate_matches_eager_gpu():
     eng_eager.config.enable_cuda_graph = True
     got = eng_eager.generate(ids, max_new_tokens=12)
     assert torch.equal(got, ref), (got, ref)
+
+
+@pytest.mark.gpu
+def test_paged_decode_batcher_gpu():
+    """ContinuousBatcher on PagedKVCache with the flash-decode kernel on
+    the decode path produces the same greedy tokens as the slot cache."""
+    from deepspeed_amd.inference.ragged import (ContinuousBatcher,
+                                                PagedKVCache, RaggedKVCache,
+                                                Request)
+    from deepspeed_amd.models import LlamaForCausalLM
+    from deepspeed_amd.models.llama import LlamaConfig
+
+    cfg = LlamaConfig(vocab_size=512, hidden_size=256, intermediate_size=384,
+                      num_layers=2, num_heads=2, num_kv_heads=1,
+                      max_seq_len=256)  # head_dim = 128: kernel-eligible
+    torch.manual_seed(8)
+    model = LlamaForCausalLM(cfg).cuda().bfloat16().eval()
+    prompts = [[1, 2, 3, 4, 5], [9, 8, 7], [20] * 40]
+
+    def run(cache_cls):
+        b = ContinuousBatcher(model, max_slots=4, dtype=torch.bfloat16,
+                              cache_cls=cache_cls)
+        for i, p in enumerate(prompts):
+            b.put(Request(uid=i, prompt=torch.tensor(p), max_new_tokens=12))
+        done = b.run_to_completion()
+        return {r.uid: list(r.generated) for r in done}
+
+    ref = run(RaggedKVCache)
+    got = run(PagedKVCache)
+    assert ref == got, (ref, got)
