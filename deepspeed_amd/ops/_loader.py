@@ -21,6 +21,10 @@ def _try_import():
     if _tried:
         return _C
     _tried = True
+    if os.environ.get("DS_AMD_DISABLE_EXT") == "1":
+        # debugging kill switch: force every op onto the torch fallback
+        _C = None
+        return None
     try:
         from . import _C as mod  # type: ignore
         _C = mod
