@@ -67,32 +67,36 @@ class FusedExperts(nn.Module):
         self.d_model, self.d_ff = d_model, d_ff
         dt = template.gate_proj.weight.dtype
         e = num_local_experts
-        self.w_gate = nn.Parameter(torch.empty(e, d_ff, d_model, dtype=dt))
-        self.w_up = nn.Parameter(torch.empty(e, d_ff, d_model, dtype=dt))
-        self.w_down = nn.Parameter(torch.empty(e, d_model, d_ff, dtype=dt))
+        # weights stored PRE-TRANSPOSED ([E, in, out]) so every bmm gets a
+        # contiguous B operand — besides skipping the strided-B path,
+        # transposed-view B at large shapes hit a hipBLASLt memory fault
+        # (r2 call 15: [8,5120,1024]x[8,1024,2816] bf16)
+        self.w_gate = nn.Parameter(torch.empty(e, d_model, d_ff, dtype=dt))
+        self.w_up = nn.Parameter(torch.empty(e, d_model, d_ff, dtype=dt))
+        self.w_down = nn.Parameter(torch.empty(e, d_ff, d_model, dtype=dt))
         self.reset_parameters()
         # expert 0 keeps the template's weights (parity with deepcopy init)
         with torch.no_grad():
-            self.w_gate[0].copy_(template.gate_proj.weight)
-            self.w_up[0].copy_(template.up_proj.weight)
-            self.w_down[0].copy_(template.down_proj.weight)
+            self.w_gate[0].copy_(template.gate_proj.weight.t())
+            self.w_up[0].copy_(template.up_proj.weight.t())
+            self.w_down[0].copy_(template.down_proj.weight.t())
         for p in self.parameters():
             p.allreduce = False
             p.group_name = expert_group_name
 
     def expert_parameters(self, j):
-        """Per-expert views in LlamaMLP parameter order
-        (gate_proj.weight, up_proj.weight, down_proj.weight) — used by
-        tests/tools that address experts individually."""
-        return [self.w_gate[j], self.w_up[j], self.w_down[j]]
+        """Per-expert views in LlamaMLP parameter order and SHAPE
+        ([out, in], transposed views of the stored [in, out] weights) —
+        used by tests/tools that address experts individually."""
+        return [self.w_gate[j].t(), self.w_up[j].t(), self.w_down[j].t()]
 
     def reset_parameters(self, std: float = 0.02):
         for w in (self.w_gate, self.w_up, self.w_down):
             w.data.normal_(0.0, std)
 
     def forward(self, x):
-        # x: [E_local, N, d_model]
-        g = torch.bmm(x, self.w_gate.transpose(1, 2))
-        u = torch.bmm(x, self.w_up.transpose(1, 2))
+        # x: [E_local, N, d_model]; weights already [E, in, out]
+        g = torch.bmm(x, self.w_gate)
+        u = torch.bmm(x, self.w_up)
         h = swiglu(g, u)
-        return torch.bmm(h, self.w_down.transpose(1, 2))
+        return torch.bmm(h, self.w_down)
