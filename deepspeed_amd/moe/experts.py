@@ -94,9 +94,23 @@ class FusedExperts(nn.Module):
         for w in (self.w_gate, self.w_up, self.w_down):
             w.data.normal_(0.0, std)
 
+    # This torch/hipBLASLt build memory-faults on bf16 bmm at large
+    # grouped shapes (probe r2_call16: [8,5120,1024]x[8,1024,2816] faults
+    # with BOTH contiguous and strided B; [8,512,1024] works). The grouped
+    # launch only matters when per-expert GEMMs are small/launch-bound, so
+    # route: bmm below the validated boundary, per-expert full-tile GEMMs
+    # (one hipBLASLt launch each, compute-bound anyway) above it.
+    _BMM_MAX_TOKENS = 512
+
     def forward(self, x):
         # x: [E_local, N, d_model]; weights already [E, in, out]
-        g = torch.bmm(x, self.w_gate)
-        u = torch.bmm(x, self.w_up)
-        h = swiglu(g, u)
-        return torch.bmm(h, self.w_down)
+        if x.size(1) <= self._BMM_MAX_TOKENS or not x.is_cuda:
+            g = torch.bmm(x, self.w_gate)
+            u = torch.bmm(x, self.w_up)
+            h = swiglu(g, u)
+            return torch.bmm(h, self.w_down)
+        outs = []
+        for e in range(self.num_local_experts):
+            h = swiglu(x[e] @ self.w_gate[e], x[e] @ self.w_up[e])
+            outs.append(h @ self.w_down[e])
+        return torch.stack(outs, dim=0)
