@@ -95,7 +95,7 @@ def graph_generate(module, input_ids, *, n_layers, kv_heads, head_dim,
     in_buf = next_tok.clone()
 
     def step():
-        pos = kv.len_t.to(torch.int32).reshape(1, 1).expand(B, 1)
+        pos = kv.len_t.to(torch.int32).repeat(B).view(B, 1)
         lg = module(in_buf, positions=pos, kv_cache=kv)
         kv.len_t.add_(1)  # in-graph: each replay self-advances
         return lg[:, -1]

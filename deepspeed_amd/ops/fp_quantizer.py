@@ -102,7 +102,7 @@ def fp_emulate_reference(x: torch.Tensor, bits: int = 6,
     g = flat.view(groups, group_size)
     amax = g.abs().amax(dim=1, keepdim=True)
     scale = torch.where(amax > 0, amax / qmax, torch.ones_like(amax))
-    y = g / scale
+    y = g * (1.0 / scale)  # reciprocal-multiply, matching the kernel
     # quantize each value to the e/m grid with RNE
     mag = y.abs()
     expo = torch.floor(torch.log2(mag.clamp_min(1e-45)))

@@ -37,7 +37,7 @@ def _curve_worker(rank, world, stage):
                for _ in range(world)]
     eng_curve, ref_curve = [], []
     for _ in range(STEPS):
-        ids = batches[rank]
+        ids = batches[rank].to(engine.device)
         loss = engine(ids, labels=ids)
         engine.backward(loss)
         engine.step()

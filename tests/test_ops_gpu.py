@@ -451,17 +451,19 @@ def test_fused_dropout_gpu():
     y = fused_bias_dropout_residual(x, bias, res, ratio=0.3, seed=7)
     y2 = fused_bias_dropout_residual(x, bias, res, ratio=0.3, seed=7)
     assert torch.equal(y, y2)  # seed-deterministic
-    kept = (y - res).abs() > 0
+    y.sum().backward()
+    # the backward reveals the exact keep mask (grad = 1/(1-p) on kept)
+    kept = x.grad.float() != 0
     rate = kept.float().mean().item()
     assert abs(rate - 0.7) < 0.03, rate
-    # kept elements are scaled by 1/(1-p)
-    ref = (x + bias).float() / 0.7 + res.float()
-    torch.testing.assert_close(y.float()[kept], ref[kept], rtol=2e-2,
-                               atol=2e-2)
-    y.sum().backward()
-    # grad flows only through kept elements, scaled
-    g = x.grad.float()
-    assert torch.equal(g != 0, kept)
+    torch.testing.assert_close(
+        x.grad.float()[kept],
+        torch.full_like(x.grad.float()[kept], 1 / 0.7), rtol=1e-2,
+        atol=1e-3)
+    # forward values: kept scaled by 1/(1-p) (+res), dropped = res
+    ref = torch.where(kept, (x.detach() + bias.detach()).float() / 0.7,
+                      torch.zeros(())) + res.float()
+    torch.testing.assert_close(y.float(), ref, rtol=3e-2, atol=3e-2)
 
 
 @pytest.mark.gpu
