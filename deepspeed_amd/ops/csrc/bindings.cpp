@@ -197,8 +197,8 @@ std::vector<at::Tensor> fp_quantize(at::Tensor x, int64_t group_size,
   const long long n = x.numel();
   const long long groups = (n + group_size - 1) / group_size;
   const int vper3 = 24 / (int)bits;
-  const long long n3 = (n + vper3 - 1) / vper3;
-  auto out = at::empty({n3 * 3}, x.options().dtype(at::kByte));
+  const long long gu = (group_size + vper3 - 1) / vper3;  // units/group
+  auto out = at::empty({groups * gu * 3}, x.options().dtype(at::kByte));
   auto scales = at::empty({groups}, x.options().dtype(at::kFloat));
   ds_fp_quantize(x.data_ptr(), dtype_code(x), out.data_ptr(),
                  scales.data_ptr<float>(), n, (int)group_size, (int)bits, 0,

@@ -106,10 +106,13 @@ __global__ void fp_quant_kernel(const T* __restrict__ x,
   if (lane == 0) scales[g] = scale;
 
   // pack: each lane handles 24 bits = LCM-friendly unit (values_per_3B:
-  // BITS=4 -> 6, 6 -> 4, 8 -> 3, 12 -> 2)
+  // BITS=4 -> 6, 6 -> 4, 8 -> 3, 12 -> 2). Groups are byte-padded to a
+  // whole number of 3-byte units so group boundaries never split a unit
+  // (group_size need not divide vper3).
   const int vper3 = 24 / BITS;
-  const int n3 = (len + vper3 - 1) / vper3;       // 3-byte units in group
-  unsigned char* gout = out + (g0 / vper3) * 3;   // group's byte base
+  const int gu = (group_size + vper3 - 1) / vper3;  // units per group
+  const int n3 = (len + vper3 - 1) / vper3;         // units in THIS group
+  unsigned char* gout = out + (long long)g * gu * 3;
   for (int u = lane; u < n3; u += 64) {
     unsigned word = 0;
     for (int t = 0; t < vper3; ++t) {
@@ -136,8 +139,9 @@ __global__ void fp_dequant_kernel(const unsigned char* __restrict__ qd,
   const int len = (int)min((long long)group_size, n - g0);
   const float scale = scales[g];
   const int vper3 = 24 / BITS;
+  const int gu = (group_size + vper3 - 1) / vper3;
   const int n3 = (len + vper3 - 1) / vper3;
-  const unsigned char* gin = qd + (g0 / vper3) * 3;
+  const unsigned char* gin = qd + (long long)g * gu * 3;
   for (int u = lane; u < n3; u += 64) {
     unsigned word = (unsigned)gin[u * 3] | ((unsigned)gin[u * 3 + 1] << 8) |
                     ((unsigned)gin[u * 3 + 2] << 16);

@@ -381,20 +381,49 @@ def test_graph_kv_cache_matches_static_eager():
 
 @pytest.mark.gpu
 def test_graph_generate_matches_eager_gpu():
-    """hipGraph-captured decode replays must equal eager greedy decode."""
+    """hipGraph-replayed decode must match the SAME static-shape decode
+    math executed eagerly (GraphKVCache without capture) — replay equals
+    eager kernel-for-kernel. (Comparing against the dynamic-slice cache
+    instead would flip greedy argmax on random-init near-ties: different
+    reduction lengths.)"""
     import torch
     import deepspeed_amd
     from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    from deepspeed_amd.inference.graph import GraphKVCache, graph_generate
 
     cfg = llama_tiny()
     torch.manual_seed(5)
     model = LlamaForCausalLM(cfg)
-    eng_eager = deepspeed_amd.init_inference(
-        model, dtype="bf16", replace_with_kernel_inject=False)
+    eng = deepspeed_amd.init_inference(model, dtype="bf16",
+                                       replace_with_kernel_inject=False)
     ids = torch.randint(0, cfg.vocab_size, (2, 8), device="cuda")
-    ref = eng_eager.generate(ids, max_new_tokens=12)
-    eng_eager.config.enable_cuda_graph = True
-    got = eng_eager.generate(ids, max_new_tokens=12)
+
+    got = graph_generate(eng.module, ids, n_layers=cfg.num_layers,
+                         kv_heads=cfg.num_kv_heads, head_dim=cfg.head_dim,
+                         max_seq=cfg.max_seq_len, dtype=torch.bfloat16,
+                         max_new_tokens=12)
+
+    # eager drive of the identical static-shape cache
+    m = eng.module
+    kvg = GraphKVCache(cfg.num_layers, 2, cfg.num_kv_heads,
+                       min(8 + 12, cfg.max_seq_len), cfg.head_dim,
+                       dtype=torch.bfloat16, device="cuda")
+    pos = torch.arange(8, dtype=torch.int32,
+                       device="cuda").expand(2, 8).contiguous()
+    with torch.no_grad():
+        logits = m(ids, positions=pos, kv_cache=kvg)
+        kvg.len_t.fill_(8)
+        kvg.advance(8)
+        tok = logits[:, -1].argmax(-1, keepdim=True)
+        outs = [ids, tok]
+        for _ in range(11):
+            p1 = kvg.len_t.to(torch.int32).repeat(2).view(2, 1)
+            logits = m(tok, positions=p1, kv_cache=kvg)
+            kvg.len_t.add_(1)
+            kvg.advance()
+            tok = logits[:, -1].argmax(-1, keepdim=True)
+            outs.append(tok)
+    ref = torch.cat(outs, dim=1)
     assert torch.equal(got, ref), (got, ref)
 
 
