@@ -113,7 +113,9 @@ def graph_generate(module, input_ids, *, n_layers, kv_heads, head_dim,
     graph = torch.cuda.CUDAGraph()
     with torch.cuda.graph(graph):
         static_logits = step()
-    # the capture itself ran one real step
+    # capture RECORDS the step without executing it — replay once so the
+    # first decode token's k/v actually lands in the cache
+    graph.replay()
     kv.advance()
     next_tok = _select_token(static_logits, do_sample, temperature, top_k,
                              top_p)
