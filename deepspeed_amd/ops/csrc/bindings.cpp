@@ -83,6 +83,10 @@ extern "C" void ds_fused_dropout(const void* x, const void* bias,
 extern "C" void ds_dropout_bwd(const void* dy, const unsigned char* mask,
                                void* dx, long long n, float ratio, int dtype,
                                void* stream);
+extern "C" void* ds_shm_open(const char* name, int rank, int world,
+                             long long max_elems);
+extern "C" int ds_shm_allreduce(void* handle, float* data, long long n);
+extern "C" void ds_shm_close(void* handle);
 extern "C" void* ds_aio_create(long long block_size, int n_threads);
 extern "C" void ds_aio_destroy(void* h);
 extern "C" int ds_aio_pwrite(void* h, const void* data, long long nbytes,
@@ -304,6 +308,27 @@ at::Tensor paged_decode(at::Tensor q, at::Tensor kpool, at::Tensor vpool,
                   cur_stream());
   return o;
 }
+
+class ShmComm {
+ public:
+  ShmComm(const std::string& name, int64_t rank, int64_t world,
+          int64_t max_elems) {
+    h_ = ds_shm_open(name.c_str(), (int)rank, (int)world, max_elems);
+    TORCH_CHECK(h_ != nullptr, "shm_open failed for group ", name);
+  }
+  ~ShmComm() {
+    if (h_) ds_shm_close(h_);
+  }
+  void all_reduce(at::Tensor t) {
+    TORCH_CHECK(t.device().is_cpu() && t.scalar_type() == at::kFloat &&
+                t.is_contiguous(), "shm all_reduce: contiguous fp32 CPU");
+    TORCH_CHECK(ds_shm_allreduce(h_, t.data_ptr<float>(), t.numel()) == 0,
+                "shm all_reduce failed (tensor larger than max_elems?)");
+  }
+
+ private:
+  void* h_ = nullptr;
+};
 
 void cpu_adam_flat(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                    c10::optional<at::Tensor> p16, double lr, double beta1,
@@ -529,6 +554,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
                                 (int)variant, cur_stream());
         },
         "dkdv-only launcher (perf diagnosis)");
+  pybind11::class_<ShmComm>(m, "ShmComm")
+      .def(pybind11::init<const std::string&, int64_t, int64_t, int64_t>())
+      .def("all_reduce", &ShmComm::all_reduce);
   m.def("paged_decode", &paged_decode,
         "split-S flash-decode over a paged KV block table");
   m.def("fused_softmax", &fused_softmax,

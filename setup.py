@@ -22,6 +22,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "bindings.cpp"),
         os.path.join(CSRC, "cpu_adam.cpp"),
         os.path.join(CSRC, "aio.cpp"),
+        os.path.join(CSRC, "shm_comm.cpp"),
         os.path.join(CSRC, "adam.hip"),
         os.path.join(CSRC, "optim.hip"),
         os.path.join(CSRC, "fp_quant.hip"),
