@@ -673,6 +673,9 @@ class Engine(torch.nn.Module):
             }
             self.checkpoint_engine.save(state, self._model_ckpt_name(ckpt_dir))
 
+        if self.has_moe_layers:
+            self._save_moe_checkpoint(ckpt_dir)
+
         if self.optimizer is not None and hasattr(self.optimizer, "state_dict") and \
                 not isinstance(self.optimizer, DummyOptim):
             opt_state = {"optimizer_state_dict": self.optimizer.state_dict()}
@@ -685,6 +688,52 @@ class Engine(torch.nn.Module):
                 f.write(tag)
         dist.barrier()
         return True
+
+    def _expert_state(self):
+        """This rank's expert parameters/buffers (tagged allreduce=False)."""
+        out = {}
+        for n, p in self.module.named_parameters():
+            if getattr(p, "allreduce", True) is False:
+                out[n] = p.detach().cpu()
+        return out
+
+    def _moe_ckpt_name(self, ckpt_dir, ep_rank):
+        return os.path.join(ckpt_dir,
+                            f"expert_ep_rank_{ep_rank}_model_states.pt")
+
+    def _save_moe_checkpoint(self, ckpt_dir):
+        """With EP > 1 every EP rank owns DIFFERENT experts, so the dense
+        mp_rank_00 file (written by dp-rank 0 only) cannot carry them:
+        each expert shard is saved once by its expert-data-parallel rank 0
+        (reference engine._save_moe_checkpoint:3319)."""
+        from ..moe.layer import MoE
+        names = {m.expert_group_name for m in self.module.modules()
+                 if isinstance(m, MoE)}
+        if not names:
+            return
+        name = next(iter(names))
+        edp = groups.get_expert_data_parallel_group(name)
+        if dist.get_rank(edp) == 0:
+            ep_rank = groups.get_expert_parallel_rank(name)
+            self.checkpoint_engine.save(self._expert_state(),
+                                        self._moe_ckpt_name(ckpt_dir,
+                                                            ep_rank))
+
+    def _load_moe_checkpoint(self, ckpt_dir):
+        from ..moe.layer import MoE
+        names = {m.expert_group_name for m in self.module.modules()
+                 if isinstance(m, MoE)}
+        if not names:
+            return
+        name = next(iter(names))
+        ep_rank = groups.get_expert_parallel_rank(name)
+        path = self._moe_ckpt_name(ckpt_dir, ep_rank)
+        if not os.path.exists(path):
+            logger.warning(f"no expert checkpoint for ep_rank {ep_rank}")
+            return
+        expert_sd = self.checkpoint_engine.load(path, map_location="cpu")
+        missing = self.module.load_state_dict(expert_sd, strict=False)
+        del missing  # only experts in this file by construction
 
     def module_state_dict(self, exclude_frozen_parameters=False):
         if self.zero_stage == 3 and hasattr(self.optimizer, "full_state_dict"):
@@ -715,7 +764,11 @@ class Engine(torch.nn.Module):
             self.optimizer.load_full_state_dict(state["module"],
                                                 strict=load_module_strict)
         else:
-            self.module.load_state_dict(state["module"], strict=load_module_strict)
+            self.module.load_state_dict(
+                state["module"],
+                strict=load_module_strict and not self.has_moe_layers)
+        if self.has_moe_layers:
+            self._load_moe_checkpoint(ckpt_dir)
         self.global_steps = state.get("global_steps", 0)
         self.global_samples = state.get("global_samples", 0)
         self.skipped_steps = state.get("skipped_steps", 0)
