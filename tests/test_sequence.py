@@ -267,3 +267,63 @@ def test_fpdt_module_in_ulysses():
     assert out.shape == (B, S, H, D)
     out.sum().backward()
     assert torch.isfinite(q.grad).all()
+
+
+def _ulysses_dp2sp2_worker(rank, world):
+    """Hybrid DP2 x SP2 mesh: Ulysses pairs split the sequence, ZeRO-1
+    runs over the sequence-DATA-parallel complement (reference mesh
+    device, deepspeed/__init__.py:156 + groups.py:591)."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM
+    from deepspeed_amd.models.llama import LlamaConfig, enable_ulysses
+    from deepspeed_amd.parallel import groups
+
+    sp = 2
+    groups.initialize_sequence_parallel(sp)
+    cfg = LlamaConfig(vocab_size=256, hidden_size=64, intermediate_size=96,
+                      num_layers=2, num_heads=4, num_kv_heads=2,
+                      max_seq_len=64)
+    torch.manual_seed(41)
+    model = LlamaForCausalLM(cfg)
+    enable_ulysses(model)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": 1, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    # by design ZeRO shards over the FULL DPxSP mesh (model replicated on
+    # every rank; averaging over world = exact grad of the global mean)
+    assert engine.dp_world_size == world
+    assert groups.get_sequence_parallel_world_size() == sp
+
+    torch.manual_seed(41)
+    ref = LlamaForCausalLM(cfg)
+    opt_ref = torch.optim.AdamW(ref.parameters(), lr=1e-3)
+
+    sp_rank = rank % sp          # contiguous SP pairs
+    dp_rank = rank // sp
+    S = 32
+    torch.manual_seed(90)
+    for _ in range(2):
+        batches = [torch.randint(0, 256, (2, S + 1)) for _ in range(2)]
+        ids = batches[dp_rank]
+        x, y = ids[:, :-1], ids[:, 1:]
+        sl = slice(sp_rank * S // sp, (sp_rank + 1) * S // sp)
+        pos = torch.arange(S, dtype=torch.int32).expand(2, S)
+        loss = engine(x[:, sl].contiguous(), labels=y[:, sl].contiguous(),
+                      positions=pos[:, sl].contiguous())
+        engine.backward(loss)
+        engine.step()
+
+        for d in range(2):
+            bx, by = batches[d][:, :-1], batches[d][:, 1:]
+            l2 = ref(bx, labels=by)
+            (l2 / 2).backward()
+        opt_ref.step()
+        opt_ref.zero_grad()
+    for (n, p), (_, pr) in zip(engine.module.named_parameters(),
+                               ref.named_parameters()):
+        torch.testing.assert_close(p, pr, rtol=3e-4, atol=5e-4), n
+
+
+def test_ulysses_dp2_sp2_mesh():
+    run_distributed(_ulysses_dp2sp2_worker, world_size=4)
