@@ -15,6 +15,7 @@
 #include <atomic>
 #include <condition_variable>
 #include <cstdint>
+#include <cstdlib>
 #include <cstring>
 #include <functional>
 #include <mutex>
@@ -85,20 +86,68 @@ class ThreadPool {
 
 class AioEngine {
  public:
-  AioEngine(int64_t block_size, int n_threads)
-      : block_size_(block_size), pool_(n_threads), errors_(0) {}
+  AioEngine(int64_t block_size, int n_threads, bool o_direct = false)
+      : block_size_(block_size), pool_(n_threads), errors_(0),
+        o_direct_(o_direct && block_size % 4096 == 0) {}
+
+  static constexpr int64_t kAlign = 4096;
+
+  // thread-local 4K-aligned bounce buffer for O_DIRECT transfers
+  char* bounce(int64_t len) {
+    thread_local char* buf = nullptr;
+    thread_local int64_t cap = 0;
+    if (cap < len) {
+      if (buf) ::free(buf);
+      cap = (len + kAlign - 1) / kAlign * kAlign;
+      buf = static_cast<char*>(::aligned_alloc(kAlign, cap));
+    }
+    return buf;
+  }
+
+  int open_write(const std::string& path) {
+    if (o_direct_) {
+      int fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_TRUNC | O_DIRECT,
+                      0644);
+      if (fd >= 0) return fd;  // else: fs without O_DIRECT -> fall back
+    }
+    return ::open(path.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
+  }
+
+  int open_read(const std::string& path) {
+    if (o_direct_) {
+      int fd = ::open(path.c_str(), O_RDONLY | O_DIRECT);
+      if (fd >= 0) return fd;
+    }
+    return ::open(path.c_str(), O_RDONLY);
+  }
 
   void pwrite(const void* data, int64_t nbytes, const std::string& path) {
-    int fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
+    int fd = open_write(path);
     if (fd < 0) throw std::runtime_error("aio: cannot open " + path);
-    if (::ftruncate(fd, nbytes) != 0) {
+    const int fl = ::fcntl(fd, F_GETFL);
+    const bool direct = (fl & O_DIRECT) != 0;
+    if (!direct && ::ftruncate(fd, nbytes) != 0) {
       ::close(fd);
       throw std::runtime_error("aio: ftruncate failed for " + path);
     }
     const char* p = static_cast<const char*>(data);
     for (int64_t off = 0; off < nbytes; off += block_size_) {
       const int64_t len = std::min(block_size_, nbytes - off);
-      pool_.submit([this, fd, p, off, len] {
+      pool_.submit([this, fd, p, off, len, direct] {
+        if (direct) {
+          // O_DIRECT: aligned bounce buffer, length rounded up to 4K
+          const int64_t wlen = (len + kAlign - 1) / kAlign * kAlign;
+          char* buf = bounce(wlen);
+          std::memcpy(buf, p + off, len);
+          if (wlen > len) std::memset(buf + len, 0, wlen - len);
+          int64_t done = 0;
+          while (done < wlen) {
+            ssize_t w = ::pwrite(fd, buf + done, wlen - done, off + done);
+            if (w <= 0) { ++errors_; return; }
+            done += w;
+          }
+          return;
+        }
         int64_t done = 0;
         while (done < len) {
           ssize_t w = ::pwrite(fd, p + off + done, len - done, off + done);
@@ -111,15 +160,32 @@ class AioEngine {
       });
     }
     fds_.push_back(fd);
+    if (direct) trunc_targets_.emplace_back(fd, nbytes);
   }
 
   void pread(void* data, int64_t nbytes, const std::string& path) {
-    int fd = ::open(path.c_str(), O_RDONLY);
+    int fd = open_read(path);
     if (fd < 0) throw std::runtime_error("aio: cannot open " + path);
+    const int fl = ::fcntl(fd, F_GETFL);
+    const bool direct = (fl & O_DIRECT) != 0;
     char* p = static_cast<char*>(data);
     for (int64_t off = 0; off < nbytes; off += block_size_) {
       const int64_t len = std::min(block_size_, nbytes - off);
-      pool_.submit([this, fd, p, off, len] {
+      pool_.submit([this, fd, p, off, len, direct] {
+        if (direct) {
+          const int64_t rlen = (len + kAlign - 1) / kAlign * kAlign;
+          char* buf = bounce(rlen);
+          int64_t done = 0;
+          while (done < rlen) {
+            ssize_t r = ::pread(fd, buf + done, rlen - done, off + done);
+            if (r < 0) { ++errors_; return; }
+            if (r == 0) break;  // EOF inside the rounded-up tail
+            done += r;
+          }
+          if (done < len) { ++errors_; return; }
+          std::memcpy(p + off, buf, len);
+          return;
+        }
         int64_t done = 0;
         while (done < len) {
           ssize_t r = ::pread(fd, p + off + done, len - done, off + done);
@@ -136,6 +202,9 @@ class AioEngine {
 
   int wait() {
     pool_.wait_all();
+    // O_DIRECT writes rounded the tail up; restore exact file sizes
+    for (auto& t : trunc_targets_) ::ftruncate(t.first, t.second);
+    trunc_targets_.clear();
     for (int fd : fds_) ::close(fd);
     fds_.clear();
     int e = errors_.exchange(0);
@@ -144,6 +213,8 @@ class AioEngine {
 
  private:
   int64_t block_size_;
+  bool o_direct_;
+  std::vector<std::pair<int, int64_t>> trunc_targets_;
   ThreadPool pool_;
   std::vector<int> fds_;
   std::atomic<int> errors_;
@@ -153,6 +224,12 @@ class AioEngine {
 
 extern "C" {
 void* ds_aio_create(long long block_size, int n_threads) {
+  // DS_AIO_O_DIRECT=1 bypasses the page cache (real-NVMe measurement
+  // mode, reference csrc/aio O_DIRECT path); transparent fallback when
+  // the filesystem refuses O_DIRECT or block_size is not 4K-aligned.
+  const char* od = ::getenv("DS_AIO_O_DIRECT");
+  if (od != nullptr && od[0] == '1')
+    return new dsaio::AioEngine(block_size, n_threads, true);
   return new dsaio::AioEngine(block_size, n_threads);
 }
 void ds_aio_destroy(void* h) { delete static_cast<dsaio::AioEngine*>(h); }
