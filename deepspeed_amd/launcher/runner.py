@@ -47,6 +47,10 @@ def parse_args(args=None):
     p.add_argument("--num_gpus", "--num-gpus", type=int, default=-1,
                    help="GPUs on this node (-1 = all visible)")
     p.add_argument("--num_nodes", type=int, default=1)
+    p.add_argument("--launcher", type=str, default="ssh",
+                   choices=["ssh", "pdsh", "slurm", "openmpi", "mpich",
+                            "impi"],
+                   help="multinode backend (reference multinode_runner.py)")
     p.add_argument("--node_rank", type=int, default=0)
     p.add_argument("--hostfile", type=str, default=None)
     p.add_argument("--master_addr", type=str, default="127.0.0.1")
@@ -170,6 +174,16 @@ def main(argv=None):
     if args.hostfile:
         hosts = parse_hostfile(args.hostfile)
         if len(hosts) > 1:
+            if args.launcher != "ssh":
+                from .multinode_runner import build_runner
+                runner = build_runner(args.launcher, args, hosts)
+                if not runner.backend_exists():
+                    raise SystemExit(
+                        f"launcher backend '{args.launcher}' not found on "
+                        "PATH")
+                user_cmd = ([args.user_script] if not args.module else
+                            ["-m", args.user_script]) + args.user_args
+                return sys.exit(runner.run(user_cmd))
             return sys.exit(launch_multinode(args, hosts))
         args.num_gpus = args.num_gpus if args.num_gpus > 0 \
             else next(iter(hosts.values()))

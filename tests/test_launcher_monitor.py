@@ -152,3 +152,61 @@ def test_launcher_elastic_restarts_exhausted(tmp_path):
         cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert out.returncode == 7
     assert "elastic restart 1/1" in out.stdout + out.stderr
+
+
+def test_multinode_runner_commands():
+    """Parse-level runner command construction (reference
+    tests/unit/launcher: command assembly without a cluster)."""
+    import argparse
+    from deepspeed_amd.launcher.multinode_runner import (build_runner,
+                                                         RUNNERS)
+    args = argparse.Namespace(master_port=29500, module=False)
+    hosts = {"node1": 8, "node2": 8}
+    user = ["train.py", "--lr", "1e-4"]
+
+    slurm = build_runner("slurm", args, hosts)
+    cmd = slurm.get_cmd(user)
+    assert cmd[0] == "srun" and "--ntasks" in cmd and "16" in cmd
+    assert "train.py" in cmd
+
+    ompi = build_runner("openmpi", args, hosts)
+    cmd = ompi.get_cmd(user)
+    assert cmd[0] == "mpirun" and "-n" in cmd and "16" in cmd
+    assert "node1:8" in cmd
+
+    mpich = build_runner("mpich", args, hosts)
+    cmd = mpich.get_cmd(user)
+    assert cmd[0] == "mpiexec" and "node1:8,node2:8" in cmd
+
+    impi = build_runner("impi", args, hosts)
+    cmd = impi.get_cmd(user)
+    assert "-ppn" in cmd
+
+    pdsh = build_runner("pdsh", args, hosts)
+    cmd = pdsh.get_cmd(user)
+    assert cmd[0] == "pdsh" and "node1,node2" in cmd
+    assert "node_rank" in " ".join(cmd) or "--node_rank" in cmd[-1]
+
+    assert set(RUNNERS) == {"pdsh", "slurm", "openmpi", "mpich", "impi"}
+
+    try:
+        build_runner("bogus", args, hosts)
+        assert False
+    except ValueError:
+        pass
+
+
+def test_bin_cli_entry_points():
+    """bin/ scripts exist, are executable, and --launcher parses."""
+    import os
+    import stat
+    from deepspeed_amd.launcher.runner import parse_args
+    root = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "bin")
+    for name in ("deepspeed", "ds_report", "ds_io", "ds_nvme_tune",
+                 "ds_bench"):
+        p = os.path.join(root, name)
+        assert os.path.exists(p), p
+        assert os.stat(p).st_mode & stat.S_IXUSR
+    a = parse_args(["--launcher", "slurm", "--num_gpus", "4", "x.py"])
+    assert a.launcher == "slurm" and a.user_script == "x.py"
