@@ -50,35 +50,44 @@ def ds_to_universal(load_dir, tag=None, out_dir=None):
     out = {"param": {}, "exp_avg": {}, "exp_avg_sq": {}, "step": 0}
     n_buckets = len(layout)
     for bi in range(n_buckets):
-        b0 = states[0]["layout"][bi]
-        gi = b0["group_idx"]
-        pg_world = b0["pg_world"]
-        # ranks of the global DP world that hold this bucket's shards, in
-        # shard order (for expert-DP buckets only a subset of files has them;
-        # match by pg_rank recorded per file)
-        shards = {}
+        # Partition files by the bucket's PARAM IDENTITY: with expert
+        # parallelism, bucket bi on different EP ranks holds DIFFERENT
+        # experts under distinct @epR-decorated names (same pg_rank values
+        # within each expert-DP group) — each identity assembles its own
+        # full flat and emits its own per-param entries.
+        by_identity = {}
         for sd in states:
             lb = sd["layout"][bi]
-            mo, ss = lb["master_offset"], lb["shard_size"]
-            flat = sd["fp32_flat_groups"][gi]
-            base = sd["base_optimizer_state"]["state"].get(gi, {})
-            shards[lb["pg_rank"]] = {
-                "param": flat[mo:mo + ss],
-                "exp_avg": base.get("exp_avg",
-                                    torch.zeros(ss))[mo:mo + ss],
-                "exp_avg_sq": base.get("exp_avg_sq",
-                                       torch.zeros(ss))[mo:mo + ss],
-            }
-            step = base.get("step", 0)
-            out["step"] = int(step.item() if torch.is_tensor(step) else step)
-        assert len(shards) == pg_world, \
-            f"bucket {bi}: found {len(shards)} shards, expected {pg_world}"
-        for kind in ("param", "exp_avg", "exp_avg_sq"):
-            full = torch.cat([shards[r][kind] for r in range(pg_world)])
-            for name, off, numel, shape in b0["params"]:
-                if name is None:
-                    continue
-                out[kind][name] = full[off:off + numel].view(shape).clone()
+            key = tuple(n for n, *_ in lb["params"])
+            by_identity.setdefault(key, []).append(sd)
+        for members in by_identity.values():
+            b0 = members[0]["layout"][bi]
+            gi = b0["group_idx"]
+            pg_world = b0["pg_world"]
+            shards = {}
+            for sd in members:
+                lb = sd["layout"][bi]
+                mo, ss = lb["master_offset"], lb["shard_size"]
+                flat = sd["fp32_flat_groups"][gi]
+                base = sd["base_optimizer_state"]["state"].get(gi, {})
+                shards[lb["pg_rank"]] = {
+                    "param": flat[mo:mo + ss],
+                    "exp_avg": base.get("exp_avg",
+                                        torch.zeros(ss))[mo:mo + ss],
+                    "exp_avg_sq": base.get("exp_avg_sq",
+                                           torch.zeros(ss))[mo:mo + ss],
+                }
+                step = base.get("step", 0)
+                out["step"] = int(step.item() if torch.is_tensor(step)
+                                  else step)
+            assert len(shards) == pg_world, \
+                f"bucket {bi}: found {len(shards)} shards, expected {pg_world}"
+            for kind in ("param", "exp_avg", "exp_avg_sq"):
+                full = torch.cat([shards[r][kind] for r in range(pg_world)])
+                for name, off, numel, shape in b0["params"]:
+                    if name is None:
+                        continue
+                    out[kind][name] = full[off:off + numel].view(shape).clone()
 
     out_dir = out_dir or os.path.join(load_dir, f"{tag}_universal")
     os.makedirs(out_dir, exist_ok=True)

@@ -557,9 +557,22 @@ class ZeroStage12Optimizer:
         return buckets
 
     def annotate_param_names(self, module):
-        """Stamp parameter names used by the universal layout manifest."""
+        """Stamp parameter names used by the universal layout manifest.
+        Expert parameters carry their EP rank in the name: with EP > 1 the
+        SAME module path holds DIFFERENT experts on each EP rank, and the
+        universal per-param files must not collide (universal resume keeps
+        the EP size fixed, like the reference)."""
+        from ...parallel import groups as pgroups
         for n, p in module.named_parameters():
-            p._ds_name = n
+            if getattr(p, "allreduce", True) is False and \
+                    getattr(p, "group_name", None):
+                try:
+                    ep_rank = pgroups.get_expert_parallel_rank(p.group_name)
+                except Exception:
+                    ep_rank = 0
+                p._ds_name = f"{n}@ep{ep_rank}"
+            else:
+                p._ds_name = n
 
     @torch.no_grad()
     def load_universal_state_dict(self, module, usd):
@@ -587,7 +600,7 @@ class ZeroStage12Optimizer:
             lo = b.pg_rank * b.shard_size        # my shard span in the bucket
             hi = lo + b.shard_size
             for p, off in zip(b.params, b.offsets):
-                name = name_of.get(p)
+                name = getattr(p, "_ds_name", None) or name_of.get(p)
                 if name is None or name not in usd["param"]:
                     continue
                 a, z = max(off, lo), min(off + p.numel(), hi)

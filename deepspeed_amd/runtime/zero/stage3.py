@@ -1237,8 +1237,22 @@ class ZeroStage3Optimizer:
         }
 
     def annotate_param_names(self, module):
+        """Stamp parameter names used by the universal layout manifest.
+        Expert parameters carry their EP rank in the name: with EP > 1 the
+        SAME module path holds DIFFERENT experts on each EP rank, and the
+        universal per-param files must not collide (universal resume keeps
+        the EP size fixed, like the reference)."""
+        from ...parallel import groups as pgroups
         for n, p in module.named_parameters():
-            p._ds_name = n
+            if getattr(p, "allreduce", True) is False and \
+                    getattr(p, "group_name", None):
+                try:
+                    ep_rank = pgroups.get_expert_parallel_rank(p.group_name)
+                except Exception:
+                    ep_rank = 0
+                p._ds_name = f"{n}@ep{ep_rank}"
+            else:
+                p._ds_name = n
 
     def layout_manifest(self):
         units = []

@@ -119,3 +119,70 @@ def test_universal_reshape_matrix(tmp_path, stage, save_ws, load_ws):
     from deepspeed_amd.checkpoint import ds_to_universal
     ds_to_universal(tmp, tag="step2")
     run_distributed(_resume_worker, world_size=load_ws, args=(tmp, stage))
+
+
+def _moe_uni_save_worker(rank, world, tmp):
+    import deepspeed_amd
+    from deepspeed_amd.models import MixtralForCausalLM, mixtral_tiny
+
+    torch.manual_seed(17)
+    cfg = mixtral_tiny(ep_size=2, num_experts=4)
+    model = MixtralForCausalLM(cfg)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": 1, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    g = torch.Generator().manual_seed(70)
+    for _ in range(2):
+        ids = torch.randint(0, cfg.vocab_size, (2, 16), generator=g)
+        loss = engine(ids, labels=ids)
+        engine.backward(loss)
+        engine.step()
+    engine.save_checkpoint(tmp, tag="moe0")
+    # dump this rank's expert params for comparison, keyed by ep rank
+    from deepspeed_amd.parallel import groups
+    name = next(m.expert_group_name for m in engine.module.modules()
+                if hasattr(m, "expert_group_name"))
+    ep_rank = groups.get_expert_parallel_rank(name)
+    if torch.distributed.get_rank(
+            groups.get_expert_data_parallel_group(name)) == 0:
+        experts = {n: p.detach().clone()
+                   for n, p in engine.module.named_parameters()
+                   if getattr(p, "allreduce", True) is False}
+        torch.save(experts, os.path.join(tmp, f"want_ep{ep_rank}.pt"))
+
+
+def _moe_uni_resume_worker(rank, world, tmp):
+    import deepspeed_amd
+    from deepspeed_amd.models import MixtralForCausalLM, mixtral_tiny
+
+    torch.manual_seed(99)  # different init: everything must come from ckpt
+    cfg = mixtral_tiny(ep_size=2, num_experts=4)
+    model = MixtralForCausalLM(cfg)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "zero_optimization": {"stage": 1, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    engine.load_checkpoint(tmp, tag="moe0", load_universal=True)
+    from deepspeed_amd.parallel import groups
+    name = next(m.expert_group_name for m in engine.module.modules()
+                if hasattr(m, "expert_group_name"))
+    ep_rank = groups.get_expert_parallel_rank(name)
+    want = torch.load(os.path.join(tmp, f"want_ep{ep_rank}.pt"),
+                      weights_only=False)
+    for n, p in engine.module.named_parameters():
+        if getattr(p, "allreduce", True) is False:
+            torch.testing.assert_close(p.detach(), want[n], rtol=1e-5,
+                                       atol=1e-6)
+
+
+def test_universal_moe_expert_dp_reshape(tmp_path):
+    """MoE universal resume: EP=2 fixed, expert-DP reshaped 2 -> 1
+    (world 4 -> 2). Expert params are keyed per EP rank in the universal
+    files (name@epR) so same-named experts from different EP ranks do not
+    collide."""
+    tmp = str(tmp_path)
+    run_distributed(_moe_uni_save_worker, world_size=4, args=(tmp,))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="moe0")
+    run_distributed(_moe_uni_resume_worker, world_size=2, args=(tmp,))
