@@ -481,6 +481,18 @@ class Engine(torch.nn.Module):
                 events.append(("Train/samples_per_sec",
                                self.tput_timer.avg_samples_per_sec(),
                                self.global_steps))
+            if getattr(self, "curriculum_scheduler", None) is not None:
+                events.append(
+                    ("Train/curriculum_difficulty",
+                     self.curriculum_scheduler.get_current_difficulty(),
+                     self.global_steps))
+            if self.has_moe_layers and hasattr(self.module, "aux_loss"):
+                try:
+                    events.append(("Train/moe_aux_loss",
+                                   float(self.module.aux_loss()),
+                                   self.global_steps))
+                except Exception:
+                    pass
             self.monitor.write_events(events)
         if self.wall_clock_breakdown:
             self.timers(STEP_GLOBAL_TIMER).stop()
