@@ -336,7 +336,7 @@ class ZeroStage12Optimizer:
         """Squared-norm of a grad shard, accumulated ON DEVICE while the
         D2H copy streams — step() then never waits for host grads just to
         decide overflow/clipping."""
-        if t.numel() == 0 or not t.is_cuda:
+        if t.numel() == 0:
             return
         nsq = t.float().pow(2).sum()
         self._dev_norm_sq = nsq if self._dev_norm_sq is None \
@@ -374,8 +374,16 @@ class ZeroStage12Optimizer:
         owned = [g for g in self.group_owned_grads if g.numel() > 0]
         norm_sq_dev = None
         if self.cpu_offload and self._dev_norm_sq is not None:
+            # per-rank shard norms are partial sums — the global norm (and
+            # the overflow decision, which MUST agree on every rank or the
+            # steps desync) needs the same reductions as the non-offload path
             norm_sq_dev = self._dev_norm_sq
             self._dev_norm_sq = None
+            if dist.is_initialized() and self.world_size > 1:
+                dist.all_reduce(norm_sq_dev, group=self.dp_group)
+            if self.mpu is not None:
+                dist.all_reduce(norm_sq_dev,
+                                group=self.mpu.get_model_parallel_group())
         elif owned:
             norms = torch._foreach_norm(owned, 2.0)
             norm_sq_dev = torch.stack([n.float() for n in norms]).pow(2).sum()

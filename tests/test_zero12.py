@@ -28,7 +28,7 @@ class TinyNet(torch.nn.Module):
 
 def _reference_mixed_precision_loop(model_fp32, data, lr, steps, gas=1,
                                     dtype=torch.bfloat16, clip=0.0,
-                                    world=1):
+                                    world=1, norms_out=None):
     """fp32 master AdamW with low-precision fwd/bwd — the semantics ZeRO-1/2
     must reproduce (modulo reduction order)."""
     master = copy.deepcopy(model_fp32).float()
@@ -55,6 +55,8 @@ def _reference_mixed_precision_loop(model_fp32, data, lr, steps, gas=1,
                 grads.append(pm.grad)
             if clip > 0:
                 norm = torch.norm(torch.stack([g.norm() for g in grads]))
+                if norms_out is not None:
+                    norms_out.append(norm.item())
                 coef = min(1.0, clip / (norm.item() + 1e-6))
                 for g in grads:
                     g.mul_(coef)
@@ -774,9 +776,12 @@ def test_gathered_params_readonly_qwz_ws2():
     run_distributed(_gathered_readonly_qwz_worker, world_size=2)
 
 
-def _offload_worker(rank, world, stage):
+def _offload_worker(rank, world, stage, clip=0.0):
     """ZeRO-Offload (optimizer states on host) parity vs the reference
-    mixed-precision loop — exercises the pipelined per-bucket step path."""
+    mixed-precision loop — exercises the pipelined per-bucket step path.
+    With clip>0 it also covers the device-accumulated grad-norm branch:
+    each rank's partial shard norm must be all-reduced before clipping
+    (a missed reduction shows up as a per-rank clip coefficient)."""
     import deepspeed_amd
     lr, steps = 1e-2, 4
     model = TinyNet()
@@ -785,20 +790,31 @@ def _offload_worker(rank, world, stage):
     engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
         "train_micro_batch_size_per_gpu": 4,
         "bf16": {"enabled": True},
+        "gradient_clipping": clip,
         "zero_optimization": {
             "stage": stage, "overlap_comm": True,
             "offload_optimizer": {"device": "cpu", "pin_memory": True}},
         "optimizer": {"type": "AdamW", "params": {"lr": lr}}})
     it = iter(data)
     engine_losses = []
+    engine_norms = []
     for _ in range(steps):
         xs, ys = next(it)
         loss = engine(xs.to(engine.device).bfloat16(), labels=ys)
         engine.backward(loss)
         engine.step()
         engine_losses.append(loss.item())
+        engine_norms.append(engine.optimizer._global_grad_norm)
+    ref_norms = []
     ref_losses, ref_master = _reference_mixed_precision_loop(
-        ref_model, data, lr, steps, 1, torch.bfloat16)
+        ref_model, data, lr, steps, 1, torch.bfloat16, clip,
+        norms_out=ref_norms)
+    if clip > 0:
+        # the recorded norm must be the GLOBAL norm (all-reduced over DP),
+        # not this rank's shard partial — Adam hides a wrongly-scaled clip
+        # coefficient, the norm value itself doesn't lie
+        for a, b in zip(engine_norms, ref_norms):
+            assert abs(a - b) / max(b, 1e-6) < 0.08, (engine_norms, ref_norms)
     for a, b in zip(engine_losses, ref_losses):
         assert abs(a - b) < 2e-2, (engine_losses, ref_losses)
     for p_e, p_r in zip(engine.module.parameters(), ref_master.parameters()):
@@ -809,3 +825,7 @@ def _offload_worker(rank, world, stage):
 @pytest.mark.parametrize("stage", [1, 2])
 def test_zero_offload_parity_ws2(stage):
     run_distributed(_offload_worker, world_size=2, args=(stage,))
+
+
+def test_zero_offload_clip_parity_ws2():
+    run_distributed(_offload_worker, world_size=2, args=(2, 0.05))
