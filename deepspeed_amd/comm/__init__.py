@@ -45,6 +45,8 @@ def init_distributed(dist_backend: Optional[str] = None,
         return
     if dist_backend is None:
         dist_backend = accel.communication_backend_name()
+    if "RANK" not in os.environ and "OMPI_COMM_WORLD_RANK" in os.environ:
+        mpi_discovery()
     if rank == -1:
         rank = int(os.environ.get("RANK", 0))
     if world_size == -1:
@@ -59,6 +61,31 @@ def init_distributed(dist_backend: Optional[str] = None,
     torch_dist.init_process_group(**kwargs)
     log_dist(f"initialized distributed: backend={dist_backend} "
              f"world_size={world_size}")
+
+
+def mpi_discovery(distributed_port: int = 29500) -> None:
+    """Derive RANK / LOCAL_RANK / WORLD_SIZE / MASTER_ADDR from an
+    mpirun-launched environment (reference comm/comm.py:694 uses mpi4py;
+    here the OpenMPI/MPICH/PMI env vars cover the same launchers without a
+    dependency — the rank-0 host is taken from the launcher's node list
+    when present, else every rank must share MASTER_ADDR already)."""
+    env = os.environ
+    rank = env.get("OMPI_COMM_WORLD_RANK", env.get("PMI_RANK"))
+    world = env.get("OMPI_COMM_WORLD_SIZE", env.get("PMI_SIZE"))
+    local = env.get("OMPI_COMM_WORLD_LOCAL_RANK",
+                    env.get("MPI_LOCALRANKID", "0"))
+    if rank is None or world is None:
+        raise RuntimeError("mpi_discovery: no OMPI_/PMI_ rank variables set")
+    env["RANK"] = str(int(rank))
+    env["WORLD_SIZE"] = str(int(world))
+    env.setdefault("LOCAL_RANK", str(int(local)))
+    if "MASTER_ADDR" not in env:
+        # OpenMPI exposes the node list; rank 0 lives on the first entry
+        nodes = env.get("OMPI_MCA_orte_node_regex") or \
+            env.get("SLURM_NODELIST") or ""
+        first = nodes.split(",")[0].split("[")[0].strip()
+        env["MASTER_ADDR"] = first if first else "127.0.0.1"
+    env.setdefault("MASTER_PORT", str(distributed_port))
 
 
 def destroy_process_group() -> None:

@@ -6,9 +6,11 @@ from .norms import RMSNorm, FusedLayerNorm, rms_norm, layer_norm
 from .rope import apply_rope, rope_tables
 from .swiglu import swiglu, geglu
 from .evoformer import DS4Sci_EvoformerAttention, EvoformerAttention
+from .spatial import nhwc_bias_add
 
 __all__ = [
     "has_ext", "get_ext", "FusedAdam", "fused_adam_step",
     "multi_tensor_adam_available", "RMSNorm", "FusedLayerNorm", "rms_norm",
     "layer_norm", "apply_rope", "rope_tables", "swiglu", "geglu",
+    "nhwc_bias_add",
 ]

@@ -38,6 +38,10 @@ extern "C" void ds_gated_act_fwd(const void* gate, const void* up, void* out,
 extern "C" void ds_gated_act_bwd(const void* dout, const void* gate,
                                  const void* up, void* dgate, void* dup,
                                  long long n, int act, int dtype, void* stream);
+extern "C" void ds_nhwc_bias_add(const void* act, const void* bias,
+                                 const void* other, const void* other_bias,
+                                 void* out, long long n, int channels,
+                                 int dtype, void* stream);
 extern "C" void ds_cpu_adam_flat(float* p, const void* g, int grad_dtype,
                                  float* m, float* v, void* p16, long long n,
                                  float lr, float beta1, float beta2, float eps,
@@ -398,6 +402,37 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> norm_bwd(
   return {dx, dw, db};
 }
 
+at::Tensor nhwc_bias_add(at::Tensor act, at::Tensor bias,
+                         c10::optional<at::Tensor> other,
+                         c10::optional<at::Tensor> other_bias) {
+  TORCH_CHECK(act.is_cuda() && act.is_contiguous() && bias.is_contiguous(),
+              "nhwc_bias_add: contiguous GPU tensors");
+  TORCH_CHECK(act.scalar_type() == at::kBFloat16 ||
+              act.scalar_type() == at::kHalf, "nhwc_bias_add: bf16/fp16");
+  const int64_t C = bias.numel();
+  TORCH_CHECK(C % 8 == 0 && act.numel() % C == 0,
+              "nhwc_bias_add: channels % 8 == 0, act a multiple of channels");
+  const void* other_p = nullptr;
+  const void* ob_p = nullptr;
+  if (other.has_value() && other->defined()) {
+    TORCH_CHECK(other->is_contiguous() && other->numel() == act.numel() &&
+                other->scalar_type() == act.scalar_type(),
+                "nhwc_bias_add: other mismatched");
+    other_p = other->data_ptr();
+  }
+  if (other_bias.has_value() && other_bias->defined()) {
+    TORCH_CHECK(other_bias->is_contiguous() && other_bias->numel() == C &&
+                other_bias->scalar_type() == act.scalar_type(),
+                "nhwc_bias_add: other_bias mismatched");
+    ob_p = other_bias->data_ptr();
+  }
+  auto out = at::empty_like(act);
+  ds_nhwc_bias_add(act.data_ptr(), bias.data_ptr(), other_p, ob_p,
+                   out.data_ptr(), act.numel(), (int)C, dtype_code(act),
+                   cur_stream());
+  return out;
+}
+
 void rope(at::Tensor x, at::Tensor cos_table, at::Tensor sin_table,
           c10::optional<at::Tensor> positions, bool backward) {
   TORCH_CHECK(x.dim() == 4 && x.is_contiguous(), "rope expects [B,S,H,D]");
@@ -591,6 +626,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "RMSNorm/LayerNorm forward");
   m.def("norm_bwd", &norm_bwd, "RMSNorm/LayerNorm backward");
   m.def("rope", &rope, "Rotary position embedding (in-place)");
+  m.def("nhwc_bias_add", &nhwc_bias_add,
+        "fused channels-last bias (+residual +residual-bias) add",
+        pybind11::arg("act"), pybind11::arg("bias"),
+        pybind11::arg("other") = c10::nullopt,
+        pybind11::arg("other_bias") = c10::nullopt);
   m.def("gated_act_fwd", &gated_act_fwd, "SwiGLU/GeGLU forward");
   m.def("gated_act_bwd", &gated_act_bwd, "SwiGLU/GeGLU backward");
   m.def("groupwise_quant", &groupwise_quant,

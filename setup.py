@@ -36,6 +36,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "norms.hip"),
         os.path.join(CSRC, "rope.hip"),
         os.path.join(CSRC, "swiglu.hip"),
+        os.path.join(CSRC, "spatial.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17", "-fopenmp", "-mavx2", "-mfma"],

@@ -171,3 +171,31 @@ def test_shm_allreduce_intranode():
     cross-process sum without touching the network stack."""
     from .common import run_distributed
     run_distributed(_shm_worker, world_size=4)
+
+
+def test_mpi_discovery_env(monkeypatch):
+    """OpenMPI env vars map to the torchrun-style rendezvous variables
+    (reference comm/comm.py:694 mpi_discovery)."""
+    import deepspeed_amd.comm as dcomm
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+              "MASTER_PORT"):
+        monkeypatch.delenv(k, raising=False)
+    monkeypatch.setenv("OMPI_COMM_WORLD_RANK", "3")
+    monkeypatch.setenv("OMPI_COMM_WORLD_SIZE", "8")
+    monkeypatch.setenv("OMPI_COMM_WORLD_LOCAL_RANK", "3")
+    dcomm.mpi_discovery(distributed_port=12345)
+    import os
+    assert os.environ["RANK"] == "3"
+    assert os.environ["WORLD_SIZE"] == "8"
+    assert os.environ["LOCAL_RANK"] == "3"
+    assert os.environ["MASTER_ADDR"] == "127.0.0.1"
+    assert os.environ["MASTER_PORT"] == "12345"
+
+
+def test_mpi_discovery_requires_mpi_env(monkeypatch):
+    import deepspeed_amd.comm as dcomm
+    for k in ("RANK", "OMPI_COMM_WORLD_RANK", "PMI_RANK"):
+        monkeypatch.delenv(k, raising=False)
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        dcomm.mpi_discovery()

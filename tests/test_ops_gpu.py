@@ -498,3 +498,23 @@ def test_paged_decode_attention_gpu():
         ref = (torch.softmax(s, -1) @ v).squeeze(1)        # [H, D]
         rel = (got[i].float() - ref).abs().max() / ref.abs().max()
         assert rel < 2e-2, (i, rel)
+
+
+@pytest.mark.gpu
+def test_nhwc_bias_add_gpu():
+    """Fused channels-last bias-add variants vs the torch fallback
+    (reference csrc/spatial/csrc/opt_bias_add.cu surface)."""
+    from deepspeed_amd.ops.spatial import nhwc_bias_add
+    torch.manual_seed(0)
+    for dt in (torch.bfloat16, torch.float16):
+        rows, C = 1024, 320
+        act = torch.randn(rows, C, device="cuda", dtype=dt)
+        bias = torch.randn(C, device="cuda", dtype=dt)
+        other = torch.randn(rows, C, device="cuda", dtype=dt)
+        obias = torch.randn(C, device="cuda", dtype=dt)
+        for args in ((act, bias), (act, bias, other),
+                     (act, bias, other, obias)):
+            got = nhwc_bias_add(*args)
+            want = nhwc_bias_add(*(a.cpu() for a in args)).to(dt)
+            torch.testing.assert_close(got.float().cpu(), want.float(),
+                                       rtol=2e-2, atol=2e-2)
