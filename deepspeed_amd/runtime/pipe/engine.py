@@ -12,9 +12,12 @@ Differences from the reference, by MI355X design:
   tied grads are correct without a separate dead-bucket path.
 """
 
+import os
+
 import torch
 
 from ... import comm as dist
+from ...parallel import groups
 from ..engine import Engine
 from . import schedule as sched
 from .module import PipelineModule
@@ -50,6 +53,26 @@ class PipelineEngine(Engine):
             sched.ReduceGrads: self._exec_reduce_grads,
             sched.OptimizerStep: self._exec_optimizer_step,
         }
+
+    # ------------------------------------------------------------- checkpoint
+
+    def _model_ckpt_name(self, dirname):
+        """Every pipeline stage owns DIFFERENT layers, so each stage's
+        dp-rank-0 writes its own stage-qualified model-states file (the
+        reference writes per-layer files, pipe/module.py ckpt_layer_path;
+        one file per stage is the same information at this granularity) —
+        without the stage qualifier PP stages overwrite each other."""
+        mp_rank = groups.get_tensor_parallel_rank()
+        return os.path.join(
+            dirname, f"mp_rank_{mp_rank:02d}_pp_rank_"
+            f"{self.stage_id:03d}_model_states.pt")
+
+    def _zero_ckpt_name(self, dirname):
+        mp_rank = groups.get_tensor_parallel_rank()
+        dp_rank = dist.get_rank(self.dp_group)
+        return os.path.join(
+            dirname, f"zero_pp_rank_{dp_rank}_mp_rank_{mp_rank:02d}"
+            f"_stage_{self.stage_id:03d}_optim_states.pt")
 
     # ----------------------------------------------------------------- driver
 
