@@ -48,46 +48,46 @@ def ds_to_universal(load_dir, tag=None, out_dir=None):
         "checkpoint has no layout manifest (saved by an older version?)"
 
     out = {"param": {}, "exp_avg": {}, "exp_avg_sq": {}, "step": 0}
-    n_buckets = len(layout)
-    for bi in range(n_buckets):
-        # Partition files by the bucket's PARAM IDENTITY: with expert
-        # parallelism, bucket bi on different EP ranks holds DIFFERENT
-        # experts under distinct @epR-decorated names (same pg_rank values
-        # within each expert-DP group) — each identity assembles its own
-        # full flat and emits its own per-param entries.
-        by_identity = {}
-        for sd in states:
-            lb = sd["layout"][bi]
+    # Partition (file, bucket) pairs by the bucket's PARAM IDENTITY.
+    # With expert parallelism, the same bucket slot on different EP ranks
+    # holds DIFFERENT experts under distinct @epR-decorated names; with
+    # pipeline parallelism, different stages have entirely different
+    # layouts (even different bucket COUNTS — never index other files by
+    # this file's bucket number). Each identity assembles its own full
+    # flat from its own members' shards and emits its per-param entries.
+    by_identity = {}
+    for sd in states:
+        for lb in sd["layout"]:
             key = tuple(n for n, *_ in lb["params"])
-            by_identity.setdefault(key, []).append(sd)
-        for members in by_identity.values():
-            b0 = members[0]["layout"][bi]
-            gi = b0["group_idx"]
-            pg_world = b0["pg_world"]
-            shards = {}
-            for sd in members:
-                lb = sd["layout"][bi]
-                mo, ss = lb["master_offset"], lb["shard_size"]
-                flat = sd["fp32_flat_groups"][gi]
-                base = sd["base_optimizer_state"]["state"].get(gi, {})
-                shards[lb["pg_rank"]] = {
-                    "param": flat[mo:mo + ss],
-                    "exp_avg": base.get("exp_avg",
-                                        torch.zeros(ss))[mo:mo + ss],
-                    "exp_avg_sq": base.get("exp_avg_sq",
-                                           torch.zeros(ss))[mo:mo + ss],
-                }
-                step = base.get("step", 0)
-                out["step"] = int(step.item() if torch.is_tensor(step)
-                                  else step)
-            assert len(shards) == pg_world, \
-                f"bucket {bi}: found {len(shards)} shards, expected {pg_world}"
-            for kind in ("param", "exp_avg", "exp_avg_sq"):
-                full = torch.cat([shards[r][kind] for r in range(pg_world)])
-                for name, off, numel, shape in b0["params"]:
-                    if name is None:
-                        continue
-                    out[kind][name] = full[off:off + numel].view(shape).clone()
+            by_identity.setdefault(key, []).append((sd, lb))
+    for members in by_identity.values():
+        _, b0 = members[0]
+        pg_world = b0["pg_world"]
+        shards = {}
+        for sd, lb in members:
+            gi = lb["group_idx"]
+            mo, ss = lb["master_offset"], lb["shard_size"]
+            flat = sd["fp32_flat_groups"][gi]
+            base = sd["base_optimizer_state"]["state"].get(gi, {})
+            shards[lb["pg_rank"]] = {
+                "param": flat[mo:mo + ss],
+                "exp_avg": base.get("exp_avg",
+                                    torch.zeros(ss))[mo:mo + ss],
+                "exp_avg_sq": base.get("exp_avg_sq",
+                                       torch.zeros(ss))[mo:mo + ss],
+            }
+            step = base.get("step", 0)
+            out["step"] = int(step.item() if torch.is_tensor(step)
+                              else step)
+        assert len(shards) == pg_world, \
+            f"bucket {b0['params'][0]}: found {len(shards)} shards, " \
+            f"expected {pg_world}"
+        for kind in ("param", "exp_avg", "exp_avg_sq"):
+            full = torch.cat([shards[r][kind] for r in range(pg_world)])
+            for name, off, numel, shape in b0["params"]:
+                if name is None:
+                    continue
+                out[kind][name] = full[off:off + numel].view(shape).clone()
 
     out_dir = out_dir or os.path.join(load_dir, f"{tag}_universal")
     os.makedirs(out_dir, exist_ok=True)

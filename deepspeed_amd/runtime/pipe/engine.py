@@ -74,6 +74,95 @@ class PipelineEngine(Engine):
             dirname, f"zero_pp_rank_{dp_rank}_mp_rank_{mp_rank:02d}"
             f"_stage_{self.stage_id:03d}_optim_states.pt")
 
+    def save_checkpoint(self, save_dir, tag=None, client_state=None,
+                        save_latest=True, exclude_frozen_parameters=False):
+        """Stage-file save (same-PP resume) PLUS per-GLOBAL-layer module
+        files (reference pipe/module.py ckpt_layer_path layer_XX files) so
+        a job at a DIFFERENT pipeline degree can reassemble its stages;
+        the optimizer state crosses degrees via the universal converter."""
+        tag = self._ckpt_tag(tag)
+        ret = super().save_checkpoint(
+            save_dir, tag=tag, client_state=client_state,
+            save_latest=save_latest,
+            exclude_frozen_parameters=exclude_frozen_parameters)
+        ckpt_dir = os.path.join(save_dir, tag)
+        mp_rank = groups.get_tensor_parallel_rank()
+        if dist.get_rank(self.dp_group) == 0:
+            for idx, mod in self.module._layers.items():
+                self.checkpoint_engine.save(
+                    mod.state_dict(), self._layer_ckpt_name(
+                        ckpt_dir, mp_rank, int(idx)))
+            owners = self.module._tied_keys_per_stage()
+            for key, mod in self.module.tied_modules.items():
+                # exactly one owning stage writes (weights are synced)
+                if self.stage_id == min(owners.get(key, [self.stage_id])):
+                    self.checkpoint_engine.save(
+                        mod.state_dict(), self._tied_ckpt_name(
+                            ckpt_dir, mp_rank, key))
+        if self.global_rank == 0:
+            torch.save({"num_stages": self.num_stages},
+                       os.path.join(ckpt_dir, "pipeline_meta.pt"))
+        dist.barrier()
+        return ret
+
+    def load_checkpoint(self, load_dir, tag=None, **kwargs):
+        if tag is None:
+            latest = os.path.join(load_dir, "latest")
+            if not os.path.exists(latest):
+                return None, {}
+            with open(latest) as f:
+                tag = f.read().strip()
+        ckpt_dir = os.path.join(load_dir, tag)
+        meta_path = os.path.join(ckpt_dir, "pipeline_meta.pt")
+        saved_stages = None
+        if os.path.exists(meta_path):
+            saved_stages = torch.load(meta_path,
+                                      weights_only=False)["num_stages"]
+        if saved_stages in (None, self.num_stages):
+            return super().load_checkpoint(load_dir, tag=tag, **kwargs)
+
+        # different pipeline degree: reassemble this stage's modules from
+        # the per-global-layer files; optimizer state must come from the
+        # universal checkpoint (native zero files are stage-shaped)
+        mp_rank = groups.get_tensor_parallel_rank()
+        for idx, mod in self.module._layers.items():
+            sd = self.checkpoint_engine.load(
+                self._layer_ckpt_name(ckpt_dir, mp_rank, int(idx)),
+                map_location="cpu")
+            mod.load_state_dict(sd)
+        for key, mod in self.module.tied_modules.items():
+            sd = self.checkpoint_engine.load(
+                self._tied_ckpt_name(ckpt_dir, mp_rank, key),
+                map_location="cpu")
+            mod.load_state_dict(sd)
+        self.module._sync_tied_weights()
+        import glob as _glob
+        metas = sorted(_glob.glob(os.path.join(
+            ckpt_dir, f"mp_rank_{mp_rank:02d}_pp_rank_*_model_states.pt")))
+        client_state = {}
+        if metas:
+            meta = self.checkpoint_engine.load(metas[0], map_location="cpu")
+            self.global_steps = meta.get("global_steps", 0)
+            self.global_samples = meta.get("global_samples", 0)
+            self.skipped_steps = meta.get("skipped_steps", 0)
+            client_state = meta.get("client_state", {})
+        if kwargs.get("load_universal"):
+            from ...checkpoint.universal import load_universal as _load_usd
+            usd = _load_usd(os.path.join(load_dir, f"{tag}_universal"))
+            assert hasattr(self.optimizer, "load_universal_state_dict"), \
+                "cross-PP-degree optimizer resume requires ZeRO + universal"
+            self.optimizer.load_universal_state_dict(self.module, usd)
+        return ckpt_dir, client_state
+
+    def _layer_ckpt_name(self, ckpt_dir, mp_rank, idx):
+        return os.path.join(
+            ckpt_dir, f"layer_{idx:04d}-mp_rank_{mp_rank:02d}"
+            "-model_states.pt")
+
+    def _tied_ckpt_name(self, ckpt_dir, mp_rank, key):
+        return os.path.join(
+            ckpt_dir, f"tied_{key}-mp_rank_{mp_rank:02d}-model_states.pt")
+
     # ----------------------------------------------------------------- driver
 
     def _reserve_buffers(self, n):

@@ -140,7 +140,13 @@ class PipelineModule(nn.Module):
         self.forward_funcs: List = []
         self.tied_modules = nn.ModuleDict()
         self.tied_weight_attrs = {}
-        self._local = nn.ModuleList()
+        # Local layers registered under their GLOBAL spec index (reference
+        # pipe/module.py names layer files by global index too): state-dict
+        # keys are then unique ACROSS stages, which same-PP resume doesn't
+        # need but universal checkpoint conversion and cross-PP-degree
+        # reshape do — local indices would collide stage 0's layer 0 with
+        # stage 1's.
+        self._layers = nn.ModuleDict()
         for idx in range(self.part_start, self.part_end):
             spec = self.specs[idx]
             if isinstance(spec, TiedLayerSpec):
@@ -160,10 +166,10 @@ class PipelineModule(nn.Module):
                 if seed_layers:
                     torch.manual_seed(base_seed + idx)
                 mod = spec.build()
-                self._local.append(mod)
+                self._layers[str(idx)] = mod
                 self.forward_funcs.append(mod)
             elif isinstance(spec, nn.Module):
-                self._local.append(spec)
+                self._layers[str(idx)] = spec
                 self.forward_funcs.append(spec)
             elif callable(spec):
                 self.forward_funcs.append(spec)

@@ -107,7 +107,9 @@ class TrainSchedule(PipeSchedule):
                 cmds.append(LoadMicroBatch(buffer_id=buf, micro_batch_id=mb))
             else:
                 cmds.append(RecvActivation(buffer_id=buf))
-            if self.is_last_stage:
+            if self.is_last_stage and not self.is_first_stage:
+                # single-stage pipelines load ONCE: the executor fills both
+                # inputs and labels from the same batch
                 cmds.append(LoadMicroBatch(buffer_id=buf, micro_batch_id=mb))
             cmds.append(ForwardPass(buffer_id=buf, micro_batch_id=mb))
             if not self.is_last_stage:
@@ -152,7 +154,7 @@ class InferenceSchedule(PipeSchedule):
                 cmds.append(LoadMicroBatch(buffer_id=buf, micro_batch_id=mb))
             else:
                 cmds.append(RecvActivation(buffer_id=buf))
-            if self.is_last_stage:
+            if self.is_last_stage and not self.is_first_stage:
                 cmds.append(LoadMicroBatch(buffer_id=buf, micro_batch_id=mb))
             cmds.append(ForwardPass(buffer_id=buf, micro_batch_id=mb))
             if not self.is_last_stage:

@@ -4,6 +4,8 @@ partitioning, 1F1B schedule shape, and exact loss/weight parity of a
 2-stage pipeline against the same model run sequentially in one process.
 """
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -247,3 +249,59 @@ def _pp_resume_worker(rank, world, ckpt_dir):
 
 def test_pipeline_checkpoint_resume_pp2dp2(tmp_path):
     run_distributed(_pp_resume_worker, world_size=4, args=(str(tmp_path),))
+
+
+def _pp2_universal_save_worker(rank, world, tmp):
+    import deepspeed_amd
+    from deepspeed_amd.runtime.pipe.module import PipelineModule
+    cfg = dict(_CONFIG)
+    cfg["zero_optimization"] = {"stage": 1, "overlap_comm": False}
+    net = PipelineModule(_make_layers(), num_stages=2,
+                         loss_fn=nn.functional.mse_loss,
+                         partition_method="parameters")
+    engine, _, _, _ = deepspeed_amd.initialize(model=net, config=cfg)
+    mbs = _CONFIG["gradient_accumulation_steps"]
+    dp = net.grid.data_parallel_id
+    it = iter(_make_data(3 * mbs, 4, seed=555 + dp))
+    for _ in range(2):
+        engine.train_batch(it)
+    engine.save_checkpoint(tmp, tag="pp2")
+    cont = engine.train_batch(it).item()
+    if rank == 0:
+        import json
+        with open(os.path.join(tmp, "cont.json"), "w") as f:
+            json.dump({"cont": cont}, f)
+
+
+def _pp1_universal_resume_worker(rank, world, tmp):
+    import deepspeed_amd
+    from deepspeed_amd.runtime.pipe.module import PipelineModule
+    cfg = dict(_CONFIG)
+    cfg["zero_optimization"] = {"stage": 1, "overlap_comm": False}
+    net = PipelineModule(_make_layers(), num_stages=1,
+                         loss_fn=nn.functional.mse_loss,
+                         partition_method="parameters")
+    engine, _, _, _ = deepspeed_amd.initialize(model=net, config=cfg)
+    tag_dir, _ = engine.load_checkpoint(tmp, tag="pp2", load_universal=True)
+    assert tag_dir is not None
+    mbs = _CONFIG["gradient_accumulation_steps"]
+    dp = net.grid.data_parallel_id
+    it = iter(_make_data(3 * mbs, 4, seed=555 + dp))
+    for _ in range(2 * mbs):
+        next(it)
+    resumed = engine.train_batch(it).item()
+    import json
+    with open(os.path.join(tmp, "cont.json")) as f:
+        cont = json.load(f)["cont"]
+    assert abs(resumed - cont) < 1e-5, (resumed, cont)
+
+
+def test_pipeline_degree_reshape_pp2_to_pp1(tmp_path):
+    """Save at PP=2 x DP=2, convert the ZeRO state to universal, resume at
+    PP=1 x DP=2: per-global-layer module files + name-keyed universal
+    optimizer state make the pipeline degree a resume-time choice."""
+    tmp = str(tmp_path)
+    run_distributed(_pp2_universal_save_worker, world_size=4, args=(tmp,))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="pp2")
+    run_distributed(_pp1_universal_resume_worker, world_size=2, args=(tmp,))
