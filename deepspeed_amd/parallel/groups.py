@@ -119,6 +119,16 @@ def get_sequence_parallel_rank() -> int:
     return dist.get_rank(g) if g is not None else 0
 
 
+def set_tensor_parallel_group(group, world_size=None, rank=None):
+    """Register an externally-built TP group (e.g. the pipeline grid's
+    per-(stage, dp) tensor groups) so Column/RowParallel layers and
+    mp_rank checkpoint naming resolve it. world_size/rank are accepted
+    for symmetry but derived from the group when queried."""
+    global _TENSOR_PARALLEL_GROUP, _MODEL_PARALLEL_GROUP
+    _TENSOR_PARALLEL_GROUP = group
+    _MODEL_PARALLEL_GROUP = group
+
+
 def initialize_tensor_parallel(tp_size: int):
     """Create TP groups: ranks [i*tp, (i+1)*tp) form one TP group; DP group is
     the strided complement."""

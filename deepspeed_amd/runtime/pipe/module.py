@@ -119,14 +119,15 @@ class PipelineModule(nn.Module):
                  loss_fn: Optional[Callable] = None,
                  partition_method: str = "parameters",
                  activation_checkpoint_interval: int = 0,
-                 seed_layers: bool = False, base_seed: int = 1234):
+                 seed_layers: bool = False, base_seed: int = 1234,
+                 tp_size: int = 1):
         super().__init__()
         if not dist.is_initialized():
             dist.init_distributed()
         self.specs = list(layers)
         if grid is None:
             assert num_stages is not None, "need num_stages or grid"
-            grid = PipelineParallelGrid(num_stages)
+            grid = PipelineParallelGrid(num_stages, tp_size=tp_size)
         self.grid = grid
         self.num_stages = grid.pipe_parallel_size
         self.stage_id = grid.stage_id
@@ -218,14 +219,17 @@ class PipelineModule(nn.Module):
         """One process group per tied key per pipe replica (all dp ids create
         the groups collectively; each rank keeps the ones it belongs to)."""
         tie_groups = {}
+        tp_size = self.grid.tensor_parallel_size
         for key, stages in self._tied_keys_per_stage().items():
             if len(stages) < 2:
                 continue
             for dp in range(self.grid.data_parallel_size):
-                ranks = [s * self.grid.data_parallel_size + dp for s in stages]
-                g = dist.new_group(ranks)
-                if self.grid.global_rank in ranks:
-                    tie_groups[key] = (ranks, g)
+                for tp in range(tp_size):
+                    ranks = [(s * self.grid.data_parallel_size + dp) *
+                             tp_size + tp for s in stages]
+                    g = dist.new_group(ranks)
+                    if self.grid.global_rank in ranks:
+                        tie_groups[key] = (ranks, g)
         return tie_groups
 
     @torch.no_grad()

@@ -269,3 +269,99 @@ def _z3_stress_worker(rank, world):
 
 def test_zero3_backpressure_stress_ws8():
     run_distributed(_z3_stress_worker, world_size=8, timeout=600)
+
+
+def _pp2tp2dp2_worker(rank, world):
+    """Full 3D: PP=2 x TP=2 x DP=2 on 8 ranks. Each pipeline stage holds
+    TP-sharded blocks (column -> row parallel pair); parity against the
+    unsharded sequential model with dp-averaged gradients."""
+    import deepspeed_amd
+    from deepspeed_amd.parallel import groups as pgroups
+    from deepspeed_amd.runtime.pipe.module import PipelineModule, LayerSpec
+    from deepspeed_amd.runtime.tensor_parallel.layers import (
+        ColumnParallelLinear, RowParallelLinear)
+
+    D, FF = 8, 16
+
+    class RefBlock(nn.Module):
+        def __init__(self, seed):
+            super().__init__()
+            torch.manual_seed(seed)
+            self.fc1 = nn.Linear(D, FF, bias=False)
+            self.fc2 = nn.Linear(FF, D, bias=False)
+
+        def forward(self, x):
+            return self.fc2(torch.relu(self.fc1(x)))
+
+    class TPBlock(nn.Module):
+        """Same math, fc1 column-sharded / fc2 row-sharded over TP."""
+
+        def __init__(self, seed):
+            super().__init__()
+            ref = RefBlock(seed)
+            g = pgroups.get_tensor_parallel_group()
+            tp = pgroups.get_tensor_parallel_world_size()
+            tr = pgroups.get_tensor_parallel_rank()
+            w1 = ref.fc1.weight.detach().chunk(tp, dim=0)[tr].clone()
+            w2 = ref.fc2.weight.detach().chunk(tp, dim=1)[tr].clone()
+            self.col = ColumnParallelLinear(w1, None, g)
+            self.row = RowParallelLinear(w2, None, g)
+
+        def forward(self, x):
+            return self.row(torch.relu(self.col(x)))
+
+    cfg = {"train_micro_batch_size_per_gpu": 4,
+           "gradient_accumulation_steps": 2,
+           "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}}
+    specs = [LayerSpec(TPBlock, 21), LayerSpec(TPBlock, 22),
+             LayerSpec(TPBlock, 23), LayerSpec(TPBlock, 24)]
+    net = PipelineModule(specs, num_stages=2, tp_size=2,
+                         loss_fn=nn.functional.mse_loss,
+                         partition_method="uniform")
+    assert net.grid.data_parallel_size == 2
+    assert net.grid.tensor_parallel_size == 2
+    engine, _, _, _ = deepspeed_amd.initialize(model=net, config=cfg)
+
+    def make_data(n, bs, seed):
+        g = torch.Generator().manual_seed(seed)
+        return [(torch.randn(bs, D, generator=g),
+                 torch.randn(bs, D, generator=g)) for _ in range(n)]
+
+    n_steps, mbs = 2, 2
+    dp = net.grid.data_parallel_id
+    data_all = [make_data(n_steps * mbs, 4, seed=77 + d) for d in range(2)]
+    it = iter(data_all[dp])
+    losses = [engine.train_batch(it).item() for _ in range(n_steps)]
+
+    ref = nn.Sequential(*[RefBlock(s) for s in (21, 22, 23, 24)])
+    opt = torch.optim.AdamW(ref.parameters(), lr=1e-3)
+    its = [iter(data_all[d]) for d in range(2)]
+    ref_losses = []
+    for _ in range(n_steps):
+        tot = 0.0
+        for d in range(2):
+            for _ in range(mbs):
+                x, y = next(its[d])
+                loss = nn.functional.mse_loss(ref(x), y)
+                (loss / (mbs * 2)).backward()
+                tot += loss.item()
+        opt.step()
+        opt.zero_grad()
+        ref_losses.append(tot / (mbs * 2))
+    for got, want in zip(losses, ref_losses):
+        assert abs(got - want) < 1e-5, (losses, ref_losses)
+
+    # my TP shards must match the reference slices after optimization
+    tr = net.grid.tensor_parallel_id
+    ref_blocks = list(ref)[net.part_start:net.part_end]
+    for m, r in zip(net.forward_funcs, ref_blocks):
+        torch.testing.assert_close(
+            m.col.weight, r.fc1.weight.detach().chunk(2, dim=0)[tr],
+            rtol=1e-4, atol=2e-5)
+        torch.testing.assert_close(
+            m.row.weight, r.fc2.weight.detach().chunk(2, dim=1)[tr],
+            rtol=1e-4, atol=2e-5)
+
+
+def test_pipeline_3d_pp2tp2dp2_ws8():
+    run_distributed(_pp2tp2dp2_worker, world_size=8, timeout=600)
