@@ -271,7 +271,7 @@ def test_zero3_backpressure_stress_ws8():
     run_distributed(_z3_stress_worker, world_size=8, timeout=600)
 
 
-def _pp2tp2dp2_worker(rank, world):
+def _pp2tp2dp2_worker(rank, world, zero_stage=0):
     """Full 3D: PP=2 x TP=2 x DP=2 on 8 ranks. Each pipeline stage holds
     TP-sharded blocks (column -> row parallel pair); parity against the
     unsharded sequential model with dp-averaged gradients."""
@@ -313,6 +313,9 @@ def _pp2tp2dp2_worker(rank, world):
     cfg = {"train_micro_batch_size_per_gpu": 4,
            "gradient_accumulation_steps": 2,
            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}}
+    if zero_stage:
+        cfg["zero_optimization"] = {"stage": zero_stage,
+                                    "overlap_comm": False}
     specs = [LayerSpec(TPBlock, 21), LayerSpec(TPBlock, 22),
              LayerSpec(TPBlock, 23), LayerSpec(TPBlock, 24)]
     net = PipelineModule(specs, num_stages=2, tp_size=2,
@@ -365,3 +368,9 @@ def _pp2tp2dp2_worker(rank, world):
 
 def test_pipeline_3d_pp2tp2dp2_ws8():
     run_distributed(_pp2tp2dp2_worker, world_size=8, timeout=600)
+
+
+def test_pipeline_3d_zero1_ws8():
+    """Same 3D mesh with ZeRO-1 partitioning over each (stage, tp) cell's
+    DP replicas — exercises the combined pipe x tensor norm group."""
+    run_distributed(_pp2tp2dp2_worker, world_size=8, timeout=600, args=(1,))
