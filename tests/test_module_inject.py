@@ -72,3 +72,76 @@ def test_injection_flash_sdpa_gpu():
         got = m(ids).logits
     assert policy.injected.get("flash_sdpa_calls", 0) >= 2, policy.injected
     torch.testing.assert_close(got.float(), ref.float(), rtol=5e-2, atol=5e-1)
+
+
+def test_diffusers_block_injection_parity():
+    """Fused diffusers transformer block vs a synthetic HF-diffusers-style
+    BasicTransformerBlock (reference diffusers_transformer_block.py)."""
+    import torch.nn as nn
+    from deepspeed_amd.module_inject.diffusers import (
+        replace_diffusers_blocks, DiffusersTransformerBlock)
+
+    D, FF, H = 32, 64, 4
+
+    class MiniAttn(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.qkv = nn.Linear(D, 3 * D, bias=False)
+            self.out = nn.Linear(D, D, bias=False)
+
+        def forward(self, x, context=None):
+            src = x if context is None else context
+            q = self.qkv(x)[..., :D]
+            k = self.qkv(src)[..., D:2 * D]
+            v = self.qkv(src)[..., 2 * D:]
+            B, S, _ = q.shape
+            Sk = k.shape[1]
+            q = q.view(B, S, H, D // H).transpose(1, 2)
+            k = k.view(B, Sk, H, D // H).transpose(1, 2)
+            v = v.view(B, Sk, H, D // H).transpose(1, 2)
+            o = torch.nn.functional.scaled_dot_product_attention(q, k, v)
+            return self.out(o.transpose(1, 2).reshape(B, S, D))
+
+    class GEGLU(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.proj = nn.Linear(D, 2 * FF)
+
+        def forward(self, x):
+            up, gate = self.proj(x).chunk(2, dim=-1)
+            return up * torch.nn.functional.gelu(gate)
+
+    class Block(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.norm1 = nn.LayerNorm(D)
+            self.norm2 = nn.LayerNorm(D)
+            self.norm3 = nn.LayerNorm(D)
+            self.attn1 = MiniAttn()
+            self.attn2 = MiniAttn()
+            self.ff = nn.Module()
+            self.ff.net = nn.ModuleList(
+                [GEGLU(), nn.Dropout(0.0), nn.Linear(FF, D)])
+
+        def forward(self, x, context=None):
+            x = self.attn1(self.norm1(x)) + x
+            x = self.attn2(self.norm2(x), context) + x
+            h = self.norm3(x)
+            for m in self.ff.net:
+                h = m(h)
+            return h + x
+
+    torch.manual_seed(5)
+    model = nn.Sequential()
+    model.block = Block()
+    x = torch.randn(2, 9, D)
+    ctx = torch.randn(2, 5, D)
+    want = model.block(x, context=ctx)
+    n = replace_diffusers_blocks(model)
+    assert n == 1 and isinstance(model.block, DiffusersTransformerBlock)
+    got = model.block(x, context=ctx)
+    # tanh-approx gelu in the fused GEGLU vs diffusers' exact gelu
+    torch.testing.assert_close(got, want, rtol=2e-3, atol=3e-4)
+    # encoder_hidden_states kwarg spelling (diffusers >= 0.11)
+    got2 = model.block(x, encoder_hidden_states=ctx)
+    torch.testing.assert_close(got2, got, rtol=0, atol=0)
