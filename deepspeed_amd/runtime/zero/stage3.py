@@ -226,6 +226,12 @@ class ZeroStage3Optimizer:
         self._recording = True
 
         self._trace_misses = 0
+        # fetch profiler (reference: partitioned_param_coordinator event
+        # counters / ZeRO-3 fetch tracing): demand = gather launched by the
+        # module's own pre-hook (prefetch arrived too late or missed),
+        # prefetched = gather launched ahead by the trace walker
+        self.fetch_stats = {"gathers": 0, "prefetched": 0, "demand": 0,
+                            "trace_misses": 0, "steps": 0}
         self._inflight_rs = []   # (handle, recv, unit, grad_full_ref)
         self._pending_release: List[_Unit] = []
         self._hooks = []
@@ -453,6 +459,7 @@ class ZeroStage3Optimizer:
     def _launch_gather(self, u: _Unit):
         if u.status != FREE:
             return
+        self.fetch_stats["gathers"] += 1
         # hpZ: gather from the node-local secondary shard over the small
         # group instead of the primary shard over the whole world
         if u.sec_shard is not None:
@@ -556,6 +563,8 @@ class ZeroStage3Optimizer:
         units = self._units_for(mod)
         for u in units:
             u.release_pending = False
+            if u.status == FREE:
+                self.fetch_stats["demand"] += 1
             self._launch_gather(u)
         self._flush_pending_releases(keep=units)
         if self._recording:
@@ -582,6 +591,8 @@ class ZeroStage3Optimizer:
         for u in units:
             u.release_pending = False
             u.in_backward = True
+            if u.status == FREE:
+                self.fetch_stats["demand"] += 1
             self._launch_gather(u)
         self._flush_pending_releases(keep=units)
         if self._trace_complete:
@@ -721,6 +732,7 @@ class ZeroStage3Optimizer:
         while i < len(trace) and budget > 0:
             u = self.units[trace[i]]
             if u.status == FREE:
+                self.fetch_stats["prefetched"] += 1
                 self._launch_gather(u)
                 budget -= u.numel
             i += 1
@@ -753,6 +765,13 @@ class ZeroStage3Optimizer:
             self._rtrace = []
             self._trace_complete = False
             self._recording = True
+        self.fetch_stats["trace_misses"] += self._trace_misses
+        self.fetch_stats["steps"] += 1
+        if os.environ.get("DS_AMD_FETCH_PROFILE") == "1":
+            s = self.fetch_stats
+            log_dist(f"ZeRO-3 fetch: gathers={s['gathers']} "
+                     f"prefetched={s['prefetched']} demand={s['demand']} "
+                     f"trace_misses={s['trace_misses']} steps={s['steps']}")
         self._trace_misses = 0
         self._fwd_cursor = 0
         self._bwd_cursor = 0
