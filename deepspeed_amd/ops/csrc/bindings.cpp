@@ -42,6 +42,9 @@ extern "C" void ds_nhwc_bias_add(const void* act, const void* bias,
                                  const void* other, const void* other_bias,
                                  void* out, long long n, int channels,
                                  int dtype, void* stream);
+extern "C" void ds_token_move(const void* x, void* y, const int* idx, int B,
+                              int S, int K, int D, int gather, int dtype,
+                              void* stream);
 extern "C" void ds_cpu_adam_flat(float* p, const void* g, int grad_dtype,
                                  float* m, float* v, void* p16, long long n,
                                  float lr, float beta1, float beta2, float eps,
@@ -402,6 +405,40 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> norm_bwd(
   return {dx, dw, db};
 }
 
+at::Tensor token_gather(at::Tensor x, at::Tensor idx) {
+  TORCH_CHECK(x.dim() == 3 && x.is_cuda() && x.is_contiguous(),
+              "token_gather: contiguous [B,S,D] GPU tensor");
+  TORCH_CHECK(idx.dim() == 2 && idx.scalar_type() == at::kInt &&
+              idx.is_contiguous() && idx.size(0) == x.size(0),
+              "token_gather: int32 [B,K] indices");
+  const int64_t D = x.size(2);
+  const int64_t vec = 16 / x.element_size();
+  TORCH_CHECK(D % vec == 0, "token_gather: D % ", vec, " == 0");
+  auto y = at::empty({x.size(0), idx.size(1), D}, x.options());
+  ds_token_move(x.data_ptr(), y.data_ptr(), idx.data_ptr<int>(),
+                (int)x.size(0), (int)x.size(1), (int)idx.size(1), (int)D,
+                1, dtype_code(x), cur_stream());
+  return y;
+}
+
+at::Tensor token_scatter(at::Tensor base, at::Tensor sub, at::Tensor idx) {
+  TORCH_CHECK(base.dim() == 3 && sub.dim() == 3 && base.is_cuda() &&
+              base.is_contiguous() && sub.is_contiguous() &&
+              base.scalar_type() == sub.scalar_type(),
+              "token_scatter: contiguous [B,S,D]/[B,K,D] GPU tensors");
+  TORCH_CHECK(idx.dim() == 2 && idx.scalar_type() == at::kInt &&
+              idx.is_contiguous() && idx.size(1) == sub.size(1),
+              "token_scatter: int32 [B,K] indices");
+  const int64_t D = base.size(2);
+  const int64_t vec = 16 / base.element_size();
+  TORCH_CHECK(D % vec == 0, "token_scatter: D % ", vec, " == 0");
+  auto y = base.clone();
+  ds_token_move(sub.data_ptr(), y.data_ptr(), idx.data_ptr<int>(),
+                (int)base.size(0), (int)base.size(1), (int)sub.size(1),
+                (int)D, 0, dtype_code(base), cur_stream());
+  return y;
+}
+
 at::Tensor nhwc_bias_add(at::Tensor act, at::Tensor bias,
                          c10::optional<at::Tensor> other,
                          c10::optional<at::Tensor> other_bias) {
@@ -626,6 +663,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_fwd", &norm_fwd, "RMSNorm/LayerNorm forward");
   m.def("norm_bwd", &norm_bwd, "RMSNorm/LayerNorm backward");
   m.def("rope", &rope, "Rotary position embedding (in-place)");
+  m.def("token_gather", &token_gather,
+        "row-coalesced token gather [B,S,D] x [B,K] -> [B,K,D]");
+  m.def("token_scatter", &token_scatter,
+        "row-coalesced token scatter-back into a cloned base");
   m.def("nhwc_bias_add", &nhwc_bias_add,
         "fused channels-last bias (+residual +residual-bias) add",
         pybind11::arg("act"), pybind11::arg("bias"),

@@ -58,15 +58,15 @@ class RandomLayerTokenDrop(nn.Module):
         keep = self.scheduler.current_seq
         if not self.training or keep >= S:
             return self.random_ltd_layer(x, *args, **kwargs)
+        from ...ops.token_ops import token_gather, token_scatter
         idx = torch.stack([
             torch.randperm(S, device=x.device)[:keep].sort().values
             for _ in range(B)])                     # [B, keep] sorted
-        gidx = idx.unsqueeze(-1).expand(B, keep, Hd)
-        sub = x.gather(1, gidx)
+        sub = token_gather(x, idx)
         out = self.random_ltd_layer(sub, *args, **kwargs)
         if isinstance(out, tuple):
             out = out[0]
-        return x.scatter(1, gidx, out)
+        return token_scatter(x, out, idx)
 
 
 def convert_to_random_ltd(model: nn.Module, layers_attr: str,

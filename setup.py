@@ -37,6 +37,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "rope.hip"),
         os.path.join(CSRC, "swiglu.hip"),
         os.path.join(CSRC, "spatial.hip"),
+        os.path.join(CSRC, "token_ops.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17", "-fopenmp", "-mavx2", "-mfma"],

@@ -518,3 +518,29 @@ def test_nhwc_bias_add_gpu():
             want = nhwc_bias_add(*(a.cpu() for a in args)).to(dt)
             torch.testing.assert_close(got.float().cpu(), want.float(),
                                        rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.gpu
+def test_token_gather_scatter_gpu():
+    """Row-coalesced token gather/scatter kernels vs torch gather/scatter
+    (random-LTD path), forward and backward."""
+    from deepspeed_amd.ops.token_ops import token_gather, token_scatter
+    torch.manual_seed(0)
+    B, S, K, D = 3, 64, 23, 256
+    for dt in (torch.bfloat16, torch.float32):
+        x = torch.randn(B, S, D, device="cuda", dtype=dt, requires_grad=True)
+        idx = torch.stack([torch.randperm(S, device="cuda")[:K].sort().values
+                           for _ in range(B)]).int()
+        sub = token_gather(x, idx)
+        g = idx.long().unsqueeze(-1).expand(B, K, D)
+        want_sub = x.gather(1, g)
+        assert torch.equal(sub, want_sub)
+        y = token_scatter(x.detach().requires_grad_(), 2 * sub.detach(), idx)
+        want_y = x.detach().scatter(1, g, 2 * sub.detach())
+        assert torch.equal(y, want_y)
+        # backward through gather
+        loss = (sub.float() ** 2).sum()
+        loss.backward()
+        xr = x.detach().clone().requires_grad_()
+        (xr.gather(1, g).float() ** 2).sum().backward()
+        torch.testing.assert_close(x.grad, xr.grad)
