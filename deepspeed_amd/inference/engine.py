@@ -8,6 +8,7 @@ serving mechanics — tensor-parallel sharding over the xGMI mesh
 (see auto_tp.py), a contiguous KV cache, and a sampling loop.
 """
 
+import os
 from dataclasses import dataclass, field
 from typing import Optional
 
@@ -53,6 +54,9 @@ class InferenceEngine(nn.Module):
             self.injection_policy = HFInjectionPolicy()
             replace_transformer_layer(self.module, self.injection_policy)
 
+        if config.checkpoint:
+            self._load_checkpoint(config.checkpoint)
+
         tp = config.tp_size
         if tp > 1:
             if not dist.is_initialized():
@@ -76,6 +80,53 @@ class InferenceEngine(nn.Module):
             self.module.to(accel.current_device())
         self.device = next(self.module.parameters()).device
         self.module.eval()
+
+    def _load_checkpoint(self, ckpt):
+        """Load weights named by ``config.checkpoint`` (reference
+        inference/engine.py _load_checkpoint): a state-dict file
+        (.pt/.bin/.safetensors), a directory of such shards, or a JSON
+        manifest {"checkpoints": [paths...]}. A model constructed on the
+        meta device is materialized empty first, so rank memory never
+        holds more than one shard + the final weights."""
+        import glob as _glob
+        import json
+
+        paths = []
+        if isinstance(ckpt, (list, tuple)):
+            paths = list(ckpt)
+        elif os.path.isdir(ckpt):
+            for pat in ("*.safetensors", "*.pt", "*.bin"):
+                paths += sorted(_glob.glob(os.path.join(ckpt, pat)))
+        elif ckpt.endswith(".json"):
+            with open(ckpt) as f:
+                manifest = json.load(f)
+            base = os.path.dirname(ckpt)
+            paths = [p if os.path.isabs(p) else os.path.join(base, p)
+                     for p in manifest["checkpoints"]]
+        else:
+            paths = [ckpt]
+        if not paths:
+            raise FileNotFoundError(f"no checkpoint files under {ckpt}")
+
+        if any(p.is_meta for p in self.module.parameters()):
+            self.module.to_empty(device="cpu")
+
+        missing = set(n for n, _ in self.module.named_parameters())
+        for path in paths:
+            if path.endswith(".safetensors"):
+                from safetensors.torch import load_file
+                sd = load_file(path)
+            else:
+                sd = torch.load(path, map_location="cpu", weights_only=True)
+                if isinstance(sd, dict) and "module" in sd and \
+                        isinstance(sd["module"], dict):
+                    sd = sd["module"]  # engine.save_checkpoint layout
+            self.module.load_state_dict(sd, strict=False)
+            missing -= set(sd)
+        if missing:
+            log_dist(f"init_inference checkpoint: {len(missing)} params not "
+                     f"found in {len(paths)} file(s), e.g. "
+                     f"{sorted(missing)[:3]}")
 
     def forward(self, *args, **kwargs):
         with torch.no_grad():

@@ -455,3 +455,61 @@ def test_paged_decode_batcher_gpu():
     ref = run(RaggedKVCache)
     got = run(PagedKVCache)
     assert ref == got, (ref, got)
+
+
+def test_init_inference_checkpoint_loading(tmp_path):
+    """config.checkpoint: single file, sharded directory, JSON manifest,
+    and meta-constructed model materialization (reference
+    inference/engine.py _load_checkpoint)."""
+    import json
+    import deepspeed_amd
+    from deepspeed_amd.models import GPT2ForCausalLM as GPT2Model, gpt2_tiny
+
+    torch.manual_seed(0)
+    src = GPT2Model(gpt2_tiny())
+    ids = torch.randint(0, 100, (1, 8))
+    with torch.no_grad():
+        want = src(ids)
+
+    sd = src.state_dict()
+    # single file
+    f1 = tmp_path / "model.pt"
+    torch.save(sd, f1)
+    torch.manual_seed(123)  # different init
+    eng = deepspeed_amd.init_inference(GPT2Model(gpt2_tiny()),
+                                       checkpoint=str(f1), dtype="fp32")
+    with torch.no_grad():
+        got = eng.module(ids.to(eng.device)).cpu()
+    torch.testing.assert_close(got, want)
+
+    # sharded directory + manifest
+    keys = sorted(sd)
+    half = len(keys) // 2
+    d = tmp_path / "shards"
+    d.mkdir()
+    torch.save({k: sd[k] for k in keys[:half]}, d / "shard_0.pt")
+    torch.save({k: sd[k] for k in keys[half:]}, d / "shard_1.pt")
+    eng = deepspeed_amd.init_inference(GPT2Model(gpt2_tiny()),
+                                       checkpoint=str(d), dtype="fp32")
+    with torch.no_grad():
+        got = eng.module(ids.to(eng.device)).cpu()
+    torch.testing.assert_close(got, want)
+
+    man = tmp_path / "ckpt.json"
+    man.write_text(json.dumps(
+        {"checkpoints": ["shards/shard_0.pt", "shards/shard_1.pt"]}))
+    eng = deepspeed_amd.init_inference(GPT2Model(gpt2_tiny()),
+                                       checkpoint=str(man), dtype="fp32")
+    with torch.no_grad():
+        got = eng.module(ids.to(eng.device)).cpu()
+    torch.testing.assert_close(got, want)
+
+    # meta-device construction: zero allocation until the load
+    with torch.device("meta"):
+        meta_model = GPT2Model(gpt2_tiny())
+    assert next(meta_model.parameters()).is_meta
+    eng = deepspeed_amd.init_inference(meta_model, checkpoint=str(f1),
+                                       dtype="fp32")
+    with torch.no_grad():
+        got = eng.module(ids.to(eng.device)).cpu()
+    torch.testing.assert_close(got, want)
