@@ -254,6 +254,9 @@ class Engine(torch.nn.Module):
                 gradient_predivide_factor=self.config.gradient_predivide_factor,
                 cpu_offload=offload,
                 offload_pin_memory=self.config.zero.offload_optimizer.pin_memory,
+                grad_accum_dtype=(torch.float32
+                                  if self.config.zero.fp32_grad_accum
+                                  else None),
                 mpu=self.mpu,
             )
         else:
@@ -326,6 +329,14 @@ class Engine(torch.nn.Module):
 
     @instrument_w_nvtx
     def forward(self, *inputs, **kwargs):
+        if self.config.fp16.enabled and self.config.fp16.auto_cast:
+            # reference fp16 auto_cast: float inputs arrive fp32, cast once
+            # at the engine boundary instead of inside every module
+            inputs = tuple(t.to(self.dtype) if torch.is_tensor(t)
+                           and t.is_floating_point() else t for t in inputs)
+            kwargs = {k: (v.to(self.dtype) if torch.is_tensor(v)
+                          and v.is_floating_point() else v)
+                      for k, v in kwargs.items()}
         if self.wall_clock_breakdown:
             self.timers(FORWARD_GLOBAL_TIMER).start()
         if self.flops_profiler is not None and \
@@ -817,6 +828,13 @@ class Engine(torch.nn.Module):
 
     def save_16bit_model(self, save_dir, save_filename="pytorch_model.bin",
                          exclude_frozen_parameters=False):
+        if self.zero_stage == 3 and \
+                not self.config.zero.stage3_gather_16bit_weights_on_model_save:
+            logger.warning(
+                "save_16bit_model skipped: ZeRO-3 params are partitioned and "
+                "stage3_gather_16bit_weights_on_model_save is false "
+                "(reference engine.save_16bit_model behavior)")
+            return False
         sd = self.module_state_dict(exclude_frozen_parameters)
         if self.global_rank == 0:
             os.makedirs(save_dir, exist_ok=True)

@@ -98,6 +98,7 @@ class ZeroStage12Optimizer:
         self.gradient_predivide_factor = gradient_predivide_factor
         self.cpu_offload = cpu_offload
         self.offload_pin_memory = offload_pin_memory
+        self.grad_accum_dtype = grad_accum_dtype
         self.mpu = mpu
         self.is_gradient_accumulation_boundary = True
         self.overflow = False
@@ -187,15 +188,22 @@ class ZeroStage12Optimizer:
                 cur_numel += p.numel()
             close_bucket()
 
+            # fp32_grad_accum: accumulate micro-step grads in an fp32 flat
+            # buffer (reference bf16_optimizer grad accumulation dtype);
+            # .grad can no longer alias the buffer (dtype differs), so the
+            # post-accumulate hook folds-and-frees each 16-bit grad instead
+            gdtype = dtype if self.grad_accum_dtype is None \
+                or dtype == torch.float32 else self.grad_accum_dtype
             shard_total = 0
             for b in group_buckets:
                 b.flat = torch.zeros(b.numel, dtype=dtype, device=device)
-                b.grad_flat = torch.zeros(b.numel, dtype=dtype, device=device)
+                b.grad_flat = torch.zeros(b.numel, dtype=gdtype, device=device)
                 for p, off in zip(b.params, b.offsets):
                     with torch.no_grad():
                         b.flat[off:off + p.numel()].copy_(p.data.view(-1))
                     p.data = b.flat[off:off + p.numel()].view_as(p.data)
-                    p.grad = b.grad_flat[off:off + p.numel()].view_as(p.data)
+                    if gdtype == dtype:
+                        p.grad = b.grad_flat[off:off + p.numel()].view_as(p.data)
                     self.param_to_bucket[p] = b
                 b.master_offset = shard_total
                 shard_total += b.shard_size
@@ -203,7 +211,7 @@ class ZeroStage12Optimizer:
 
             master_device = torch.device("cpu") if self.cpu_offload else device
             master = torch.empty(shard_total, dtype=torch.float32, device=master_device)
-            owned = torch.empty(shard_total, dtype=dtype, device=master_device)
+            owned = torch.empty(shard_total, dtype=gdtype, device=master_device)
             if self.cpu_offload and self.offload_pin_memory and accel.available():
                 master = master.pin_memory()
                 owned = owned.pin_memory()
@@ -239,7 +247,14 @@ class ZeroStage12Optimizer:
             # autograd accumulates directly into the preset flat view; if torch
             # ever replaced .grad (it should not), fold it back in.
             expected = bucket.grad_flat[offset:offset + param.numel()]
-            if param.grad is not None and \
+            if expected.dtype != param.dtype:
+                # fp32 accumulation mode: fold the fresh 16-bit grad into the
+                # fp32 flat and free it
+                if param.grad is not None:
+                    expected.add_(param.grad.detach().view(-1)
+                                  .to(expected.dtype))
+                    param.grad = None
+            elif param.grad is not None and \
                     param.grad.data_ptr() != expected.data_ptr():
                 expected.add_(param.grad.detach().view(-1))
                 param.grad = expected.view_as(param)

@@ -132,6 +132,7 @@ class ZeroStage3Optimizer:
         self.prefetch_bucket_size = int(zc.stage3_prefetch_bucket_size)
         self.persistence_threshold = int(zc.stage3_param_persistence_threshold)
         self.max_live_parameters = int(zc.stage3_max_live_parameters)
+        self.max_reuse_distance = int(zc.stage3_max_reuse_distance)
         self.clip_grad = config.gradient_clipping
         self.overlap_comm = zc.overlap_comm
         # ZeRO-Infinity: "nvme" keeps the fp32 master + Adam moments in swap
@@ -523,7 +524,7 @@ class ZeroStage3Optimizer:
         u.status = FREE
         u.release_pending = False
 
-    def _flush_pending_releases(self, keep=()):
+    def _flush_pending_releases(self, keep=(), phase=None):
         if not self._pending_release:
             return
         still = []
@@ -531,8 +532,29 @@ class ZeroStage3Optimizer:
             if u in keep or u.in_backward:
                 still.append(u)
             elif u.release_pending:
-                self._release(u)
+                if phase is not None and self._reuse_within(u, phase):
+                    still.append(u)   # next use is close — keep resident
+                else:
+                    self._release(u)
         self._pending_release = still
+
+    def _reuse_within(self, u, phase):
+        """stage3_max_reuse_distance (reference partitioned_param_coordinator
+        __params_to_release): skip releasing a unit whose next use in the
+        recorded trace is within the distance budget (elements) — tied/
+        chunk-reused modules would otherwise re-gather immediately."""
+        if not self._trace_complete or self.max_reuse_distance <= 0:
+            return False
+        t, c = (self._trace, self._fwd_cursor) if phase == "fwd" else \
+            (self._rtrace, self._bwd_cursor)
+        elems = 0
+        for i in range(c, len(t)):
+            if t[i] == u.index:
+                return elems < self.max_reuse_distance
+            elems += self.units[t[i]].numel
+            if elems >= self.max_reuse_distance:
+                return False
+        return False
 
     # ------------------------------------------------------------------ hooks
 
@@ -566,7 +588,7 @@ class ZeroStage3Optimizer:
             if u.status == FREE:
                 self.fetch_stats["demand"] += 1
             self._launch_gather(u)
-        self._flush_pending_releases(keep=units)
+        self._flush_pending_releases(keep=units, phase="fwd")
         if self._recording:
             for u in units:
                 self._trace.append(u.index)
@@ -594,7 +616,7 @@ class ZeroStage3Optimizer:
             if u.status == FREE:
                 self.fetch_stats["demand"] += 1
             self._launch_gather(u)
-        self._flush_pending_releases(keep=units)
+        self._flush_pending_releases(keep=units, phase="bwd")
         if self._trace_complete:
             self._advance_bwd_cursor(units)
             self._prefetch(self._rtrace, self._bwd_cursor)

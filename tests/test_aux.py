@@ -644,3 +644,64 @@ def test_contiguous_allocator_defrag():
     b.defragment()
     assert torch.equal(payload.tensor,
                        torch.arange(6000, dtype=torch.float32))
+
+
+def test_fp16_auto_cast_inputs():
+    """fp16.auto_cast: fp32 float inputs are cast to the engine dtype at
+    the engine boundary (reference fp16 auto_cast)."""
+    import deepspeed_amd
+
+    seen = {}
+
+    class Probe(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            seen["dtype"] = x.dtype
+            y = self.fc(x)
+            return torch.nn.functional.mse_loss(y.float(),
+                                                labels.float())
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=Probe(), config={
+        "train_micro_batch_size_per_gpu": 2,
+        "fp16": {"enabled": True, "auto_cast": True,
+                 "initial_scale_power": 4},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    loss = eng(torch.randn(2, 8), labels=torch.randn(2, 1))
+    assert seen["dtype"] == torch.float16
+    eng.backward(loss)
+    eng.step()
+
+
+def test_save_16bit_model_zero3_gate(tmp_path):
+    """stage3_gather_16bit_weights_on_model_save=false refuses the
+    consolidated save instead of silently gathering."""
+    import deepspeed_amd
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            return torch.nn.functional.mse_loss(
+                self.fc(x).float(), labels.float())
+
+    def build(gather):
+        return deepspeed_amd.initialize(model=M(), config={
+            "train_micro_batch_size_per_gpu": 2,
+            "bf16": {"enabled": True},
+            "zero_optimization": {
+                "stage": 3,
+                "stage3_gather_16bit_weights_on_model_save": gather},
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})[0]
+
+    eng = build(False)
+    assert eng.save_16bit_model(str(tmp_path)) is False
+    assert not (tmp_path / "pytorch_model.bin").exists()
+    eng = build(True)
+    assert eng.save_16bit_model(str(tmp_path)) is True
+    sd = torch.load(tmp_path / "pytorch_model.bin", weights_only=True)
+    assert "fc.weight" in sd and sd["fc.weight"].shape == (1, 8)

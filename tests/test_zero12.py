@@ -861,3 +861,102 @@ def _fetch_stats_worker(rank=0, world=1):
     assert s["trace_misses"] == 0
     # steady state (steps 2-4) runs off the trace: prefetch does the work
     assert s["prefetched"] > 0
+
+
+def test_zero_fp32_grad_accum_ws2():
+    """fp32_grad_accum: micro-step grads accumulate in an fp32 flat buffer
+    (.grad can't alias it, the hook folds-and-frees each bf16 grad)."""
+    run_distributed(_fp32_accum_worker, world_size=2)
+
+
+def _fp32_accum_worker(rank, world):
+    import deepspeed_amd
+    lr, steps, gas = 1e-2, 3, 3
+    model = TinyNet()
+    ref_model = copy.deepcopy(model)
+    data = _make_data(steps * gas)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 4,
+        "gradient_accumulation_steps": gas,
+        "bf16": {"enabled": True},
+        "zero_optimization": {"stage": 2, "overlap_comm": False,
+                              "fp32_grad_accum": True},
+        "optimizer": {"type": "AdamW", "params": {"lr": lr}}})
+    assert engine.optimizer.buckets[0].grad_flat.dtype == torch.float32
+    it = iter(data)
+    losses = []
+    for _ in range(steps):
+        micro = []
+        for _ in range(gas):
+            xs, ys = next(it)
+            loss = engine(xs.to(engine.device).bfloat16(), labels=ys)
+            engine.backward(loss)
+            micro.append(loss.item())
+        engine.step()
+        losses.append(sum(micro) / len(micro))
+    ref_losses, ref_master = _reference_mixed_precision_loop(
+        ref_model, data, lr, steps, gas, torch.bfloat16)
+    for a, b in zip(losses, ref_losses):
+        assert abs(a - b) < 2e-2, (losses, ref_losses)
+    for p_e, p_r in zip(engine.module.parameters(), ref_master.parameters()):
+        assert torch.allclose(p_e.float().cpu(), p_r.to(p_e.dtype).float(),
+                              atol=3e-2, rtol=3e-2)
+
+
+class _ReuseNet(torch.nn.Module):
+    """a -> b -> a again: with max_reuse_distance=0 the 'a' unit is
+    released when 'b' fetches and must re-gather; with a large distance it
+    stays resident through the second use."""
+
+    def __init__(self):
+        super().__init__()
+        torch.manual_seed(9)
+        self.a = torch.nn.Linear(32, 32)
+        self.b = torch.nn.Linear(32, 32)
+
+    def forward(self, x, labels=None):
+        y = self.a(torch.tanh(self.b(self.a(x))))
+        if labels is not None:
+            return torch.nn.functional.mse_loss(y.float(), labels.float())
+        return y
+
+
+def test_zero3_max_reuse_distance():
+    run_local(_reuse_worker)
+
+
+def _steady_gathers(reuse_distance):
+    import deepspeed_amd
+    model = _ReuseNet()
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 4,
+        "bf16": {"enabled": True},
+        "zero_optimization": {"stage": 3, "overlap_comm": False,
+                              "stage3_max_reuse_distance": reuse_distance,
+                              "stage3_param_persistence_threshold": 0,
+                              "stage3_prefetch_bucket_size": 0},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    data = _make_data(3, hidden=32)
+    losses = []
+    for xs, ys in data:
+        loss = engine(xs.bfloat16(), labels=ys)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    before = dict(engine.optimizer.fetch_stats)
+    xs, ys = data[0]
+    loss = engine(xs.bfloat16(), labels=ys)
+    engine.backward(loss)
+    engine.step()
+    steady = engine.optimizer.fetch_stats["gathers"] - before["gathers"]
+    return steady, losses
+
+
+def _reuse_worker(rank, world):
+    # the reuse hold must save at least one re-gather of 'a' per step
+    g_nohold, losses = _steady_gathers(0)
+    g_hold, losses_hold = _steady_gathers(1_000_000_000)
+    assert g_hold < g_nohold, (g_hold, g_nohold)
+    for a, b in zip(losses, losses_hold):
+        assert abs(a - b) < 1e-6
+
