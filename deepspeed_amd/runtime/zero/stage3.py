@@ -142,6 +142,21 @@ class ZeroStage3Optimizer:
         self.nvme_offload = zc.offload_optimizer.device == "nvme"
         self.cpu_offload = zc.offload_optimizer.device in ("cpu", "nvme")
         self.offload_pin_memory = zc.offload_optimizer.pin_memory
+        # ZeRO-Infinity parameter tier: permanent shards in pinned host
+        # memory (reference offload_config OffloadParamConfig; nvme params
+        # ride the same host path — the optimizer-state NVMe tier is where
+        # the capacity win is, see _nvme_chunks)
+        self.param_offload = zc.offload_param.device in ("cpu", "nvme")
+        if self.param_offload:
+            if zc.zero_quantized_weights or \
+                    int(zc.zero_hpz_partition_size) > 1:
+                raise ValueError("offload_param does not compose with "
+                                 "qwZ/hpZ (host shards cannot feed the "
+                                 "quantized/secondary gather paths)")
+            if zc.offload_param.device == "nvme":
+                log_dist("offload_param.device=nvme: parameter shards are "
+                         "held in pinned host memory (the NVMe tier applies "
+                         "to optimizer state via offload_optimizer)")
         self._swapper = None
         if self.nvme_offload:
             from ..swap_tensor.swapper import AsyncTensorSwapper
@@ -368,6 +383,17 @@ class ZeroStage3Optimizer:
             else:
                 # ws=1: the shard IS the full buffer — keep it, no clone
                 u.shard = full
+            if self.param_offload and not u.persist:
+                # ZeRO-Infinity parameter tier: the permanent shard lives in
+                # pinned host memory; gathers stage it H2D first. Persistent
+                # (small) units stay device-resident — offloading them
+                # would cost a host round-trip per step for ~no memory.
+                host = torch.empty(u.shard.shape, dtype=u.shard.dtype,
+                                   device="cpu")
+                if self.offload_pin_memory and accel.available():
+                    host = host.pin_memory()
+                host.copy_(u.shard)
+                u.shard = host
             for p in u.params:
                 p.ds_shape = p.shape
                 p.ds_numel = p.numel()
@@ -467,6 +493,10 @@ class ZeroStage3Optimizer:
             src, group, gworld = u.sec_shard, self.hpz_group, self.hpz_world
         else:
             src, group, gworld = u.shard, self.dp_group, self.world_size
+        if src.device.type == "cpu" and self._device.type != "cpu":
+            # offload_param: stage the pinned host shard H2D; stream order
+            # sequences the copy before the allgather / first use
+            src = src.to(self._device, non_blocking=True)
         if self.world_size > 1 and self.quantized_weights:
             from ...ops.quantizer import quantize
             gs = self._unit_quant_group(src.numel())
@@ -488,8 +518,10 @@ class ZeroStage3Optimizer:
                                                    group=group,
                                                    async_op=True)
         else:
-            # ws=1: shard covers the whole unit — alias, no alloc, no copy
-            u.full = u.shard
+            # ws=1: shard covers the whole unit — alias when device-resident
+            # (no alloc, no copy); with offload_param, src is the fresh H2D
+            # staging copy
+            u.full = src
             u.handle = None
         u.status = INFLIGHT
 
@@ -888,6 +920,11 @@ class ZeroStage3Optimizer:
                 if u.group_idx != gi or not u.trainable:
                     continue
                 out16 = u.shard if u.shard.dtype == torch.bfloat16 else None
+                if out16 is not None and out16.device != master.device:
+                    # param-only offload (host shard, device master): the
+                    # fused kernel can't write across; fall back to the
+                    # post-step D2H copy
+                    out16 = None
                 if out16 is None:
                     wrote_params = False
                 if self.direct_grad:

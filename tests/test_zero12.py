@@ -960,3 +960,48 @@ def _reuse_worker(rank, world):
     for a, b in zip(losses, losses_hold):
         assert abs(a - b) < 1e-6
 
+
+
+@pytest.mark.parametrize("opt_offload", [True, False])
+def test_zero3_param_offload_parity_ws2(opt_offload):
+    """ZeRO-Infinity parameter tier: permanent shards in host memory,
+    staged H2D per gather — with and without the optimizer-state tier."""
+    run_distributed(_param_offload_worker, world_size=2,
+                    args=(opt_offload,))
+
+
+def _param_offload_worker(rank, world, opt_offload):
+    import deepspeed_amd
+    lr, steps = 1e-2, 4
+    model = TinyNet()
+    ref_model = copy.deepcopy(model)
+    data = _make_data(steps)
+    zconf = {"stage": 3, "overlap_comm": False,
+             "stage3_param_persistence_threshold": 64,
+             "offload_param": {"device": "cpu", "pin_memory": True}}
+    if opt_offload:
+        zconf["offload_optimizer"] = {"device": "cpu", "pin_memory": True}
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 4,
+        "bf16": {"enabled": True},
+        "zero_optimization": zconf,
+        "optimizer": {"type": "AdamW", "params": {"lr": lr}}})
+    big = [u for u in engine.optimizer.units if not u.persist]
+    assert big and all(u.shard.device.type == "cpu" for u in big)
+    losses = []
+    for xs, ys in data:
+        loss = engine(xs.to(engine.device).bfloat16(), labels=ys)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    ref_losses, ref_master = _reference_mixed_precision_loop(
+        ref_model, data, lr, steps, 1, torch.bfloat16)
+    for a, b in zip(losses, ref_losses):
+        assert abs(a - b) < 2e-2, (losses, ref_losses)
+    sd = engine.optimizer.get_full_state_dict()
+    if rank == 0:
+        ref_sd = ref_master.state_dict()
+        for name, t in sd.items():
+            assert torch.allclose(t.float().cpu(),
+                                  ref_sd[name].to(t.dtype).float(),
+                                  atol=3e-2, rtol=3e-2), name
