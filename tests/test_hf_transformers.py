@@ -174,3 +174,33 @@ def _autotp_gpt2_worker(rank, world):
 
 def test_autotp_hf_gpt2_forward_parity_ws2():
     run_distributed(_autotp_gpt2_worker, world_size=2)
+
+
+@pytest.mark.parametrize("model_type", ["llama", "mistral", "qwen2"])
+def test_kernel_inject_hf_arch_zoo(model_type):
+    """Kernel injection across the HF decoder zoo (reference
+    module_inject/containers/*): RMSNorm + SwiGLU swaps must fire on every
+    arch and preserve forward numerics."""
+    from transformers import AutoConfig, AutoModelForCausalLM
+    from deepspeed_amd.module_inject import replace_transformer_layer, \
+        HFInjectionPolicy
+    from deepspeed_amd.ops.norms import RMSNorm
+
+    cfg = AutoConfig.for_model(
+        model_type, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        vocab_size=256, eos_token_id=0, pad_token_id=0)
+    torch.manual_seed(0)
+    model = AutoModelForCausalLM.from_config(cfg)
+    model.eval()
+    ids = torch.randint(0, 256, (1, 12))
+    with torch.no_grad():
+        want = model(ids).logits
+    policy = HFInjectionPolicy()
+    replace_transformer_layer(model, policy)
+    n = sum(policy.injected.values())
+    assert n >= 2 * cfg.num_hidden_layers, policy.injected
+    assert any(isinstance(m, RMSNorm) for m in model.modules())
+    with torch.no_grad():
+        got = model(ids).logits
+    torch.testing.assert_close(got, want, rtol=2e-3, atol=2e-3)
