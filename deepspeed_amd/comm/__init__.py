@@ -270,12 +270,20 @@ class _ChainedHandle:
 class CommsLogger:
     """Times collectives and estimates algorithmic bandwidth."""
 
-    def __init__(self, enabled=False, verbose=False):
+    def __init__(self, enabled=False, verbose=False, prof_all=True,
+                 prof_ops=(), debug=False):
         self.enabled = enabled
-        self.verbose = verbose
+        self.verbose = verbose or debug
+        self.prof_all = prof_all
+        self.prof_ops = set(prof_ops)
         self.records = {}  # name -> [count, total_bytes, total_time_s]
 
+    def wants(self, name):
+        return self.prof_all or name in self.prof_ops
+
     def timed(self, name, nbytes, world_size, fn):
+        if not self.wants(name):
+            return fn()
         accel.synchronize()
         t0 = time.perf_counter()
         out = fn()
@@ -318,9 +326,12 @@ class CommsLogger:
         return "\n".join(lines)
 
 
-def configure_comms_logger(enabled=False, verbose=False) -> CommsLogger:
+def configure_comms_logger(enabled=False, verbose=False, prof_all=True,
+                           prof_ops=(), debug=False) -> CommsLogger:
     global _comms_logger
-    _comms_logger = CommsLogger(enabled=enabled, verbose=verbose)
+    _comms_logger = CommsLogger(enabled=enabled, verbose=verbose,
+                                prof_all=prof_all, prof_ops=prof_ops,
+                                debug=debug)
     return _comms_logger
 
 

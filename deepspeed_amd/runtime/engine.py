@@ -84,8 +84,11 @@ class Engine(torch.nn.Module):
         self.wall_clock_breakdown = self.config.wall_clock_breakdown
 
         if self.config.comms_logger.enabled:
-            dist.configure_comms_logger(enabled=True,
-                                        verbose=self.config.comms_logger.verbose)
+            dist.configure_comms_logger(
+                enabled=True,
+                verbose=self.config.comms_logger.verbose,
+                prof_all=self.config.comms_logger.prof_all,
+                debug=self.config.comms_logger.debug)
 
         self.monitor = None
         if self.config.monitor.enabled:
@@ -351,6 +354,7 @@ class Engine(torch.nn.Module):
             if self.global_rank == 0:
                 self.flops_profiler.print_model_profile(
                     profile_step=self.global_steps,
+                    module_depth=self.config.flops_profiler.module_depth,
                     top_modules=self.config.flops_profiler.top_modules,
                     detailed=self.config.flops_profiler.detailed,
                     output_file=self.config.flops_profiler.output_file)
