@@ -1,8 +1,8 @@
 """Model zoo: Llama-3 (flagship), GPT-2, Mixtral-style MoE."""
 
 from .llama import (LlamaConfig, LlamaForCausalLM, llama3_8b, llama3_70b,
-                    llama_tiny, llama_mini, phi3_mini)
-from .gpt2 import GPT2Config, GPT2ForCausalLM, gpt2_small, gpt2_tiny
+                    llama_tiny, llama_mini, phi3_mini, qwen2_7b, qwen2_mini)
+from .gpt2 import GPT2Config, GPT2ForCausalLM, gpt2_small, gpt2_tiny, opt_125m, opt_mini
 from .bert import (BertConfig, BertForPreTraining, BertModel, bert_base,
                    bert_large, bert_tiny)
 from .mixtral import (MixtralConfig, MixtralForCausalLM, mixtral_8x7b,
@@ -10,8 +10,8 @@ from .mixtral import (MixtralConfig, MixtralForCausalLM, mixtral_8x7b,
 
 __all__ = [
     "LlamaConfig", "LlamaForCausalLM", "llama3_8b", "llama3_70b",
-    "llama_tiny", "llama_mini", "phi3_mini", "GPT2Config", "GPT2ForCausalLM",
-    "gpt2_small", "gpt2_tiny", "BertConfig", "BertModel",
+    "llama_tiny", "llama_mini", "phi3_mini", "qwen2_7b", "qwen2_mini", "GPT2Config", "GPT2ForCausalLM",
+    "gpt2_small", "gpt2_tiny", "opt_125m", "opt_mini", "BertConfig", "BertModel",
     "BertForPreTraining", "bert_base", "bert_large", "bert_tiny",
     "MixtralConfig", "MixtralForCausalLM",
     "mixtral_8x7b", "mixtral_tiny", "mixtral_mini",

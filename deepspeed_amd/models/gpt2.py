@@ -24,6 +24,8 @@ class GPT2Config:
     max_seq_len: int = 1024
     ln_eps: float = 1e-5
     initializer_range: float = 0.02
+    activation: str = "gelu"   # "relu" for OPT-family
+    mlp_ratio: int = 4
 
     @property
     def head_dim(self):
@@ -32,6 +34,18 @@ class GPT2Config:
 
 def gpt2_small():
     return GPT2Config()
+
+
+def opt_125m():
+    """OPT-125M shapes: GPT-2 topology with ReLU MLPs (reference
+    inference/v2/model_implementations/opt)."""
+    return GPT2Config(vocab_size=50272, hidden_size=768, num_layers=12,
+                      num_heads=12, max_seq_len=2048, activation="relu")
+
+
+def opt_mini():
+    return GPT2Config(vocab_size=512, hidden_size=64, num_layers=2,
+                      num_heads=4, max_seq_len=128, activation="relu")
 
 
 def gpt2_tiny():
@@ -47,8 +61,9 @@ class GPT2Block(nn.Module):
         self.attn = nn.Linear(h, 3 * h)
         self.attn_out = nn.Linear(h, h)
         self.ln_2 = FusedLayerNorm(h, eps=cfg.ln_eps)
-        self.mlp_fc = nn.Linear(h, 4 * h)
-        self.mlp_proj = nn.Linear(4 * h, h)
+        self.cfg = cfg
+        self.mlp_fc = nn.Linear(h, cfg.mlp_ratio * h)
+        self.mlp_proj = nn.Linear(cfg.mlp_ratio * h, h)
         self.num_heads = cfg.num_heads
         self.head_dim = cfg.head_dim
 
@@ -62,7 +77,10 @@ class GPT2Block(nn.Module):
         o = sdpa_gqa(q, k, v, causal=True)
         o = o.transpose(1, 2).reshape(B, S, H)
         x = x + self.attn_out(o)
-        x = x + self.mlp_proj(F.gelu(self.mlp_fc(self.ln_2(x)),
+        h = self.mlp_fc(self.ln_2(x))
+        if self.cfg.activation == "relu":
+            return x + self.mlp_proj(F.relu(h))
+        x = x + self.mlp_proj(F.gelu(h,
                                      approximate="tanh"))
         return x
 

@@ -34,6 +34,7 @@ class LlamaConfig:
     rms_eps: float = 1e-5
     tie_embeddings: bool = False
     initializer_range: float = 0.02
+    attention_bias: bool = False  # qwen2-style qkv bias
 
     @property
     def head_dim(self):
@@ -66,6 +67,21 @@ def phi3_mini():
                        tie_embeddings=False)
 
 
+def qwen2_7b():
+    """Qwen2-7B shapes: Llama architecture + qkv bias, 152k vocab
+    (reference inference/v2/model_implementations/qwen_v2)."""
+    return LlamaConfig(vocab_size=152064, hidden_size=3584,
+                       intermediate_size=18944, num_layers=28, num_heads=28,
+                       num_kv_heads=4, max_seq_len=32768,
+                       rope_theta=1000000.0, attention_bias=True)
+
+
+def qwen2_mini():
+    return LlamaConfig(vocab_size=512, hidden_size=64, intermediate_size=128,
+                       num_layers=2, num_heads=4, num_kv_heads=2,
+                       max_seq_len=128, attention_bias=True)
+
+
 def llama_mini():
     """~0.5B for single-GPU smoke runs."""
     return LlamaConfig(vocab_size=32000, hidden_size=1024,
@@ -81,9 +97,10 @@ class LlamaAttention(nn.Module):
         self.num_kv_heads = cfg.num_kv_heads
         self.head_dim = cfg.head_dim
         h = cfg.hidden_size
-        self.q_proj = nn.Linear(h, cfg.num_heads * self.head_dim, bias=False)
-        self.k_proj = nn.Linear(h, cfg.num_kv_heads * self.head_dim, bias=False)
-        self.v_proj = nn.Linear(h, cfg.num_kv_heads * self.head_dim, bias=False)
+        ab = cfg.attention_bias
+        self.q_proj = nn.Linear(h, cfg.num_heads * self.head_dim, bias=ab)
+        self.k_proj = nn.Linear(h, cfg.num_kv_heads * self.head_dim, bias=ab)
+        self.v_proj = nn.Linear(h, cfg.num_kv_heads * self.head_dim, bias=ab)
         self.o_proj = nn.Linear(cfg.num_heads * self.head_dim, h, bias=False)
 
     def forward(self, x, cos, sin, positions=None, kv_cache=None,
