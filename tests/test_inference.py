@@ -525,9 +525,13 @@ def test_model_zoo_qwen2_opt_train_and_generate():
 def _zoo_worker(rank=0, world=1):
     import deepspeed_amd
     from deepspeed_amd.models import (LlamaForCausalLM, qwen2_mini,
-                                      GPT2ForCausalLM, opt_mini)
+                                      GPT2ForCausalLM, opt_mini,
+                                      FalconForCausalLM, falcon_mini,
+                                      falcon_mini_gqa)
     for build in (lambda: LlamaForCausalLM(qwen2_mini()),
-                  lambda: GPT2ForCausalLM(opt_mini())):
+                  lambda: GPT2ForCausalLM(opt_mini()),
+                  lambda: FalconForCausalLM(falcon_mini()),
+                  lambda: FalconForCausalLM(falcon_mini_gqa())):
         torch.manual_seed(0)
         model = build()
         engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
