@@ -11,6 +11,18 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 # are copied locked). Keep the pytest parent single-threaded so tests may
 # freely run torch ops in-process AND fork afterwards.
 os.environ.setdefault("OMP_NUM_THREADS", "1")
+
+# Tests that initialize the engine IN the pytest process rendezvous on
+# MASTER_PORT (default 29500). Pick a random free port once per pytest
+# session so two concurrent pytest runs (or a stray daemon on 29500)
+# cannot cross-connect to each other's TCPStore and deadlock.
+if "MASTER_PORT" not in os.environ:
+    import socket
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        os.environ["MASTER_PORT"] = str(_s.getsockname()[1])
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+
 import torch  # noqa: E402
 
 torch.set_num_threads(1)

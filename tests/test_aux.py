@@ -649,6 +649,11 @@ def test_contiguous_allocator_defrag():
 def test_fp16_auto_cast_inputs():
     """fp16.auto_cast: fp32 float inputs are cast to the engine dtype at
     the engine boundary (reference fp16 auto_cast)."""
+    from .common import run_local
+    run_local(_auto_cast_worker)
+
+
+def _auto_cast_worker(rank=0, world=1):
     import deepspeed_amd
 
     seen = {}
@@ -678,6 +683,13 @@ def test_fp16_auto_cast_inputs():
 def test_save_16bit_model_zero3_gate(tmp_path):
     """stage3_gather_16bit_weights_on_model_save=false refuses the
     consolidated save instead of silently gathering."""
+    from .common import run_local
+    run_local(_save16_gate_worker, args=(str(tmp_path),))
+
+
+def _save16_gate_worker(rank, world, tmp_path):
+    from pathlib import Path
+    tmp_path = Path(tmp_path)
     import deepspeed_amd
 
     class M(torch.nn.Module):
