@@ -37,10 +37,24 @@ class CurriculumLearningConfig(ConfigModel):
     metric_path: Optional[str] = None
 
 
+class RandomLTDConfig(ConfigModel):
+    """Random layer-token-drop (reference data_pipeline/data_routing):
+    middle layers run on a sampled token subset that grows on a schedule."""
+    enabled: bool = False
+    layers_attr: str = "model.layers"   # path to the decoder ModuleList
+    skip_first: int = 1
+    skip_last: int = 1
+    min_value: int = 128                # starting kept-token count
+    max_value: int = 4096               # full sequence by schedule end
+    seq_per_step: int = 16
+    total_ltd_steps: int = 1000
+
+
 class DataEfficiencyConfig(ConfigModel):
     enabled: bool = False
     seed: int = 1234
     curriculum_learning: CurriculumLearningConfig = CurriculumLearningConfig()
+    random_ltd: RandomLTDConfig = RandomLTDConfig()
 
 
 class FP16Config(ConfigModel):

@@ -122,6 +122,31 @@ class Engine(torch.nn.Module):
         self.checkpoint_engine = create_checkpoint_engine(
             self.config.raw.get("checkpoint", {}).get("engine", "torch"))
 
+        # random-LTD (reference data_routing/basic_layer.py): wrap middle
+        # decoder layers BEFORE the optimizer partitions params (wrapping
+        # preserves param identity, but ZeRO-3 records module units)
+        self.random_ltd_scheduler = None
+        ltd = self.config.data_efficiency.random_ltd
+        if self.config.data_efficiency.enabled and ltd.enabled:
+            from .data_pipeline.random_ltd import (RandomLTDScheduler,
+                                                   convert_to_random_ltd)
+            obj = self.module
+            for part in ltd.layers_attr.split("."):
+                obj = getattr(obj, part)
+            self.random_ltd_scheduler = RandomLTDScheduler(
+                total_layers=len(obj),
+                random_ltd_layer_num=len(obj) - ltd.skip_first -
+                ltd.skip_last,
+                start_seq=ltd.min_value, max_seq=ltd.max_value,
+                step_size=ltd.seq_per_step,
+                schedule_steps=ltd.total_ltd_steps)
+            n = convert_to_random_ltd(self.module, ltd.layers_attr,
+                                      self.random_ltd_scheduler,
+                                      skip_first=ltd.skip_first,
+                                      skip_last=ltd.skip_last)
+            log_dist(f"random-LTD: wrapped {n} layers, kept tokens "
+                     f"{ltd.min_value}->{ltd.max_value}")
+
         self._configure_distributed_model(dont_change_device)
         self._configure_optimizer(model_parameters)
         self._configure_lr_scheduler()
@@ -482,6 +507,8 @@ class Engine(torch.nn.Module):
                 self.lr_scheduler.step()
 
         self.global_steps += 1
+        if self.random_ltd_scheduler is not None:
+            self.random_ltd_scheduler.update_seq(self.global_steps)
         if self.tput_timer:
             self.tput_timer.stop(global_step=True)
         if self.monitor is not None and self.global_rank == 0:
@@ -695,6 +722,9 @@ class Engine(torch.nn.Module):
                 "global_samples": self.global_samples,
                 "skipped_steps": self.skipped_steps,
                 "dp_world_size": self.dp_world_size,
+                "random_ltd": (self.random_ltd_scheduler.state_dict()
+                               if self.random_ltd_scheduler is not None
+                               else None),
                 "ds_config": self.config.raw,
                 "client_state": client_state or {},
             }
@@ -799,6 +829,9 @@ class Engine(torch.nn.Module):
         self.global_steps = state.get("global_steps", 0)
         self.global_samples = state.get("global_samples", 0)
         self.skipped_steps = state.get("skipped_steps", 0)
+        if self.random_ltd_scheduler is not None and \
+                state.get("random_ltd") is not None:
+            self.random_ltd_scheduler.load_state_dict(state["random_ltd"])
 
         if load_universal:
             # elastic load: per-param fp32 state sliced at THIS world size

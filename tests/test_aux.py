@@ -717,3 +717,43 @@ def _save16_gate_worker(rank, world, tmp_path):
     assert eng.save_16bit_model(str(tmp_path)) is True
     sd = torch.load(tmp_path / "pytorch_model.bin", weights_only=True)
     assert "fc.weight" in sd and sd["fc.weight"].shape == (1, 8)
+
+
+def test_random_ltd_from_config():
+    """data_efficiency.random_ltd wraps middle decoder layers and the
+    kept-token count follows the schedule across engine steps."""
+    from .common import run_local
+    run_local(_random_ltd_worker)
+
+
+def _random_ltd_worker(rank=0, world=1):
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    from deepspeed_amd.runtime.data_pipeline.random_ltd import \
+        RandomLayerTokenDrop
+
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "bf16": {"enabled": True},
+        "zero_optimization": {"stage": 2, "overlap_comm": False},
+        "data_efficiency": {
+            "enabled": True,
+            "random_ltd": {"enabled": True, "layers_attr": "model.layers",
+                           "skip_first": 1, "skip_last": 0,
+                           "min_value": 8, "max_value": 32,
+                           "seq_per_step": 8, "total_ltd_steps": 4}},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    wrapped = [m for m in engine.module.modules()
+               if isinstance(m, RandomLayerTokenDrop)]
+    assert len(wrapped) == 1  # 2 layers, skip_first=1
+    seqs = []
+    for _ in range(5):
+        ids = torch.randint(0, 500, (2, 32))
+        loss = engine(ids.to(engine.device), labels=ids.to(engine.device))
+        engine.backward(loss)
+        engine.step()
+        seqs.append(engine.random_ltd_scheduler.current_seq)
+        assert torch.isfinite(loss)
+    assert seqs[0] < seqs[-1] and seqs[-1] == 32, seqs
