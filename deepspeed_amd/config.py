@@ -50,6 +50,14 @@ class RandomLTDConfig(ConfigModel):
     total_ltd_steps: int = 1000
 
 
+class ProgressiveLayerDropConfig(ConfigModel):
+    """theta(t) keep-probability schedule (reference runtime/
+    progressive_layer_drop.py; config key progressive_layer_drop)."""
+    enabled: bool = False
+    theta: float = 0.5
+    gamma: float = 0.001
+
+
 class DataEfficiencyConfig(ConfigModel):
     enabled: bool = False
     seed: int = 1234
@@ -251,6 +259,8 @@ class Config:
         self.aio = AIOConfig(**g("aio", {}))
         self.data_efficiency = DataEfficiencyConfig(
             **g("data_efficiency", {}))
+        self.progressive_layer_drop = ProgressiveLayerDropConfig(
+            **g("progressive_layer_drop", {}))
 
         self.data_types_grad_accum_dtype = (
             g("data_types", {}).get("grad_accum_dtype", None))

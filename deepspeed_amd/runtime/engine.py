@@ -147,6 +147,13 @@ class Engine(torch.nn.Module):
             log_dist(f"random-LTD: wrapped {n} layers, kept tokens "
                      f"{ltd.min_value}->{ltd.max_value}")
 
+        self.progressive_layer_drop = None
+        if self.config.progressive_layer_drop.enabled:
+            from .progressive_layer_drop import ProgressiveLayerDrop
+            self.progressive_layer_drop = ProgressiveLayerDrop(
+                theta=self.config.progressive_layer_drop.theta,
+                gamma=self.config.progressive_layer_drop.gamma)
+
         self._configure_distributed_model(dont_change_device)
         self._configure_optimizer(model_parameters)
         self._configure_lr_scheduler()
@@ -509,6 +516,8 @@ class Engine(torch.nn.Module):
         self.global_steps += 1
         if self.random_ltd_scheduler is not None:
             self.random_ltd_scheduler.update_seq(self.global_steps)
+        if self.progressive_layer_drop is not None:
+            self.progressive_layer_drop.update_state(self.global_steps)
         if self.tput_timer:
             self.tput_timer.stop(global_step=True)
         if self.monitor is not None and self.global_rank == 0:

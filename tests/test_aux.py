@@ -757,3 +757,37 @@ def _random_ltd_worker(rank=0, world=1):
         seqs.append(engine.random_ltd_scheduler.current_seq)
         assert torch.isfinite(loss)
     assert seqs[0] < seqs[-1] and seqs[-1] == 32, seqs
+
+
+def test_progressive_layer_drop_from_config():
+    from .common import run_local
+    run_local(_pld_worker)
+
+
+def _pld_worker(rank=0, world=1):
+    import deepspeed_amd
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            return torch.nn.functional.mse_loss(self.fc(x).float(),
+                                                labels.float())
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=M(), config={
+        "train_micro_batch_size_per_gpu": 2,
+        "progressive_layer_drop": {"enabled": True, "theta": 0.5,
+                                   "gamma": 0.1},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    thetas = [eng.progressive_layer_drop.get_theta()]
+    for _ in range(3):
+        loss = eng(torch.randn(2, 8).to(eng.device),
+                   labels=torch.randn(2, 1).to(eng.device))
+        eng.backward(loss)
+        eng.step()
+        thetas.append(eng.progressive_layer_drop.get_theta())
+    # theta decays from 1.0 toward theta=0.5
+    assert thetas[0] == 1.0 and all(a > b for a, b in zip(thetas, thetas[1:]))
+    assert thetas[-1] > 0.5
