@@ -55,11 +55,12 @@ def falcon_mini_gqa():
 
 
 class FalconBlock(nn.Module):
-    def __init__(self, cfg: FalconConfig):
+    def __init__(self, cfg: FalconConfig, layer_idx: int = 0):
         super().__init__()
         h = cfg.hidden_size
         d = cfg.head_dim
         self.cfg = cfg
+        self.layer_idx = layer_idx
         self.ln_attn = FusedLayerNorm(h, eps=cfg.ln_eps)
         self.ln_mlp = FusedLayerNorm(h, eps=cfg.ln_eps) \
             if cfg.parallel_attn_norms else None
@@ -69,7 +70,7 @@ class FalconBlock(nn.Module):
         self.mlp_fc = nn.Linear(h, 4 * h, bias=False)
         self.mlp_proj = nn.Linear(4 * h, h, bias=False)
 
-    def forward(self, x, cos, sin, positions=None):
+    def forward(self, x, cos, sin, positions=None, kv_cache=None):
         cfg = self.cfg
         B, S, _ = x.shape
         d = cfg.head_dim
@@ -84,8 +85,14 @@ class FalconBlock(nn.Module):
         v = qkv[..., nq + nk:].view(B, S, cfg.num_kv_heads, d)
         q = apply_rope(q, cos, sin, positions)
         k = apply_rope(k, cos, sin, positions)
-        o = sdpa_gqa(q.transpose(1, 2), k.transpose(1, 2),
-                     v.transpose(1, 2), causal=True)
+        q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+        attn_mask = None
+        if kv_cache is not None:
+            k, v = kv_cache.update(self.layer_idx, k, v)
+            attn_mask = getattr(kv_cache, "last_mask", None)
+        o = sdpa_gqa(q, k, v,
+                     causal=attn_mask is None and q.size(2) == k.size(2),
+                     attn_mask=attn_mask)
         attn_out = self.dense(o.transpose(1, 2).reshape(B, S, -1))
 
         mlp_out = self.mlp_proj(F.gelu(self.mlp_fc(m_in)))
@@ -99,19 +106,19 @@ class FalconModel(nn.Module):
         self.cfg = cfg
         self.wte = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
         self.blocks = nn.ModuleList(
-            [FalconBlock(cfg) for _ in range(cfg.num_layers)])
+            [FalconBlock(cfg, i) for i in range(cfg.num_layers)])
         self.ln_f = FusedLayerNorm(cfg.hidden_size, eps=cfg.ln_eps)
         cos, sin = rope_tables(cfg.head_dim, cfg.max_seq_len,
                                theta=cfg.rope_theta)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
-    def forward(self, input_ids, positions=None):
+    def forward(self, input_ids, positions=None, kv_cache=None):
         x = self.wte(input_ids)
         cos = self.rope_cos.to(x.device)
         sin = self.rope_sin.to(x.device)
         for b in self.blocks:
-            x = b(x, cos, sin, positions)
+            x = b(x, cos, sin, positions, kv_cache=kv_cache)
         return self.ln_f(x)
 
 
@@ -129,8 +136,9 @@ class FalconForCausalLM(nn.Module):
             if getattr(module, "bias", None) is not None:
                 module.bias.data.zero_()
 
-    def forward(self, input_ids, labels=None, positions=None):
-        hidden = self.transformer(input_ids, positions)
+    def forward(self, input_ids, labels=None, positions=None,
+                kv_cache=None):
+        hidden = self.transformer(input_ids, positions, kv_cache=kv_cache)
         if labels is None:
             return self.lm_head(hidden)
         from .llama import chunked_cross_entropy

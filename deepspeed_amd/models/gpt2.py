@@ -54,9 +54,10 @@ def gpt2_tiny():
 
 
 class GPT2Block(nn.Module):
-    def __init__(self, cfg: GPT2Config):
+    def __init__(self, cfg: GPT2Config, layer_idx: int = 0):
         super().__init__()
         h = cfg.hidden_size
+        self.layer_idx = layer_idx
         self.ln_1 = FusedLayerNorm(h, eps=cfg.ln_eps)
         self.attn = nn.Linear(h, 3 * h)
         self.attn_out = nn.Linear(h, h)
@@ -67,14 +68,20 @@ class GPT2Block(nn.Module):
         self.num_heads = cfg.num_heads
         self.head_dim = cfg.head_dim
 
-    def forward(self, x):
+    def forward(self, x, kv_cache=None):
         B, S, H = x.shape
         qkv = self.attn(self.ln_1(x))
         q, k, v = qkv.split(H, dim=-1)
         q = q.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
         k = k.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
         v = v.view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
-        o = sdpa_gqa(q, k, v, causal=True)
+        attn_mask = None
+        if kv_cache is not None:
+            k, v = kv_cache.update(self.layer_idx, k, v)
+            attn_mask = getattr(kv_cache, "last_mask", None)
+        o = sdpa_gqa(q, k, v,
+                     causal=attn_mask is None and q.size(2) == k.size(2),
+                     attn_mask=attn_mask)
         o = o.transpose(1, 2).reshape(B, S, H)
         x = x + self.attn_out(o)
         h = self.mlp_fc(self.ln_2(x))
@@ -91,7 +98,8 @@ class GPT2ForCausalLM(nn.Module):
         self.cfg = cfg
         self.wte = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
         self.wpe = nn.Embedding(cfg.max_seq_len, cfg.hidden_size)
-        self.blocks = nn.ModuleList([GPT2Block(cfg) for _ in range(cfg.num_layers)])
+        self.blocks = nn.ModuleList([GPT2Block(cfg, i)
+                                     for i in range(cfg.num_layers)])
         self.ln_f = FusedLayerNorm(cfg.hidden_size, eps=cfg.ln_eps)
         self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
         self.lm_head.weight = self.wte.weight  # tied
@@ -104,12 +112,15 @@ class GPT2ForCausalLM(nn.Module):
             if isinstance(module, nn.Linear) and module.bias is not None:
                 module.bias.data.zero_()
 
-    def forward(self, input_ids, labels=None):
+    def forward(self, input_ids, labels=None, positions=None, kv_cache=None):
         B, S = input_ids.shape
-        pos = torch.arange(S, device=input_ids.device)
-        x = self.wte(input_ids) + self.wpe(pos)[None]
+        if positions is None:
+            pos_emb = self.wpe(torch.arange(S, device=input_ids.device))[None]
+        else:
+            pos_emb = self.wpe(positions.long())
+        x = self.wte(input_ids) + pos_emb
         for blk in self.blocks:
-            x = blk(x)
+            x = blk(x, kv_cache=kv_cache)
         x = self.ln_f(x)
         logits = self.lm_head(x)
         if labels is not None:

@@ -545,13 +545,11 @@ def _zoo_worker(rank=0, world=1):
         engine.step()
         assert torch.isfinite(loss)
         inf = deepspeed_amd.init_inference(build(), dtype="fp32")
-        if hasattr(inf.module, "model"):   # llama-family: KV-cache generate
-            out = inf.generate(ids[:1, :4].to(inf.device), max_new_tokens=4)
-            assert out.shape[1] == 8
-        else:                              # gpt2-family: plain greedy loop
-            cur = ids[:1, :4].to(inf.device)
-            with torch.no_grad():
-                for _ in range(4):
-                    nxt = inf.module(cur)[:, -1:].argmax(-1)
-                    cur = torch.cat([cur, nxt], dim=1)
-            assert cur.shape[1] == 8
+        out = inf.generate(ids[:1, :4].to(inf.device), max_new_tokens=4)
+        # KV-cached generate must match the full-context greedy loop
+        cur = ids[:1, :4].to(inf.device)
+        with torch.no_grad():
+            for _ in range(4):
+                nxt = inf.module(cur)[:, -1:].argmax(-1)
+                cur = torch.cat([cur, nxt], dim=1)
+        assert torch.equal(out, cur)
