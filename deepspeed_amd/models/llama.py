@@ -67,6 +67,15 @@ def phi3_mini():
                        tie_embeddings=False)
 
 
+def mistral_7b():
+    """Mistral-7B shapes (llama architecture; the 4k-window sliding
+    attention is omitted — full causal attention is a superset and the
+    paged KV cache caps context by max_seq_len)."""
+    return LlamaConfig(vocab_size=32000, hidden_size=4096,
+                       intermediate_size=14336, num_layers=32, num_heads=32,
+                       num_kv_heads=8, max_seq_len=8192, rope_theta=10000.0)
+
+
 def qwen2_7b():
     """Qwen2-7B shapes: Llama architecture + qkv bias, 152k vocab
     (reference inference/v2/model_implementations/qwen_v2)."""
