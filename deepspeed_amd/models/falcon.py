@@ -9,7 +9,6 @@ shot (one residual add per block instead of two). New-decoder variants
 ``parallel_attn_norms`` selects that. Reuses this framework's fused
 LayerNorm, RoPE tables, and GQA sdpa routing."""
 
-import math
 from dataclasses import dataclass
 
 import torch
