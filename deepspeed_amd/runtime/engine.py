@@ -335,6 +335,48 @@ class Engine(torch.nn.Module):
             return self.optimizer.get_global_grad_norm()
         return getattr(self, "_fallback_grad_norm", 0.0)
 
+    def get_mom(self):
+        """Momentum / betas of the first param group (reference
+        engine.get_mom:2474)."""
+        if self.optimizer is None:
+            return []
+        key = "momentum" if "momentum" in self.optimizer.param_groups[0] \
+            else "betas"
+        return [g.get(key) for g in self.optimizer.param_groups]
+
+    def get_pld_theta(self):
+        return (self.progressive_layer_drop.get_theta()
+                if self.progressive_layer_drop is not None else None)
+
+    def set_train_batch_size(self, train_batch_size: int):
+        """Change the global batch size by adjusting gradient accumulation
+        (micro-batch size and DP world are fixed; reference
+        engine.set_train_batch_size:524). Used by the autotuner and
+        curriculum schedules."""
+        denom = self.train_micro_batch_size_per_gpu * self.dp_world_size
+        if train_batch_size % denom != 0:
+            raise ValueError(
+                f"train_batch_size {train_batch_size} not divisible by "
+                f"micro_batch x dp = {denom}")
+        self.gradient_accumulation_steps = train_batch_size // denom
+        self.train_batch_size = train_batch_size
+        self.config.gradient_accumulation_steps = \
+            self.gradient_accumulation_steps
+        self.config.train_batch_size = train_batch_size
+        if self.tput_timer:
+            self.tput_timer.batch_size = train_batch_size
+
+    def set_train_micro_batch_size(self, micro_batch_size: int):
+        """Change the per-GPU micro batch; gradient accumulation steps stay
+        fixed, so the global batch scales with it."""
+        self.train_micro_batch_size_per_gpu = micro_batch_size
+        self.train_batch_size = micro_batch_size * \
+            self.gradient_accumulation_steps * self.dp_world_size
+        self.config.train_micro_batch_size_per_gpu = micro_batch_size
+        self.config.train_batch_size = self.train_batch_size
+        if self.tput_timer:
+            self.tput_timer.batch_size = self.train_batch_size
+
     @property
     def loss_scale(self):
         if hasattr(self.optimizer, "loss_scale"):

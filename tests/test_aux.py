@@ -791,3 +791,41 @@ def _pld_worker(rank=0, world=1):
     # theta decays from 1.0 toward theta=0.5
     assert thetas[0] == 1.0 and all(a > b for a, b in zip(thetas, thetas[1:]))
     assert thetas[-1] > 0.5
+
+
+def test_engine_dynamic_batch_api():
+    from .common import run_local
+    run_local(_dyn_batch_worker)
+
+
+def _dyn_batch_worker(rank=0, world=1):
+    import deepspeed_amd
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            return torch.nn.functional.mse_loss(self.fc(x).float(),
+                                                labels.float())
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=M(), config={
+        "train_micro_batch_size_per_gpu": 2,
+        "gradient_accumulation_steps": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    assert eng.train_batch_size == 4
+    eng.set_train_batch_size(8)  # gas 2 -> 4
+    assert eng.gradient_accumulation_steps == 4
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        eng.set_train_batch_size(7)
+    # train one full accumulation window under the new gas
+    for i in range(4):
+        loss = eng(torch.randn(2, 8).to(eng.device),
+                   labels=torch.randn(2, 1).to(eng.device))
+        eng.backward(loss)
+        eng.step()
+    assert eng.global_steps == 1
+    assert eng.get_mom()[0] == (0.9, 0.999)
+    assert eng.get_pld_theta() is None
