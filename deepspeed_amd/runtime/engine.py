@@ -335,6 +335,185 @@ class Engine(torch.nn.Module):
             return self.optimizer.get_global_grad_norm()
         return getattr(self, "_fallback_grad_norm", 0.0)
 
+    # ---- reference-parity config accessors (engine.py:588-1010). The
+    # reference exposes these as METHODS; frameworks built on DeepSpeed
+    # (Megatron-DeepSpeed, trainer integrations) call them, so they are
+    # methods here too. train_batch_size / gradient_accumulation_steps /
+    # train_micro_batch_size_per_gpu remain plain attributes (documented
+    # delta in docs/migrating_from_deepspeed.md).
+
+    def fp16_enabled(self):
+        return self.config.fp16.enabled
+
+    def bfloat16_enabled(self):
+        return self.config.bf16.enabled
+
+    def fp16_auto_cast(self):
+        return self.config.fp16.auto_cast
+
+    def amp_enabled(self):
+        return False  # torch.amp path not used; bf16/fp16 policies instead
+
+    def dynamic_loss_scale(self):
+        return self.config.fp16.loss_scale == 0
+
+    def initial_dynamic_scale(self):
+        return 2.0 ** self.config.fp16.initial_scale_power
+
+    def zero_optimization_stage(self):
+        return self.zero_stage
+
+    def zero_optimization_partition_gradients(self):
+        return self.zero_stage >= 2
+
+    def zero_optimization_partition_weights(self):
+        return self.zero_stage >= 3
+
+    def zero_overlap_comm(self):
+        return self.config.zero.overlap_comm
+
+    def zero_reduce_bucket_size(self):
+        return self.config.zero.reduce_bucket_size
+
+    def zero_allgather_bucket_size(self):
+        return self.config.zero.allgather_bucket_size
+
+    def zero_offload_optimizer(self):
+        return self.config.zero.offload_optimizer
+
+    def zero_offload_param(self):
+        return self.config.zero.offload_param
+
+    def zero_cpu_offload(self):
+        return self.config.zero.offload_optimizer.device in ("cpu", "nvme")
+
+    def zero_prefetch_bucket_size(self):
+        return self.config.zero.stage3_prefetch_bucket_size
+
+    def zero_param_persistence_threshold(self):
+        return self.config.zero.stage3_param_persistence_threshold
+
+    def zero_max_live_parameters(self):
+        return self.config.zero.stage3_max_live_parameters
+
+    def zero_max_reuse_distance(self):
+        return self.config.zero.stage3_max_reuse_distance
+
+    def zero_gather_16bit_weights_on_model_save(self):
+        return self.config.zero.stage3_gather_16bit_weights_on_model_save
+
+    def zero_hpz_partition_size(self):
+        return self.config.zero.zero_hpz_partition_size
+
+    def zero_quantized_weights(self):
+        return self.config.zero.zero_quantized_weights
+
+    def zero_quantized_gradients(self):
+        return self.config.zero.zero_quantized_gradients
+
+    def mics_shard_size(self):
+        return self.config.zero.mics_shard_size
+
+    def gradient_clipping(self):
+        return self.config.gradient_clipping
+
+    def gradient_predivide_factor(self):
+        return self.config.gradient_predivide_factor
+
+    def steps_per_print(self):
+        return self.config.steps_per_print
+
+    def memory_breakdown(self):
+        return self.config.memory_breakdown
+
+    def sparse_gradients_enabled(self):
+        return self.config.sparse_gradients_enabled
+
+    def optimizer_name(self):
+        if self.client_optimizer is not None:
+            return type(self.client_optimizer).__name__
+        return self.config.optimizer.type
+
+    def optimizer_params(self):
+        return self.config.optimizer.params
+
+    def scheduler_name(self):
+        return self.config.scheduler.type if self.config.scheduler else None
+
+    def scheduler_params(self):
+        return self.config.scheduler.params if self.config.scheduler else None
+
+    def flops_profiler_enabled(self):
+        return self.config.flops_profiler.enabled
+
+    def curriculum_learning_enabled(self):
+        return self.config.data_efficiency.curriculum_learning.enabled
+
+    def data_efficiency_enabled(self):
+        return self.config.data_efficiency.enabled
+
+    def random_ltd_enabled(self):
+        return self.random_ltd_scheduler is not None
+
+    def pld_enabled(self):
+        return self.progressive_layer_drop is not None
+
+    def pld_theta(self):
+        return self.get_pld_theta()
+
+    def aio_config(self):
+        return self.config.aio
+
+    def communication_data_type(self):
+        return self._comm_dtype() or self.dtype
+
+    def get_batch_info(self):
+        return (self.train_batch_size, self.train_micro_batch_size_per_gpu,
+                self.gradient_accumulation_steps)
+
+    def get_data_types(self):
+        return (self.dtype, torch.float32)
+
+    def was_step_applied(self) -> bool:
+        """False when the last step was skipped (fp16 overflow)."""
+        return not bool(getattr(self.optimizer, "overflow", False))
+
+    def zero_grad(self):
+        for p in self.module.parameters():
+            if p.grad is not None:
+                p.grad.detach_()
+                p.grad.zero_()
+
+    def empty_partition_cache(self):
+        """Release every gathered ZeRO-3 full buffer (reference
+        engine.empty_partition_cache) and return cached HBM to the pool."""
+        opt = self.optimizer
+        if hasattr(opt, "units") and hasattr(opt, "_release"):
+            for u in opt.units:
+                if not u.persist:
+                    opt._release(u)
+        if accel.available():
+            torch.cuda.empty_cache()
+
+    def save_fp16_model(self, save_dir, save_filename="pytorch_model.bin"):
+        """Reference-compat alias of save_16bit_model."""
+        return self.save_16bit_model(save_dir, save_filename)
+
+    def load_module_state_dict(self, state_dict, strict=True):
+        self.module.load_state_dict(state_dict, strict=strict)
+
+    def destroy(self):
+        """Detach grad hooks and break engine<->optimizer cycles so the
+        model can be reused outside the engine."""
+        opt = self.optimizer
+        for attr in ("_grad_hooks", "_hooks"):
+            for h in getattr(opt, attr, []) or []:
+                try:
+                    h.remove()
+                except Exception:
+                    pass
+        self.optimizer = None
+
     def get_mom(self):
         """Momentum / betas of the first param group (reference
         engine.get_mom:2474)."""

@@ -829,3 +829,46 @@ def _dyn_batch_worker(rank=0, world=1):
     assert eng.global_steps == 1
     assert eng.get_mom()[0] == (0.9, 0.999)
     assert eng.get_pld_theta() is None
+
+
+def test_engine_config_accessors():
+    from .common import run_local
+    run_local(_accessor_worker)
+
+
+def _accessor_worker(rank=0, world=1):
+    import deepspeed_amd
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            return torch.nn.functional.mse_loss(self.fc(x).float(),
+                                                labels.float())
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=M(), config={
+        "train_micro_batch_size_per_gpu": 2,
+        "bf16": {"enabled": True},
+        "gradient_clipping": 0.7,
+        "zero_optimization": {"stage": 2, "overlap_comm": False,
+                              "reduce_bucket_size": 1234},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    assert eng.bfloat16_enabled() and not eng.fp16_enabled()
+    assert eng.zero_optimization_stage() == 2
+    assert eng.zero_optimization_partition_gradients()
+    assert not eng.zero_optimization_partition_weights()
+    assert eng.zero_reduce_bucket_size() == 1234
+    assert eng.gradient_clipping() == 0.7
+    assert eng.optimizer_name() == "AdamW"
+    assert eng.get_batch_info() == (2, 2, 1)
+    assert eng.was_step_applied()
+    loss = eng(torch.randn(2, 8).to(eng.device).bfloat16(),
+               labels=torch.randn(2, 1).to(eng.device))
+    eng.backward(loss)
+    eng.step()
+    eng.empty_partition_cache()
+    eng.zero_grad()
+    eng.destroy()
+    assert eng.optimizer is None
