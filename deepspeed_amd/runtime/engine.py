@@ -479,10 +479,12 @@ class Engine(torch.nn.Module):
         return not bool(getattr(self.optimizer, "overflow", False))
 
     def zero_grad(self):
-        for p in self.module.parameters():
-            if p.grad is not None:
-                p.grad.detach_()
-                p.grad.zero_()
+        with torch.no_grad():
+            # grads are VIEWS into the ZeRO flat buffers; zero in place
+            # (detach_/None would break the view scheme)
+            for p in self.module.parameters():
+                if p.grad is not None:
+                    p.grad.zero_()
 
     def empty_partition_cache(self):
         """Release every gathered ZeRO-3 full buffer (reference
