@@ -326,6 +326,17 @@ class CommsLogger:
         return "\n".join(lines)
 
 
+def initialize_mesh_device(mesh_shape, mesh_dim_names=("data_parallel",
+                                                        "sequence_parallel")):
+    """torch DeviceMesh over the initialized world (reference
+    comm/comm.py:609 initialize_mesh_device) — DP x SP style grids for
+    mesh-parameterized models."""
+    from torch.distributed.device_mesh import init_device_mesh
+    device_type = "cuda" if accel.available() else "cpu"
+    return init_device_mesh(device_type, tuple(mesh_shape),
+                            mesh_dim_names=tuple(mesh_dim_names))
+
+
 def configure_comms_logger(enabled=False, verbose=False, prof_all=True,
                            prof_ops=(), debug=False) -> CommsLogger:
     global _comms_logger

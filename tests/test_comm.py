@@ -199,3 +199,22 @@ def test_mpi_discovery_requires_mpi_env(monkeypatch):
     import pytest as _pytest
     with _pytest.raises(RuntimeError):
         dcomm.mpi_discovery()
+
+
+def test_initialize_mesh_device_ws4():
+    from .common import run_distributed
+    run_distributed(_mesh_worker, world_size=4)
+
+
+def _mesh_worker(rank, world):
+    import deepspeed_amd.comm as dcomm
+    mesh = dcomm.initialize_mesh_device((2, 2))
+    dp = mesh.get_group("data_parallel")
+    sp = mesh.get_group("sequence_parallel")
+    import torch.distributed as td
+    assert td.get_world_size(dp) == 2 and td.get_world_size(sp) == 2
+    # the two axes partition the 4 ranks consistently
+    t = torch.tensor([float(rank)])
+    td.all_reduce(t, group=sp)
+    expected = {0: 1.0, 1: 1.0, 2: 5.0, 3: 5.0}[rank]
+    assert t.item() == expected, (rank, t.item())
