@@ -73,13 +73,48 @@ class InferenceEngine(nn.Module):
 
         if isinstance(config.dtype, str):
             config.dtype = {"fp32": torch.float32, "fp16": torch.float16,
-                            "bf16": torch.bfloat16}[config.dtype]
-        if config.dtype != torch.float32:
+                            "bf16": torch.bfloat16,
+                            "int8": torch.int8}[config.dtype]
+        self._weight_quantized = False
+        if config.dtype == torch.int8:
+            # weight-only int8 (reference init_inference(dtype=torch.int8)
+            # -> GroupQuantizer): int8 weights + group scales resident,
+            # bf16 activations, dequant on the fly per linear
+            self.module.to(torch.bfloat16)
+            config.dtype = torch.bfloat16
+            n = self._quantize_linear_weights()
+            self._weight_quantized = True
+            log_dist(f"init_inference: int8 weight-only quantized "
+                     f"{n} linears")
+        elif config.dtype != torch.float32:
             self.module.to(config.dtype)
         if accel.available():
             self.module.to(accel.current_device())
         self.device = next(self.module.parameters()).device
         self.module.eval()
+
+    def _quantize_linear_weights(self, group_size: int = 2048) -> int:
+        from ..linear.optimized_linear import (QuantizationConfig,
+                                               QuantizedParameter)
+        qcfg = QuantizationConfig(group_size=group_size)
+
+        class WOQLinear(nn.Module):
+            def __init__(self, lin: nn.Linear):
+                super().__init__()
+                self.qweight = QuantizedParameter(lin.weight.data, qcfg)
+                self.bias = lin.bias
+
+            def forward(self, x):
+                w = self.qweight.dequantized().to(x.dtype)
+                return torch.nn.functional.linear(x, w, self.bias)
+
+        n = 0
+        for parent in list(self.module.modules()):
+            for name, child in list(parent.named_children()):
+                if type(child) is nn.Linear:
+                    setattr(parent, name, WOQLinear(child))
+                    n += 1
+        return n
 
     def _load_checkpoint(self, ckpt):
         """Load weights named by ``config.checkpoint`` (reference

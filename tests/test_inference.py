@@ -553,3 +553,31 @@ def _zoo_worker(rank=0, world=1):
                 nxt = inf.module(cur)[:, -1:].argmax(-1)
                 cur = torch.cat([cur, nxt], dim=1)
         assert torch.equal(out, cur)
+
+
+def test_init_inference_int8_weight_only():
+    """dtype=int8: weight-only groupwise quantization (int8 weights +
+    scales resident, bf16 activations, dequant per linear) — generation
+    stays close to the bf16 engine (reference init_inference int8 path)."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+
+    torch.manual_seed(0)
+    m16 = LlamaForCausalLM(llama_tiny())
+    torch.manual_seed(0)
+    m8 = LlamaForCausalLM(llama_tiny())
+    ids = torch.randint(0, 500, (1, 12))
+    inf16 = deepspeed_amd.init_inference(m16, dtype="bf16")
+    inf8 = deepspeed_amd.init_inference(m8, dtype="int8")
+    assert inf8._weight_quantized
+    with torch.no_grad():
+        l16 = inf16.module(ids.to(inf16.device)).float()
+        l8 = inf8.module(ids.to(inf8.device)).float()
+    # int8 grouped quantization error is small relative to logit scale
+    rel = (l16 - l8).abs().max() / l16.abs().max()
+    assert rel < 0.05, float(rel)
+    # linears hold int8 buffers, not bf16 weights
+    from deepspeed_amd.linear.optimized_linear import QuantizedParameter
+    qps = [mod for mod in inf8.module.modules()
+           if isinstance(mod, QuantizedParameter)]
+    assert qps and all(qp.q.dtype == torch.int8 for qp in qps)
