@@ -516,6 +516,23 @@ class Engine(torch.nn.Module):
                     pass
         self.optimizer = None
 
+    def set_custom_curriculum_learning_schedule(self, fn):
+        """Replace the pacing function of the curriculum scheduler
+        (curriculum_type "custom"): fn(global_steps) -> difficulty
+        (reference engine.set_custom_curriculum_learning_schedule)."""
+        sched = getattr(self, "curriculum_scheduler", None)
+        assert sched is not None, \
+            "build the dataloader via engine.deepspeed_io with " \
+            "curriculum_learning enabled first"
+        sched.set_custom_get_difficulty(fn)
+
+    def set_data_post_process_func(self, fn):
+        """Post-process hook fn(batch, difficulty) applied to every batch
+        the curriculum dataloader yields (reference
+        engine.set_data_post_process_func — e.g. truncate samples to the
+        current difficulty). Call BEFORE engine.deepspeed_io."""
+        self._data_post_process_func = fn
+
     def get_mom(self):
         """Momentum / betas of the first param group (reference
         engine.get_mom:2474)."""
@@ -905,6 +922,14 @@ class Engine(torch.nn.Module):
                 dp_size=self.dp_world_size, seed=de.seed)
             self.curriculum_sampler = sampler
             from torch.utils.data import DataLoader
+            post = getattr(self, "_data_post_process_func", None)
+            if post is not None:
+                from torch.utils.data._utils.collate import default_collate
+                base = collate_fn or default_collate
+
+                def collate_fn(items, _base=base, _sched=sched):
+                    return post(_base(items),
+                                _sched.get_current_difficulty())
             return DataLoader(dataset, batch_sampler=sampler,
                               num_workers=num_workers,
                               collate_fn=collate_fn)

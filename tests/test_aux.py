@@ -872,3 +872,63 @@ def _accessor_worker(rank=0, world=1):
     eng.zero_grad()
     eng.destroy()
     assert eng.optimizer is None
+
+
+def test_curriculum_custom_schedule_and_post_process():
+    from .common import run_local
+    run_local(_curriculum_custom_worker)
+
+
+def _curriculum_custom_worker(rank=0, world=1):
+    import deepspeed_amd
+
+    class DS(torch.utils.data.Dataset):
+        def __len__(self):
+            return 64
+
+        def __getitem__(self, i):
+            return torch.arange(4 + (i % 13)), i
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(4, 1)
+
+        def forward(self, x, labels=None):
+            return self.fc(x.float()[:, :4]).sum()
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=M(), config={
+        "train_micro_batch_size_per_gpu": 4,
+        "data_efficiency": {
+            "enabled": True,
+            "curriculum_learning": {"enabled": True,
+                                    "curriculum_type": "custom",
+                                    "min_difficulty": 4,
+                                    "max_difficulty": 16}},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+
+    seen = []
+
+    def post(batch, difficulty):
+        xs, ys = batch
+        seen.append(difficulty)
+        return xs[:, :difficulty], ys
+
+    eng.set_data_post_process_func(post)
+    # pad-collate variable-length rows
+    def pad_collate(items):
+        xs = torch.nn.utils.rnn.pad_sequence([x for x, _ in items],
+                                             batch_first=True)
+        ys = torch.tensor([y for _, y in items])
+        return xs, ys
+
+    loader = eng.deepspeed_io(DS(), collate_fn=pad_collate)
+    eng.set_custom_curriculum_learning_schedule(
+        lambda step: min(16, 4 + 2 * step))
+    # the sampler advances the schedule itself, one step per batch
+    it = iter(loader)
+    xs, _ = next(it)
+    assert xs.shape[1] <= 4
+    xs, _ = next(it)
+    assert xs.shape[1] <= 6
+    assert seen[0] == 4 and seen[1] == 6, seen
