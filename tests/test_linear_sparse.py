@@ -192,3 +192,36 @@ def test_longformer_and_variable_layouts():
         want = F.scaled_dot_product_attention(q, q, q,
                                               attn_mask=mask.unsqueeze(0))
         torch.testing.assert_close(got, want, atol=2e-5, rtol=1e-4)
+
+
+def test_sparse_self_attention_module():
+    """SparseSelfAttention module (reference sparse_self_attention.py):
+    layout-cached forward equals dense SDPA under the block mask; key
+    padding masks fold in exactly; pad/unpad helpers round-trip."""
+    import math
+    from deepspeed_amd.ops.sparse_attention import (
+        SparseSelfAttention, FixedSparsityConfig, SparseAttentionUtils,
+        layout_to_dense_mask)
+    torch.manual_seed(0)
+    B, H, S, D = 2, 4, 64, 16
+    cfg = FixedSparsityConfig(num_heads=H, block=16)
+    attn = SparseSelfAttention(cfg)
+    q, k, v = (torch.randn(B, H, S, D) for _ in range(3))
+    out = attn(q, k, v)
+    mask = layout_to_dense_mask(cfg.make_layout(S), 16)
+    scores = (q.float() @ k.float().transpose(-1, -2)) / math.sqrt(D)
+    scores = scores.masked_fill(~mask[None], float("-inf"))
+    ref = (torch.softmax(scores, -1) @ v.float()).to(q.dtype)
+    torch.testing.assert_close(out, ref, rtol=2e-3, atol=2e-3)
+    kpm = torch.ones(B, S)
+    kpm[:, 50:] = 0
+    out2 = attn(q, k, v, key_padding_mask=kpm)
+    scores2 = scores.masked_fill(~kpm.bool()[:, None, None, :],
+                                 float("-inf"))
+    ref2 = torch.nan_to_num(torch.softmax(scores2, -1)) @ v.float()
+    torch.testing.assert_close(out2.float(), ref2, rtol=2e-3, atol=2e-3)
+    pad, ids, am = SparseAttentionUtils.pad_to_block_size(
+        16, torch.ones(2, 60, dtype=torch.long), torch.ones(2, 60))
+    assert pad == 4 and ids.shape[1] == 64 and am.shape[1] == 64
+    assert SparseAttentionUtils.unpad_sequence_output(
+        pad, torch.zeros(2, 64, 8)).shape[1] == 60
