@@ -268,68 +268,23 @@ class SparseSelfAttention(torch.nn.Module):
             self._masks[seq_len] = mask
         return self._masks[seq_len].to(device)
 
-    def forward(self, query, key, value, attn_mask: Optional[torch.Tensor] = None):
+    def forward(self, query, key, value,
+                attn_mask: Optional[torch.Tensor] = None,
+                key_padding_mask: Optional[torch.Tensor] = None):
+        """attn_mask: [S, S]-broadcastable bool; key_padding_mask: [B, S]
+        truthy = keep (reference sparse_self_attention.py mask pair)."""
         B, H, S, D = query.shape
-        if attn_mask is None:
+        if attn_mask is None and key_padding_mask is None:
             # gather path: compute only the live blocks
             return block_sparse_attention(query, key, value,
                                           self._layout(S), self.config.block)
         mask = self._mask(S, query.device).unsqueeze(0)  # [1, H, S, S]
-        mask = mask & attn_mask.bool()
+        if attn_mask is not None:
+            mask = mask & attn_mask.bool()
+        if key_padding_mask is not None:
+            mask = mask & key_padding_mask.bool()[:, None, None, :]
         return F.scaled_dot_product_attention(query, key, value,
                                               attn_mask=mask)
-
-
-class SparseSelfAttention(torch.nn.Module):
-    """User-facing sparse attention module (reference
-    deepspeed/ops/sparse_attention/sparse_self_attention.py): caches the
-    block layout per sequence length, applies optional attention masks by
-    folding masked columns out of the layout, and routes through the
-    live-block gather path."""
-
-    def __init__(self, sparsity_config: SparsityConfig,
-                 softmax_scale: Optional[float] = None,
-                 attn_mask_mode: str = "mul"):
-        super().__init__()
-        assert attn_mask_mode in ("mul", "add")
-        self.config = sparsity_config
-        self.softmax_scale = softmax_scale
-        self.attn_mask_mode = attn_mask_mode
-        self._layouts = {}
-
-    def _layout(self, seq_len: int, device) -> torch.Tensor:
-        lay = self._layouts.get(seq_len)
-        if lay is None:
-            assert seq_len % self.config.block == 0, \
-                f"seq {seq_len} % block {self.config.block} != 0 — pad " \
-                "with SparseAttentionUtils.pad_to_block_size"
-            lay = self.config.make_layout(seq_len)
-            self._layouts[seq_len] = lay
-        return lay.to(device)
-
-    def forward(self, query, key, value, key_padding_mask=None):
-        """query/key/value [B, H, S, D]; key_padding_mask [B, S] with 0/
-        False (mul mode) or -inf (add mode) marking masked keys."""
-        B, H, S, D = query.shape
-        layout = self._layout(S, query.device)
-        out = block_sparse_attention(query, key, value, layout,
-                                     self.config.block, self.softmax_scale)
-        if key_padding_mask is not None:
-            # masked keys excluded exactly: recompute with the mask folded
-            # into the scores via a dense equivalence on the layout mask
-            dense_mask = layout_to_dense_mask(layout, self.config.block)
-            if self.attn_mask_mode == "mul":
-                keep = key_padding_mask.to(torch.bool)
-            else:
-                keep = key_padding_mask > -1e30
-            full = dense_mask[None] & keep[:, None, None, :]
-            scale = self.softmax_scale or 1.0 / math.sqrt(D)
-            scores = (query.float() @ key.float().transpose(-1, -2)) * scale
-            scores = scores.masked_fill(~full, float("-inf"))
-            p = torch.softmax(scores, -1)
-            p = torch.nan_to_num(p)     # fully-masked rows -> zeros
-            out = (p @ value.float()).to(query.dtype)
-        return out
 
 
 def layout_to_dense_mask(layout: torch.Tensor, block: int) -> torch.Tensor:
