@@ -71,11 +71,23 @@ class InferenceEngine(nn.Module):
             self.tp_group = None
             self.tp_rank = 0
 
+        self._weight_quantized = False
+        if isinstance(config.dtype, str) and config.dtype in ("fp4", "fp6",
+                                                              "fp12"):
+            # FP6-style weight-only float quantization (reference
+            # DeepSpeed-FP6 / fp_quantizer): tight-packed weights + group
+            # scales on GPU via fp_quant.hip; bit-accurate emulation on CPU
+            bits = int(config.dtype[2:])
+            self.module.to(torch.bfloat16)
+            config.dtype = torch.bfloat16
+            n = self._quantize_linear_weights_fp(bits)
+            self._weight_quantized = True
+            log_dist(f"init_inference: fp{bits} weight-only quantized "
+                     f"{n} linears")
         if isinstance(config.dtype, str):
             config.dtype = {"fp32": torch.float32, "fp16": torch.float16,
                             "bf16": torch.bfloat16,
                             "int8": torch.int8}[config.dtype]
-        self._weight_quantized = False
         if config.dtype == torch.int8:
             # weight-only int8 (reference init_inference(dtype=torch.int8)
             # -> GroupQuantizer): int8 weights + group scales resident,
@@ -113,6 +125,50 @@ class InferenceEngine(nn.Module):
             for name, child in list(parent.named_children()):
                 if type(child) is nn.Linear:
                     setattr(parent, name, WOQLinear(child))
+                    n += 1
+        return n
+
+    def _quantize_linear_weights_fp(self, bits: int,
+                                    group_size: int = 2048) -> int:
+        from ..ops import fp_quantizer as fpq
+        from ..ops._loader import get_ext
+
+        class FPWOQLinear(nn.Module):
+            def __init__(self, lin: nn.Linear):
+                super().__init__()
+                self.bits = bits
+                self.group_size = group_size
+                self.shape = lin.weight.shape
+                w = lin.weight.data
+                if w.is_cuda and get_ext() is not None:
+                    q, s = fpq.fp_quantize(w, bits, group_size)
+                    self.register_buffer("q", q)
+                    self.register_buffer("scales", s)
+                    self.weight_emu = None
+                else:
+                    # CPU: identical numerics, dense storage (the packed
+                    # layout is the GPU kernel's memory-format win)
+                    self.weight_emu = nn.Parameter(
+                        fpq.fp_emulate_reference(w.float(), bits,
+                                                 group_size).to(w.dtype),
+                        requires_grad=False)
+                self.bias = lin.bias
+
+            def forward(self, x):
+                if self.weight_emu is not None:
+                    w = self.weight_emu.to(x.dtype)
+                else:
+                    w = fpq.fp_dequantize(
+                        self.q, self.scales,
+                        int(self.shape[0] * self.shape[1]), self.bits,
+                        self.group_size, x.dtype).view(self.shape)
+                return torch.nn.functional.linear(x, w, self.bias)
+
+        n = 0
+        for parent in list(self.module.modules()):
+            for name, child in list(parent.named_children()):
+                if type(child) is nn.Linear:
+                    setattr(parent, name, FPWOQLinear(child))
                     n += 1
         return n
 

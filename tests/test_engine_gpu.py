@@ -154,3 +154,29 @@ def test_zero3_param_offload_gpu():
     off = run(True)
     for a, b in zip(base, off):
         assert abs(a - b) < 1e-3, (base, off)
+
+
+def test_init_inference_fp6_weight_only_gpu():
+    """fp6 weight-only serving on device: packed fp_quant.hip weights,
+    per-forward dequant; logits close to the bf16 engine and byte size
+    ~6/16 of bf16."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    _init_env()
+    torch.manual_seed(0)
+    m16 = LlamaForCausalLM(llama_tiny()).cuda()
+    torch.manual_seed(0)
+    m6 = LlamaForCausalLM(llama_tiny()).cuda()
+    ids = torch.randint(0, 500, (1, 12), device="cuda")
+    inf16 = deepspeed_amd.init_inference(m16, dtype="bf16")
+    inf6 = deepspeed_amd.init_inference(m6, dtype="fp6")
+    with torch.no_grad():
+        l16 = inf16.module(ids).float()
+        l6 = inf6.module(ids).float()
+    rel = (l16 - l6).abs().max() / l16.abs().max()
+    assert float(rel) < 0.12, float(rel)
+    woq = [m for m in inf6.module.modules()
+           if type(m).__name__ == "FPWOQLinear"][0]
+    assert woq.weight_emu is None and woq.q.dtype == torch.uint8
+    n_w = woq.shape[0] * woq.shape[1]
+    assert woq.q.numel() <= n_w  # 6 bits packed vs 16: well under 1 B/elem

@@ -581,3 +581,33 @@ def test_init_inference_int8_weight_only():
     qps = [mod for mod in inf8.module.modules()
            if isinstance(mod, QuantizedParameter)]
     assert qps and all(qp.q.dtype == torch.int8 for qp in qps)
+
+
+def test_init_inference_fp6_weight_only():
+    """dtype='fp6': weight-only FP6 (e3m2) quantization — CPU path uses
+    the bit-accurate emulation the HIP kernel is tested against; logits
+    stay close to bf16."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    from deepspeed_amd.ops.fp_quantizer import fp_emulate_reference
+
+    torch.manual_seed(0)
+    m16 = LlamaForCausalLM(llama_tiny())
+    torch.manual_seed(0)
+    m6 = LlamaForCausalLM(llama_tiny())
+    ids = torch.randint(0, 500, (1, 12))
+    inf16 = deepspeed_amd.init_inference(m16, dtype="bf16")
+    inf6 = deepspeed_amd.init_inference(m6, dtype="fp6")
+    assert inf6._weight_quantized
+    with torch.no_grad():
+        l16 = inf16.module(ids.to(inf16.device)).float()
+        l6 = inf6.module(ids.to(inf6.device)).float()
+    rel = (l16 - l6).abs().max() / l16.abs().max()
+    assert rel < 0.12, float(rel)
+    # the quantized weight IS the emulation of the original
+    woq = [m for m in inf6.module.modules()
+           if type(m).__name__ == "FPWOQLinear"][0]
+    src = [m for m in inf16.module.modules()
+           if isinstance(m, torch.nn.Linear)][0]
+    want = fp_emulate_reference(src.weight.float(), 6, 2048).bfloat16()
+    torch.testing.assert_close(woq.weight_emu.data, want)
