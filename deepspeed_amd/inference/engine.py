@@ -79,6 +79,10 @@ class InferenceEngine(nn.Module):
             # scales on GPU via fp_quant.hip; bit-accurate emulation on CPU
             bits = int(config.dtype[2:])
             self.module.to(torch.bfloat16)
+            if accel.available():
+                # quantize ON device so the packed kernel (not the dense
+                # emulation) produces the resident weights
+                self.module.to(accel.current_device())
             config.dtype = torch.bfloat16
             n = self._quantize_linear_weights_fp(bits)
             self._weight_quantized = True
@@ -93,6 +97,8 @@ class InferenceEngine(nn.Module):
             # -> GroupQuantizer): int8 weights + group scales resident,
             # bf16 activations, dequant on the fly per linear
             self.module.to(torch.bfloat16)
+            if accel.available():
+                self.module.to(accel.current_device())
             config.dtype = torch.bfloat16
             n = self._quantize_linear_weights()
             self._weight_quantized = True
