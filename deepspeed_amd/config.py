@@ -50,6 +50,19 @@ class RandomLTDConfig(ConfigModel):
     total_ltd_steps: int = 1000
 
 
+class EigenvalueConfig(ConfigModel):
+    """Block Hessian eigenvalue estimation for MoQ quantization
+    scheduling (reference runtime/eigenvalue.py + config key eigenvalue)."""
+    enabled: bool = False
+    verbose: bool = False
+    max_iter: int = 100
+    tol: float = 1e-2
+    stability: float = 1e-6
+    gas_boundary_resolution: int = 1
+    layer_name: str = "layers"
+    layer_num: int = 0
+
+
 class ProgressiveLayerDropConfig(ConfigModel):
     """theta(t) keep-probability schedule (reference runtime/
     progressive_layer_drop.py; config key progressive_layer_drop)."""
@@ -261,6 +274,7 @@ class Config:
             **g("data_efficiency", {}))
         self.progressive_layer_drop = ProgressiveLayerDropConfig(
             **g("progressive_layer_drop", {}))
+        self.eigenvalue = EigenvalueConfig(**g("eigenvalue", {}))
 
         self.data_types_grad_accum_dtype = (
             g("data_types", {}).get("grad_accum_dtype", None))

@@ -147,6 +147,23 @@ class Engine(torch.nn.Module):
             log_dist(f"random-LTD: wrapped {n} layers, kept tokens "
                      f"{ltd.min_value}->{ltd.max_value}")
 
+        self.eigenvalue = None
+        ev = self.config.eigenvalue
+        if ev.enabled:
+            from .eigenvalue import Eigenvalue
+            self.eigenvalue = Eigenvalue(
+                verbose=ev.verbose, max_iter=ev.max_iter, tol=ev.tol,
+                stability=ev.stability,
+                gas_boundary_resolution=ev.gas_boundary_resolution,
+                layer_name=ev.layer_name, layer_num=ev.layer_num)
+            # tag the blocks whose sensitivity MoQ schedules by: children
+            # of the ModuleList/attr named layer_name
+            for name, mod in self.module.named_modules():
+                parts = name.split(".")
+                if len(parts) >= 2 and parts[-2] == ev.layer_name and \
+                        parts[-1].isdigit():
+                    mod._deepspeed_eigenvalue_block = True
+
         self.progressive_layer_drop = None
         if self.config.progressive_layer_drop.enabled:
             from .progressive_layer_drop import ProgressiveLayerDrop

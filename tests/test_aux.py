@@ -932,3 +932,53 @@ def _curriculum_custom_worker(rank=0, world=1):
     xs, _ = next(it)
     assert xs.shape[1] <= 6
     assert seen[0] == 4 and seen[1] == 6, seen
+
+
+def test_eigenvalue_from_config():
+    """eigenvalue config section: engine tags decoder blocks and the
+    power iteration returns one positive eigenvalue per block after a
+    create_graph backward (MoQ sensitivity input)."""
+    from .common import run_local
+    run_local(_eigenvalue_worker)
+
+
+def _eigenvalue_worker(rank=0, world=1):
+    import deepspeed_amd
+
+    class Blk(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(8, 8)
+
+        def forward(self, x):
+            return torch.tanh(self.fc(x))
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.layers = torch.nn.ModuleList([Blk(), Blk()])
+            self.out = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            for l in self.layers:
+                x = l(x)
+            y = self.out(x)
+            if labels is not None:
+                return torch.nn.functional.mse_loss(y.float(),
+                                                    labels.float())
+            return y
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=M(), config={
+        "train_micro_batch_size_per_gpu": 4,
+        "eigenvalue": {"enabled": True, "max_iter": 20, "tol": 1e-2,
+                       "layer_name": "layers", "layer_num": 2},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    assert eng.eigenvalue is not None
+    tagged = [m for m in eng.module.modules()
+              if getattr(m, "_deepspeed_eigenvalue_block", False)]
+    assert len(tagged) == 2
+    loss = eng(torch.randn(4, 8).to(eng.device),
+               labels=torch.randn(4, 1).to(eng.device))
+    loss.backward(create_graph=True)
+    evs = eng.eigenvalue.compute_eigenvalue(eng.module)
+    assert len(evs) == 2 and all(e >= 0 for e in evs)
