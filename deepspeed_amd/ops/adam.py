@@ -190,3 +190,59 @@ def fused_adam_step(optimizer, group, master_param, grad_shard, combined_scale,
             if out16 is not None:
                 out16.view(-1).copy_(master[sl])
     return True
+
+
+class DeepSpeedCPUAdam(FusedAdam):
+    """Host-resident Adam stepped by the AVX kernel (reference
+    ops/adam/cpu_adam.py DeepSpeedCPUAdam — the user-facing optimizer for
+    hand-rolled ZeRO-Offload setups). Params must live on CPU; 16-bit
+    params keep an fp32 master in state like FusedAdam."""
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        ext = get_ext()
+        for group in self.param_groups:
+            lr = group["lr"]
+            beta1, beta2 = group["betas"]
+            eps = group["eps"]
+            wd = group["weight_decay"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                assert not p.is_cuda, \
+                    "DeepSpeedCPUAdam steps host params (use FusedAdam on GPU)"
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    master = p if p.dtype == torch.float32 else p.float()
+                    if p.dtype != torch.float32:
+                        state["master"] = master.detach().clone()
+                    state["exp_avg"] = torch.zeros_like(
+                        master, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(
+                        master, dtype=torch.float32)
+                state["step"] += 1
+                master = state.get("master", p)
+                g = p.grad.contiguous().view(-1)
+                if ext is not None and hasattr(ext, "cpu_adam_flat") and \
+                        g.dtype in (torch.float32, torch.bfloat16):
+                    w16 = p.view(-1) if p.dtype == torch.bfloat16 else None
+                    ext.cpu_adam_flat(master.view(-1), g,
+                                      state["exp_avg"].view(-1),
+                                      state["exp_avg_sq"].view(-1), w16,
+                                      lr, beta1, beta2, eps, wd,
+                                      state["step"], 1.0, self.adam_w_mode)
+                    if p.dtype not in (torch.bfloat16, torch.float32):
+                        p.copy_(master)
+                else:
+                    _torch_adam_step(master, p.grad, state["exp_avg"],
+                                     state["exp_avg_sq"], lr, beta1, beta2,
+                                     eps, wd, state["step"],
+                                     self.adam_w_mode)
+                    if master is not p:
+                        p.copy_(master)
+        return loss

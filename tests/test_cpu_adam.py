@@ -119,3 +119,40 @@ def test_lamb_via_engine_config():
         assert torch.isfinite(loss)
 
     run_local(worker)
+
+
+def test_deepspeed_cpu_adam_user_api():
+    """DeepSpeedCPUAdam (reference ops/adam/cpu_adam.py): host params step
+    through the AVX kernel, matching torch AdamW; bf16 params keep an fp32
+    master. GPU params are rejected."""
+    from deepspeed_amd.ops.adam import DeepSpeedCPUAdam
+    torch.manual_seed(0)
+    p = torch.nn.Parameter(torch.randn(1000))
+    ref = torch.nn.Parameter(p.detach().clone())
+    opt = DeepSpeedCPUAdam([p], lr=1e-2, weight_decay=0.01)
+    ropt = torch.optim.AdamW([ref], lr=1e-2, weight_decay=0.01)
+    for _ in range(5):
+        g = torch.randn(1000)
+        p.grad = g.clone()
+        ref.grad = g.clone()
+        opt.step()
+        ropt.step()
+    torch.testing.assert_close(p, ref, rtol=1e-4, atol=1e-5)
+    # bf16 params: fp32 master in state, bf16 write-back
+    pb = torch.nn.Parameter(torch.randn(256).bfloat16())
+    ob = DeepSpeedCPUAdam([pb], lr=1e-2)
+    pb.grad = torch.randn(256).bfloat16()
+    ob.step()
+    assert ob.state[pb]["master"].dtype == torch.float32
+
+
+def test_reference_namespace_shims():
+    from deepspeed_amd.utils import (logger, log_dist, groups,
+                                     RepeatingLoader, see_memory_usage)
+    from deepspeed_amd.pipe import PipelineModule, LayerSpec, TiedLayerSpec
+    from deepspeed_amd.accelerator import get_accelerator
+    a = get_accelerator()
+    assert a.device_name() and a.communication_backend_name()
+    loader = RepeatingLoader([1, 2])
+    it = iter(loader)
+    assert [next(it) for _ in range(4)] == [1, 2, 1, 2]
