@@ -3,6 +3,10 @@
     python -m deepspeed_amd.launcher.runner --num_gpus 2 examples/generate.py
 """
 
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import torch
 
 import deepspeed_amd

@@ -4,6 +4,10 @@ cache, SplitFuse token packing, and the paged flash-decode kernel.
     python examples/serve_continuous_batching.py          # offline demo
     python examples/serve_continuous_batching.py --http   # FastAPI server
 """
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import argparse
 
 import torch

@@ -1,6 +1,10 @@
 """Curriculum-learning example: data_efficiency config drives the
 difficulty schedule; the engine dataloader samples only sequences at or
 below the current difficulty (difficulty = sample length by default)."""
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import torch
 
 import deepspeed_amd

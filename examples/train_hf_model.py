@@ -10,6 +10,10 @@ Multi-node with elastic restarts:
         --max_restarts 3 examples/train_hf_model.py
 """
 
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import torch
 
 import deepspeed_amd

@@ -4,6 +4,10 @@ Launch on one MI355X node:
     python -m deepspeed_amd.launcher.runner --num_gpus 8 examples/train_llama.py
 """
 
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import torch
 
 import deepspeed_amd

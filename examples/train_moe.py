@@ -3,6 +3,10 @@ parallelism (EP over the xGMI mesh when launched with
 `python -m deepspeed_amd.launcher.runner --num_gpus 8 examples/train_moe.py`),
 grouped expert GEMMs, expert-DP-aware ZeRO-2, and per-EP-rank expert
 checkpoints."""
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 import torch
 
 import deepspeed_amd
