@@ -72,6 +72,9 @@ class InferenceEngine(nn.Module):
             self.tp_rank = 0
 
         self._weight_quantized = False
+        # NOTE: weight-only quantization targets plain nn.Linear modules;
+        # with tp_size > 1 the TP-sharded linear wrappers keep their 16-bit
+        # shards (quantized TP serving = follow-up, docs/roadmap.md)
         if isinstance(config.dtype, str) and config.dtype in ("fp4", "fp6",
                                                               "fp12"):
             # FP6-style weight-only float quantization (reference
