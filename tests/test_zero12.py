@@ -863,13 +863,15 @@ def _fetch_stats_worker(rank=0, world=1):
     assert s["prefetched"] > 0
 
 
-def test_zero_fp32_grad_accum_ws2():
+@pytest.mark.parametrize("offload", [False, True])
+def test_zero_fp32_grad_accum_ws2(offload):
     """fp32_grad_accum: micro-step grads accumulate in an fp32 flat buffer
-    (.grad can't alias it, the hook folds-and-frees each bf16 grad)."""
-    run_distributed(_fp32_accum_worker, world_size=2)
+    (.grad can't alias it, the hook folds-and-frees each bf16 grad) —
+    alone and composed with the optimizer offload tier."""
+    run_distributed(_fp32_accum_worker, world_size=2, args=(offload,))
 
 
-def _fp32_accum_worker(rank, world):
+def _fp32_accum_worker(rank, world, offload=False):
     import deepspeed_amd
     lr, steps, gas = 1e-2, 3, 3
     model = TinyNet()
@@ -880,7 +882,9 @@ def _fp32_accum_worker(rank, world):
         "gradient_accumulation_steps": gas,
         "bf16": {"enabled": True},
         "zero_optimization": {"stage": 2, "overlap_comm": False,
-                              "fp32_grad_accum": True},
+                              "fp32_grad_accum": True,
+                              **({"offload_optimizer": {"device": "cpu"}}
+                                 if offload else {})},
         "optimizer": {"type": "AdamW", "params": {"lr": lr}}})
     assert engine.optimizer.buckets[0].grad_flat.dtype == torch.float32
     it = iter(data)
