@@ -374,3 +374,72 @@ def test_pipeline_3d_zero1_ws8():
     """Same 3D mesh with ZeRO-1 partitioning over each (stage, tp) cell's
     DP replicas — exercises the combined pipe x tensor norm group."""
     run_distributed(_pp2tp2dp2_worker, world_size=8, timeout=600, args=(1,))
+
+
+def _pp2tp2dp2_ckpt_worker(rank, world, tmp):
+    """3D (PP2xTP2xDP2) checkpoint save/resume: model files are
+    (mp_rank, pp_rank)-qualified so no two of the 4 model-owning groups
+    collide; resume restores exact weights."""
+    import deepspeed_amd
+    from deepspeed_amd.parallel import groups as pgroups
+    from deepspeed_amd.runtime.pipe.module import PipelineModule, LayerSpec
+    from deepspeed_amd.runtime.tensor_parallel.layers import (
+        ColumnParallelLinear, RowParallelLinear)
+
+    D, FF = 8, 16
+
+    class TPBlock(nn.Module):
+        def __init__(self, seed):
+            super().__init__()
+            torch.manual_seed(seed)
+            w1 = torch.randn(FF, D)
+            w2 = torch.randn(D, FF)
+            g = pgroups.get_tensor_parallel_group()
+            tp = pgroups.get_tensor_parallel_world_size()
+            tr = pgroups.get_tensor_parallel_rank()
+            self.col = ColumnParallelLinear(
+                w1.chunk(tp, dim=0)[tr].clone(), None, g)
+            self.row = RowParallelLinear(
+                w2.chunk(tp, dim=1)[tr].clone(), None, g)
+
+        def forward(self, x):
+            return self.row(torch.relu(self.col(x)))
+
+    def build():
+        net = PipelineModule([LayerSpec(TPBlock, 31), LayerSpec(TPBlock, 32),
+                              LayerSpec(TPBlock, 33), LayerSpec(TPBlock, 34)],
+                             num_stages=2, tp_size=2,
+                             loss_fn=nn.functional.mse_loss,
+                             partition_method="uniform")
+        eng, _, _, _ = deepspeed_amd.initialize(model=net, config={
+            "train_micro_batch_size_per_gpu": 4,
+            "gradient_accumulation_steps": 2,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+        return net, eng
+
+    def data(n, seed):
+        g = torch.Generator().manual_seed(seed)
+        return [(torch.randn(4, D, generator=g),
+                 torch.randn(4, D, generator=g)) for _ in range(n)]
+
+    net, eng = build()
+    dp = net.grid.data_parallel_id
+    it = iter(data(6, seed=99 + dp))
+    for _ in range(2):
+        eng.train_batch(it)
+    eng.save_checkpoint(tmp, tag="t3d")
+    cont = eng.train_batch(it).item()
+
+    net2, eng2 = build()
+    tag, _ = eng2.load_checkpoint(tmp, tag="t3d")
+    assert tag is not None
+    it2 = iter(data(6, seed=99 + dp))
+    for _ in range(4):
+        next(it2)
+    resumed = eng2.train_batch(it2).item()
+    assert abs(resumed - cont) < 1e-6, (resumed, cont)
+
+
+def test_pipeline_3d_checkpoint_resume_ws8(tmp_path):
+    run_distributed(_pp2tp2dp2_ckpt_worker, world_size=8, timeout=600,
+                    args=(str(tmp_path),))
