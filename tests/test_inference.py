@@ -611,3 +611,26 @@ def test_init_inference_fp6_weight_only():
            if isinstance(m, torch.nn.Linear)][0]
     want = fp_emulate_reference(src.weight.float(), 6, 2048).bfloat16()
     torch.testing.assert_close(woq.weight_emu.data, want)
+
+
+def test_continuous_batcher_model_zoo():
+    """The batcher's kv_cache interface is model-agnostic: GPT-2/OPT and
+    Falcon families serve through the same slot pool, matching
+    engine.generate exactly."""
+    import deepspeed_amd
+    from deepspeed_amd.inference.ragged import ContinuousBatcher, Request
+    from deepspeed_amd.models import (GPT2ForCausalLM, opt_mini,
+                                      FalconForCausalLM, falcon_mini_gqa)
+    for build in (lambda: GPT2ForCausalLM(opt_mini()),
+                  lambda: FalconForCausalLM(falcon_mini_gqa())):
+        torch.manual_seed(4)
+        model = build()
+        engine = deepspeed_amd.init_inference(model, dtype=torch.float32)
+        prompt = torch.randint(0, 500, (1, 8),
+                               generator=torch.Generator().manual_seed(1))
+        want = engine.generate(prompt, max_new_tokens=6)[0, 8:].tolist()
+        batcher = ContinuousBatcher(model, max_slots=4)
+        batcher.put(Request(uid=0, prompt=prompt[0], max_new_tokens=6))
+        done = batcher.run_to_completion()
+        assert len(done) == 1 and done[0].generated == want, \
+            (type(model).__name__, done[0].generated, want)
