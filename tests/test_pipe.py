@@ -305,3 +305,62 @@ def test_pipeline_degree_reshape_pp2_to_pp1(tmp_path):
     from deepspeed_amd.checkpoint import ds_to_universal
     ds_to_universal(tmp, tag="pp2")
     run_distributed(_pp1_universal_resume_worker, world_size=2, args=(tmp,))
+
+
+def _tied_reshape_worker_save(rank, world, tmp):
+    """Save a PP2 pipeline with tied first/last layers; the tied module
+    file is written once by the lowest owning stage."""
+    import deepspeed_amd
+    from deepspeed_amd.runtime.pipe.module import (PipelineModule,
+                                                   TiedLayerSpec, LayerSpec)
+    cfg = dict(_CONFIG)
+    cfg["zero_optimization"] = {"stage": 1, "overlap_comm": False}
+    specs = [TiedLayerSpec("emb", nn.Linear, 8, 8),
+             LayerSpec(nn.Linear, 8, 8),
+             LayerSpec(nn.Linear, 8, 8),
+             TiedLayerSpec("emb", nn.Linear, 8, 8)]
+    net = PipelineModule(specs, num_stages=2,
+                         loss_fn=nn.functional.mse_loss,
+                         partition_method="uniform")
+    engine, _, _, _ = deepspeed_amd.initialize(model=net, config=cfg)
+    g = torch.Generator().manual_seed(7)
+    data = [(torch.randn(4, 8, generator=g),
+             torch.randn(4, 8, generator=g)) for _ in range(6)]
+    it = iter(data)
+    for _ in range(2):
+        engine.train_batch(it)
+    engine.save_checkpoint(tmp, tag="tied")
+    if rank == 0:
+        torch.save(net.tied_modules["emb"].weight.detach().clone(),
+                   os.path.join(tmp, "tied_w.pt"))
+
+
+def _tied_reshape_worker_load(rank, world, tmp):
+    import deepspeed_amd
+    from deepspeed_amd.runtime.pipe.module import (PipelineModule,
+                                                   TiedLayerSpec, LayerSpec)
+    cfg = dict(_CONFIG)
+    cfg["zero_optimization"] = {"stage": 1, "overlap_comm": False}
+    specs = [TiedLayerSpec("emb", nn.Linear, 8, 8),
+             LayerSpec(nn.Linear, 8, 8),
+             LayerSpec(nn.Linear, 8, 8),
+             TiedLayerSpec("emb", nn.Linear, 8, 8)]
+    net = PipelineModule(specs, num_stages=1,
+                         loss_fn=nn.functional.mse_loss,
+                         partition_method="uniform")
+    engine, _, _, _ = deepspeed_amd.initialize(model=net, config=cfg)
+    tag, _ = engine.load_checkpoint(tmp, tag="tied", load_universal=True)
+    assert tag is not None
+    want = torch.load(os.path.join(tmp, "tied_w.pt"), weights_only=True)
+    torch.testing.assert_close(net.tied_modules["emb"].weight.detach(),
+                               want)
+
+
+def test_pipeline_tied_reshape_pp2_to_pp1(tmp_path):
+    """Cross-PP-degree resume with TIED layers: the per-key tied file
+    restores the single shared module at PP=1."""
+    tmp = str(tmp_path)
+    run_distributed(_tied_reshape_worker_save, world_size=2, args=(tmp,))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="tied")
+    run_distributed(_tied_reshape_worker_load, world_size=1, args=(tmp,))
