@@ -247,6 +247,11 @@ class Engine(torch.nn.Module):
                 adam_w = True
             from ..ops.adam import FusedAdam
             return FusedAdam(model_parameters, adam_w_mode=adam_w, **params)
+        if name in ("deepspeedcpuadam", "cpuadam", "cpu_adam"):
+            from ..ops.adam import DeepSpeedCPUAdam
+            adam_w = params.pop("adam_w_mode", True)
+            return DeepSpeedCPUAdam(model_parameters, adam_w_mode=adam_w,
+                                    **params)
         if name == "sgd":
             return torch.optim.SGD(model_parameters, **params)
         if name == "lion":
