@@ -634,3 +634,27 @@ def test_continuous_batcher_model_zoo():
         done = batcher.run_to_completion()
         assert len(done) == 1 and done[0].generated == want, \
             (type(model).__name__, done[0].generated, want)
+
+
+def test_checkpoint_load_then_weight_quantize():
+    """checkpoint= composes with dtype='int8': weights load first, then
+    quantize — served logits track the SOURCE model, not the random
+    init."""
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    import tempfile, os
+    torch.manual_seed(0)
+    src = LlamaForCausalLM(llama_tiny())
+    ids = torch.randint(0, 500, (1, 10))
+    with torch.no_grad():
+        want = src(ids).float()
+    with tempfile.TemporaryDirectory() as d:
+        f = os.path.join(d, "m.pt")
+        torch.save(src.state_dict(), f)
+        torch.manual_seed(1234)   # different init
+        inf = deepspeed_amd.init_inference(LlamaForCausalLM(llama_tiny()),
+                                           checkpoint=f, dtype="int8")
+        with torch.no_grad():
+            got = inf.module(ids.to(inf.device)).float()
+    rel = (got - want).abs().max() / want.abs().max()
+    assert float(rel) < 0.05, float(rel)
