@@ -658,3 +658,27 @@ def test_checkpoint_load_then_weight_quantize():
             got = inf.module(ids.to(inf.device)).float()
     rel = (got - want).abs().max() / want.abs().max()
     assert float(rel) < 0.05, float(rel)
+
+
+def test_autotp_qwen2_bias_parity_ws2():
+    """AutoTP over a qkv-bias architecture (Qwen2): column-sharded
+    linears must shard their bias rows too."""
+    run_distributed(_autotp_qwen2_worker, world_size=2)
+
+
+def _autotp_qwen2_worker(rank, world):
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, qwen2_mini
+    torch.manual_seed(9)
+    model = LlamaForCausalLM(qwen2_mini())
+    torch.manual_seed(9)
+    ref_model = LlamaForCausalLM(qwen2_mini())
+    ids = torch.randint(0, 500, (2, 12),
+                        generator=torch.Generator().manual_seed(3))
+    with torch.no_grad():
+        ref = ref_model(ids)
+    engine = deepspeed_amd.init_inference(
+        model, dtype=torch.float32, tensor_parallel={"tp_size": world})
+    with torch.no_grad():
+        out = engine(ids)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
