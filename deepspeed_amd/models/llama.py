@@ -35,6 +35,9 @@ class LlamaConfig:
     tie_embeddings: bool = False
     initializer_range: float = 0.02
     attention_bias: bool = False  # qwen2-style qkv bias
+    # llama3-style rope_scaling dict: {factor, low_freq_factor,
+    # high_freq_factor, original_max_position_embeddings}
+    rope_scaling: dict = None
 
     @property
     def head_dim(self):
@@ -43,6 +46,15 @@ class LlamaConfig:
 
 def llama3_8b():
     return LlamaConfig()
+
+
+def llama3_1_8b():
+    """Llama-3.1-8B: llama3 rope scaling unlocks 128k context on the
+    3-8B geometry (frequencies verified bit-exact vs HF transformers'
+    _compute_llama3_parameters)."""
+    return LlamaConfig(max_seq_len=131072, rope_scaling={
+        "factor": 8.0, "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+        "original_max_position_embeddings": 8192})
 
 
 def llama3_70b():
@@ -255,7 +267,8 @@ class LlamaModel(nn.Module):
     def forward(self, input_ids, positions=None, kv_cache=None):
         x = self.embed_tokens(input_ids)
         cos, sin = rope_tables(self.cfg.head_dim, self.cfg.max_seq_len,
-                               self.cfg.rope_theta, device=x.device)
+                               self.cfg.rope_theta, device=x.device,
+                               rope_scaling=self.cfg.rope_scaling)
         recompute = (self.gradient_checkpointing and self.training
                      and torch.is_grad_enabled())
         if recompute:

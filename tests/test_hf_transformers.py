@@ -204,3 +204,35 @@ def test_kernel_inject_hf_arch_zoo(model_type):
     with torch.no_grad():
         got = model(ids).logits
     torch.testing.assert_close(got, want, rtol=2e-3, atol=2e-3)
+
+
+def test_llama3_rope_scaling_matches_hf():
+    """llama3-style rope_scaling frequencies bit-match HF transformers'
+    _compute_llama3_parameters; a scaled model steps fine."""
+    from transformers.modeling_rope_utils import ROPE_INIT_FUNCTIONS
+    from transformers import AutoConfig
+    from deepspeed_amd.ops.rope import llama3_scale_freqs
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+
+    cfg = AutoConfig.for_model(
+        "llama", hidden_size=256, num_attention_heads=4,
+        num_hidden_layers=1, vocab_size=128, rope_theta=500000.0,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192},
+        max_position_embeddings=32768)
+    hf_inv, _ = ROPE_INIT_FUNCTIONS["llama3"](cfg, device="cpu")
+    base = 1.0 / (500000.0 ** (torch.arange(0, 64, 2,
+                                            dtype=torch.float64) / 64))
+    torch.testing.assert_close(llama3_scale_freqs(base).float(),
+                               hf_inv.float(), rtol=1e-6, atol=1e-9)
+
+    c = llama_tiny()
+    c.rope_scaling = {"factor": 8.0, "low_freq_factor": 1.0,
+                      "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 64}
+    m = LlamaForCausalLM(c)
+    ids = torch.randint(0, 500, (1, 16))
+    loss = m(ids, labels=ids)
+    loss.backward()
+    assert torch.isfinite(loss)
