@@ -91,6 +91,62 @@ def _fused_qkv_rows(out: int, tp_rank: int, tp_size: int) -> torch.Tensor:
 
 
 @torch.no_grad()
+def _is_falcon_block(m: nn.Module) -> bool:
+    return all(hasattr(m, a) for a in ("qkv", "dense", "mlp_fc",
+                                       "mlp_proj", "num_heads",
+                                       "num_kv_heads", "head_dim"))
+
+
+def _shard_falcon_block(blk: nn.Module, tp_group, tp_rank: int,
+                        tp_size: int) -> int:
+    """Falcon's fused qkv concatenates [H*d | Hkv*d | Hkv*d] — NOT three
+    equal thirds, so the generic fused-qkv split cannot apply. TP plan:
+    split the q heads across ranks, REPLICATE the (grouped/multi-query)
+    kv heads, row-shard `dense` over each rank's q-head columns, and
+    column/row-shard the MLP pair. The block's cfg is replaced by a
+    per-block copy with the local head count."""
+    H, Hkv, d = blk.num_heads, blk.num_kv_heads, blk.head_dim
+    assert H % tp_size == 0, f"falcon heads {H} not divisible by {tp_size}"
+    hl = H // tp_size
+    q_rows = torch.arange(tp_rank * hl * d, (tp_rank + 1) * hl * d)
+    if Hkv % tp_size == 0:
+        # GQA: each rank takes the kv heads that serve its q-head block
+        # (splitting kv keeps the q->kv group mapping intact)
+        kvl = Hkv // tp_size
+        k0 = H * d + tp_rank * kvl * d
+        v0 = (H + Hkv) * d + tp_rank * kvl * d
+        kv_rows = torch.cat([torch.arange(k0, k0 + kvl * d),
+                             torch.arange(v0, v0 + kvl * d)])
+        new_kv = kvl
+    else:
+        # MQA (or kv not divisible): replicate every kv head; the local
+        # q block maps onto the full (replicated) kv set
+        assert Hkv == 1 or hl % Hkv == 0, \
+            f"local heads {hl} not grouped by kv {Hkv}"
+        kv_rows = torch.arange(H * d, (H + 2 * Hkv) * d)
+        new_kv = Hkv
+    rows = torch.cat([q_rows, kv_rows])
+    W = blk.qkv.weight.data
+    b = blk.qkv.bias.data if blk.qkv.bias is not None else None
+    blk.qkv = LinearLayer(W[rows].clone(),
+                          b[rows].clone() if b is not None else None)
+    Wd = blk.dense.weight.data                     # [h, H*d]
+    bd = blk.dense.bias.data if blk.dense.bias is not None else None
+    blk.dense = LinearAllreduce(Wd[:, q_rows].clone(), bd, tp_group)
+    Wf = blk.mlp_fc.weight.data                    # [4h, h]
+    out = Wf.size(0)
+    sl = slice(tp_rank * out // tp_size, (tp_rank + 1) * out // tp_size)
+    bf = blk.mlp_fc.bias.data[sl].clone() \
+        if blk.mlp_fc.bias is not None else None
+    blk.mlp_fc = LinearLayer(Wf[sl].clone(), bf)
+    Wp = blk.mlp_proj.weight.data                  # [h, 4h]
+    bp = blk.mlp_proj.bias.data if blk.mlp_proj.bias is not None else None
+    blk.mlp_proj = LinearAllreduce(Wp[:, sl].clone(), bp, tp_group)
+    blk.num_heads = hl
+    blk.num_kv_heads = new_kv
+    return 4
+
+
 def shard_model(model: nn.Module, tp_group, tp_rank: int, tp_size: int,
                 column_patterns=COLUMN_PATTERNS, row_patterns=ROW_PATTERNS):
     """Replace matching nn.Linear modules with TP-sharded versions in place.
@@ -101,8 +157,15 @@ def shard_model(model: nn.Module, tp_group, tp_rank: int, tp_size: int,
     if tp_size == 1:
         return 0
     replaced = 0
+    falcon_children = set()
+    for m in model.modules():
+        if _is_falcon_block(m):
+            replaced += _shard_falcon_block(m, tp_group, tp_rank, tp_size)
+            falcon_children.update(id(c) for c in m.modules() if c is not m)
     for parent_name, parent in list(model.named_modules()):
         for child_name, child in list(parent._modules.items()):
+            if id(child) in falcon_children:
+                continue
             is_conv1d = _is_hf_conv1d(child)
             if not isinstance(child, nn.Linear) and not is_conv1d:
                 continue
@@ -149,6 +212,10 @@ def shard_attention_heads(model: nn.Module, tp_rank: int, tp_size: int):
     if tp_size == 1:
         return
     for mod in model.modules():
+        if _is_falcon_block(mod):
+            # already sharded by _shard_falcon_block: q heads split,
+            # kv heads REPLICATED — the generic halving does not apply
+            continue
         for attr in ("num_heads", "num_kv_heads", "num_attention_heads",
                      "num_key_value_heads"):
             n = getattr(mod, attr, None)

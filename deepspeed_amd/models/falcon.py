@@ -60,6 +60,11 @@ class FalconBlock(nn.Module):
         d = cfg.head_dim
         self.cfg = cfg
         self.layer_idx = layer_idx
+        # explicit geometry (TP sharding patches num_heads per rank;
+        # cfg.head_dim is derived from the GLOBAL head count)
+        self.num_heads = cfg.num_heads
+        self.num_kv_heads = cfg.num_kv_heads
+        self.head_dim = cfg.head_dim
         self.ln_attn = FusedLayerNorm(h, eps=cfg.ln_eps)
         self.ln_mlp = FusedLayerNorm(h, eps=cfg.ln_eps) \
             if cfg.parallel_attn_norms else None
@@ -70,18 +75,17 @@ class FalconBlock(nn.Module):
         self.mlp_proj = nn.Linear(4 * h, h, bias=False)
 
     def forward(self, x, cos, sin, positions=None, kv_cache=None):
-        cfg = self.cfg
         B, S, _ = x.shape
-        d = cfg.head_dim
+        d = self.head_dim
         a_in = self.ln_attn(x)
         m_in = self.ln_mlp(x) if self.ln_mlp is not None else a_in
 
         qkv = self.qkv(a_in)
-        nq = cfg.num_heads * d
-        nk = cfg.num_kv_heads * d
-        q = qkv[..., :nq].view(B, S, cfg.num_heads, d)
-        k = qkv[..., nq:nq + nk].view(B, S, cfg.num_kv_heads, d)
-        v = qkv[..., nq + nk:].view(B, S, cfg.num_kv_heads, d)
+        nq = self.num_heads * d
+        nk = self.num_kv_heads * d
+        q = qkv[..., :nq].view(B, S, self.num_heads, d)
+        k = qkv[..., nq:nq + nk].view(B, S, self.num_kv_heads, d)
+        v = qkv[..., nq + nk:].view(B, S, self.num_kv_heads, d)
         q = apply_rope(q, cos, sin, positions)
         k = apply_rope(k, cos, sin, positions)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))

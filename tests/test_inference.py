@@ -682,3 +682,33 @@ def _autotp_qwen2_worker(rank, world):
     with torch.no_grad():
         out = engine(ids)
     torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_autotp_falcon_parity_ws2():
+    """Falcon TP: non-uniform fused qkv (q heads split across ranks, MQA/
+    GQA kv REPLICATED), dense row-sharded over the local q-head columns,
+    MLP column/row pair — exact forward parity vs the unsharded model."""
+    run_distributed(_autotp_falcon_worker, world_size=2)
+
+
+def _autotp_falcon_worker(rank, world):
+    import deepspeed_amd
+    from deepspeed_amd.models import (FalconForCausalLM, falcon_mini,
+                                      falcon_mini_gqa)
+    for cfg_fn in (falcon_mini, falcon_mini_gqa):
+        torch.manual_seed(8)
+        model = FalconForCausalLM(cfg_fn())
+        torch.manual_seed(8)
+        ref_model = FalconForCausalLM(cfg_fn())
+        ids = torch.randint(0, 500, (2, 12),
+                            generator=torch.Generator().manual_seed(3))
+        with torch.no_grad():
+            ref = ref_model(ids)
+        engine = deepspeed_amd.init_inference(
+            model, dtype=torch.float32, tensor_parallel={"tp_size": world})
+        blocks = [m for m in engine.module.modules()
+                  if hasattr(m, "num_kv_heads") and hasattr(m, "qkv")]
+        assert blocks and all(b.num_heads == 2 for b in blocks)  # 4 heads/2
+        with torch.no_grad():
+            out = engine(ids)
+        torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
