@@ -982,3 +982,42 @@ def _eigenvalue_worker(rank=0, world=1):
     loss.backward(create_graph=True)
     evs = eng.eigenvalue.compute_eigenvalue(eng.module)
     assert len(evs) == 2 and all(e >= 0 for e in evs)
+
+
+def test_save_checkpoint_exclude_frozen(tmp_path):
+    """exclude_frozen_parameters (LoRA-style): the model-states file keeps
+    only trainable params (reference engine.save_checkpoint kwarg)."""
+    from .common import run_local
+    run_local(_frozen_worker, args=(str(tmp_path),))
+
+
+def _frozen_worker(rank, world, tmp):
+    import deepspeed_amd
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.base = torch.nn.Linear(8, 8)
+            self.base.weight.requires_grad_(False)
+            self.base.bias.requires_grad_(False)
+            self.adapter = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            return torch.nn.functional.mse_loss(
+                self.adapter(self.base(x)).float(), labels.float())
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=M(), config={
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    loss = eng(torch.randn(2, 8).to(eng.device),
+               labels=torch.randn(2, 1).to(eng.device))
+    eng.backward(loss)
+    eng.step()
+    eng.save_checkpoint(tmp, tag="fz", exclude_frozen_parameters=True)
+    import os
+    state = torch.load(os.path.join(tmp, "fz",
+                                    "mp_rank_00_model_states.pt"),
+                       weights_only=False)
+    keys = set(state["module"])
+    assert "adapter.weight" in keys and "adapter.bias" in keys
+    assert not any(k.startswith("base.") for k in keys), keys
