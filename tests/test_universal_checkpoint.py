@@ -186,3 +186,21 @@ def test_universal_moe_expert_dp_reshape(tmp_path):
     from deepspeed_amd.checkpoint import ds_to_universal
     ds_to_universal(tmp, tag="moe0")
     run_distributed(_moe_uni_resume_worker, world_size=2, args=(tmp,))
+
+
+def test_universal_cross_stage_z2_to_z3(tmp_path):
+    """Save under ZeRO-2, convert, resume under ZeRO-3 (same DP): the
+    universal layout is stage-agnostic (per-param fp32 + moments)."""
+    tmp = str(tmp_path)
+    run_distributed(_save_worker, world_size=2, args=(tmp, 2))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="step2")
+    run_distributed(_resume_worker, world_size=2, args=(tmp, 3))
+
+
+def test_universal_cross_stage_z3_to_z1(tmp_path):
+    tmp = str(tmp_path)
+    run_distributed(_save_worker, world_size=2, args=(tmp, 3))
+    from deepspeed_amd.checkpoint import ds_to_universal
+    ds_to_universal(tmp, tag="step2")
+    run_distributed(_resume_worker, world_size=2, args=(tmp, 1))
