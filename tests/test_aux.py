@@ -1021,3 +1021,53 @@ def _frozen_worker(rank, world, tmp):
     keys = set(state["module"])
     assert "adapter.weight" in keys and "adapter.bias" in keys
     assert not any(k.startswith("base.") for k in keys), keys
+
+
+def test_load_module_only_and_scheduler_resume(tmp_path):
+    """load_module_only=True restores weights but leaves the optimizer
+    fresh; a WarmupLR scheduler's position rides the checkpoint."""
+    from .common import run_local
+    run_local(_module_only_worker, args=(str(tmp_path),))
+
+
+def _module_only_worker(rank, world, tmp):
+    import deepspeed_amd
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(3)
+            self.fc = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            return torch.nn.functional.mse_loss(self.fc(x).float(),
+                                                labels.float())
+
+    cfg = {"train_micro_batch_size_per_gpu": 2,
+           "optimizer": {"type": "AdamW", "params": {"lr": 1e-2}},
+           "scheduler": {"type": "WarmupLR",
+                         "params": {"warmup_num_steps": 10,
+                                    "warmup_max_lr": 1e-2}}}
+    eng, _, _, sched = deepspeed_amd.initialize(model=M(), config=cfg)
+    for _ in range(4):
+        loss = eng(torch.randn(2, 8).to(eng.device),
+                   labels=torch.randn(2, 1).to(eng.device))
+        eng.backward(loss)
+        eng.step()
+    lr_after4 = eng.get_lr()[0]
+    eng.save_checkpoint(tmp, tag="s4")
+    w = eng.module.fc.weight.detach().clone()
+
+    # full resume: scheduler position restored
+    eng2, _, _, _ = deepspeed_amd.initialize(model=M(), config=cfg)
+    eng2.load_checkpoint(tmp, tag="s4")
+    assert abs(eng2.get_lr()[0] - lr_after4) < 1e-9
+    torch.testing.assert_close(eng2.module.fc.weight.detach(), w)
+
+    # module-only: weights restored, optimizer state empty
+    eng3, _, _, _ = deepspeed_amd.initialize(model=M(), config=cfg)
+    eng3.load_checkpoint(tmp, tag="s4", load_module_only=True)
+    torch.testing.assert_close(eng3.module.fc.weight.detach(), w)
+    inner = getattr(eng3.optimizer, "optimizer", eng3.optimizer)
+    assert all(len(s) == 0 or s.get("step", 0) == 0
+               for s in inner.state.values()) or not inner.state
