@@ -712,3 +712,27 @@ def _autotp_falcon_worker(rank, world):
         with torch.no_grad():
             out = engine(ids)
         torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_continuous_batcher_paged_model_zoo():
+    """Paged-block KV cache under the batcher for GPT-2/OPT and Falcon —
+    block tables and validity masks are model-agnostic too."""
+    import deepspeed_amd
+    from deepspeed_amd.inference.ragged import (ContinuousBatcher, Request,
+                                                PagedKVCache)
+    from deepspeed_amd.models import (GPT2ForCausalLM, opt_mini,
+                                      FalconForCausalLM, falcon_mini_gqa)
+    for build in (lambda: GPT2ForCausalLM(opt_mini()),
+                  lambda: FalconForCausalLM(falcon_mini_gqa())):
+        torch.manual_seed(4)
+        model = build()
+        engine = deepspeed_amd.init_inference(model, dtype=torch.float32)
+        prompt = torch.randint(0, 500, (1, 8),
+                               generator=torch.Generator().manual_seed(1))
+        want = engine.generate(prompt, max_new_tokens=6)[0, 8:].tolist()
+        batcher = ContinuousBatcher(model, max_slots=4,
+                                    cache_cls=PagedKVCache)
+        batcher.put(Request(uid=0, prompt=prompt[0], max_new_tokens=6))
+        done = batcher.run_to_completion()
+        assert len(done) == 1 and done[0].generated == want, \
+            (type(model).__name__, done[0].generated, want)
