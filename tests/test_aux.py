@@ -1071,3 +1071,35 @@ def _module_only_worker(rank, world, tmp):
     inner = getattr(eng3.optimizer, "optimizer", eng3.optimizer)
     assert all(len(s) == 0 or s.get("step", 0) == 0
                for s in inner.state.values()) or not inner.state
+
+
+def test_random_ltd_with_zero3():
+    """random-LTD wrapping composes with ZeRO-3 module-unit partitioning
+    (wrap happens before the optimizer builds units)."""
+    from .common import run_local
+    run_local(_ltd_zero3_worker)
+
+
+def _ltd_zero3_worker(rank=0, world=1):
+    import deepspeed_amd
+    from deepspeed_amd.models import LlamaForCausalLM, llama_tiny
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny())
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "bf16": {"enabled": True},
+        "zero_optimization": {"stage": 3, "overlap_comm": False},
+        "data_efficiency": {
+            "enabled": True,
+            "random_ltd": {"enabled": True, "layers_attr": "model.layers",
+                           "skip_first": 1, "skip_last": 0,
+                           "min_value": 8, "max_value": 32,
+                           "seq_per_step": 8, "total_ltd_steps": 4}},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    for _ in range(3):
+        ids = torch.randint(0, 500, (2, 32))
+        loss = engine(ids.to(engine.device), labels=ids.to(engine.device))
+        engine.backward(loss)
+        engine.step()
+        assert torch.isfinite(loss)
+    assert engine.random_ltd_scheduler.current_seq > 8
