@@ -260,11 +260,17 @@ class Config:
         self.flops_profiler = FlopsProfilerConfig(**g("flops_profiler", {}))
         # reference accepts the writer blocks at the top level of ds_config
         # (tensorboard/wandb/comet/csv_monitor) as well as grouped
-        mon = dict(g("monitor_config", {}))
+        mon = dict(g("monitor_config", g("monitor", {})))
         for k in ("tensorboard", "wandb", "comet", "csv_monitor"):
             blk = g(k, None)
             if blk is not None and k not in mon:
                 mon[k] = blk
+        # any enabled writer block activates the monitor (reference
+        # monitor_config semantics: the writers carry their own enables)
+        if not mon.get("enabled") and any(
+                isinstance(mon.get(k), dict) and mon[k].get("enabled")
+                for k in ("tensorboard", "wandb", "comet", "csv_monitor")):
+            mon["enabled"] = True
         self.monitor = MonitorConfig(**mon)
         self.moe = MoEConfig(**g("moe", {}))
         self.pipeline = PipelineConfig(**g("pipeline", {}))

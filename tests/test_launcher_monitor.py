@@ -210,3 +210,48 @@ def test_bin_cli_entry_points():
         assert os.stat(p).st_mode & stat.S_IXUSR
     a = parse_args(["--launcher", "slurm", "--num_gpus", "4", "x.py"])
     assert a.launcher == "slurm" and a.user_script == "x.py"
+
+
+def test_engine_monitor_csv_and_comms_logger(tmp_path):
+    """Engine-integrated observability: monitor csv files appear with
+    Train/ events, and the comms logger records collectives (ws2)."""
+    from .common import run_distributed
+    run_distributed(_monitored_worker, world_size=2,
+                    args=(str(tmp_path),))
+
+
+def _monitored_worker(rank, world, tmp):
+    import os
+    import deepspeed_amd
+    import deepspeed_amd.comm as dcomm
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = torch.nn.Linear(8, 1)
+
+        def forward(self, x, labels=None):
+            return torch.nn.functional.mse_loss(self.fc(x).float(),
+                                                labels.float())
+
+    eng, _, _, _ = deepspeed_amd.initialize(model=M(), config={
+        "train_micro_batch_size_per_gpu": 2,
+        "steps_per_print": 1,
+        "csv_monitor": {"enabled": True, "output_path": tmp,
+                        "job_name": "t"},
+        "comms_logger": {"enabled": True},
+        "zero_optimization": {"stage": 2, "overlap_comm": False},
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}}})
+    for _ in range(2):
+        loss = eng(torch.randn(2, 8).to(eng.device),
+                   labels=torch.randn(2, 1).to(eng.device))
+        eng.backward(loss)
+        eng.step()
+    cl = dcomm.get_comms_logger()
+    assert cl is not None and cl.records, cl.records if cl else None
+    if rank == 0:
+        files = []
+        for root, _, fs in os.walk(tmp):
+            files += [os.path.join(root, f) for f in fs if f.endswith(".csv")]
+        assert files, f"no csv monitor output under {tmp}"
+        assert any("Train_lr" in os.path.basename(f) for f in files)
