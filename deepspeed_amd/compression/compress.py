@@ -24,6 +24,10 @@ def init_compression(model: nn.Module, compression_config: dict):
      "row_pruning": {...}, "activation_quantization": {...}}
     """
     cc = compression_config or {}
+    # accept a full ds_config (reference init_compression takes the whole
+    # config and reads its compression_training section)
+    if "compression_training" in cc:
+        cc = cc["compression_training"] or {}
 
     def groups_of(section):
         return (cc.get(section, {}) or {}).get("different_groups", {}) or {}
